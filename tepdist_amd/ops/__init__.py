@@ -1,0 +1,11 @@
+from tepdist_amd.ops.interface import (  # noqa: F401
+    linear,
+    matmul,
+    layernorm,
+    softmax,
+    attention,
+    embedding,
+    cross_entropy,
+    dropout,
+    adamw_step,
+)

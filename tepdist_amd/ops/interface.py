@@ -1,0 +1,237 @@
+"""Autograd-facing op layer with device dispatch.
+
+Every op the planner shards goes through here. Dispatch:
+  - CUDA (= ROCm/MI355X) tensors -> tepdist_amd.ops.hip (hand-written CDNA4
+    HIP kernels; raises ImportError loudly if the extension is missing on a
+    GPU machine — no silent eager fallback).
+  - CPU tensors -> tepdist_amd.ops.reference (pure torch; used by CPU tests
+    and as the numerics baseline).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from tepdist_amd.ops import reference as ref
+
+
+def _backend(t: torch.Tensor):
+    if t.is_cuda:
+        from tepdist_amd.ops import hip as be  # loud ImportError if .so missing
+        return be
+    return ref
+
+
+# --------------------------------------------------------------------------
+
+
+class LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias, act: str):
+        be = _backend(x)
+        x2d = x.reshape(-1, x.shape[-1])
+        y, pre_act = be.linear_fwd(x2d, w, bias, act)
+        ctx.save_for_backward(x2d, w, pre_act if pre_act is not None else torch.empty(0))
+        ctx.has_bias = bias is not None
+        ctx.act = act
+        ctx.in_shape = x.shape
+        return y.reshape(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, w, pre_act = ctx.saved_tensors
+        if ctx.act == "none":
+            pre_act = None
+        be = _backend(dy)
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx, dw, db = be.linear_bwd(dy2d, x2d, w, ctx.has_bias, ctx.act, pre_act)
+        return dx.reshape(ctx.in_shape), dw, db, None
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None,
+           act: str = "none") -> torch.Tensor:
+    """y = act(x @ w^T + bias); w in [out, in] layout."""
+    return LinearFn.apply(x, w, bias, act)
+
+
+class MatmulFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        be = _backend(a)
+        ctx.save_for_backward(a, b)
+        return be.matmul(a, b)
+
+    @staticmethod
+    def backward(ctx, dc):
+        a, b = ctx.saved_tensors
+        be = _backend(dc)
+        dc = dc.contiguous()
+        da = be.matmul(dc, b.transpose(-1, -2).contiguous())
+        db = be.matmul(a.transpose(-1, -2).contiguous(), dc)
+        return da, db
+
+
+def matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    return MatmulFn.apply(a, b)
+
+
+# --------------------------------------------------------------------------
+
+
+class LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, beta, eps):
+        be = _backend(x)
+        x2d = x.reshape(-1, x.shape[-1])
+        y, mean, rstd = be.layernorm_fwd(x2d, gamma, beta, eps)
+        ctx.save_for_backward(x2d, gamma, mean, rstd)
+        ctx.in_shape = x.shape
+        return y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, gamma, mean, rstd = ctx.saved_tensors
+        be = _backend(dy)
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx, dgamma, dbeta = be.layernorm_bwd(dy2d, x2d, gamma, mean, rstd)
+        return dx.reshape(ctx.in_shape), dgamma, dbeta, None
+
+
+def layernorm(x, gamma, beta, eps: float = 1e-5):
+    return LayerNormFn.apply(x, gamma, beta, eps)
+
+
+# --------------------------------------------------------------------------
+
+
+class SoftmaxFn(torch.autograd.Function):
+    """Fused scale + optional causal mask + softmax over the last dim."""
+
+    @staticmethod
+    def forward(ctx, scores, scale, causal):
+        be = _backend(scores)
+        p = be.softmax_fwd(scores, scale=scale, causal=causal)
+        ctx.save_for_backward(p)
+        ctx.scale = scale
+        return p
+
+    @staticmethod
+    def backward(ctx, dp):
+        (p,) = ctx.saved_tensors
+        be = _backend(dp)
+        ds = be.softmax_bwd(dp.contiguous(), p, scale=ctx.scale)
+        return ds, None, None
+
+
+def softmax(scores, scale: float = 1.0, causal: bool = False):
+    return SoftmaxFn.apply(scores, scale, causal)
+
+
+# --------------------------------------------------------------------------
+
+
+class AttentionFn(torch.autograd.Function):
+    """Causal multi-head attention on [B, H, S, D] tensors."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        be = _backend(q)
+        out, p = be.attention_fwd(q, k, v, causal=causal)
+        ctx.save_for_backward(q, k, v, p)
+        ctx.causal = causal
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, p = ctx.saved_tensors
+        be = _backend(dout)
+        dq, dk, dv = be.attention_bwd(dout.contiguous(), q, k, v, p, causal=ctx.causal)
+        return dq, dk, dv, None
+
+
+def attention(q, k, v, causal: bool = True):
+    return AttentionFn.apply(q, k, v, causal)
+
+
+# --------------------------------------------------------------------------
+
+
+class EmbeddingFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ids, table):
+        be = _backend(table)
+        ctx.save_for_backward(ids)
+        ctx.vocab = table.shape[0]
+        return be.embedding_fwd(ids, table)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids,) = ctx.saved_tensors
+        be = _backend(dy)
+        return None, be.embedding_bwd(dy.contiguous(), ids, ctx.vocab)
+
+
+def embedding(ids, table):
+    return EmbeddingFn.apply(ids, table)
+
+
+# --------------------------------------------------------------------------
+
+
+class CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        be = _backend(logits)
+        loss, lse = be.cross_entropy_fwd(logits, targets, ignore_index)
+        ctx.save_for_backward(logits, targets, lse)
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse = ctx.saved_tensors
+        be = _backend(logits)
+        dl = be.cross_entropy_bwd(dloss, logits, targets, lse, ctx.ignore_index)
+        return dl, None, None
+
+
+def cross_entropy(logits, targets, ignore_index: int = -1):
+    """Mean NLL over non-ignored targets; logits [M, V], targets [M]."""
+    return CrossEntropyFn.apply(logits, targets, ignore_index)
+
+
+# --------------------------------------------------------------------------
+
+
+class DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed, offset):
+        be = _backend(x)
+        y, mask = be.dropout_fwd(x, p, seed, offset)
+        ctx.save_for_backward(mask if mask is not None else torch.empty(0))
+        ctx.p = p
+        ctx.has_mask = mask is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (mask,) = ctx.saved_tensors
+        be = _backend(dy)
+        dx = be.dropout_bwd(dy.contiguous(), mask if ctx.has_mask else None, ctx.p)
+        return dx, None, None, None
+
+
+def dropout(x, p: float, seed: int = 0, offset: int = 0):
+    if p == 0.0:
+        return x
+    return DropoutFn.apply(x, p, seed, offset)
+
+
+def adamw_step(param, master, grad, exp_avg, exp_avg_sq, *, lr, beta1=0.9,
+               beta2=0.999, eps=1e-8, weight_decay=0.01, step=1):
+    be = _backend(param)
+    be.adamw_step(param, master, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
+                  eps, weight_decay, step)

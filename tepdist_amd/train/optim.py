@@ -1,0 +1,67 @@
+"""AdamW with fp32 master weights over bf16 model params.
+
+The per-tensor update is the fused HIP kernel (ops.adamw_step) on GPU and
+the torch reference on CPU. Plays the role of the reference's server-side
+apply-gradients (AG) module (SURVEY.md §2.3 SyncFreeDecomposition: the AG
+DefContext runs the optimizer on the server).
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, List
+
+import torch
+
+from tepdist_amd import ops
+
+
+class AdamW:
+    def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-4,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.01,
+                 no_decay_1d: bool = True):
+        self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.no_decay_1d = no_decay_1d
+        self.step_count = 0
+        self.state = {}
+        for p in self.params:
+            self.state[p] = {
+                "master": p.detach().float().clone(),
+                "exp_avg": torch.zeros_like(p, dtype=torch.float32),
+                "exp_avg_sq": torch.zeros_like(p, dtype=torch.float32),
+            }
+
+    @torch.no_grad()
+    def step(self):
+        self.step_count += 1
+        for p in self.params:
+            if p.grad is None:
+                continue
+            st = self.state[p]
+            wd = 0.0 if (self.no_decay_1d and p.dim() == 1) else self.weight_decay
+            ops.adamw_step(p.data, st["master"], p.grad, st["exp_avg"],
+                           st["exp_avg_sq"], lr=self.lr, beta1=self.beta1,
+                           beta2=self.beta2, eps=self.eps, weight_decay=wd,
+                           step=self.step_count)
+
+    def zero_grad(self):
+        for p in self.params:
+            p.grad = None
+
+    def state_dict(self):
+        return {
+            "step": self.step_count,
+            "state": [
+                {k: v for k, v in self.state[p].items()} for p in self.params
+            ],
+        }
+
+    def load_state_dict(self, sd):
+        self.step_count = sd["step"]
+        for p, st in zip(self.params, sd["state"]):
+            for k in ("master", "exp_avg", "exp_avg_sq"):
+                self.state[p][k].copy_(st[k])
