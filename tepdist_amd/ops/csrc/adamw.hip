@@ -1,0 +1,90 @@
+// Fused AdamW step for gfx950: bf16 param + fp32 master/exp_avg/exp_avg_sq,
+// grad bf16 or fp32. One pass, f32x4-vectorized.
+
+#include <algorithm>
+#include <stdexcept>
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tepdist {
+
+namespace {
+
+constexpr int NT = 256;
+
+template <bool GRAD_BF16>
+__launch_bounds__(NT) __global__
+void adamw_kernel(bf16_t* __restrict__ param, float* __restrict__ master,
+                  const bf16_t* __restrict__ gb, const float* __restrict__ gf,
+                  float* __restrict__ m, float* __restrict__ v, int64_t n,
+                  float lr, float beta1, float beta2, float eps, float wd,
+                  float inv_bc1, float inv_bc2) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i0 < n; i0 += stride) {
+    const int cnt = (int)std::min<int64_t>(4, n - i0);
+    if (cnt == 4) {
+      f32x4 mv = *reinterpret_cast<f32x4*>(m + i0);
+      f32x4 vv = *reinterpret_cast<f32x4*>(v + i0);
+      f32x4 ma = *reinterpret_cast<f32x4*>(master + i0);
+      float g[4];
+      if (GRAD_BF16) {
+        const bf16x4 gv = *reinterpret_cast<const bf16x4*>(gb + i0);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) g[e] = bf2f(gv[e]);
+      } else {
+        const f32x4 gv = *reinterpret_cast<const f32x4*>(gf + i0);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) g[e] = gv[e];
+      }
+      bf16x4 pv;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        mv[e] = beta1 * mv[e] + (1.f - beta1) * g[e];
+        vv[e] = beta2 * vv[e] + (1.f - beta2) * g[e] * g[e];
+        const float denom = sqrtf(vv[e] * inv_bc2) + eps;
+        ma[e] = ma[e] * (1.f - lr * wd) - lr * (mv[e] * inv_bc1) / denom;
+        pv[e] = f2bf(ma[e]);
+      }
+      *reinterpret_cast<f32x4*>(m + i0) = mv;
+      *reinterpret_cast<f32x4*>(v + i0) = vv;
+      *reinterpret_cast<f32x4*>(master + i0) = ma;
+      *reinterpret_cast<bf16x4*>(param + i0) = pv;
+    } else {
+      for (int e = 0; e < cnt; ++e) {
+        const int64_t i = i0 + e;
+        const float g = GRAD_BF16 ? bf2f(gb[i]) : gf[i];
+        m[i] = beta1 * m[i] + (1.f - beta1) * g;
+        v[i] = beta2 * v[i] + (1.f - beta2) * g * g;
+        const float denom = sqrtf(v[i] * inv_bc2) + eps;
+        master[i] = master[i] * (1.f - lr * wd) - lr * (m[i] * inv_bc1) / denom;
+        param[i] = f2bf(master[i]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void adamw_bf16(void* param, float* master, const void* grad_bf16,
+                const float* grad_f32, float* exp_avg, float* exp_avg_sq,
+                int64_t n, float lr, float beta1, float beta2, float eps,
+                float weight_decay, float bc1, float bc2, hipStream_t stream) {
+  const int blocks = (int)std::min<int64_t>((n / 4 + NT - 1) / NT, 2048);
+  const float inv_bc1 = 1.0f / bc1, inv_bc2 = 1.0f / bc2;
+  if (grad_bf16 != nullptr) {
+    hipLaunchKernelGGL(adamw_kernel<true>, dim3(std::max(blocks, 1)), dim3(NT),
+                       0, stream, static_cast<bf16_t*>(param), master,
+                       static_cast<const bf16_t*>(grad_bf16), nullptr,
+                       exp_avg, exp_avg_sq, n, lr, beta1, beta2, eps,
+                       weight_decay, inv_bc1, inv_bc2);
+  } else {
+    hipLaunchKernelGGL(adamw_kernel<false>, dim3(std::max(blocks, 1)),
+                       dim3(NT), 0, stream, static_cast<bf16_t*>(param),
+                       master, nullptr, grad_f32, exp_avg, exp_avg_sq, n, lr,
+                       beta1, beta2, eps, weight_decay, inv_bc1, inv_bc2);
+  }
+}
+
+}  // namespace tepdist

@@ -1,0 +1,180 @@
+// pybind11 bindings for the CDNA4 kernel library. Deliberately torch-free:
+// tensors cross as raw device pointers + shapes (tepdist_amd/ops/hip.py owns
+// allocation via torch and passes torch.cuda.current_stream().cuda_stream).
+// This keeps the extension a pure hipcc build (no hipify, no libtorch ABI).
+
+#include <pybind11/pybind11.h>
+
+#include <hip/hip_runtime.h>
+
+#include "common.h"
+#include "kernels.h"
+
+namespace py = pybind11;
+using namespace tepdist;
+
+namespace {
+
+hipStream_t S(uintptr_t s) { return reinterpret_cast<hipStream_t>(s); }
+
+void check_launch() { HIP_CHECK(hipGetLastError()); }
+
+}  // namespace
+
+PYBIND11_MODULE(_tepdist_hip, m) {
+  m.doc() = "TePDist-AMD gfx950 kernel library";
+
+  m.def("gemm", [](uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t Cpre,
+                   uintptr_t bias, int M, int N, int K, int lda, int ldb,
+                   int ldc, int64_t sa, int64_t sb, int64_t sc, int batch,
+                   bool a_kc, bool b_kc, int epi, uintptr_t stream) {
+    gemm_bf16(reinterpret_cast<void*>(A), reinterpret_cast<void*>(B),
+              reinterpret_cast<void*>(C), reinterpret_cast<void*>(Cpre),
+              reinterpret_cast<void*>(bias), M, N, K, lda, ldb, ldc, sa, sb,
+              sc, batch, a_kc, b_kc, epi, S(stream));
+    check_launch();
+  });
+
+  m.def("layernorm_fwd", [](uintptr_t x, uintptr_t g, uintptr_t b, uintptr_t y,
+                            uintptr_t mean, uintptr_t rstd, int rows, int cols,
+                            float eps, uintptr_t stream) {
+    layernorm_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(g),
+                       reinterpret_cast<void*>(b), reinterpret_cast<void*>(y),
+                       reinterpret_cast<float*>(mean),
+                       reinterpret_cast<float*>(rstd), rows, cols, eps,
+                       S(stream));
+    check_launch();
+  });
+
+  m.def("layernorm_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t g,
+                            uintptr_t mean, uintptr_t rstd, uintptr_t dx,
+                            uintptr_t dg_part, uintptr_t db_part,
+                            uintptr_t dgamma, uintptr_t dbeta, int rows,
+                            int cols, int part_rows, uintptr_t stream) {
+    layernorm_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),
+                       reinterpret_cast<void*>(g),
+                       reinterpret_cast<float*>(mean),
+                       reinterpret_cast<float*>(rstd),
+                       reinterpret_cast<void*>(dx),
+                       reinterpret_cast<float*>(dg_part),
+                       reinterpret_cast<float*>(db_part), rows, cols,
+                       part_rows, S(stream));
+    check_launch();
+    layernorm_bwd_reduce(reinterpret_cast<float*>(dg_part),
+                         reinterpret_cast<float*>(db_part),
+                         reinterpret_cast<void*>(dgamma),
+                         reinterpret_cast<void*>(dbeta), part_rows, cols,
+                         S(stream));
+    check_launch();
+  });
+
+  m.def("softmax_fwd", [](uintptr_t x, uintptr_t p, int64_t rows, int cols,
+                          int sq, float scale, bool causal, uintptr_t stream) {
+    softmax_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(p),
+                     rows, cols, sq, scale, causal, S(stream));
+    check_launch();
+  });
+
+  m.def("softmax_bwd", [](uintptr_t dp, uintptr_t p, uintptr_t ds,
+                          int64_t rows, int cols, float scale,
+                          uintptr_t stream) {
+    softmax_bwd_bf16(reinterpret_cast<void*>(dp), reinterpret_cast<void*>(p),
+                     reinterpret_cast<void*>(ds), rows, cols, scale,
+                     S(stream));
+    check_launch();
+  });
+
+  m.def("embedding_fwd", [](uintptr_t ids, uintptr_t table, uintptr_t out,
+                            int64_t n_ids, int dim, uintptr_t stream) {
+    embedding_fwd_bf16(reinterpret_cast<const int64_t*>(ids),
+                       reinterpret_cast<void*>(table),
+                       reinterpret_cast<void*>(out), n_ids, dim, S(stream));
+    check_launch();
+  });
+
+  m.def("embedding_bwd", [](uintptr_t dy, uintptr_t ids, uintptr_t grad_f32,
+                            uintptr_t grad_bf16, int64_t n_ids, int vocab,
+                            int dim, uintptr_t stream) {
+    embedding_bwd_bf16(reinterpret_cast<void*>(dy),
+                       reinterpret_cast<const int64_t*>(ids),
+                       reinterpret_cast<float*>(grad_f32),
+                       reinterpret_cast<void*>(grad_bf16), n_ids, vocab, dim,
+                       S(stream));
+    check_launch();
+  });
+
+  m.def("cross_entropy_fwd", [](uintptr_t logits, uintptr_t targets,
+                                uintptr_t nll, uintptr_t lse, int64_t rows,
+                                int cols, int ignore, uintptr_t stream) {
+    cross_entropy_fwd_bf16(reinterpret_cast<void*>(logits),
+                           reinterpret_cast<const int64_t*>(targets),
+                           reinterpret_cast<float*>(nll),
+                           reinterpret_cast<float*>(lse), rows, cols, ignore,
+                           S(stream));
+    check_launch();
+  });
+
+  m.def("cross_entropy_bwd", [](uintptr_t logits, uintptr_t targets,
+                                uintptr_t lse, float dloss_over_n,
+                                uintptr_t dlogits, int64_t rows, int cols,
+                                int ignore, uintptr_t stream) {
+    cross_entropy_bwd_bf16(reinterpret_cast<void*>(logits),
+                           reinterpret_cast<const int64_t*>(targets),
+                           reinterpret_cast<const float*>(lse), dloss_over_n,
+                           reinterpret_cast<void*>(dlogits), rows, cols,
+                           ignore, S(stream));
+    check_launch();
+  });
+
+  m.def("dropout_fwd", [](uintptr_t x, uintptr_t y, uintptr_t mask, int64_t n,
+                          float p, uint64_t seed, uint64_t offset,
+                          uintptr_t stream) {
+    dropout_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(y),
+                     reinterpret_cast<uint8_t*>(mask), n, p, seed, offset,
+                     S(stream));
+    check_launch();
+  });
+
+  m.def("dropout_bwd", [](uintptr_t dy, uintptr_t mask, uintptr_t dx,
+                          int64_t n, float p, uintptr_t stream) {
+    dropout_bwd_bf16(reinterpret_cast<void*>(dy),
+                     reinterpret_cast<const uint8_t*>(mask),
+                     reinterpret_cast<void*>(dx), n, p, S(stream));
+    check_launch();
+  });
+
+  m.def("gelu_fwd", [](uintptr_t x, uintptr_t y, int64_t n, uintptr_t stream) {
+    gelu_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(y), n,
+                  S(stream));
+    check_launch();
+  });
+
+  m.def("gelu_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t dx, int64_t n,
+                       uintptr_t stream) {
+    gelu_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),
+                  reinterpret_cast<void*>(dx), n, S(stream));
+    check_launch();
+  });
+
+  m.def("bias_sum", [](uintptr_t dy, uintptr_t db, uintptr_t ws, int64_t rows,
+                       int cols, uintptr_t stream) {
+    bias_sum_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(db),
+                  reinterpret_cast<float*>(ws), rows, cols, S(stream));
+    check_launch();
+  });
+
+  m.def("adamw", [](uintptr_t param, uintptr_t master, uintptr_t grad_bf16,
+                    uintptr_t grad_f32, uintptr_t m_, uintptr_t v_, int64_t n,
+                    float lr, float b1, float b2, float eps, float wd,
+                    float bc1, float bc2, uintptr_t stream) {
+    adamw_bf16(reinterpret_cast<void*>(param),
+               reinterpret_cast<float*>(master),
+               reinterpret_cast<void*>(grad_bf16),
+               reinterpret_cast<const float*>(grad_f32),
+               reinterpret_cast<float*>(m_), reinterpret_cast<float*>(v_), n,
+               lr, b1, b2, eps, wd, bc1, bc2, S(stream));
+    check_launch();
+  });
+
+  m.def("device_sync", []() { HIP_CHECK(hipDeviceSynchronize()); });
+}

@@ -1,0 +1,106 @@
+// Elementwise kernels: GELU fwd/bwd, bias-grad column sum. Vectorized bf16x8
+// grid-stride loops (guide G13).
+
+#include <algorithm>
+#include <stdexcept>
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tepdist {
+
+namespace {
+
+constexpr int NT = 256;
+
+__launch_bounds__(NT) __global__
+void gelu_fwd_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
+                     int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+       i0 < n; i0 += stride) {
+    if (i0 + 8 <= n) {
+      const bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i0);
+      bf16x8 yv;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) yv[e] = f2bf(gelu_f(bf2f(xv[e])));
+      *reinterpret_cast<bf16x8*>(y + i0) = yv;
+    } else {
+      for (int64_t i = i0; i < n; ++i) y[i] = f2bf(gelu_f(bf2f(x[i])));
+    }
+  }
+}
+
+__launch_bounds__(NT) __global__
+void gelu_bwd_kernel(const bf16_t* __restrict__ dy,
+                     const bf16_t* __restrict__ x, bf16_t* __restrict__ dx,
+                     int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+       i0 < n; i0 += stride) {
+    if (i0 + 8 <= n) {
+      const bf16x8 dv = *reinterpret_cast<const bf16x8*>(dy + i0);
+      const bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i0);
+      bf16x8 o;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        o[e] = f2bf(bf2f(dv[e]) * gelu_grad_f(bf2f(xv[e])));
+      *reinterpret_cast<bf16x8*>(dx + i0) = o;
+    } else {
+      for (int64_t i = i0; i < n; ++i)
+        dx[i] = f2bf(bf2f(dy[i]) * gelu_grad_f(bf2f(x[i])));
+    }
+  }
+}
+
+// db[c] = sum_r dy[r][c]. Each block owns a 256-column stripe; threads own a
+// column each and walk a row slice; fp32 atomics combine row slices.
+__launch_bounds__(NT) __global__
+void bias_sum_kernel(const bf16_t* __restrict__ dy, float* __restrict__ ws,
+                     int64_t rows, int cols) {
+  const int c = blockIdx.x * NT + threadIdx.x;
+  if (c >= cols) return;
+  const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
+  const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
+  float s = 0.f;
+  for (int64_t r = r0; r < r1; ++r) s += bf2f(dy[r * cols + c]);
+  atomicAdd(&ws[c], s);
+}
+
+__global__ void cast_ws_kernel(const float* __restrict__ ws,
+                               bf16_t* __restrict__ db, int cols) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c < cols) db[c] = f2bf(ws[c]);
+}
+
+}  // namespace
+
+void gelu_fwd_bf16(const void* x, void* y, int64_t n, hipStream_t stream) {
+  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 2048);
+  hipLaunchKernelGGL(gelu_fwd_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
+                     stream, static_cast<const bf16_t*>(x),
+                     static_cast<bf16_t*>(y), n);
+}
+
+void gelu_bwd_bf16(const void* dy, const void* x, void* dx, int64_t n,
+                   hipStream_t stream) {
+  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 2048);
+  hipLaunchKernelGGL(gelu_bwd_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
+                     stream, static_cast<const bf16_t*>(dy),
+                     static_cast<const bf16_t*>(x), static_cast<bf16_t*>(dx),
+                     n);
+}
+
+void bias_sum_bf16(const void* dy, void* db_out, float* ws_zeroed,
+                   int64_t rows, int cols, hipStream_t stream) {
+  bf16_t* db = static_cast<bf16_t*>(db_out);
+  float* ws = ws_zeroed;  // caller-zeroed fp32 workspace of `cols`
+  const int ysplit = (int)std::min<int64_t>((rows + 255) / 256, 64);
+  dim3 grid((cols + NT - 1) / NT, std::max(ysplit, 1));
+  hipLaunchKernelGGL(bias_sum_kernel, grid, dim3(NT), 0, stream,
+                     static_cast<const bf16_t*>(dy), ws, rows, cols);
+  hipLaunchKernelGGL(cast_ws_kernel, dim3((cols + NT - 1) / NT), dim3(NT), 0,
+                     stream, ws, db, cols);
+}
+
+}  // namespace tepdist

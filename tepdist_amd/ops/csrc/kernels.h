@@ -1,0 +1,81 @@
+// Host-side entry points of the CDNA4 kernel library (implemented in *.hip).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+namespace tepdist {
+
+using bf16 = __hip_bfloat16;
+
+// --- GEMM ------------------------------------------------------------------
+// C[M,N] = A @ B (+bias, +gelu epilogue). Layout flags:
+//   a_kc: A stored row-major [M,K] (k-contiguous); else stored [K,M].
+//   b_kc: B stored row-major [N,K] (k-contiguous, i.e. the operand is a
+//         transposed weight W[N,K]); else stored [K,N].
+// epi: 0 = none, 1 = +bias, 2 = +bias+gelu (writes pre-act to c_pre),
+//      3 = gelu only (writes pre-act to c_pre).
+// Batched over `batch` with element strides; stride 0 broadcasts.
+void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
+               const void* bias, int M, int N, int K, int lda, int ldb,
+               int ldc, int64_t stride_a, int64_t stride_b, int64_t stride_c,
+               int batch, bool a_kc, bool b_kc, int epi, hipStream_t stream);
+
+// --- LayerNorm -------------------------------------------------------------
+void layernorm_fwd_bf16(const void* x, const void* gamma, const void* beta,
+                        void* y, float* mean, float* rstd, int rows, int cols,
+                        float eps, hipStream_t stream);
+void layernorm_bwd_bf16(const void* dy, const void* x, const void* gamma,
+                        const float* mean, const float* rstd, void* dx,
+                        float* dgamma_part, float* dbeta_part, int rows,
+                        int cols, int part_rows, hipStream_t stream);
+void layernorm_bwd_reduce(const float* dgamma_part, const float* dbeta_part,
+                          void* dgamma, void* dbeta, int part_rows, int cols,
+                          hipStream_t stream);
+
+// --- Softmax (fused scale + causal mask) -----------------------------------
+// x: [rows, cols] rows = B*H*Sq, cols = Sk; causal masks col > row_in_tile
+// (with q_offset = Sk - Sq so the last query attends to everything).
+void softmax_fwd_bf16(const void* x, void* p, int64_t rows, int cols, int sq,
+                      float scale, bool causal, hipStream_t stream);
+void softmax_bwd_bf16(const void* dp, const void* p, void* ds, int64_t rows,
+                      int cols, float scale, hipStream_t stream);
+
+// --- Embedding -------------------------------------------------------------
+void embedding_fwd_bf16(const int64_t* ids, const void* table, void* out,
+                        int64_t n_ids, int dim, hipStream_t stream);
+void embedding_bwd_bf16(const void* dy, const int64_t* ids, float* grad_f32,
+                        void* grad_bf16, int64_t n_ids, int vocab, int dim,
+                        hipStream_t stream);
+
+// --- Cross entropy ---------------------------------------------------------
+void cross_entropy_fwd_bf16(const void* logits, const int64_t* targets,
+                            float* nll, float* lse, int64_t rows, int cols,
+                            int ignore_index, hipStream_t stream);
+void cross_entropy_bwd_bf16(const void* logits, const int64_t* targets,
+                            const float* lse, float dloss_over_n,
+                            void* dlogits, int64_t rows, int cols,
+                            int ignore_index, hipStream_t stream);
+
+// --- Dropout ---------------------------------------------------------------
+void dropout_fwd_bf16(const void* x, void* y, uint8_t* mask, int64_t n,
+                      float p, uint64_t seed, uint64_t offset,
+                      hipStream_t stream);
+void dropout_bwd_bf16(const void* dy, const uint8_t* mask, void* dx, int64_t n,
+                      float p, hipStream_t stream);
+
+// --- Elementwise -----------------------------------------------------------
+void gelu_fwd_bf16(const void* x, void* y, int64_t n, hipStream_t stream);
+void gelu_bwd_bf16(const void* dy, const void* x, void* dx, int64_t n,
+                   hipStream_t stream);
+void bias_sum_bf16(const void* dy, void* db, float* ws_zeroed, int64_t rows,
+                   int cols, hipStream_t stream);
+
+// --- AdamW fused step ------------------------------------------------------
+void adamw_bf16(void* param, float* master, const void* grad_bf16,
+                const float* grad_f32, float* exp_avg, float* exp_avg_sq,
+                int64_t n, float lr, float beta1, float beta2, float eps,
+                float weight_decay, float bc1, float bc2, hipStream_t stream);
+
+}  // namespace tepdist

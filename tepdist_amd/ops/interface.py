@@ -69,8 +69,8 @@ class MatmulFn(torch.autograd.Function):
         a, b = ctx.saved_tensors
         be = _backend(dc)
         dc = dc.contiguous()
-        da = be.matmul(dc, b.transpose(-1, -2).contiguous())
-        db = be.matmul(a.transpose(-1, -2).contiguous(), dc)
+        da = be.matmul(dc, b.transpose(-1, -2))
+        db = be.matmul(a.transpose(-1, -2), dc)
         return da, db
 
 

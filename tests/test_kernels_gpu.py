@@ -1,0 +1,254 @@
+"""GPU numerics tests: every hand-written gfx950 kernel vs a plain PyTorch
+fp32 reference of the same op (SURVEY.md §4 test strategy)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from tepdist_amd.ops import hip
+from tepdist_amd import ops
+
+BF16 = torch.bfloat16
+
+
+def _randn(*shape, seed=0, dtype=BF16):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randn(*shape, generator=g).to(dtype).cuda()
+
+
+def _assert_close_bf16(ours, ref32, atol=2e-2, rtol=2e-2, scale=None):
+    """Compare a bf16 kernel output against an fp32 reference with a
+    tolerance scaled to the reference magnitude."""
+    ours32 = ours.float()
+    err = (ours32 - ref32).abs()
+    denom = scale if scale is not None else ref32.abs().max().clamp_min(1.0)
+    rel = (err / denom).max().item()
+    assert rel < rtol, f"max scaled err {rel} (atol ref {err.max().item()})"
+
+
+# --------------------------------------------------------------------------
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (200, 136, 72),
+                                   (256, 512, 1024), (33, 17, 9)])
+def test_gemm_nt_layout(M, N, K):
+    # asymmetric inputs (catches transposed-output bugs)
+    a = _randn(M, K, seed=1)
+    w = _randn(N, K, seed=2)
+    y, _ = hip.linear_fwd(a, w, None, "none")
+    torch.cuda.synchronize()
+    ref = a.float() @ w.float().t()
+    _assert_close_bf16(y, ref, rtol=3e-2, scale=math.sqrt(K))
+
+
+@pytest.mark.parametrize("ta,tb", [(False, False), (False, True),
+                                   (True, False)])
+def test_gemm_transposes(ta, tb):
+    M, N, K = 96, 144, 80
+    a_base = _randn(K, M, seed=3) if ta else _randn(M, K, seed=3)
+    b_base = _randn(N, K, seed=4) if tb else _randn(K, N, seed=4)
+    a = a_base.t() if ta else a_base
+    b = b_base.t() if tb else b_base
+    c = hip.matmul(a, b)
+    torch.cuda.synchronize()
+    ref = a.float() @ b.float()
+    _assert_close_bf16(c, ref, rtol=3e-2, scale=math.sqrt(K))
+
+
+def test_gemm_batched():
+    B, M, N, K = 6, 64, 96, 32
+    a = _randn(B, M, K, seed=5)
+    b = _randn(B, K, N, seed=6)
+    c = hip.matmul(a, b)
+    torch.cuda.synchronize()
+    ref = a.float() @ b.float()
+    _assert_close_bf16(c, ref, rtol=3e-2, scale=math.sqrt(K))
+
+
+def test_linear_bias_gelu():
+    M, N, K = 128, 96, 64
+    x = _randn(M, K, seed=7)
+    w = _randn(N, K, seed=8)
+    b = _randn(N, seed=9)
+    y, pre = hip.linear_fwd(x, w, b, "gelu")
+    torch.cuda.synchronize()
+    pre_ref = (x.float() @ w.float().t() + b.float())
+    y_ref = torch.nn.functional.gelu(pre_ref, approximate="tanh")
+    _assert_close_bf16(pre, pre_ref, rtol=3e-2, scale=math.sqrt(K))
+    _assert_close_bf16(y, y_ref, rtol=3e-2, scale=math.sqrt(K))
+
+
+def test_linear_bwd():
+    M, N, K = 192, 80, 112
+    x = _randn(M, K, seed=10)
+    w = _randn(N, K, seed=11)
+    dy = _randn(M, N, seed=12)
+    dx, dw, db = hip.linear_bwd(dy, x, w, True, "none", None)
+    torch.cuda.synchronize()
+    _assert_close_bf16(dx, dy.float() @ w.float(), rtol=3e-2,
+                       scale=math.sqrt(N))
+    _assert_close_bf16(dw, dy.float().t() @ x.float(), rtol=3e-2,
+                       scale=math.sqrt(M))
+    _assert_close_bf16(db, dy.float().sum(0), rtol=3e-2, scale=math.sqrt(M))
+
+
+# --------------------------------------------------------------------------
+
+
+def test_layernorm_fwd_bwd():
+    rows, cols = 512, 1024
+    x = _randn(rows, cols, seed=13)
+    g = _randn(cols, seed=14)
+    b = _randn(cols, seed=15)
+    y, mean, rstd = hip.layernorm_fwd(x, g, b, 1e-5)
+    dy = _randn(rows, cols, seed=16)
+    dx, dg, db = hip.layernorm_bwd(dy, x, g, mean, rstd)
+    torch.cuda.synchronize()
+
+    x32 = x.float().requires_grad_()
+    g32 = g.float().requires_grad_()
+    b32 = b.float().requires_grad_()
+    y32 = torch.nn.functional.layer_norm(x32, (cols,), g32, b32, 1e-5)
+    y32.backward(dy.float())
+    _assert_close_bf16(y, y32.detach(), rtol=2e-2)
+    _assert_close_bf16(dx, x32.grad, rtol=3e-2)
+    _assert_close_bf16(dg, g32.grad, rtol=3e-2, scale=math.sqrt(rows))
+    _assert_close_bf16(db, b32.grad, rtol=3e-2, scale=math.sqrt(rows))
+
+
+def test_softmax_causal_fwd_bwd():
+    B, H, S = 2, 3, 256
+    s = _randn(B, H, S, S, seed=17)
+    scale = 0.125
+    p = hip.softmax_fwd(s, scale=scale, causal=True)
+    dp = _randn(B, H, S, S, seed=18)
+    ds = hip.softmax_bwd(dp, p, scale=scale)
+    torch.cuda.synchronize()
+
+    s32 = (s.float() * scale).requires_grad_()
+    mask = torch.ones(S, S, dtype=torch.bool, device="cuda").tril()
+    sm = s32.masked_fill(~mask, float("-inf"))
+    # recompute graph for grads wrt pre-scale scores
+    s2 = s.float().requires_grad_()
+    p_ref = torch.softmax((s2 * scale).masked_fill(~mask, float("-inf")), -1)
+    p_ref.backward(dp.float())
+    _assert_close_bf16(p, p_ref.detach(), rtol=2e-2)
+    _assert_close_bf16(ds, s2.grad, rtol=2e-2)
+
+
+def test_attention_fwd_bwd():
+    B, H, S, D = 2, 4, 128, 64
+    q = _randn(B, H, S, D, seed=19)
+    k = _randn(B, H, S, D, seed=20)
+    v = _randn(B, H, S, D, seed=21)
+    out, p = hip.attention_fwd(q, k, v, causal=True)
+    dout = _randn(B, H, S, D, seed=22)
+    dq, dk, dv = hip.attention_bwd(dout, q, k, v, p, causal=True)
+    torch.cuda.synchronize()
+
+    q2, k2, v2 = (t.float().requires_grad_() for t in (q, k, v))
+    ref = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2,
+                                                           is_causal=True)
+    ref.backward(dout.float())
+    _assert_close_bf16(out, ref.detach(), rtol=3e-2)
+    _assert_close_bf16(dq, q2.grad, rtol=4e-2)
+    _assert_close_bf16(dk, k2.grad, rtol=4e-2)
+    _assert_close_bf16(dv, v2.grad, rtol=4e-2)
+
+
+def test_embedding_fwd_bwd():
+    V, D, n = 1000, 256, 4096
+    table = _randn(V, D, seed=23)
+    ids = torch.randint(0, V, (n,), device="cuda")
+    out = hip.embedding_fwd(ids, table)
+    dy = _randn(n, D, seed=24)
+    grad = hip.embedding_bwd(dy, ids, V)
+    torch.cuda.synchronize()
+
+    t32 = table.float().requires_grad_()
+    out_ref = torch.nn.functional.embedding(ids, t32)
+    out_ref.backward(dy.float())
+    assert torch.equal(out.float(), table[ids].float())
+    _assert_close_bf16(grad, t32.grad, rtol=3e-2,
+                       scale=torch.tensor(4.0, device="cuda"))
+
+
+def test_cross_entropy_fwd_bwd():
+    M, V = 512, 50304
+    logits = _randn(M, V, seed=25)
+    targets = torch.randint(0, 50257, (M,), device="cuda")
+    targets[5] = -1
+    loss, lse = hip.cross_entropy_fwd(logits, targets, -1)
+    dlogits = hip.cross_entropy_bwd(torch.ones((), device="cuda"), logits,
+                                    targets, lse, -1)
+    torch.cuda.synchronize()
+
+    l32 = logits.float().requires_grad_()
+    loss_ref = torch.nn.functional.cross_entropy(l32, targets,
+                                                 ignore_index=-1)
+    loss_ref.backward()
+    assert abs(loss.item() - loss_ref.item()) < 2e-2 * loss_ref.item()
+    _assert_close_bf16(dlogits, l32.grad, rtol=5e-2,
+                       scale=l32.grad.abs().max())
+
+
+def test_dropout():
+    x = torch.ones(1 << 20, dtype=BF16, device="cuda")
+    y, mask = hip.dropout_fwd(x, 0.3, seed=7, offset=3)
+    y2, mask2 = hip.dropout_fwd(x, 0.3, seed=7, offset=3)
+    torch.cuda.synchronize()
+    assert torch.equal(mask, mask2), "philox mask not deterministic"
+    keep = mask.float().mean().item()
+    assert abs(keep - 0.7) < 0.01
+    kept_vals = y.float()[mask.bool()]
+    assert torch.allclose(kept_vals,
+                          torch.full_like(kept_vals, 1 / 0.7), atol=1e-2)
+    dy = _randn(1 << 20, seed=26)
+    dx = hip.dropout_bwd(dy, mask, 0.3)
+    torch.cuda.synchronize()
+    ref = dy.float() * mask.float() / 0.7
+    _assert_close_bf16(dx, ref, rtol=2e-2)
+
+
+def test_adamw_matches_torch():
+    n = 10007
+    g0 = torch.Generator().manual_seed(0)
+    p32 = torch.randn(n, generator=g0).cuda()
+    p_ref = torch.nn.Parameter(p32.clone())
+    opt_ref = torch.optim.AdamW([p_ref], lr=1e-2, betas=(0.9, 0.999),
+                                eps=1e-8, weight_decay=0.01)
+    param = p32.to(BF16)
+    master = p32.clone()
+    m = torch.zeros_like(master)
+    v = torch.zeros_like(master)
+    for step in range(1, 6):
+        g = torch.randn(n, generator=torch.Generator().manual_seed(step)).cuda()
+        p_ref.grad = g.clone()
+        opt_ref.step()
+        hip.adamw_step(param, master, g, m, v, lr=1e-2, beta1=0.9,
+                       beta2=0.999, eps=1e-8, weight_decay=0.01, step=step)
+    torch.cuda.synchronize()
+    assert (master - p_ref.detach()).abs().max().item() < 1e-5
+
+
+# --------------------------------------------------------------------------
+
+
+def test_gpt2_tiny_gpu_trains():
+    from tepdist_amd.models import GPT2, GPT2_CONFIGS
+    from tepdist_amd.train import AdamW, Trainer
+    cfg = GPT2_CONFIGS["gpt2-test"]
+    torch.manual_seed(0)
+    model = GPT2(cfg, dtype=BF16).cuda()
+    opt = AdamW(model.parameters(), lr=1e-3)
+    trainer = Trainer(model, opt)
+    g = torch.Generator().manual_seed(42)
+    ids = torch.randint(0, cfg.vocab_size, (4, 33), generator=g).cuda()
+    batch = (ids[:, :-1], ids[:, 1:])
+    losses = [trainer.train_step(lambda i: batch) for _ in range(12)]
+    assert losses[-1] < losses[0] * 0.9, losses
+    assert all(math.isfinite(l) for l in losses)
