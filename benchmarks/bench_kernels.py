@@ -5,7 +5,11 @@ Run on the GPU box:  python benchmarks/bench_kernels.py
 """
 
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

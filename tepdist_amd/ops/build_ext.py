@@ -31,6 +31,7 @@ SOURCES = [
     "dropout.hip",
     "elementwise.hip",
     "adamw.hip",
+    "debug.hip",
     "bindings.cpp",
 ]
 

@@ -176,5 +176,10 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
+  m.def("tr16_probe", [](uintptr_t a, uintptr_t b, uintptr_t stream) {
+    tr16_probe(reinterpret_cast<float*>(a), reinterpret_cast<float*>(b),
+               S(stream));
+  });
+
   m.def("device_sync", []() { HIP_CHECK(hipDeviceSynchronize()); });
 }

@@ -53,18 +53,33 @@ void gelu_bwd_kernel(const bf16_t* __restrict__ dy,
   }
 }
 
-// db[c] = sum_r dy[r][c]. Each block owns a 256-column stripe; threads own a
-// column each and walk a row slice; fp32 atomics combine row slices.
+// db[c] = sum_r dy[r][c]. Each thread owns an 8-column group (one bf16x8
+// load per row -> coalesced) and walks a row slice; fp32 atomics combine
+// the row slices in ws.
 __launch_bounds__(NT) __global__
 void bias_sum_kernel(const bf16_t* __restrict__ dy, float* __restrict__ ws,
                      int64_t rows, int cols) {
-  const int c = blockIdx.x * NT + threadIdx.x;
-  if (c >= cols) return;
+  const int ncg = (cols + 7) / 8;
+  const int cg = blockIdx.x * NT + threadIdx.x;
+  if (cg >= ncg) return;
+  const int c0 = cg * 8;
   const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
   const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
-  float s = 0.f;
-  for (int64_t r = r0; r < r1; ++r) s += bf2f(dy[r * cols + c]);
-  atomicAdd(&ws[c], s);
+  float acc[8] = {};
+  if (c0 + 8 <= cols) {
+    for (int64_t r = r0; r < r1; ++r) {
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(dy + r * cols + c0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += bf2f(v[e]);
+    }
+  } else {
+    for (int64_t r = r0; r < r1; ++r)
+      for (int e = 0; e < 8 && c0 + e < cols; ++e)
+        acc[e] += bf2f(dy[r * cols + c0 + e]);
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e)
+    if (c0 + e < cols) atomicAdd(&ws[c0 + e], acc[e]);
 }
 
 __global__ void cast_ws_kernel(const float* __restrict__ ws,
@@ -95,8 +110,9 @@ void bias_sum_bf16(const void* dy, void* db_out, float* ws_zeroed,
                    int64_t rows, int cols, hipStream_t stream) {
   bf16_t* db = static_cast<bf16_t*>(db_out);
   float* ws = ws_zeroed;  // caller-zeroed fp32 workspace of `cols`
-  const int ysplit = (int)std::min<int64_t>((rows + 255) / 256, 64);
-  dim3 grid((cols + NT - 1) / NT, std::max(ysplit, 1));
+  const int ysplit = (int)std::min<int64_t>((rows + 63) / 64, 128);
+  const int ncg = (cols + 7) / 8;
+  dim3 grid((ncg + NT - 1) / NT, std::max(ysplit, 1));
   hipLaunchKernelGGL(bias_sum_kernel, grid, dim3(NT), 0, stream,
                      static_cast<const bf16_t*>(dy), ws, rows, cols);
   hipLaunchKernelGGL(cast_ws_kernel, dim3((cols + NT - 1) / NT), dim3(NT), 0,

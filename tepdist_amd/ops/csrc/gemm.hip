@@ -1,21 +1,27 @@
 // bf16 MFMA GEMM for gfx950 (MI355X), fp32 accumulation.
 //
-// Structure (v1, register-staged): 128x128 tile, BK=32, 4 waves per block
-// (2x2 wave grid, 64x64 per wave as 4x4 fragments of 16x16), double-buffered
-// LDS with the load-early/write-late split (guide T14), mfma_f32_16x16x32_bf16
-// inner loop. LDS rows padded +16B so the column fragment reads
-// (ds_read_b128, one row per lane in a 16-lane group) are bank-conflict-free
-// without an XOR swizzle.
+// Structure: 128x128 tile, BK=32, 4 waves per block (2x2 wave grid, 64x64
+// per wave as 4x4 fragments of 16x16), double-buffered LDS with the
+// load-early/write-late split (guide T14), mfma_f32_16x16x32_bf16 inner loop.
 //
-// Layout handling: both operands are staged into canonical k-contiguous LDS
-// images ([rows][BK]); operands whose storage is k-outer ([K,F]) are
-// transposed during staging via coalesced 2B loads + packed b128 LDS writes.
-// This serves all three cases training needs (fwd NT, dgrad NN, wgrad TN)
-// with one inner loop.
+// Layout handling without transpose cost: the MFMA contraction is invariant
+// under any k-permutation applied consistently to both operands' fragments
+// (sum over (g,e) of A[m][pi(g,e)]*B[pi(g,e)][n] == sum over k). So:
+//  - "KC" operands (k fastest-varying in memory: A stored [M,K], B stored
+//    [N,K]) stage into a row-padded [F][BK] image with ds_read_b128 fragment
+//    reads (one row per lane in a 16-lane group; +16B row pad keeps it
+//    bank-conflict-free).
+//  - "KO" operands (k outermost: stored [K,F], i.e. dgrad's weight and both
+//    wgrad operands) stage UNtransposed into a [BK/4][BM/16][4][16]-subtiled
+//    image using only 16B loads + 16B LDS writes, and fragments are read with
+//    the gfx950 hardware transpose-read ds_read_b64_tr_b16 (guide T10).
+//  When layouts mix, the KC operand's staging permutes columns so both sides
+//  share the tr-read's k order pi(g,e) = e<4 ? 4g+e : 16+4g+(e-4).
 //
 // Replaces the reference's XLA-codegen GEMMs (all device code
 // in TePDist is XLA-generated PTX; here it is hand-written CDNA4).
 
+#include <algorithm>
 #include <stdexcept>
 #include <string>
 
@@ -28,22 +34,30 @@ namespace {
 
 constexpr int BM = 128, BN = 128, BK = 32;
 constexpr int LDS_ROW = BK + 8;  // +8 bf16 = +16B row pad: conflict-free b128
+constexpr int SLOT = BM * LDS_ROW;  // elements per operand slot (KO uses 4K)
 constexpr int NTHREADS = 256;
 
 enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3 };
 
-// --- staging: k-contiguous operand (stored [F][ld], k inner) ---------------
+typedef __attribute__((address_space(3))) bf16x4* lds_tr_ptr;
 
+DEV_INLINE bf16x4 tr_read(const bf16_t* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_tr_ptr)p);
+}
+
+// --- staging loads ---------------------------------------------------------
+
+// KC operand (stored [F][ld], k inner): 2 x 16B per thread.
 DEV_INLINE void stage_kc_load(const bf16_t* __restrict__ src, int F, int K,
                               int ld, int f0, int k0, bool aligned,
                               bf16x8 regs[2]) {
 #pragma unroll
   for (int u = 0; u < 2; ++u) {
-    int idx = threadIdx.x + u * NTHREADS;  // 0..511
-    int row = idx >> 2;                    // 0..127
-    int kc = (idx & 3) * 8;
+    const int idx = threadIdx.x + u * NTHREADS;  // 0..511
+    const int row = idx >> 2;                    // 0..127
+    const int kc = (idx & 3) * 8;
     bf16x8 v = {};
-    int f = f0 + row, k = k0 + kc;
+    const int f = f0 + row, k = k0 + kc;
     if (f < F && k < K) {
       const bf16_t* p = src + (int64_t)f * ld + k;
       if (aligned && k + 8 <= K) {
@@ -58,42 +72,91 @@ DEV_INLINE void stage_kc_load(const bf16_t* __restrict__ src, int F, int K,
   }
 }
 
-// --- staging: k-outer operand (stored [K][ld]) -> transpose in registers ---
-
+// KO operand (stored [K][ld], f inner): 2 x 16B per thread, no transpose.
 DEV_INLINE void stage_ko_load(const bf16_t* __restrict__ src, int F, int K,
-                              int ld, int f0, int k0, bf16x8 regs[2]) {
-  int f = threadIdx.x & 127;
-  int kh = threadIdx.x >> 7;  // 0/1: k halves of 16
-  int fg = f0 + f;
+                              int ld, int f0, int k0, bool aligned,
+                              bf16x8 regs[2]) {
 #pragma unroll
-  for (int h = 0; h < 2; ++h) {
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NTHREADS;  // 0..511
+    const int k = idx >> 4;                      // 0..31
+    const int f8 = (idx & 15) * 8;               // 0..120
     bf16x8 v = {};
-    if (fg < F) {
+    const int kg = k0 + k, fg = f0 + f8;
+    if (kg < K && fg < F) {
+      const bf16_t* p = src + (int64_t)kg * ld + fg;
+      if (aligned && fg + 8 <= F) {
+        v = *reinterpret_cast<const bf16x8*>(p);
+      } else {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        int k = k0 + kh * 16 + h * 8 + e;
-        if (k < K) v[e] = src[(int64_t)k * ld + fg];
+        for (int e = 0; e < 8; ++e)
+          if (fg + e < F) v[e] = p[e];
       }
     }
-    regs[h] = v;
+    regs[u] = v;
   }
 }
 
-DEV_INLINE void stage_write_kc(bf16_t* dst, const bf16x8 regs[2]) {
+// --- staging writes --------------------------------------------------------
+
+// natural k order (used when both operands are KC)
+DEV_INLINE void stage_write_kc_natural(bf16_t* dst, const bf16x8 regs[2]) {
 #pragma unroll
   for (int u = 0; u < 2; ++u) {
-    int idx = threadIdx.x + u * NTHREADS;
-    int row = idx >> 2;
-    int kc = (idx & 3) * 8;
+    const int idx = threadIdx.x + u * NTHREADS;
+    const int row = idx >> 2;
+    const int kc = (idx & 3) * 8;
     *reinterpret_cast<bf16x8*>(dst + row * LDS_ROW + kc) = regs[u];
   }
 }
 
+// pi-permuted columns (KC operand mixed with a tr-read operand): global k
+// run [8j,8j+8) splits into two b64 writes at cols 16*(j%2)+4*(j/2) and +8.
+DEV_INLINE void stage_write_kc_pi(bf16_t* dst, const bf16x8 regs[2]) {
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NTHREADS;
+    const int row = idx >> 2;
+    const int j = idx & 3;
+    const int col0 = 16 * (j & 1) + 4 * (j >> 1);
+    const bf16x8 v = regs[u];
+    bf16x4 lo = {v[0], v[1], v[2], v[3]};
+    bf16x4 hi = {v[4], v[5], v[6], v[7]};
+    *reinterpret_cast<bf16x4*>(dst + row * LDS_ROW + col0) = lo;
+    *reinterpret_cast<bf16x4*>(dst + row * LDS_ROW + col0 + 8) = hi;
+  }
+}
+
+// KO operand into the [BK/4][F/16][4][16]-subtiled image (128B subtiles).
 DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x8 regs[2]) {
-  int f = threadIdx.x & 127;
-  int kh = threadIdx.x >> 7;
-  *reinterpret_cast<bf16x8*>(dst + f * LDS_ROW + kh * 16) = regs[0];
-  *reinterpret_cast<bf16x8*>(dst + f * LDS_ROW + kh * 16 + 8) = regs[1];
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NTHREADS;
+    const int k = idx >> 4;
+    const int f8 = (idx & 15) * 8;
+    const int off = ((k >> 2) * (BM / 16) + (f8 >> 4)) * 64 + (k & 3) * 16 +
+                    (f8 & 15);
+    *reinterpret_cast<bf16x8*>(dst + off) = regs[u];
+  }
+}
+
+// --- fragment loads --------------------------------------------------------
+
+// KC image (natural or pi — content differs, read pattern identical)
+DEV_INLINE bf16x8 frag_kc(const bf16_t* s, int fbase, int lane) {
+  return *reinterpret_cast<const bf16x8*>(
+      s + (fbase + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
+}
+
+// KO image via two hardware transpose-reads; k order = pi(g,e).
+DEV_INLINE bf16x8 frag_ko(const bf16_t* s, int fbase, int lane) {
+  const int g = lane >> 4;
+  const int fsub = fbase >> 4;
+  const bf16_t* p0 = s + (g * (BM / 16) + fsub) * 64 + (lane & 15);
+  const bf16_t* p1 = s + ((g + 4) * (BM / 16) + fsub) * 64 + (lane & 15);
+  const bf16x4 lo = tr_read(p0);
+  const bf16x4 hi = tr_read(p1);
+  return bf16x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
 }
 
 template <bool A_KC, bool B_KC, int EPI>
@@ -103,6 +166,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
                  const bf16_t* __restrict__ bias, int M, int N, int K,
                  int lda, int ldb, int ldc, int64_t strideA, int64_t strideB,
                  int64_t strideC) {
+  constexpr bool NATURAL = A_KC && B_KC;
   A += blockIdx.z * strideA;
   B += blockIdx.z * strideB;
   C += blockIdx.z * strideC;
@@ -110,49 +174,56 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
 
   const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
 
-  __shared__ bf16_t smem[2][2][BM * LDS_ROW];  // [buf][A=0/B=1]
+  __shared__ bf16_t smem[2][2][SLOT];  // [buf][A=0/B=1]
 
   const int lane = threadIdx.x & 63;
   const int wm = (threadIdx.x >> 7) * 64;        // wave row (wid>>1)*64
   const int wn = ((threadIdx.x >> 6) & 1) * 64;  // wave col (wid&1)*64
 
-  const bool a_al = A_KC && ((lda & 7) == 0);
-  const bool b_al = B_KC && ((ldb & 7) == 0);
+  const bool a_al = (lda & 7) == 0;
+  const bool b_al = (ldb & 7) == 0;
 
   f32x4 acc[4][4] = {};
 
   const int nk = (K + BK - 1) / BK;
   bf16x8 ra[2], rb[2];
 
-  // prologue: stage tile 0
-  if (A_KC) stage_kc_load(A, M, K, lda, m0, 0, a_al, ra);
-  else      stage_ko_load(A, M, K, lda, m0, 0, ra);
-  if (B_KC) stage_kc_load(B, N, K, ldb, n0, 0, b_al, rb);
-  else      stage_ko_load(B, N, K, ldb, n0, 0, rb);
-  if (A_KC) stage_write_kc(smem[0][0], ra); else stage_write_ko(smem[0][0], ra);
-  if (B_KC) stage_write_kc(smem[0][1], rb); else stage_write_ko(smem[0][1], rb);
+  auto load_tiles = [&](int k0) {
+    if (A_KC) stage_kc_load(A, M, K, lda, m0, k0, a_al, ra);
+    else      stage_ko_load(A, M, K, lda, m0, k0, a_al, ra);
+    if (B_KC) stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb);
+    else      stage_ko_load(B, N, K, ldb, n0, k0, b_al, rb);
+  };
+  auto write_tiles = [&](int buf) {
+    if (A_KC) {
+      if (NATURAL) stage_write_kc_natural(smem[buf][0], ra);
+      else         stage_write_kc_pi(smem[buf][0], ra);
+    } else {
+      stage_write_ko(smem[buf][0], ra);
+    }
+    if (B_KC) {
+      if (NATURAL) stage_write_kc_natural(smem[buf][1], rb);
+      else         stage_write_kc_pi(smem[buf][1], rb);
+    } else {
+      stage_write_ko(smem[buf][1], rb);
+    }
+  };
+
+  load_tiles(0);
+  write_tiles(0);
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     const int cur = t & 1;
-    // issue next tile's global loads early (latency hides under the MFMAs)
-    if (t + 1 < nk) {
-      const int k0 = (t + 1) * BK;
-      if (A_KC) stage_kc_load(A, M, K, lda, m0, k0, a_al, ra);
-      else      stage_ko_load(A, M, K, lda, m0, k0, ra);
-      if (B_KC) stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb);
-      else      stage_ko_load(B, N, K, ldb, n0, k0, rb);
-    }
+    if (t + 1 < nk) load_tiles((t + 1) * BK);  // issue early (T14)
 
     const bf16_t* sa = smem[cur][0];
     const bf16_t* sb = smem[cur][1];
     bf16x8 af[4], bfr[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      af[i] = *reinterpret_cast<const bf16x8*>(
-          sa + (wm + i * 16 + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
-      bfr[i] = *reinterpret_cast<const bf16x8*>(
-          sb + (wn + i * 16 + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
+      af[i] = A_KC ? frag_kc(sa, wm + i * 16, lane) : frag_ko(sa, wm + i * 16, lane);
+      bfr[i] = B_KC ? frag_kc(sb, wn + i * 16, lane) : frag_ko(sb, wn + i * 16, lane);
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
@@ -161,13 +232,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
 
-    if (t + 1 < nk) {
-      const int nxt = cur ^ 1;
-      if (A_KC) stage_write_kc(smem[nxt][0], ra);
-      else      stage_write_ko(smem[nxt][0], ra);
-      if (B_KC) stage_write_kc(smem[nxt][1], rb);
-      else      stage_write_ko(smem[nxt][1], rb);
-    }
+    if (t + 1 < nk) write_tiles(cur ^ 1);
     __syncthreads();
   }
 
@@ -177,7 +242,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) {
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
-      int n = n0 + wn + ni * 16 + (lane & 15);
+      const int n = n0 + wn + ni * 16 + (lane & 15);
       bv[ni] = (n < N) ? bf2f(bias[n]) : 0.0f;
     }
   }
@@ -196,7 +261,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
         const int64_t off = (int64_t)m * ldc + n;
         if (EPI >= 2) {
           // round pre-act to bf16 first (matches ops/reference.py semantics)
-          bf16_t pre = f2bf(v);
+          const bf16_t pre = f2bf(v);
           Cpre[off] = pre;
           v = gelu_f(bf2f(pre));
         }

@@ -78,4 +78,7 @@ void adamw_bf16(void* param, float* master, const void* grad_bf16,
                 int64_t n, float lr, float beta1, float beta2, float eps,
                 float weight_decay, float bc1, float bc2, hipStream_t stream);
 
+// --- diagnostics -----------------------------------------------------------
+void tr16_probe(float* out_pattern, float* out_uniform, hipStream_t stream);
+
 }  // namespace tepdist

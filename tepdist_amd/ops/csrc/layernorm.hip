@@ -183,20 +183,26 @@ void ln_bwd_kernel(const bf16_t* __restrict__ dy, const bf16_t* __restrict__ x,
     }
 }
 
+// one wave per column: lanes stride the partial rows, wave-reduce.
 __global__ void ln_bwd_reduce_kernel(const float* __restrict__ dg_part,
                                      const float* __restrict__ db_part,
                                      bf16_t* __restrict__ dgamma,
                                      bf16_t* __restrict__ dbeta,
                                      int part_rows, int cols) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int lane = threadIdx.x & 63;
+  const int c = blockIdx.x * (NT / WAVE) + (threadIdx.x >> 6);
   if (c >= cols) return;
   float sg = 0.f, sb = 0.f;
-  for (int r = 0; r < part_rows; ++r) {
+  for (int r = lane; r < part_rows; r += WAVE) {
     sg += dg_part[(int64_t)r * cols + c];
     sb += db_part[(int64_t)r * cols + c];
   }
-  dgamma[c] = f2bf(sg);
-  dbeta[c] = f2bf(sb);
+  sg = wave_allreduce_sum(sg);
+  sb = wave_allreduce_sum(sb);
+  if (lane == 0) {
+    dgamma[c] = f2bf(sg);
+    dbeta[c] = f2bf(sb);
+  }
 }
 
 int nv_for(int cols) { return (cols + WAVE * 8 - 1) / (WAVE * 8); }
@@ -255,7 +261,7 @@ void layernorm_bwd_bf16(const void* dy, const void* x, const void* gamma,
 void layernorm_bwd_reduce(const float* dgamma_part, const float* dbeta_part,
                           void* dgamma, void* dbeta, int part_rows, int cols,
                           hipStream_t stream) {
-  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((cols + 255) / 256), dim3(256),
+  hipLaunchKernelGGL(ln_bwd_reduce_kernel, dim3((cols + 3) / 4), dim3(NT),
                      0, stream, dgamma_part, dbeta_part,
                      static_cast<bf16_t*>(dgamma), static_cast<bf16_t*>(dbeta),
                      part_rows, cols);

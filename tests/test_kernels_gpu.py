@@ -252,3 +252,21 @@ def test_gpt2_tiny_gpu_trains():
     losses = [trainer.train_step(lambda i: batch) for _ in range(12)]
     assert losses[-1] < losses[0] * 0.9, losses
     assert all(math.isfinite(l) for l in losses)
+
+
+def test_tr16_probe_mapping():
+    """Nails the exact lane->element mapping of ds_read_b64_tr_b16 (the GEMM
+    transposed-operand fragment loads depend on it). On failure the assert
+    message prints the actual mapping."""
+    from tepdist_amd.ops import _tepdist_hip as ext
+    pat = torch.zeros(256, device="cuda")
+    uni = torch.zeros(256, device="cuda")
+    ext.tr16_probe(pat.data_ptr(), uni.data_ptr(),
+                   torch.cuda.current_stream().cuda_stream)
+    torch.cuda.synchronize()
+    pat = pat.reshape(64, 4).cpu().int()
+    expect = torch.tensor([[(l & 15) + j * 16 + (l >> 4) * 64
+                            for j in range(4)] for l in range(64)],
+                          dtype=torch.int32)
+    assert torch.equal(pat, expect), \
+        f"tr16 actual mapping:\n{pat}\nuniform-addr result:\n{uni.reshape(64,4).cpu().int()}"
