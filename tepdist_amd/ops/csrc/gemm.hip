@@ -127,16 +127,31 @@ DEV_INLINE void stage_write_kc_pi(bf16_t* dst, const bf16x8 regs[2]) {
   }
 }
 
-// KO operand into the [BK/4][F/16][4][16]-subtiled image (128B subtiles).
+// KO operand image: [F/4 panels][BK rows][4 f] — element (k,f) at
+// panel=f/4, offset (panel*BK + k')*4 + f%4 where k' swizzles 4-row blocks
+// (k' = (k/4 ^ (panel&7))*4 + k%4) to spread panels over LDS banks.
+// ds_read_b64_tr_b16 semantics (measured, tests/test_kernels_gpu.py
+// tr16 probe): each QUAD of lanes reads a 4x4 bf16 block at the quad's
+// address with 8-byte row stride; lane i of the quad receives column i.
+// A panel's 4-row block is exactly such a block (rows 8B apart), so the
+// quad read returns 4 k's at a fixed f with zero staging transpose.
+DEV_INLINE int ko_elem_off(int k, int f) {
+  const int panel = f >> 2;
+  const int kb = ((k >> 2) ^ (panel & 7)) << 2;
+  return (panel * BK + kb + (k & 3)) * 4 + (f & 3);
+}
+
 DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x8 regs[2]) {
 #pragma unroll
   for (int u = 0; u < 2; ++u) {
     const int idx = threadIdx.x + u * NTHREADS;
     const int k = idx >> 4;
     const int f8 = (idx & 15) * 8;
-    const int off = ((k >> 2) * (BM / 16) + (f8 >> 4)) * 64 + (k & 3) * 16 +
-                    (f8 & 15);
-    *reinterpret_cast<bf16x8*>(dst + off) = regs[u];
+    const bf16x8 v = regs[u];
+    const bf16x4 lo = {v[0], v[1], v[2], v[3]};
+    const bf16x4 hi = {v[4], v[5], v[6], v[7]};
+    *reinterpret_cast<bf16x4*>(dst + ko_elem_off(k, f8)) = lo;
+    *reinterpret_cast<bf16x4*>(dst + ko_elem_off(k, f8 + 4)) = hi;
   }
 }
 
@@ -148,12 +163,16 @@ DEV_INLINE bf16x8 frag_kc(const bf16_t* s, int fbase, int lane) {
       s + (fbase + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
 }
 
-// KO image via two hardware transpose-reads; k order = pi(g,e).
+// KO image via two quad transpose-reads; k order = pi(g,e): read 1 gives
+// k = 4g+j (logical block g), read 2 gives k = 16+4g+j (block 4+g).
 DEV_INLINE bf16x8 frag_ko(const bf16_t* s, int fbase, int lane) {
   const int g = lane >> 4;
-  const int fsub = fbase >> 4;
-  const bf16_t* p0 = s + (g * (BM / 16) + fsub) * 64 + (lane & 15);
-  const bf16_t* p1 = s + ((g + 4) * (BM / 16) + fsub) * 64 + (lane & 15);
+  const int i = lane & 3;                    // column within the quad
+  const int panel = (fbase >> 2) + ((lane >> 2) & 3);
+  const int kb1 = (g ^ (panel & 7)) << 2;
+  const int kb2 = ((4 + g) ^ (panel & 7)) << 2;
+  const bf16_t* p0 = s + (panel * BK + kb1) * 4 + i;
+  const bf16_t* p1 = s + (panel * BK + kb2) * 4 + i;
   const bf16x4 lo = tr_read(p0);
   const bf16x4 hi = tr_read(p1);
   return bf16x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};

@@ -255,9 +255,11 @@ def test_gpt2_tiny_gpu_trains():
 
 
 def test_tr16_probe_mapping():
-    """Nails the exact lane->element mapping of ds_read_b64_tr_b16 (the GEMM
-    transposed-operand fragment loads depend on it). On failure the assert
-    message prints the actual mapping."""
+    """Verifies the measured semantics of ds_read_b64_tr_b16: each QUAD of
+    lanes reads a 4x4 bf16 block at the quad's address (8-byte row stride);
+    lane i of the quad receives column i. The GEMM transposed-operand
+    fragment loads depend on exactly this. On failure the assert message
+    prints the actual mapping."""
     from tepdist_amd.ops import _tepdist_hip as ext
     pat = torch.zeros(256, device="cuda")
     uni = torch.zeros(256, device="cuda")
@@ -265,8 +267,10 @@ def test_tr16_probe_mapping():
                    torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     pat = pat.reshape(64, 4).cpu().int()
-    expect = torch.tensor([[(l & 15) + j * 16 + (l >> 4) * 64
-                            for j in range(4)] for l in range(64)],
-                          dtype=torch.int32)
+    # probe addresses: lane l at element (l&15) + (l>>4)*64; quad leader's
+    # address anchors the 4x4 block; lane l%4 gets column l%4, rows +4j.
+    expect = torch.tensor(
+        [[(4 * (l // 4)) % 16 + (l // 16) * 64 + (l % 4) + 4 * j
+          for j in range(4)] for l in range(64)], dtype=torch.int32)
     assert torch.equal(pat, expect), \
-        f"tr16 actual mapping:\n{pat}\nuniform-addr result:\n{uni.reshape(64,4).cpu().int()}"
+        f"tr16 actual mapping:\n{pat}\nuniform-addr result:\n{uni.reshape(64, 4).cpu().int()}"
