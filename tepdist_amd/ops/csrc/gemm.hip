@@ -4,19 +4,20 @@
 // per wave as 4x4 fragments of 16x16), double-buffered LDS with the
 // load-early/write-late split (guide T14), mfma_f32_16x16x32_bf16 inner loop.
 //
-// Layout handling without transpose cost: the MFMA contraction is invariant
-// under any k-permutation applied consistently to both operands' fragments
-// (sum over (g,e) of A[m][pi(g,e)]*B[pi(g,e)][n] == sum over k). So:
+// Layout handling: both operands stage into the canonical k-contiguous
+// row-padded [F][BK] LDS image (ds_read_b128 fragment reads, one row per
+// lane in a 16-lane group; the +16B row pad keeps them bank-conflict-free).
 //  - "KC" operands (k fastest-varying in memory: A stored [M,K], B stored
-//    [N,K]) stage into a row-padded [F][BK] image with ds_read_b128 fragment
-//    reads (one row per lane in a 16-lane group; +16B row pad keeps it
-//    bank-conflict-free).
-//  - "KO" operands (k outermost: stored [K,F], i.e. dgrad's weight and both
-//    wgrad operands) stage UNtransposed into a [BK/4][BM/16][4][16]-subtiled
-//    image using only 16B loads + 16B LDS writes, and fragments are read with
-//    the gfx950 hardware transpose-read ds_read_b64_tr_b16 (guide T10).
-//  When layouts mix, the KC operand's staging permutes columns so both sides
-//  share the tr-read's k order pi(g,e) = e<4 ? 4g+e : 16+4g+(e-4).
+//    [N,K]): straight 16B loads + 16B LDS writes.
+//  - "KO" operands (k outermost: stored [K,F] — dgrad's weight and both
+//    wgrad operands): each thread loads an 8(k) x 2(f) slab with eight 4B
+//    loads (f-coalesced across the wave), transposes it in registers with
+//    eight v_perm_b32, and writes two k-contiguous 16B rows. No scalar LDS
+//    traffic. (ds_read_b64_tr_b16 was measured — see debug.hip tr16_probe —
+//    to deliver only 16 distinct values per 16-lane group: each quad reads
+//    the (lane&3)-th bf16 of 4 rows based at the group's quad-LEADER
+//    addresses, so it cannot feed a 16x16 MFMA fragment; the register
+//    transpose path is the fast one.)
 //
 // Replaces the reference's XLA-codegen GEMMs (all device code
 // in TePDist is XLA-generated PTX; here it is hand-written CDNA4).
@@ -38,12 +39,6 @@ constexpr int SLOT = BM * LDS_ROW;  // elements per operand slot (KO uses 4K)
 constexpr int NTHREADS = 256;
 
 enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3 };
-
-typedef __attribute__((address_space(3))) bf16x4* lds_tr_ptr;
-
-DEV_INLINE bf16x4 tr_read(const bf16_t* p) {
-  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_tr_ptr)p);
-}
 
 // --- staging loads ---------------------------------------------------------
 
@@ -72,28 +67,27 @@ DEV_INLINE void stage_kc_load(const bf16_t* __restrict__ src, int F, int K,
   }
 }
 
-// KO operand (stored [K][ld], f inner): 2 x 16B per thread, no transpose.
+// KO operand (stored [K][ld], f inner): thread t owns the 8(k) x 2(f) slab
+// at f = 2*(t&63), k-block = t>>6; eight bf16x2 loads, f-coalesced.
 DEV_INLINE void stage_ko_load(const bf16_t* __restrict__ src, int F, int K,
                               int ld, int f0, int k0, bool aligned,
-                              bf16x8 regs[2]) {
+                              bf16x2 regs[8]) {
+  const int f = f0 + 2 * (threadIdx.x & 63);
+  const int kt = k0 + 8 * (threadIdx.x >> 6);
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    const int idx = threadIdx.x + u * NTHREADS;  // 0..511
-    const int k = idx >> 4;                      // 0..31
-    const int f8 = (idx & 15) * 8;               // 0..120
-    bf16x8 v = {};
-    const int kg = k0 + k, fg = f0 + f8;
-    if (kg < K && fg < F) {
-      const bf16_t* p = src + (int64_t)kg * ld + fg;
-      if (aligned && fg + 8 <= F) {
-        v = *reinterpret_cast<const bf16x8*>(p);
+  for (int j = 0; j < 8; ++j) {
+    const int k = kt + j;
+    bf16x2 v = {};
+    if (k < K && f < F) {
+      const bf16_t* p = src + (int64_t)k * ld + f;
+      if (aligned && f + 2 <= F) {
+        v = *reinterpret_cast<const bf16x2*>(p);
       } else {
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          if (fg + e < F) v[e] = p[e];
+        v[0] = p[0];
+        if (f + 1 < F) v[1] = p[1];
       }
     }
-    regs[u] = v;
+    regs[j] = v;
   }
 }
 
@@ -110,72 +104,33 @@ DEV_INLINE void stage_write_kc_natural(bf16_t* dst, const bf16x8 regs[2]) {
   }
 }
 
-// pi-permuted columns (KC operand mixed with a tr-read operand): global k
-// run [8j,8j+8) splits into two b64 writes at cols 16*(j%2)+4*(j/2) and +8.
-DEV_INLINE void stage_write_kc_pi(bf16_t* dst, const bf16x8 regs[2]) {
+// KO slab write: transpose 8x2 in registers (v_perm_b32), two b128 writes.
+DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x2 regs[8]) {
+  const int f = 2 * (threadIdx.x & 63);
+  const int kc = 8 * (threadIdx.x >> 6);
+  uint32_t r[8];
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    const int idx = threadIdx.x + u * NTHREADS;
-    const int row = idx >> 2;
-    const int j = idx & 3;
-    const int col0 = 16 * (j & 1) + 4 * (j >> 1);
-    const bf16x8 v = regs[u];
-    bf16x4 lo = {v[0], v[1], v[2], v[3]};
-    bf16x4 hi = {v[4], v[5], v[6], v[7]};
-    *reinterpret_cast<bf16x4*>(dst + row * LDS_ROW + col0) = lo;
-    *reinterpret_cast<bf16x4*>(dst + row * LDS_ROW + col0 + 8) = hi;
-  }
-}
-
-// KO operand image: [F/4 panels][BK rows][4 f] — element (k,f) at
-// panel=f/4, offset (panel*BK + k')*4 + f%4 where k' swizzles 4-row blocks
-// (k' = (k/4 ^ (panel&7))*4 + k%4) to spread panels over LDS banks.
-// ds_read_b64_tr_b16 semantics (measured, tests/test_kernels_gpu.py
-// tr16 probe): each QUAD of lanes reads a 4x4 bf16 block at the quad's
-// address with 8-byte row stride; lane i of the quad receives column i.
-// A panel's 4-row block is exactly such a block (rows 8B apart), so the
-// quad read returns 4 k's at a fixed f with zero staging transpose.
-DEV_INLINE int ko_elem_off(int k, int f) {
-  const int panel = f >> 2;
-  const int kb = ((k >> 2) ^ (panel & 7)) << 2;
-  return (panel * BK + kb + (k & 3)) * 4 + (f & 3);
-}
-
-DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x8 regs[2]) {
+  for (int j = 0; j < 8; ++j)
+    r[j] = __builtin_bit_cast(uint32_t, regs[j]);
+  uint32_t o0[4], o1[4];
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    const int idx = threadIdx.x + u * NTHREADS;
-    const int k = idx >> 4;
-    const int f8 = (idx & 15) * 8;
-    const bf16x8 v = regs[u];
-    const bf16x4 lo = {v[0], v[1], v[2], v[3]};
-    const bf16x4 hi = {v[4], v[5], v[6], v[7]};
-    *reinterpret_cast<bf16x4*>(dst + ko_elem_off(k, f8)) = lo;
-    *reinterpret_cast<bf16x4*>(dst + ko_elem_off(k, f8 + 4)) = hi;
+  for (int d = 0; d < 4; ++d) {
+    // result bytes [A0,A1,B0,B1] with A=r[2d] (src1=bytes 0-3), B=r[2d+1]
+    o0[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x05040100u);
+    o1[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x07060302u);
   }
+  *reinterpret_cast<uint4*>(dst + f * LDS_ROW + kc) =
+      make_uint4(o0[0], o0[1], o0[2], o0[3]);
+  *reinterpret_cast<uint4*>(dst + (f + 1) * LDS_ROW + kc) =
+      make_uint4(o1[0], o1[1], o1[2], o1[3]);
 }
 
 // --- fragment loads --------------------------------------------------------
 
-// KC image (natural or pi — content differs, read pattern identical)
-DEV_INLINE bf16x8 frag_kc(const bf16_t* s, int fbase, int lane) {
+// fragment read from the canonical [F][LDS_ROW] image
+DEV_INLINE bf16x8 frag(const bf16_t* s, int fbase, int lane) {
   return *reinterpret_cast<const bf16x8*>(
       s + (fbase + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
-}
-
-// KO image via two quad transpose-reads; k order = pi(g,e): read 1 gives
-// k = 4g+j (logical block g), read 2 gives k = 16+4g+j (block 4+g).
-DEV_INLINE bf16x8 frag_ko(const bf16_t* s, int fbase, int lane) {
-  const int g = lane >> 4;
-  const int i = lane & 3;                    // column within the quad
-  const int panel = (fbase >> 2) + ((lane >> 2) & 3);
-  const int kb1 = (g ^ (panel & 7)) << 2;
-  const int kb2 = ((4 + g) ^ (panel & 7)) << 2;
-  const bf16_t* p0 = s + (panel * BK + kb1) * 4 + i;
-  const bf16_t* p1 = s + (panel * BK + kb2) * 4 + i;
-  const bf16x4 lo = tr_read(p0);
-  const bf16x4 hi = tr_read(p1);
-  return bf16x8{lo[0], lo[1], lo[2], lo[3], hi[0], hi[1], hi[2], hi[3]};
 }
 
 template <bool A_KC, bool B_KC, int EPI>
@@ -185,7 +140,6 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
                  const bf16_t* __restrict__ bias, int M, int N, int K,
                  int lda, int ldb, int ldc, int64_t strideA, int64_t strideB,
                  int64_t strideC) {
-  constexpr bool NATURAL = A_KC && B_KC;
   A += blockIdx.z * strideA;
   B += blockIdx.z * strideB;
   C += blockIdx.z * strideC;
@@ -199,33 +153,28 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   const int wm = (threadIdx.x >> 7) * 64;        // wave row (wid>>1)*64
   const int wn = ((threadIdx.x >> 6) & 1) * 64;  // wave col (wid&1)*64
 
-  const bool a_al = (lda & 7) == 0;
+  const bool a_al = (lda & 7) == 0;   // 16B loads (KC staging)
   const bool b_al = (ldb & 7) == 0;
+  const bool a_al2 = (lda & 1) == 0;  // 4B loads (KO staging)
+  const bool b_al2 = (ldb & 1) == 0;
 
   f32x4 acc[4][4] = {};
 
   const int nk = (K + BK - 1) / BK;
-  bf16x8 ra[2], rb[2];
+  bf16x8 ra_kc[2], rb_kc[2];
+  bf16x2 ra_ko[8], rb_ko[8];
 
   auto load_tiles = [&](int k0) {
-    if (A_KC) stage_kc_load(A, M, K, lda, m0, k0, a_al, ra);
-    else      stage_ko_load(A, M, K, lda, m0, k0, a_al, ra);
-    if (B_KC) stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb);
-    else      stage_ko_load(B, N, K, ldb, n0, k0, b_al, rb);
+    if (A_KC) stage_kc_load(A, M, K, lda, m0, k0, a_al, ra_kc);
+    else      stage_ko_load(A, M, K, lda, m0, k0, a_al2, ra_ko);
+    if (B_KC) stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb_kc);
+    else      stage_ko_load(B, N, K, ldb, n0, k0, b_al2, rb_ko);
   };
   auto write_tiles = [&](int buf) {
-    if (A_KC) {
-      if (NATURAL) stage_write_kc_natural(smem[buf][0], ra);
-      else         stage_write_kc_pi(smem[buf][0], ra);
-    } else {
-      stage_write_ko(smem[buf][0], ra);
-    }
-    if (B_KC) {
-      if (NATURAL) stage_write_kc_natural(smem[buf][1], rb);
-      else         stage_write_kc_pi(smem[buf][1], rb);
-    } else {
-      stage_write_ko(smem[buf][1], rb);
-    }
+    if (A_KC) stage_write_kc_natural(smem[buf][0], ra_kc);
+    else      stage_write_ko(smem[buf][0], ra_ko);
+    if (B_KC) stage_write_kc_natural(smem[buf][1], rb_kc);
+    else      stage_write_ko(smem[buf][1], rb_ko);
   };
 
   load_tiles(0);
@@ -241,8 +190,8 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16x8 af[4], bfr[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-      af[i] = A_KC ? frag_kc(sa, wm + i * 16, lane) : frag_ko(sa, wm + i * 16, lane);
-      bfr[i] = B_KC ? frag_kc(sb, wn + i * 16, lane) : frag_ko(sb, wn + i * 16, lane);
+      af[i] = frag(sa, wm + i * 16, lane);
+      bfr[i] = frag(sb, wn + i * 16, lane);
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)

@@ -255,11 +255,12 @@ def test_gpt2_tiny_gpu_trains():
 
 
 def test_tr16_probe_mapping():
-    """Verifies the measured semantics of ds_read_b64_tr_b16: each QUAD of
-    lanes reads a 4x4 bf16 block at the quad's address (8-byte row stride);
-    lane i of the quad receives column i. The GEMM transposed-operand
-    fragment loads depend on exactly this. On failure the assert message
-    prints the actual mapping."""
+    """Documents the MEASURED semantics of ds_read_b64_tr_b16 on gfx950:
+    within each 16-lane group, the four QUAD LEADERS' addresses (lanes
+    16g+4j) define 4 row bases; lane l reads the (l&3)-th bf16 of each row
+    (so data is duplicated across the 4 quads of a group). This is why the
+    GEMM's transposed-operand path uses a v_perm register transpose instead
+    (see gemm.hip header)."""
     from tepdist_amd.ops import _tepdist_hip as ext
     pat = torch.zeros(256, device="cuda")
     uni = torch.zeros(256, device="cuda")
@@ -267,10 +268,12 @@ def test_tr16_probe_mapping():
                    torch.cuda.current_stream().cuda_stream)
     torch.cuda.synchronize()
     pat = pat.reshape(64, 4).cpu().int()
-    # probe addresses: lane l at element (l&15) + (l>>4)*64; quad leader's
-    # address anchors the 4x4 block; lane l%4 gets column l%4, rows +4j.
-    expect = torch.tensor(
-        [[(4 * (l // 4)) % 16 + (l // 16) * 64 + (l % 4) + 4 * j
-          for j in range(4)] for l in range(64)], dtype=torch.int32)
-    assert torch.equal(pat, expect), \
-        f"tr16 actual mapping:\n{pat}\nuniform-addr result:\n{uni.reshape(64, 4).cpu().int()}"
+    uni = uni.reshape(64, 4).cpu().int()
+    # probe addresses: lane l at element (l&15) + (l>>4)*64
+    expect_pat = torch.tensor(
+        [[(l >> 4) * 64 + 4 * j + (l & 3) for j in range(4)]
+         for l in range(64)], dtype=torch.int32)
+    expect_uni = torch.tensor(
+        [[l & 3 for _ in range(4)] for l in range(64)], dtype=torch.int32)
+    assert torch.equal(pat, expect_pat), f"tr16 pattern mapping:\n{pat}"
+    assert torch.equal(uni, expect_uni), f"tr16 uniform mapping:\n{uni}"
