@@ -145,7 +145,15 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   C += blockIdx.z * strideC;
   if (EPI >= 2) Cpre += blockIdx.z * strideC;
 
-  const int m0 = blockIdx.y * BM, n0 = blockIdx.x * BN;
+  // XCD-aware bijective block swizzle (guide T1): the dispatcher places
+  // block b on XCD b%8; remap so each XCD gets a CONTIGUOUS run of tiles
+  // (consecutive n-tiles share the A panel -> L2 hits stay on one XCD).
+  const int nbx = (N + BN - 1) / BN;
+  const int nwg = gridDim.x;
+  const int q = nwg >> 3, r = nwg & 7;
+  const int xcd = blockIdx.x & 7, idx = blockIdx.x >> 3;
+  const int swz = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  const int m0 = (swz / nbx) * BM, n0 = (swz % nbx) * BN;
 
   __shared__ bf16_t smem[2][2][SLOT];  // [buf][A=0/B=1]
 
@@ -245,7 +253,8 @@ void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
                const void* bias, int M, int N, int K, int lda, int ldb,
                int ldc, int64_t stride_a, int64_t stride_b, int64_t stride_c,
                int batch, bool a_kc, bool b_kc, int epi, hipStream_t stream) {
-  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM, batch);
+  const int nbx = (N + BN - 1) / BN, nby = (M + BM - 1) / BM;
+  dim3 grid(nbx * nby, 1, batch);
   dim3 block(NTHREADS);
   const bf16_t* a = static_cast<const bf16_t*>(A);
   const bf16_t* b = static_cast<const bf16_t*>(B);
