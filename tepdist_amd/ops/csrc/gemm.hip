@@ -34,9 +34,18 @@ namespace tepdist {
 namespace {
 
 constexpr int BM = 128, BN = 128, BK = 32;
-constexpr int LDS_ROW = BK + 8;  // +8 bf16 = +16B row pad: conflict-free b128
-constexpr int SLOT = BM * LDS_ROW;  // elements per operand slot (KO uses 4K)
+constexpr int SLOT = BM * BK;  // elements per operand slot
 constexpr int NTHREADS = 256;
+
+// LDS addressing: unpadded 64B rows + an XOR swizzle of the 16B column slot
+// by x(row) = ((row>>1) ^ (row>>3)) & 3. Measured (SQ_LDS_BANK_CONFLICT):
+// the padded-row layout cost ~8 extra cycles per ds_read_b128; this swizzle
+// makes the fragment reads conflict-free under both contiguous and
+// quad-strided 16-lane service groupings and leaves writes at <=2-way.
+DEV_INLINE int lds_off(int row, int col_e) {
+  const int x = ((row >> 1) ^ (row >> 3)) & 3;
+  return row * BK + (col_e ^ (x << 3));
+}
 
 enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3 };
 
@@ -100,7 +109,7 @@ DEV_INLINE void stage_write_kc_natural(bf16_t* dst, const bf16x8 regs[2]) {
     const int idx = threadIdx.x + u * NTHREADS;
     const int row = idx >> 2;
     const int kc = (idx & 3) * 8;
-    *reinterpret_cast<bf16x8*>(dst + row * LDS_ROW + kc) = regs[u];
+    *reinterpret_cast<bf16x8*>(dst + lds_off(row, kc)) = regs[u];
   }
 }
 
@@ -119,9 +128,9 @@ DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x2 regs[8]) {
     o0[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x05040100u);
     o1[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x07060302u);
   }
-  *reinterpret_cast<uint4*>(dst + f * LDS_ROW + kc) =
+  *reinterpret_cast<uint4*>(dst + lds_off(f, kc)) =
       make_uint4(o0[0], o0[1], o0[2], o0[3]);
-  *reinterpret_cast<uint4*>(dst + (f + 1) * LDS_ROW + kc) =
+  *reinterpret_cast<uint4*>(dst + lds_off(f + 1, kc)) =
       make_uint4(o1[0], o1[1], o1[2], o1[3]);
 }
 
@@ -130,7 +139,7 @@ DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x2 regs[8]) {
 // fragment read from the canonical [F][LDS_ROW] image
 DEV_INLINE bf16x8 frag(const bf16_t* s, int fbase, int lane) {
   return *reinterpret_cast<const bf16x8*>(
-      s + (fbase + (lane & 15)) * LDS_ROW + 8 * (lane >> 4));
+      s + lds_off(fbase + (lane & 15), 8 * (lane >> 4)));
 }
 
 template <bool A_KC, bool B_KC, int EPI>
