@@ -100,6 +100,32 @@ DEV_INLINE void stage_ko_load(const bf16_t* __restrict__ src, int F, int K,
   }
 }
 
+// KC operand via global_load_lds (direct HBM->LDS DMA, 16B per lane; the
+// guide's step-3 lever). The glds destination is wave-uniform base + lane*16B
+// (lane-linear), so the XOR swizzle moves to the per-lane SOURCE address
+// (rule 21): LDS element (row, c) receives global k = c ^ (x(row)<<3).
+// Only for fully-interior tiles (no bounds handling in the DMA).
+typedef __attribute__((address_space(1))) const void* glds_src_t;
+typedef __attribute__((address_space(3))) void* glds_dst_t;
+
+DEV_INLINE void stage_kc_glds(const bf16_t* __restrict__ src, int ld, int f0,
+                              int k0, bf16_t* dst) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NTHREADS;
+    const int row = idx >> 2;                 // dest row (lane-linear)
+    const int c = (idx & 3) * 8;              // dest column (elements)
+    const int x = ((row >> 1) ^ (row >> 3)) & 3;
+    const int ksrc = k0 + (c ^ (x << 3));     // inverse swizzle on the source
+    const bf16_t* g = src + (int64_t)(f0 + row) * ld + ksrc;
+    bf16_t* l = dst + (16 * wave + 64 * u) * BK;  // wave-uniform base
+    __builtin_amdgcn_global_load_lds((glds_src_t)g, (glds_dst_t)l, 16,
+                                     /*offset=*/0, /*aux=*/0);
+  }
+}
+
 // --- staging writes --------------------------------------------------------
 
 // natural k order (used when both operands are KC)
@@ -181,26 +207,46 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   bf16x8 ra_kc[2], rb_kc[2];
   bf16x2 ra_ko[8], rb_ko[8];
 
-  auto load_tiles = [&](int k0) {
-    if (A_KC) stage_kc_load(A, M, K, lda, m0, k0, a_al, ra_kc);
-    else      stage_ko_load(A, M, K, lda, m0, k0, a_al2, ra_ko);
-    if (B_KC) stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb_kc);
-    else      stage_ko_load(B, N, K, ldb, n0, k0, b_al2, rb_ko);
+  // interior tiles of aligned KC operands go by global_load_lds DMA
+  const bool a_glds = A_KC && a_al && (m0 + BM <= M);
+  const bool b_glds = B_KC && b_al && (n0 + BN <= N);
+
+  auto load_tiles = [&](int k0, int buf) {
+    const bool kin = k0 + BK <= K;
+    if (A_KC) {
+      if (a_glds && kin) stage_kc_glds(A, lda, m0, k0, smem[buf][0]);
+      else               stage_kc_load(A, M, K, lda, m0, k0, a_al, ra_kc);
+    } else {
+      stage_ko_load(A, M, K, lda, m0, k0, a_al2, ra_ko);
+    }
+    if (B_KC) {
+      if (b_glds && kin) stage_kc_glds(B, ldb, n0, k0, smem[buf][1]);
+      else               stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb_kc);
+    } else {
+      stage_ko_load(B, N, K, ldb, n0, k0, b_al2, rb_ko);
+    }
   };
-  auto write_tiles = [&](int buf) {
-    if (A_KC) stage_write_kc_natural(smem[buf][0], ra_kc);
-    else      stage_write_ko(smem[buf][0], ra_ko);
-    if (B_KC) stage_write_kc_natural(smem[buf][1], rb_kc);
-    else      stage_write_ko(smem[buf][1], rb_ko);
+  auto write_tiles = [&](int buf, int k0) {
+    const bool kin = k0 + BK <= K;
+    if (A_KC) {
+      if (!(a_glds && kin)) stage_write_kc_natural(smem[buf][0], ra_kc);
+    } else {
+      stage_write_ko(smem[buf][0], ra_ko);
+    }
+    if (B_KC) {
+      if (!(b_glds && kin)) stage_write_kc_natural(smem[buf][1], rb_kc);
+    } else {
+      stage_write_ko(smem[buf][1], rb_ko);
+    }
   };
 
-  load_tiles(0);
-  write_tiles(0);
+  load_tiles(0, 0);
+  write_tiles(0, 0);
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     const int cur = t & 1;
-    if (t + 1 < nk) load_tiles((t + 1) * BK);  // issue early (T14)
+    if (t + 1 < nk) load_tiles((t + 1) * BK, cur ^ 1);  // issue early (T14)
 
     const bf16_t* sa = smem[cur][0];
     const bf16_t* sb = smem[cur][1];
@@ -217,7 +263,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
 
-    if (t + 1 < nk) write_tiles(cur ^ 1);
+    if (t + 1 < nk) write_tiles(cur ^ 1, (t + 1) * BK);
     __syncthreads();
   }
 
