@@ -33,17 +33,19 @@ namespace tepdist {
 
 namespace {
 
-constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int BM = 128, BN = 128, BK = 64;
 constexpr int SLOT = BM * BK;  // elements per operand slot
 constexpr int NTHREADS = 256;
 
-// LDS addressing: unpadded 64B rows + an XOR swizzle of the 16B column slot
-// by x(row) = ((row>>1) ^ (row>>3)) & 3. Measured (SQ_LDS_BANK_CONFLICT):
-// the padded-row layout cost ~8 extra cycles per ds_read_b128; this swizzle
-// makes the fragment reads conflict-free under both contiguous and
-// quad-strided 16-lane service groupings and leaves writes at <=2-way.
+// LDS addressing: unpadded 128B rows + an XOR swizzle of the 16B column
+// slot by x(row) = bit-reversed (row>>1)&7. The bit reversal puts the
+// fragment-read service groups' row bit (row bit 1 under quad-strided
+// 16-lane grouping, row bits 1..3 under contiguous grouping) into slot
+// bits that the reads' own column spread does not already cover, making
+// ds_read_b128 column reads conflict-free; writes stay <=2-way.
 DEV_INLINE int lds_off(int row, int col_e) {
-  const int x = ((row >> 1) ^ (row >> 3)) & 3;
+  const int r1 = row >> 1;
+  const int x = ((r1 & 1) << 2) | (r1 & 2) | ((r1 >> 2) & 1);
   return row * BK + (col_e ^ (x << 3));
 }
 
@@ -54,12 +56,12 @@ enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3 };
 // KC operand (stored [F][ld], k inner): 2 x 16B per thread.
 DEV_INLINE void stage_kc_load(const bf16_t* __restrict__ src, int F, int K,
                               int ld, int f0, int k0, bool aligned,
-                              bf16x8 regs[2]) {
+                              bf16x8 regs[4]) {
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
-    const int idx = threadIdx.x + u * NTHREADS;  // 0..511
-    const int row = idx >> 2;                    // 0..127
-    const int kc = (idx & 3) * 8;
+  for (int u = 0; u < 4; ++u) {
+    const int idx = threadIdx.x + u * NTHREADS;  // 0..1023
+    const int row = idx >> 3;                    // 0..127
+    const int kc = (idx & 7) * 8;
     bf16x8 v = {};
     const int f = f0 + row, k = k0 + kc;
     if (f < F && k < K) {
@@ -80,23 +82,26 @@ DEV_INLINE void stage_kc_load(const bf16_t* __restrict__ src, int F, int K,
 // at f = 2*(t&63), k-block = t>>6; eight bf16x2 loads, f-coalesced.
 DEV_INLINE void stage_ko_load(const bf16_t* __restrict__ src, int F, int K,
                               int ld, int f0, int k0, bool aligned,
-                              bf16x2 regs[8]) {
+                              bf16x2 regs[2][8]) {
   const int f = f0 + 2 * (threadIdx.x & 63);
-  const int kt = k0 + 8 * (threadIdx.x >> 6);
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    const int k = kt + j;
-    bf16x2 v = {};
-    if (k < K && f < F) {
-      const bf16_t* p = src + (int64_t)k * ld + f;
-      if (aligned && f + 2 <= F) {
-        v = *reinterpret_cast<const bf16x2*>(p);
-      } else {
-        v[0] = p[0];
-        if (f + 1 < F) v[1] = p[1];
+  for (int h = 0; h < 2; ++h) {
+    const int kt = k0 + 8 * ((threadIdx.x >> 6) + 4 * h);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = kt + j;
+      bf16x2 v = {};
+      if (k < K && f < F) {
+        const bf16_t* p = src + (int64_t)k * ld + f;
+        if (aligned && f + 2 <= F) {
+          v = *reinterpret_cast<const bf16x2*>(p);
+        } else {
+          v[0] = p[0];
+          if (f + 1 < F) v[1] = p[1];
+        }
       }
+      regs[h][j] = v;
     }
-    regs[j] = v;
   }
 }
 
@@ -113,14 +118,15 @@ DEV_INLINE void stage_kc_glds(const bf16_t* __restrict__ src, int ld, int f0,
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
+  for (int u = 0; u < 4; ++u) {
     const int idx = threadIdx.x + u * NTHREADS;
-    const int row = idx >> 2;                 // dest row (lane-linear)
-    const int c = (idx & 3) * 8;              // dest column (elements)
-    const int x = ((row >> 1) ^ (row >> 3)) & 3;
+    const int row = idx >> 3;                 // dest row (lane-linear)
+    const int c = (idx & 7) * 8;              // dest column (elements)
+    const int r1 = row >> 1;
+    const int x = ((r1 & 1) << 2) | (r1 & 2) | ((r1 >> 2) & 1);
     const int ksrc = k0 + (c ^ (x << 3));     // inverse swizzle on the source
     const bf16_t* g = src + (int64_t)(f0 + row) * ld + ksrc;
-    bf16_t* l = dst + (16 * wave + 64 * u) * BK;  // wave-uniform base
+    bf16_t* l = dst + (8 * wave + 32 * u) * BK;  // wave-uniform base
     __builtin_amdgcn_global_load_lds((glds_src_t)g, (glds_dst_t)l, 16,
                                      /*offset=*/0, /*aux=*/0);
   }
@@ -129,43 +135,46 @@ DEV_INLINE void stage_kc_glds(const bf16_t* __restrict__ src, int ld, int f0,
 // --- staging writes --------------------------------------------------------
 
 // natural k order (used when both operands are KC)
-DEV_INLINE void stage_write_kc_natural(bf16_t* dst, const bf16x8 regs[2]) {
+DEV_INLINE void stage_write_kc_natural(bf16_t* dst, const bf16x8 regs[4]) {
 #pragma unroll
-  for (int u = 0; u < 2; ++u) {
+  for (int u = 0; u < 4; ++u) {
     const int idx = threadIdx.x + u * NTHREADS;
-    const int row = idx >> 2;
-    const int kc = (idx & 3) * 8;
+    const int row = idx >> 3;
+    const int kc = (idx & 7) * 8;
     *reinterpret_cast<bf16x8*>(dst + lds_off(row, kc)) = regs[u];
   }
 }
 
 // KO slab write: transpose 8x2 in registers (v_perm_b32), two b128 writes.
-DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x2 regs[8]) {
+DEV_INLINE void stage_write_ko(bf16_t* dst, const bf16x2 regs[2][8]) {
   const int f = 2 * (threadIdx.x & 63);
-  const int kc = 8 * (threadIdx.x >> 6);
-  uint32_t r[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    r[j] = __builtin_bit_cast(uint32_t, regs[j]);
-  uint32_t o0[4], o1[4];
+  for (int h = 0; h < 2; ++h) {
+    const int kc = 8 * ((threadIdx.x >> 6) + 4 * h);
+    uint32_t r[8];
 #pragma unroll
-  for (int d = 0; d < 4; ++d) {
-    // result bytes [A0,A1,B0,B1] with A=r[2d] (src1=bytes 0-3), B=r[2d+1]
-    o0[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x05040100u);
-    o1[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x07060302u);
+    for (int j = 0; j < 8; ++j)
+      r[j] = __builtin_bit_cast(uint32_t, regs[h][j]);
+    uint32_t o0[4], o1[4];
+#pragma unroll
+    for (int d = 0; d < 4; ++d) {
+      // result bytes [A0,A1,B0,B1]: A=r[2d] (src1=bytes 0-3), B=r[2d+1]
+      o0[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x05040100u);
+      o1[d] = __builtin_amdgcn_perm(r[2 * d + 1], r[2 * d], 0x07060302u);
+    }
+    *reinterpret_cast<uint4*>(dst + lds_off(f, kc)) =
+        make_uint4(o0[0], o0[1], o0[2], o0[3]);
+    *reinterpret_cast<uint4*>(dst + lds_off(f + 1, kc)) =
+        make_uint4(o1[0], o1[1], o1[2], o1[3]);
   }
-  *reinterpret_cast<uint4*>(dst + lds_off(f, kc)) =
-      make_uint4(o0[0], o0[1], o0[2], o0[3]);
-  *reinterpret_cast<uint4*>(dst + lds_off(f + 1, kc)) =
-      make_uint4(o1[0], o1[1], o1[2], o1[3]);
 }
 
 // --- fragment loads --------------------------------------------------------
 
 // fragment read from the canonical [F][LDS_ROW] image
-DEV_INLINE bf16x8 frag(const bf16_t* s, int fbase, int lane) {
+DEV_INLINE bf16x8 frag(const bf16_t* s, int fbase, int lane, int kk) {
   return *reinterpret_cast<const bf16x8*>(
-      s + lds_off(fbase + (lane & 15), 8 * (lane >> 4)));
+      s + lds_off(fbase + (lane & 15), 8 * (lane >> 4) + 32 * kk));
 }
 
 template <bool A_KC, bool B_KC, int EPI>
@@ -204,8 +213,8 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   f32x4 acc[4][4] = {};
 
   const int nk = (K + BK - 1) / BK;
-  bf16x8 ra_kc[2], rb_kc[2];
-  bf16x2 ra_ko[8], rb_ko[8];
+  bf16x8 ra_kc[4], rb_kc[4];
+  bf16x2 ra_ko[2][8], rb_ko[2][8];
 
   // interior tiles of aligned KC operands go by global_load_lds DMA
   const bool a_glds = A_KC && a_al && (m0 + BM <= M);
@@ -250,18 +259,21 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
 
     const bf16_t* sa = smem[cur][0];
     const bf16_t* sb = smem[cur][1];
-    bf16x8 af[4], bfr[4];
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      af[i] = frag(sa, wm + i * 16, lane);
-      bfr[i] = frag(sb, wn + i * 16, lane);
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        af[i] = frag(sa, wm + i * 16, lane, kk);
+        bfr[i] = frag(sb, wn + i * 16, lane, kk);
+      }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
     }
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
 
     if (t + 1 < nk) write_tiles(cur ^ 1, (t + 1) * BK);
     __syncthreads();
