@@ -27,11 +27,19 @@ PYBIND11_MODULE(_tepdist_hip, m) {
   m.def("gemm", [](uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t Cpre,
                    uintptr_t bias, int M, int N, int K, int lda, int ldb,
                    int ldc, int64_t sa, int64_t sb, int64_t sc, int batch,
-                   bool a_kc, bool b_kc, int epi, uintptr_t stream) {
+                   bool a_kc, bool b_kc, int epi, int split_k,
+                   uintptr_t stream) {
     gemm_bf16(reinterpret_cast<void*>(A), reinterpret_cast<void*>(B),
               reinterpret_cast<void*>(C), reinterpret_cast<void*>(Cpre),
               reinterpret_cast<void*>(bias), M, N, K, lda, ldb, ldc, sa, sb,
-              sc, batch, a_kc, b_kc, epi, S(stream));
+              sc, batch, a_kc, b_kc, epi, split_k, S(stream));
+    check_launch();
+  });
+
+  m.def("splitk_reduce", [](uintptr_t parts, uintptr_t out, int nparts,
+                            int64_t mn, uintptr_t stream) {
+    splitk_reduce(reinterpret_cast<const float*>(parts),
+                  reinterpret_cast<void*>(out), nparts, mn, S(stream));
     check_launch();
   });
 

@@ -177,17 +177,29 @@ DEV_INLINE bf16x8 frag(const bf16_t* s, int fbase, int lane, int kk) {
       s + lds_off(fbase + (lane & 15), 8 * (lane >> 4) + 32 * kk));
 }
 
-template <bool A_KC, bool B_KC, int EPI>
+// F32OUT: split-K mode — blockIdx.z selects a K-chunk (k_chunk elements)
+// of batch-0 operands and the epilogue writes fp32 partials at
+// C + z*strideC (reduced by splitk_reduce_kernel).
+template <bool A_KC, bool B_KC, int EPI, bool F32OUT = false>
 __launch_bounds__(NTHREADS) __global__
 void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
-                 bf16_t* __restrict__ C, bf16_t* __restrict__ Cpre,
+                 void* __restrict__ Cv, bf16_t* __restrict__ Cpre,
                  const bf16_t* __restrict__ bias, int M, int N, int K,
                  int lda, int ldb, int ldc, int64_t strideA, int64_t strideB,
-                 int64_t strideC) {
-  A += blockIdx.z * strideA;
-  B += blockIdx.z * strideB;
-  C += blockIdx.z * strideC;
-  if (EPI >= 2) Cpre += blockIdx.z * strideC;
+                 int64_t strideC, int k_chunk) {
+  bf16_t* C = static_cast<bf16_t*>(Cv);
+  float* Cf = static_cast<float*>(Cv);
+  int kbeg = 0, kend = K;
+  if (F32OUT) {
+    kbeg = blockIdx.z * k_chunk;
+    kend = min(K, kbeg + k_chunk);
+    Cf += blockIdx.z * strideC;
+  } else {
+    A += blockIdx.z * strideA;
+    B += blockIdx.z * strideB;
+    C += blockIdx.z * strideC;
+    if (EPI >= 2) Cpre += blockIdx.z * strideC;
+  }
 
   // XCD-aware bijective block swizzle (guide T1): the dispatcher places
   // block b on XCD b%8; remap so each XCD gets a CONTIGUOUS run of tiles
@@ -212,7 +224,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
 
   f32x4 acc[4][4] = {};
 
-  const int nk = (K + BK - 1) / BK;
+  const int nk = (kend - kbeg + BK - 1) / BK;
   bf16x8 ra_kc[4], rb_kc[4];
   bf16x2 ra_ko[2][8], rb_ko[2][8];
 
@@ -221,22 +233,22 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   const bool b_glds = B_KC && b_al && (n0 + BN <= N);
 
   auto load_tiles = [&](int k0, int buf) {
-    const bool kin = k0 + BK <= K;
+    const bool kin = k0 + BK <= kend;
     if (A_KC) {
       if (a_glds && kin) stage_kc_glds(A, lda, m0, k0, smem[buf][0]);
-      else               stage_kc_load(A, M, K, lda, m0, k0, a_al, ra_kc);
+      else               stage_kc_load(A, M, kend, lda, m0, k0, a_al, ra_kc);
     } else {
-      stage_ko_load(A, M, K, lda, m0, k0, a_al2, ra_ko);
+      stage_ko_load(A, M, kend, lda, m0, k0, a_al2, ra_ko);
     }
     if (B_KC) {
       if (b_glds && kin) stage_kc_glds(B, ldb, n0, k0, smem[buf][1]);
-      else               stage_kc_load(B, N, K, ldb, n0, k0, b_al, rb_kc);
+      else               stage_kc_load(B, N, kend, ldb, n0, k0, b_al, rb_kc);
     } else {
-      stage_ko_load(B, N, K, ldb, n0, k0, b_al2, rb_ko);
+      stage_ko_load(B, N, kend, ldb, n0, k0, b_al2, rb_ko);
     }
   };
   auto write_tiles = [&](int buf, int k0) {
-    const bool kin = k0 + BK <= K;
+    const bool kin = k0 + BK <= kend;
     if (A_KC) {
       if (!(a_glds && kin)) stage_write_kc_natural(smem[buf][0], ra_kc);
     } else {
@@ -249,13 +261,14 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     }
   };
 
-  load_tiles(0, 0);
-  write_tiles(0, 0);
+  load_tiles(kbeg, 0);
+  write_tiles(0, kbeg);
   __syncthreads();
 
   for (int t = 0; t < nk; ++t) {
     const int cur = t & 1;
-    if (t + 1 < nk) load_tiles((t + 1) * BK, cur ^ 1);  // issue early (T14)
+    if (t + 1 < nk)
+      load_tiles(kbeg + (t + 1) * BK, cur ^ 1);  // issue early (T14)
 
     const bf16_t* sa = smem[cur][0];
     const bf16_t* sb = smem[cur][1];
@@ -275,7 +288,7 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
               af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
     }
 
-    if (t + 1 < nk) write_tiles(cur ^ 1, (t + 1) * BK);
+    if (t + 1 < nk) write_tiles(cur ^ 1, kbeg + (t + 1) * BK);
     __syncthreads();
   }
 
@@ -302,6 +315,10 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
         float v = acc[mi][ni][e];
         if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) v += bv[ni];
         const int64_t off = (int64_t)m * ldc + n;
+        if (F32OUT) {
+          Cf[off] = v;
+          continue;
+        }
         if (EPI >= 2) {
           // round pre-act to bf16 first (matches ops/reference.py semantics)
           const bf16_t pre = f2bf(v);
@@ -314,25 +331,78 @@ void gemm_kernel(const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
   }
 }
 
+__global__ void splitk_reduce_kernel(const float* __restrict__ parts,
+                                     bf16_t* __restrict__ out, int nparts,
+                                     int64_t mn) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i < mn; i += stride) {
+    if (i + 4 <= mn) {
+      f32x4 sv = *reinterpret_cast<const f32x4*>(parts + i);
+      for (int p = 1; p < nparts; ++p) {
+        const f32x4 v = *reinterpret_cast<const f32x4*>(parts + p * mn + i);
+#pragma unroll
+        for (int e = 0; e < 4; ++e) sv[e] += v[e];
+      }
+      bf16x4 o;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) o[e] = f2bf(sv[e]);
+      *reinterpret_cast<bf16x4*>(out + i) = o;
+    } else {
+      for (int64_t j = i; j < mn; ++j) {
+        float acc = 0.f;
+        for (int p = 0; p < nparts; ++p) acc += parts[p * mn + j];
+        out[j] = f2bf(acc);
+      }
+    }
+  }
+}
+
 }  // namespace
+
+void splitk_reduce(const float* parts, void* out, int nparts, int64_t mn,
+                   hipStream_t stream) {
+  const int blocks = (int)std::min<int64_t>(
+      (mn / 4 + NTHREADS - 1) / NTHREADS, 2048);
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(std::max(blocks, 1)),
+                     dim3(NTHREADS), 0, stream, parts,
+                     static_cast<bf16_t*>(out), nparts, mn);
+}
 
 void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
                const void* bias, int M, int N, int K, int lda, int ldb,
                int ldc, int64_t stride_a, int64_t stride_b, int64_t stride_c,
-               int batch, bool a_kc, bool b_kc, int epi, hipStream_t stream) {
+               int batch, bool a_kc, bool b_kc, int epi, int split_k,
+               hipStream_t stream) {
   const int nbx = (N + BN - 1) / BN, nby = (M + BM - 1) / BM;
-  dim3 grid(nbx * nby, 1, batch);
   dim3 block(NTHREADS);
   const bf16_t* a = static_cast<const bf16_t*>(A);
   const bf16_t* b = static_cast<const bf16_t*>(B);
-  bf16_t* c = static_cast<bf16_t*>(C);
   bf16_t* cp = static_cast<bf16_t*>(c_pre);
   const bf16_t* bi = static_cast<const bf16_t*>(bias);
 
+  if (split_k > 1) {
+    if (batch != 1 || epi != 0)
+      throw std::runtime_error("split_k requires batch=1, epi=0");
+    const int k_chunk = ((K + split_k - 1) / split_k + BK - 1) / BK * BK;
+    dim3 gridk(nbx * nby, 1, split_k);
+#define GEMM_SK(AKC, BKC)                                                   \
+    hipLaunchKernelGGL((gemm_kernel<AKC, BKC, 0, true>), gridk, block, 0,   \
+                       stream, a, b, C, cp, bi, M, N, K, lda, ldb, ldc,     \
+                       stride_a, stride_b, stride_c, k_chunk)
+    if (a_kc && b_kc) { GEMM_SK(true, true); }
+    else if (a_kc && !b_kc) { GEMM_SK(true, false); }
+    else if (!a_kc && b_kc) { GEMM_SK(false, true); }
+    else { GEMM_SK(false, false); }
+#undef GEMM_SK
+    return;
+  }
+
+  dim3 grid(nbx * nby, 1, batch);
 #define GEMM_LAUNCH(AKC, BKC, E)                                            \
   hipLaunchKernelGGL((gemm_kernel<AKC, BKC, E>), grid, block, 0, stream, a, \
-                     b, c, cp, bi, M, N, K, lda, ldb, ldc, stride_a,        \
-                     stride_b, stride_c)
+                     b, C, cp, bi, M, N, K, lda, ldb, ldc, stride_a,        \
+                     stride_b, stride_c, 0)
 
 #define GEMM_EPI(AKC, BKC)                         \
   switch (epi) {                                   \

@@ -51,14 +51,26 @@ def _gemm_raw(a: torch.Tensor, b_stored: torch.Tensor, a_kc: bool, b_kc: bool,
               M: int, N: int, K: int, lda: int, ldb: int,
               sa: int, sb: int, batch: int, bias=None, epi: int = 0,
               out=None, out_pre=None, device=None):
+    dev = device if device is not None else a.device
     c = out if out is not None else torch.empty(
-        (batch, M, N) if batch > 1 else (M, N), dtype=BF16,
-        device=device if device is not None else a.device)
+        (batch, M, N) if batch > 1 else (M, N), dtype=BF16, device=dev)
     cp = out_pre
     if epi >= 2 and cp is None:
         cp = torch.empty_like(c)
+
+    # split-K when the (M,N) tile grid underfills the 256-CU chip and K deep
+    blocks = ((M + 127) // 128) * ((N + 127) // 128)
+    if batch == 1 and epi == 0 and K >= 2048 and blocks < 384:
+        split_k = min(8, max(2, (512 + blocks - 1) // blocks))
+        parts = torch.empty(split_k, M * N, dtype=torch.float32, device=dev)
+        ext.gemm(a.data_ptr(), b_stored.data_ptr(), parts.data_ptr(), 0, 0,
+                 M, N, K, lda, ldb, N, 0, 0, M * N, 1, a_kc, b_kc, 0,
+                 split_k, _stream())
+        ext.splitk_reduce(parts.data_ptr(), c.data_ptr(), split_k, M * N,
+                          _stream())
+        return c, cp
     ext.gemm(a.data_ptr(), b_stored.data_ptr(), c.data_ptr(), _p(cp), _p(bias),
-             M, N, K, lda, ldb, N, sa, sb, M * N, batch, a_kc, b_kc, epi,
+             M, N, K, lda, ldb, N, sa, sb, M * N, batch, a_kc, b_kc, epi, 1,
              _stream())
     return c, cp
 
