@@ -6,6 +6,11 @@ tied embedding / LM head, learned positional embeddings), re-implemented
 natively on our op layer: every matmul / layernorm / softmax / embedding /
 cross-entropy call dispatches to a hand-written CDNA4 HIP kernel on GPU.
 
+Tensor parallelism: pass a ParallelEnv with tp_size>1 and the model builds
+with Megatron-style sharded layers (qkv column-parallel split per head, proj
+row-parallel, MLP column+row, vocab-parallel embedding and cross entropy) —
+the execution form of the auto-planner's tensor sharding strategies.
+
 Weights are bf16; optimizer keeps fp32 masters (see train/optim.py).
 """
 
@@ -19,57 +24,94 @@ import torch.nn as nn
 
 from tepdist_amd import ops
 from tepdist_amd.models.configs import GPT2Config
+from tepdist_amd.parallel.tp import (
+    ColumnParallelLinear,
+    ParallelEnv,
+    RowParallelLinear,
+    VocabParallelEmbedding,
+    vocab_parallel_cross_entropy,
+)
 
 
 class GPT2Block(nn.Module):
-    def __init__(self, cfg: GPT2Config, dtype=torch.bfloat16):
+    def __init__(self, cfg: GPT2Config, dtype=torch.bfloat16,
+                 env: Optional[ParallelEnv] = None):
         super().__init__()
+        env = env or ParallelEnv.single()
         d = cfg.n_embd
+        tp = env.tp_size
+        assert cfg.n_head % tp == 0, "n_head must divide tp_size"
         self.cfg = cfg
-        self.n_head = cfg.n_head
+        self.env = env
+        self.n_head_local = cfg.n_head // tp
+        self.d_local = d // tp
         self.ln1_g = nn.Parameter(torch.ones(d, dtype=dtype))
         self.ln1_b = nn.Parameter(torch.zeros(d, dtype=dtype))
         self.ln2_g = nn.Parameter(torch.ones(d, dtype=dtype))
         self.ln2_b = nn.Parameter(torch.zeros(d, dtype=dtype))
-        # weights in [out, in] layout (ops.linear computes x @ w^T)
-        self.w_qkv = nn.Parameter(torch.empty(3 * d, d, dtype=dtype))
-        self.b_qkv = nn.Parameter(torch.zeros(3 * d, dtype=dtype))
-        self.w_proj = nn.Parameter(torch.empty(d, d, dtype=dtype))
-        self.b_proj = nn.Parameter(torch.zeros(d, dtype=dtype))
-        self.w_fc = nn.Parameter(torch.empty(4 * d, d, dtype=dtype))
-        self.b_fc = nn.Parameter(torch.zeros(4 * d, dtype=dtype))
-        self.w_out = nn.Parameter(torch.empty(d, 4 * d, dtype=dtype))
-        self.b_out = nn.Parameter(torch.zeros(d, dtype=dtype))
+        if tp == 1:
+            # weights in [out, in] layout (ops.linear computes x @ w^T)
+            self.w_qkv = nn.Parameter(torch.empty(3 * d, d, dtype=dtype))
+            self.b_qkv = nn.Parameter(torch.zeros(3 * d, dtype=dtype))
+            self.w_proj = nn.Parameter(torch.empty(d, d, dtype=dtype))
+            self.b_proj = nn.Parameter(torch.zeros(d, dtype=dtype))
+            self.w_fc = nn.Parameter(torch.empty(4 * d, d, dtype=dtype))
+            self.b_fc = nn.Parameter(torch.zeros(4 * d, dtype=dtype))
+            self.w_out = nn.Parameter(torch.empty(d, 4 * d, dtype=dtype))
+            self.b_out = nn.Parameter(torch.zeros(d, dtype=dtype))
+        else:
+            # qkv shard is [3, H/tp, hd, d] flattened: q,k,v of THIS rank's
+            # heads (see shard_qkv_weight for the master->shard mapping)
+            self.qkv = ColumnParallelLinear(d, 3 * d, env, bias=True,
+                                            dtype=dtype)
+            self.proj = RowParallelLinear(d, d, env, bias=True, dtype=dtype)
+            self.fc = ColumnParallelLinear(d, 4 * d, env, bias=True,
+                                           act="gelu", dtype=dtype)
+            self.out = RowParallelLinear(4 * d, d, env, bias=True, dtype=dtype)
+
+    def _attn(self, qkv: torch.Tensor, B: int, S: int) -> torch.Tensor:
+        Hl, hd = self.n_head_local, self.cfg.n_embd // self.cfg.n_head
+        dl = Hl * hd
+        q, k, v = qkv.split(dl, dim=-1)
+        q = q.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
+        k = k.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
+        v = v.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
+        a = ops.attention(q, k, v, causal=True)  # [B,Hl,S,hd]
+        return a.transpose(1, 2).reshape(B, S, dl).contiguous()
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, d = x.shape
-        H = self.n_head
-        hd = d // H
-
         h = ops.layernorm(x, self.ln1_g, self.ln1_b, self.cfg.ln_eps)
-        qkv = ops.linear(h, self.w_qkv, self.b_qkv)          # [B,S,3d]
-        q, k, v = qkv.split(d, dim=-1)
-        q = q.reshape(B, S, H, hd).transpose(1, 2).contiguous()
-        k = k.reshape(B, S, H, hd).transpose(1, 2).contiguous()
-        v = v.reshape(B, S, H, hd).transpose(1, 2).contiguous()
-        a = ops.attention(q, k, v, causal=True)              # [B,H,S,hd]
-        a = a.transpose(1, 2).reshape(B, S, d).contiguous()
-        x = x + ops.linear(a, self.w_proj, self.b_proj)
-
-        h = ops.layernorm(x, self.ln2_g, self.ln2_b, self.cfg.ln_eps)
-        h = ops.linear(h, self.w_fc, self.b_fc, act="gelu")  # fused bias+gelu
-        x = x + ops.linear(h, self.w_out, self.b_out)
+        if self.env.tp_size == 1:
+            qkv = ops.linear(h, self.w_qkv, self.b_qkv)
+            a = self._attn(qkv, B, S)
+            x = x + ops.linear(a, self.w_proj, self.b_proj)
+            h = ops.layernorm(x, self.ln2_g, self.ln2_b, self.cfg.ln_eps)
+            h = ops.linear(h, self.w_fc, self.b_fc, act="gelu")
+            x = x + ops.linear(h, self.w_out, self.b_out)
+        else:
+            qkv = self.qkv(h)                     # [B,S,3*d/tp]
+            a = self._attn(qkv, B, S)
+            x = x + self.proj(a)
+            h = ops.layernorm(x, self.ln2_g, self.ln2_b, self.cfg.ln_eps)
+            x = x + self.out(self.fc(h))
         return x
 
 
 class GPT2(nn.Module):
-    def __init__(self, cfg: GPT2Config, dtype=torch.bfloat16):
+    def __init__(self, cfg: GPT2Config, dtype=torch.bfloat16,
+                 env: Optional[ParallelEnv] = None):
         super().__init__()
         self.cfg = cfg
+        self.env = env or ParallelEnv.single()
         V, d = cfg.padded_vocab, cfg.n_embd
-        self.wte = nn.Parameter(torch.empty(V, d, dtype=dtype))
+        if self.env.tp_size == 1:
+            self.wte = nn.Parameter(torch.empty(V, d, dtype=dtype))
+        else:
+            self.wte_mod = VocabParallelEmbedding(V, d, self.env, dtype=dtype)
         self.wpe = nn.Parameter(torch.empty(cfg.n_ctx, d, dtype=dtype))
-        self.blocks = nn.ModuleList(GPT2Block(cfg, dtype) for _ in range(cfg.n_layer))
+        self.blocks = nn.ModuleList(
+            GPT2Block(cfg, dtype, self.env) for _ in range(cfg.n_layer))
         self.lnf_g = nn.Parameter(torch.ones(d, dtype=dtype))
         self.lnf_b = nn.Parameter(torch.zeros(d, dtype=dtype))
         self.reset_parameters()
@@ -81,35 +123,65 @@ class GPT2(nn.Module):
         proj_std = std / math.sqrt(2 * self.cfg.n_layer)
         for name, p in self.named_parameters():
             if p.dim() == 2:
-                s = proj_std if ("w_proj" in name or "w_out" in name) else std
+                s = proj_std if ("w_proj" in name or "w_out" in name or
+                                 "proj.weight" in name or
+                                 "out.weight" in name) else std
                 p.copy_(torch.randn(p.shape, generator=g) * s)
-            elif name.endswith("_b") or "b_" in name:
-                pass  # biases stay zero
-        # zero the padded vocab rows so they never win the softmax
-        self.wte[self.cfg.vocab_size:].zero_()
+        if self.env.tp_size == 1:
+            # zero the padded vocab rows so they never win the softmax
+            self.wte[self.cfg.vocab_size:].zero_()
+        else:
+            emb = self.wte_mod
+            lo = max(0, self.cfg.vocab_size - emb.vocab_start)
+            if lo < emb.vocab_local:
+                emb.weight[lo:].zero_()
 
     def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None):
         """input_ids [B, S]; labels [B, S] (next-token ids, -1 = ignore).
-        Returns loss (scalar f32) if labels given, else logits [B,S,V]."""
+        Returns loss (scalar f32) if labels given, else logits [B,S,V]
+        ([B,S,V/tp] local shard under tensor parallelism)."""
         B, S = input_ids.shape
         pos = torch.arange(S, device=input_ids.device)
-        x = ops.embedding(input_ids, self.wte) + ops.embedding(pos, self.wpe)
+        if self.env.tp_size == 1:
+            x = ops.embedding(input_ids, self.wte) + ops.embedding(pos, self.wpe)
+        else:
+            x = self.wte_mod(input_ids) + ops.embedding(pos, self.wpe)
         for blk in self.blocks:
             x = blk(x)
         x = ops.layernorm(x, self.lnf_g, self.lnf_b, self.cfg.ln_eps)
-        logits = ops.linear(x, self.wte)  # tied LM head: x @ wte^T
+        if self.env.tp_size == 1:
+            logits = ops.linear(x, self.wte)  # tied LM head: x @ wte^T
+            if labels is None:
+                return logits
+            return ops.cross_entropy(
+                logits.reshape(-1, logits.shape[-1]), labels.reshape(-1),
+                ignore_index=-1)
+        # vocab-parallel tied LM head + cross entropy
+        from tepdist_amd.parallel.mappings import copy_to_group
+        x = copy_to_group(x, self.env.tp_group)
+        logits_local = ops.linear(x, self.wte_mod.weight)
         if labels is None:
-            return logits
-        loss = ops.cross_entropy(
-            logits.reshape(-1, logits.shape[-1]), labels.reshape(-1),
-            ignore_index=-1)
-        return loss
+            return logits_local
+        return vocab_parallel_cross_entropy(
+            logits_local.reshape(-1, logits_local.shape[-1]),
+            labels.reshape(-1), self.wte_mod.vocab_start,
+            self.wte_mod.vocab_local, self.env.tp_group, ignore_index=-1)
 
     def flops_per_token(self) -> float:
         """Approximate training FLOPs per token (fwd+bwd, 6N + attention)."""
         cfg = self.cfg
-        n = sum(p.numel() for p in self.parameters()) - cfg.padded_vocab * cfg.n_embd
-        # 6*N matmul flops + attention 12*L*d*S
-        return 6 * (n + cfg.padded_vocab * cfg.n_embd) + \
-            12 * cfg.n_layer * cfg.n_embd * cfg.n_ctx
+        n = sum(p.numel() for p in self.parameters())
+        return 6 * n + 12 * cfg.n_layer * cfg.n_embd * cfg.n_ctx
+
+
+def shard_qkv_weight(w_qkv: torch.Tensor, b_qkv: torch.Tensor, n_head: int,
+                     tp_rank: int, tp_size: int):
+    """Maps a full packed qkv weight [3d, d] / bias [3d] to rank tp_rank's
+    ColumnParallel shard (contiguous q,k,v of that rank's heads)."""
+    d3, d = w_qkv.shape
+    hd = d // n_head
+    hl = n_head // tp_size
+    w = w_qkv.reshape(3, n_head, hd, d)[:, tp_rank * hl:(tp_rank + 1) * hl]
+    b = b_qkv.reshape(3, n_head, hd)[:, tp_rank * hl:(tp_rank + 1) * hl]
+    return w.reshape(3 * hl * hd, d).contiguous(), b.reshape(-1).contiguous()

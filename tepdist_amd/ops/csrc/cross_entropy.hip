@@ -57,7 +57,9 @@ void ce_fwd_kernel(const bf16_t* __restrict__ logits,
       const float lse = m + __logf(s);
       lse_out[row] = lse;
       const int64_t t = targets[row];
-      nll[row] = (t == ignore_index) ? 0.f : lse - bf2f(lr[t]);
+      // t < 0: either the ignore index or a "no target in this vocab
+      // shard" sentinel (vocab-parallel CE) -> no target-logit read
+      nll[row] = (t < 0) ? 0.f : lse - bf2f(lr[t]);
     }
     __syncthreads();
   }
@@ -72,7 +74,8 @@ void ce_bwd_kernel(const bf16_t* __restrict__ logits,
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16_t* lr = logits + row * cols;
     bf16_t* dr = dlogits + row * cols;
-    const int64_t t = targets[row];
+    const int64_t t = targets[row];  // t==ignore: zero row; other t<0
+    // (vocab-parallel "target not in shard"): softmax grad, no onehot
     if (t == ignore_index) {
       for (int c0 = threadIdx.x * 8; c0 < cols; c0 += NT * 8) {
         bf16x8 z = {};

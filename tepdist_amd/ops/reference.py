@@ -190,9 +190,10 @@ def cross_entropy_fwd(logits: torch.Tensor, targets: torch.Tensor,
     l32 = logits.float()
     lse = torch.logsumexp(l32, dim=-1)
     valid = targets != ignore_index
+    has_tgt = targets >= 0  # negative non-ignore = "no target here"
     tgt = targets.clamp_min(0)
     nll = lse - l32.gather(-1, tgt.unsqueeze(-1)).squeeze(-1)
-    nll = torch.where(valid, nll, torch.zeros_like(nll))
+    nll = torch.where(valid & has_tgt, nll, torch.zeros_like(nll))
     n = valid.sum().clamp_min(1)
     return nll.sum() / n, lse
 
@@ -203,9 +204,10 @@ def cross_entropy_bwd(dloss: torch.Tensor, logits: torch.Tensor,
     l32 = logits.float()
     p = torch.exp(l32 - lse.unsqueeze(-1))
     valid = (targets != ignore_index)
+    has_tgt = targets >= 0  # negative non-ignore: softmax grad, no onehot
     tgt = targets.clamp_min(0)
     p.scatter_add_(-1, tgt.unsqueeze(-1),
-                   -torch.ones_like(tgt, dtype=torch.float32).unsqueeze(-1))
+                   -has_tgt.to(torch.float32).unsqueeze(-1))
     n = valid.sum().clamp_min(1).float()
     p = p * (dloss.float() / n)
     p = torch.where(valid.unsqueeze(-1), p, torch.zeros_like(p))
