@@ -185,3 +185,80 @@ def shard_qkv_weight(w_qkv: torch.Tensor, b_qkv: torch.Tensor, n_head: int,
     w = w_qkv.reshape(3, n_head, hd, d)[:, tp_rank * hl:(tp_rank + 1) * hl]
     b = b_qkv.reshape(3, n_head, hd)[:, tp_rank * hl:(tp_rank + 1) * hl]
     return w.reshape(3 * hl * hd, d).contiguous(), b.reshape(-1).contiguous()
+
+
+class GPT2Stage(nn.Module):
+    """One pipeline stage of GPT-2 (a contiguous layer range; the first
+    stage owns the embeddings, the last owns the final LN + LM head and the
+    loss). The head is untied from the embedding across stages (the tied
+    form only exists when both live on one stage). Composable with tensor
+    parallelism via env."""
+
+    def __init__(self, cfg: GPT2Config, layer_start: int, layer_end: int,
+                 is_first: bool, is_last: bool, dtype=torch.bfloat16,
+                 env: Optional[ParallelEnv] = None):
+        super().__init__()
+        self.cfg = cfg
+        self.env = env or ParallelEnv.single()
+        self.is_first, self.is_last = is_first, is_last
+        V, d = cfg.padded_vocab, cfg.n_embd
+        if is_first:
+            self.wte = nn.Parameter(torch.empty(V, d, dtype=dtype))
+            self.wpe = nn.Parameter(torch.empty(cfg.n_ctx, d, dtype=dtype))
+        self.blocks = nn.ModuleList(
+            GPT2Block(cfg, dtype, self.env)
+            for _ in range(layer_start, layer_end))
+        if is_last:
+            self.lnf_g = nn.Parameter(torch.ones(d, dtype=dtype))
+            self.lnf_b = nn.Parameter(torch.zeros(d, dtype=dtype))
+            self.lm_head = nn.Parameter(torch.empty(V, d, dtype=dtype))
+        self.reset_parameters()
+
+    @torch.no_grad()
+    def reset_parameters(self, seed: int = 1234):
+        g = torch.Generator().manual_seed(seed)
+        std = 0.02
+        proj_std = std / math.sqrt(2 * self.cfg.n_layer)
+        for name, p in self.named_parameters():
+            if p.dim() == 2:
+                s = proj_std if ("w_proj" in name or "w_out" in name or
+                                 "proj.weight" in name or
+                                 "out.weight" in name) else std
+                p.copy_(torch.randn(p.shape, generator=g) * s)
+        if self.is_first:
+            self.wte[self.cfg.vocab_size:].zero_()
+        if self.is_last:
+            self.lm_head[self.cfg.vocab_size:].zero_()
+
+    def forward(self, x, labels=None):
+        if self.is_first:
+            input_ids = x
+            B, S = input_ids.shape
+            pos = torch.arange(S, device=input_ids.device)
+            x = ops.embedding(input_ids, self.wte) + \
+                ops.embedding(pos, self.wpe)
+        for blk in self.blocks:
+            x = blk(x)
+        if not self.is_last:
+            return x
+        x = ops.layernorm(x, self.lnf_g, self.lnf_b, self.cfg.ln_eps)
+        logits = ops.linear(x, self.lm_head)
+        if labels is None:
+            return logits
+        return ops.cross_entropy(logits.reshape(-1, logits.shape[-1]),
+                                 labels.reshape(-1), ignore_index=-1)
+
+
+def layer_ranges(n_layer: int, num_stages: int,
+                 layer_stage: Optional[list] = None):
+    """Contiguous [start, end) layer range per stage (from a plan's
+    layer->stage map or balanced)."""
+    if layer_stage is None:
+        per = (n_layer + num_stages - 1) // num_stages
+        return [(s * per, min((s + 1) * per, n_layer))
+                for s in range(num_stages)]
+    ranges = []
+    for s in range(num_stages):
+        ls = [l for l, st in enumerate(layer_stage) if st == s]
+        ranges.append((min(ls), max(ls) + 1) if ls else (0, 0))
+    return ranges

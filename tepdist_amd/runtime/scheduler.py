@@ -1,0 +1,155 @@
+"""Compile-time event-driven list scheduler over the task DAG.
+
+Re-implements the reference TaskScheduler's structure
+(task_scheduler.h:86-374): a ClusterState -> DevState
+simulation that assigns each task a schedule index on its device, with
+  - per-device memory accounting against the HBM cap (288 GB on MI355X;
+    the reference's default cap is at task_scheduler.h:177),
+  - a micro_num_limit bound on in-flight forward micro-batches per device
+    (the 1F1B memory bound),
+  - task durations from flops / bytes via the cost model,
+  - `sched_cnt` candidate schedules (different ready-queue policies), best
+    makespan wins (GROUP_SCHED_COUNT).
+The resulting per-device ordered lists drive the executor."""
+
+from __future__ import annotations
+
+import heapq
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from tepdist_amd.config import get_env
+from tepdist_amd.planner.cost_model import CostModel
+from tepdist_amd.runtime.task_graph import TaskDAG, TaskNode, TaskType
+
+
+@dataclass
+class ScheduleResult:
+    order: Dict[int, List[int]]          # device -> task ids in order
+    makespan: float
+    peak_mem: Dict[int, float]
+    policy: str
+
+
+class TaskScheduler:
+    def __init__(self, dag: TaskDAG, cm: CostModel = None,
+                 micro_num_limit: int = 0, mem_cap_bytes: float = None):
+        self.dag = dag
+        self.cm = cm or CostModel()
+        env = get_env()
+        self.micro_limit = micro_num_limit or env.micro_num_limit
+        self.mem_cap = mem_cap_bytes if mem_cap_bytes is not None \
+            else self.cm.hw.hbm_bytes
+
+    def _duration(self, t: TaskNode) -> float:
+        if t.type in (TaskType.COMPUTE_FW, TaskType.COMPUTE_BW):
+            return t.flops / (self.cm.hw.bf16_tflops * 1e12) + 10e-6
+        if t.type in (TaskType.SEND, TaskType.RECV):
+            return self.cm.p2p(t.out_bytes)
+        if t.type == TaskType.AR:
+            return self.cm.all_reduce(t.out_bytes, 2)
+        if t.type in (TaskType.GA, TaskType.AG):
+            return t.out_bytes * 3 / (self.cm.hw.hbm_gbps * 1e9) + 5e-6
+        return 1e-6
+
+    def schedule(self, sched_cnt: int = None) -> ScheduleResult:
+        cnt = sched_cnt or get_env().group_sched_count
+        policies = ["bw_first", "fifo"][:max(cnt, 1)]
+        best = None
+        for pol in policies:
+            r = self._simulate(pol)
+            if r is not None and (best is None or r.makespan < best.makespan):
+                best = r
+        assert best is not None, "no feasible schedule"
+        # write sched_idx back
+        for dev, ids in best.order.items():
+            for i, tid in enumerate(ids):
+                self.dag.tasks[tid].sched_idx = i
+        return best
+
+    def _priority(self, t: TaskNode, policy: str):
+        if policy == "bw_first":
+            # 1F1B flavor: backward of older micros before new forwards
+            kind = 0 if t.type == TaskType.COMPUTE_BW else \
+                (1 if t.type == TaskType.COMPUTE_FW else -1)
+            return (kind, t.split.micro, t.id)
+        return (t.id,)
+
+    def _simulate(self, policy: str) -> Optional[ScheduleResult]:
+        dag = self.dag
+        indeg = {i: len(t.parents) for i, t in dag.tasks.items()}
+        ready: Dict[int, list] = {}
+        dev_free: Dict[int, float] = {}
+        mem: Dict[int, float] = {}
+        peak: Dict[int, float] = {}
+        inflight_fw: Dict[int, int] = {}
+        order: Dict[int, List[int]] = {}
+        finish: Dict[int, float] = {}
+
+        def push(tid):
+            t = dag.tasks[tid]
+            ready.setdefault(t.device, [])
+            heapq.heappush(ready[t.device], (self._priority(t, policy), tid))
+
+        for i, d in indeg.items():
+            if d == 0:
+                push(i)
+        events: list = []  # (time, tid) completions
+        time_now = 0.0
+        scheduled = 0
+        total = len(dag.tasks)
+        guard = 0
+        while scheduled < total:
+            guard += 1
+            if guard > 10 * total + 100:
+                return None
+            progressed = False
+            for dev in list(ready.keys()):
+                q = ready[dev]
+                while q:
+                    # respect micro_num_limit: delay new forwards when too
+                    # many activations are live
+                    _, tid = q[0]
+                    t = dag.tasks[tid]
+                    if (self.micro_limit > 0 and
+                            t.type == TaskType.COMPUTE_FW and
+                            inflight_fw.get(dev, 0) >= self.micro_limit):
+                        break
+                    heapq.heappop(q)
+                    start = max(dev_free.get(dev, 0.0), time_now)
+                    dur = self._duration(t)
+                    end = start + dur
+                    dev_free[dev] = end
+                    finish[tid] = end
+                    order.setdefault(dev, []).append(tid)
+                    if t.type == TaskType.COMPUTE_FW:
+                        inflight_fw[dev] = inflight_fw.get(dev, 0) + 1
+                        mem[dev] = mem.get(dev, 0.0) + t.out_bytes
+                        peak[dev] = max(peak.get(dev, 0.0), mem[dev])
+                        if mem[dev] > self.mem_cap:
+                            return None
+                    heapq.heappush(events, (end, tid))
+                    scheduled += 1
+                    progressed = True
+            if not progressed:
+                if not events:
+                    return None
+                time_now, done = heapq.heappop(events)
+                t = dag.tasks[done]
+                if t.type == TaskType.COMPUTE_BW:
+                    inflight_fw[t.device] = max(
+                        inflight_fw.get(t.device, 0) - 1, 0)
+                    mem[t.device] = max(
+                        mem.get(t.device, 0.0) - t.out_bytes, 0.0)
+                    # free the matching forward's activation
+                    for p in t.parents:
+                        pt = dag.tasks[p]
+                        if pt.release_at == done:
+                            mem[t.device] = max(
+                                mem.get(t.device, 0.0) - pt.out_bytes, 0.0)
+                for c in t.children:
+                    indeg[c] -= 1
+                    if indeg[c] == 0:
+                        push(c)
+        makespan = max(finish.values()) if finish else 0.0
+        return ScheduleResult(order, makespan, peak, policy)
