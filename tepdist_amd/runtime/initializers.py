@@ -87,11 +87,13 @@ def init_shard(name: str, full_shape: Tuple[int, ...], spec: InitSpec,
     seed = _var_seed(global_seed, name)
     idx = _global_indices(tuple(full_shape), shard_dim, shard_index,
                           num_shards)
-    u = _uniform_at(seed, idx)
+    # one counter stream, two values per element (keeps shard consistency
+    # AND independence of the Box-Muller pair)
+    u = _uniform_at(seed, idx * 2)
     if spec.kind == "random_uniform":
         out = spec.low + (spec.high - spec.low) * u
     elif spec.kind in ("random_normal", "truncated_normal"):
-        u2 = _uniform_at(seed ^ 0x5DEECE66D, idx)
+        u2 = _uniform_at(seed, idx * 2 + 1)
         r = torch.sqrt(-2.0 * torch.log(u.clamp_min(1e-12)))
         out = spec.mean + spec.std * r * torch.cos(2 * math.pi * u2)
         if spec.kind == "truncated_normal":
