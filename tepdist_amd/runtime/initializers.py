@@ -36,13 +36,18 @@ def _var_seed(global_seed: int, name: str) -> int:
 
 def _uniform_at(seed: int, idx: torch.Tensor) -> torch.Tensor:
     """Deterministic uniform [0,1) at arbitrary global linear indices
-    (counter-based splitmix64 keyed by (seed, index))."""
-    M = (1 << 64) - 1
-    z = (idx * 0x9E3779B97F4A7C15 + seed) & M
-    z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & M
-    z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & M
-    z = z ^ (z >> 31)
-    return ((z >> 11).to(torch.float64) / float(1 << 53)).to(torch.float32)
+    (counter-based splitmix64 keyed by (seed, index)). Computed in numpy
+    uint64 — torch int64 is signed and its arithmetic shifts would break
+    the mix."""
+    import numpy as np
+    z = idx.numpy().astype(np.uint64)
+    with np.errstate(over="ignore"):
+        z = z * np.uint64(0x9E3779B97F4A7C15) + np.uint64(seed)
+        z = (z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        z = (z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        z = z ^ (z >> np.uint64(31))
+    u = (z >> np.uint64(11)).astype(np.float64) / float(1 << 53)
+    return torch.from_numpy(u.astype(np.float32))
 
 
 def _global_indices(full_shape: Tuple[int, ...], shard_dim: int,
