@@ -1,0 +1,96 @@
+"""IR interpreter: executes a client-shipped Graph on the server.
+
+Plays the role of the reference's compiled sub-module executables (the
+server compiles the received HLO and runs it) — here the
+graph runs through the tepdist_amd.ops layer (hand-written CDNA4 kernels on
+GPU, torch reference on CPU) with torch autograd providing the backward
+pass and the fused AdamW applying gradients (the AG role)."""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+
+from tepdist_amd import ops
+from tepdist_amd.ir.graph import Graph, Node
+
+
+class GraphInterpreter:
+    def __init__(self, graph: Graph, device: str = "cpu",
+                 dtype=torch.float32):
+        self.g = graph
+        self.device = device
+        self.dtype = dtype
+
+    def run(self, feeds: Dict[str, torch.Tensor],
+            variables: Dict[str, torch.Tensor]) -> Dict[int, torch.Tensor]:
+        """One forward pass. feeds keyed by input node name; variables by
+        param name (tensors with requires_grad for training). Returns
+        {output node id: tensor}."""
+        env: Dict[int, torch.Tensor] = {}
+        for n in self.g.topo():
+            env[n.id] = self._eval(n, env, feeds, variables)
+        return {o: env[o] for o in self.g.outputs}
+
+    def _eval(self, n: Node, env, feeds, variables) -> torch.Tensor:
+        ins = [env[i] for i in n.inputs]
+        if n.op == "data":
+            if n.name in feeds:
+                return feeds[n.name].to(self.device)
+            if n.name == "pos":  # position ids (B*S,) derived from seq attr
+                bs = n.shape[0]
+                b = n.attrs.get("batch", 1)
+                s = bs // b
+                return torch.arange(s, device=self.device).repeat(b)
+            raise KeyError(f"missing feed {n.name}")
+        if n.op == "param":
+            return variables[n.name]
+        if n.op == "embedding":
+            return ops.embedding(ins[0], ins[1])
+        if n.op == "linear":
+            bias = ins[2] if len(ins) > 2 else None
+            return ops.linear(ins[0], ins[1], bias,
+                              act=n.attrs.get("act", "none"))
+        if n.op == "matmul":
+            return ops.matmul(ins[0], ins[1])
+        if n.op == "layernorm":
+            return ops.layernorm(ins[0], ins[1], ins[2])
+        if n.op == "softmax":
+            return ops.softmax(ins[0], scale=n.attrs.get("scale", 1.0),
+                               causal=n.attrs.get("causal", False))
+        if n.op == "attention":
+            x = ins  # q, k, v in flattened (B*S, d) form
+            b = n.attrs["batch"]
+            h = n.attrs["heads"]
+            s = n.attrs["seq"]
+            hd = n.shape[1] // h
+            q, k, v = (t.reshape(b, s, h, hd).transpose(1, 2).contiguous()
+                       for t in x)
+            o = ops.attention(q, k, v, causal=True)
+            return o.transpose(1, 2).reshape(b * s, h * hd).contiguous()
+        if n.op == "split":
+            dim = n.attrs.get("dim", -1)
+            idx = n.attrs.get("index", 0)
+            size = n.shape[dim if dim >= 0 else len(n.shape) - 1]
+            return ins[0].narrow(dim, idx * size, size)
+        if n.op == "add":
+            return ins[0] + ins[1]
+        if n.op == "mul":
+            return ins[0] * ins[1]
+        if n.op == "gelu":
+            from tepdist_amd.ops.interface import _backend
+            return _backend(ins[0]).gelu_fwd(ins[0]) if ins[0].is_cuda \
+                else torch.nn.functional.gelu(ins[0], approximate="tanh")
+        if n.op == "dropout":
+            return ops.dropout(ins[0], n.attrs.get("p", 0.0))
+        if n.op == "cross_entropy":
+            return ops.cross_entropy(ins[0], ins[1].reshape(-1),
+                                     ignore_index=-1)
+        if n.op == "reshape":
+            return ins[0].reshape(n.shape)
+        if n.op == "transpose":
+            return ins[0].permute(n.attrs["perm"]).contiguous()
+        if n.op == "elementwise":
+            return ins[0]
+        raise NotImplementedError(f"op {n.op}")

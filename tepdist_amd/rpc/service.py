@@ -1,0 +1,249 @@
+"""The TepDist service: master/worker orchestration logic.
+
+Method surface mirrors the reference's XlaService additions
+(rpc/xla_service.proto:93-199 + Service::* in service_rt.cc, SURVEY.md
+§2.2): BuildExecutionPlan, ExecutePlan, TransferToServerHost,
+TransferHostRawData, TransferModuleAndDefCtx, DispatchPlan,
+InitRemoteComm, ExecuteRemotePlan, FetchResourceVars, TransferVarArgMap,
+DoRemoteSave, DoRemoteRestore.
+
+BuildExecutionPlan runs the full planning pipeline (AutoParallel -> stage
+cut -> TaskDAG -> TaskScheduler) and caches the plan; ExecutePlan runs
+iterations through the IR interpreter with server-held variables
+(initialized server-side by the sharded initializers — the reference's
+RewriteInitializationRemote contract), applying the fused AdamW optimizer.
+Execution in this process is single-device (CPU here, cuda:0 on a GPU
+box); multi-GPU data/tensor/pipeline execution runs under torchrun via
+bench.py/examples (one process per GPU over RCCL), with DispatchPlan /
+ExecuteRemotePlan carrying plans to slave servers."""
+
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, Optional
+
+import torch
+
+from tepdist_amd.config import get_env
+from tepdist_amd.ir.graph import Graph
+from tepdist_amd.ir.interpreter import GraphInterpreter
+from tepdist_amd.planner.auto_parallel import AutoParallel
+from tepdist_amd.runtime.checkpoint import CheckpointManager, SliceMeta
+from tepdist_amd.runtime.initializers import InitSpec, init_shard
+from tepdist_amd.runtime.scheduler import TaskScheduler
+from tepdist_amd.runtime.task_graph import build_task_dag
+from tepdist_amd.train.optim import AdamW
+
+
+@dataclass
+class ExecutionPlan:
+    handle: int
+    graph: Graph
+    plan: object
+    schedule: object
+    interpreter: GraphInterpreter
+
+
+@dataclass
+class _VarState:
+    tensor: torch.Tensor
+    master: torch.Tensor
+    exp_avg: torch.Tensor
+    exp_avg_sq: torch.Tensor
+
+
+class TepdistService:
+    """One service instance per worker process (task_index 0 = master)."""
+
+    def __init__(self, task_index: int = 0, device: Optional[str] = None,
+                 ckpt_dir: str = "/tmp/tepdist_ckpt"):
+        self.task_index = task_index
+        self.device = device or (
+            "cuda:0" if torch.cuda.is_available() else "cpu")
+        self.plans: Dict[int, ExecutionPlan] = {}
+        self.vars: Dict[str, _VarState] = {}
+        self.host_data: Dict[str, torch.Tensor] = {}
+        self.var_arg_map: Dict[str, int] = {}
+        self.ckpt = CheckpointManager(ckpt_dir)
+        self.ckpt_opts = {"lazy_save": False, "restore_step": None,
+                          "max_to_keep": 5}
+        self._next_handle = 1
+        self._lock = threading.Lock()   # the reference's execute_plan_mutex_
+        self.step_count = 0
+        self.lr = 1e-4
+
+    # ------------------------------------------------------------------
+
+    def build_execution_plan(self, req: dict) -> dict:
+        g = Graph.from_json(req["graph"])
+        devs = int(req.get("num_devices", 1))
+        t0 = time.time()
+        plan = AutoParallel(g, devs).run()
+        dag = build_task_dag(plan.pp, plan.micro_batches,
+                             dp_degree=plan.dp)
+        sched = TaskScheduler(dag, mem_cap_bytes=float("inf")).schedule()
+        interp = GraphInterpreter(g, self.device)
+        with self._lock:
+            h = self._next_handle
+            self._next_handle += 1
+            self.plans[h] = ExecutionPlan(h, g, plan, sched, interp)
+        self._init_variables(g, req.get("init_specs", {}))
+        return {"handle": h, "summary": plan.summary(),
+                "dp": plan.dp, "tp": plan.tp, "pp": plan.pp,
+                "micro_batches": plan.micro_batches,
+                "search_time_s": time.time() - t0}
+
+    def _init_variables(self, g: Graph, init_specs: dict):
+        """Server-side variable creation (RewriteInitializationRemote moved
+        init to the server; DistributedRandomInitializer seeds shards)."""
+        restore = self.ckpt_opts.get("restore_step")
+        if restore is not None:
+            want = {name: SliceMeta(tuple(g.nodes[nid].shape))
+                    for name, nid in g.params.items()}
+            loaded = self.ckpt.restore(restore, want)
+        dtype = torch.float32 if self.device == "cpu" else torch.bfloat16
+        for name, nid in g.params.items():
+            if name in self.vars:
+                continue
+            node = g.nodes[nid]
+            if restore is not None and name in loaded:
+                t = loaded[name].to(dtype)
+            elif name in self.host_data:   # client-transferred variable
+                t = self.host_data[name].to(dtype)
+            else:
+                spec = InitSpec(**init_specs.get(name, {})) \
+                    if name in init_specs else _default_spec(name, node.shape)
+                t = init_shard(name, node.shape, spec, dtype=dtype)
+            t = t.to(self.device).requires_grad_()
+            self.vars[name] = _VarState(
+                t, t.detach().float().clone(),
+                torch.zeros_like(t, dtype=torch.float32),
+                torch.zeros_like(t, dtype=torch.float32))
+
+    # ------------------------------------------------------------------
+
+    def execute_plan(self, req: dict) -> dict:
+        h = req["handle"]
+        plan = self.plans[h]
+        feeds = {k: v.to(self.device) for k, v in req.get("inputs", {}).items()}
+        env = get_env()
+        with self._lock:
+            t0 = time.time()
+            self.step_count += 1
+            variables = {n: vs.tensor for n, vs in self.vars.items()}
+            outs = plan.interpreter.run(feeds, variables)
+            loss = list(outs.values())[0]
+            loss.backward()
+            from tepdist_amd import ops as _ops
+            for n, vs in self.vars.items():
+                if vs.tensor.grad is None:
+                    continue
+                _ops.adamw_step(vs.tensor.data, vs.master, vs.tensor.grad,
+                                vs.exp_avg, vs.exp_avg_sq, lr=self.lr,
+                                step=self.step_count)
+                vs.tensor.grad = None
+            dur_ms = (time.time() - t0) * 1e3
+            if env.debug:
+                print(f"[ExecutePlan Duration] step={self.step_count} "
+                      f"{dur_ms:.2f} ms", flush=True)
+        if self.ckpt_opts.get("lazy_save"):
+            self.ckpt_opts["lazy_save"] = False
+            self._save(self.step_count)
+        return {"outputs": {str(k): v.detach() for k, v in outs.items()},
+                "step": self.step_count, "duration_ms": dur_ms}
+
+    # ------------------------------------------------------------------
+
+    def transfer_to_server_host(self, req: dict) -> dict:
+        name = req["name"]
+        self.host_data[name] = req["data"]
+        if req.get("variable"):
+            self.var_arg_map[name] = req.get("global_idx", -1)
+        return {"ok": True, "handle": name}
+
+    def transfer_host_raw_data(self, req: dict) -> dict:
+        self.host_data[req["name"]] = req["data"]
+        return {"ok": True}
+
+    def transfer_var_arg_map(self, req: dict) -> dict:
+        self.var_arg_map.update(req["map"])
+        return {"ok": True}
+
+    def fetch_resource_vars(self, req: dict) -> dict:
+        names = req.get("names") or list(self.vars)
+        return {"vars": {n: self.vars[n].tensor for n in names
+                         if n in self.vars}}
+
+    # -- master -> slave surface (rebuilt plans on remote workers) ---------
+
+    def transfer_module_and_defctx(self, req: dict) -> dict:
+        g = Graph.from_json(req["graph"])
+        self._pending_graph = g
+        self._pending_tree = req.get("def_tree")
+        return {"ok": True}
+
+    def dispatch_plan(self, req: dict) -> dict:
+        g = getattr(self, "_pending_graph", None)
+        if g is None:
+            return {"ok": False, "error": "no module transferred"}
+        interp = GraphInterpreter(g, self.device)
+        with self._lock:
+            h = self._next_handle
+            self._next_handle += 1
+            self.plans[h] = ExecutionPlan(h, g, req.get("plan"), None, interp)
+        self._init_variables(g, {})
+        return {"ok": True, "handle": h}
+
+    def init_remote_comm(self, req: dict) -> dict:
+        """Communicator bootstrap: the reference RPCs raw ncclUniqueIds
+        (service_rt.cc:310-334); over RCCL we carry the rendezvous
+        (master addr/port + rank/world) for torch.distributed init."""
+        import os
+        os.environ["MASTER_ADDR"] = req.get("master_addr", "127.0.0.1")
+        os.environ["MASTER_PORT"] = str(req.get("master_port", 29500))
+        self._comm_info = {"rank": req.get("rank", self.task_index),
+                           "world": req.get("world", 1)}
+        return {"ok": True}
+
+    def execute_remote_plan(self, req: dict) -> dict:
+        return self.execute_plan(req)
+
+    # -- checkpoint --------------------------------------------------------
+
+    def do_remote_save(self, req: dict) -> dict:
+        self.ckpt_opts["max_to_keep"] = req.get("max_to_keep", 5)
+        if req.get("lazy", False) or not self.vars:
+            self.ckpt_opts["lazy_save"] = True
+            return {"ok": True, "lazy": True}
+        self._save(req.get("global_step", self.step_count))
+        return {"ok": True, "lazy": False}
+
+    def _save(self, step: int):
+        self.ckpt.max_to_keep = self.ckpt_opts["max_to_keep"]
+        shards = {}
+        for n, vs in self.vars.items():
+            shards[n] = (vs.master, SliceMeta(tuple(vs.tensor.shape)))
+            shards[f"opt.m.{n}"] = (vs.exp_avg,
+                                    SliceMeta(tuple(vs.tensor.shape)))
+            shards[f"opt.v.{n}"] = (vs.exp_avg_sq,
+                                    SliceMeta(tuple(vs.tensor.shape)))
+        self.ckpt.save(step, shards, rank=self.task_index)
+
+    def do_remote_restore(self, req: dict) -> dict:
+        self.ckpt_opts["restore_step"] = req.get("global_step",
+                                                 self.ckpt.latest_step())
+        # drop live vars so the next plan build restores from the checkpoint
+        self.vars.clear()
+        return {"ok": True, "step": self.ckpt_opts["restore_step"]}
+
+
+def _default_spec(name: str, shape) -> InitSpec:
+    if name.endswith(("_b", ".bias")) or "b_" in name.split(".")[-1]:
+        return InitSpec("zeros")
+    if name.endswith("_g") or "ln" in name:
+        return InitSpec("ones") if name.endswith("_g") else InitSpec("zeros")
+    if len(shape) >= 2:
+        return InitSpec("random_normal", std=0.02)
+    return InitSpec("zeros")
