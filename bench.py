@@ -1,4 +1,4 @@
-"""Flagship benchmark: GPT-2 auto-parallel training throughput on MI355X.
+"""Flagship benchmark: GPT-2 AUTO-PARALLEL training throughput on MI355X.
 
 Contract (driver-facing):
   python bench.py --gpus N --steps K --warmup W
@@ -8,9 +8,13 @@ W untimed warmup steps, then exactly K timed steps bracketed by a barrier +
 torch.cuda.synchronize on both sides; elapsed time is MAX over ranks; rank 0
 prints one JSON line with the whole-node aggregate tokens/sec.
 
+Auto-parallel: the model's IR goes through the AutoParallel planner
+(exploration mode), whose chosen dp x tp x pp x micro plan is mapped onto
+process groups (TP sharded layers / DP bucketed all-reduce / 1F1B pipeline).
+Plan-search wall-clock is reported (BASELINE.json: "plan-search sec").
+
 Metric/config per BASELINE.json: tokens/sec (whole node), GPT-2
-auto-parallel, synthetic data, random-init weights, bf16.
-"""
+auto-parallel, synthetic data, random-init weights, bf16."""
 
 from __future__ import annotations
 
@@ -19,10 +23,32 @@ import json
 import time
 
 import torch
+import torch.distributed as dist
 
+from tepdist_amd.ir import gpt2_ir
 from tepdist_amd.models import GPT2, GPT2_CONFIGS
 from tepdist_amd.parallel import GradReducer, init_distributed
 from tepdist_amd.train import AdamW, Trainer
+
+
+def plan_parallelism(cfg, world: int, global_batch: int, seq: int,
+                     override: str):
+    """Returns (dp, tp, pp, micro, search_s)."""
+    if override != "auto":
+        dp, tp, pp = world, 1, 1
+        if override.startswith("tp"):
+            tp = int(override[2:]); dp = world // tp
+        elif override.startswith("pp"):
+            pp = int(override[2:]); dp = world // pp
+        elif override != "dp":
+            raise ValueError(override)
+        return dp, tp, pp, 1, 0.0
+    from tepdist_amd.planner import AutoParallel
+    t0 = time.time()
+    g = gpt2_ir(cfg, batch=global_batch, seq=seq)
+    plan = AutoParallel(g, world).run()
+    search = time.time() - t0
+    return plan.dp, plan.tp, plan.pp, plan.micro_batches, search
 
 
 def main():
@@ -31,9 +57,11 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", type=str, default="gpt2-345m")
-    ap.add_argument("--micro-batch", type=int, default=8)
-    ap.add_argument("--grad-accum", type=int, default=1)
+    ap.add_argument("--micro-batch", type=int, default=16,
+                    help="per-GPU batch rows per step (weak scaling)")
     ap.add_argument("--seq", type=int, default=1024)
+    ap.add_argument("--parallel", type=str, default="auto",
+                    help="auto | dp | tp<N> | pp<N>")
     ap.add_argument("--device", type=str, default=None,
                     help="override device (cpu for plumbing tests)")
     args = ap.parse_args()
@@ -47,26 +75,36 @@ def main():
 
     cfg = GPT2_CONFIGS[args.model]
     seq = min(args.seq, cfg.n_ctx)
-    dtype = torch.bfloat16
-    torch.manual_seed(1234)
-    model = GPT2(cfg, dtype=dtype).to(device)
-    opt = AdamW(model.parameters(), lr=1e-4)
-    reducer = None
+    dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
+    global_batch = args.micro_batch * world
+
+    # ---- plan (deterministic across ranks; rank 0's choice broadcast) ----
+    search_s = 0.0
+    if rank == 0:
+        dp, tp, pp, micro, search_s = plan_parallelism(
+            cfg, world, global_batch, seq, args.parallel)
     if world > 1:
-        reducer = GradReducer(model.parameters())
-    trainer = Trainer(model, opt, grad_accum_steps=args.grad_accum,
-                      reducer=reducer)
+        t = torch.tensor([dp, tp, pp, micro] if rank == 0 else [0, 0, 0, 0],
+                         dtype=torch.int64)
+        if device.type == "cuda":
+            t = t.to(device)
+        dist.broadcast(t, 0)
+        dp, tp, pp, micro = (int(v) for v in t.tolist())
+    if rank == 0:
+        print(f"# plan: dp={dp} tp={tp} pp={pp} micro={micro} "
+              f"search={search_s:.2f}s", flush=True)
+    search_b = search_s if rank == 0 else 0.0
 
-    # synthetic data of the benchmark shape (no network for datasets)
+    local_batch = max(global_batch // dp, 1)
+    grad_accum = max(min(micro, local_batch), 1) if pp == 1 else 1
+    micro_size = max(local_batch // grad_accum, 1)
+
+    torch.manual_seed(1234)
     g = torch.Generator().manual_seed(4321 + rank)
-    def make_batch(_i):
-        ids = torch.randint(0, cfg.vocab_size, (args.micro_batch, seq + 1),
-                            generator=g)
-        x = ids[:, :-1].to(device)
-        y = ids[:, 1:].to(device)
-        return x, y
 
-    import torch.distributed as dist
+    def make_ids(n_rows):
+        return torch.randint(0, cfg.vocab_size, (n_rows, seq + 1),
+                             generator=g)
 
     def barrier_sync():
         if device.type == "cuda":
@@ -76,25 +114,81 @@ def main():
         if device.type == "cuda":
             torch.cuda.synchronize(device)
 
-    for _ in range(args.warmup):
-        trainer.train_step(make_batch)
+    # ---- build the distributed model per the plan ------------------------
+    if pp == 1:
+        from tepdist_amd.parallel.tp import ParallelEnv
+        env = ParallelEnv.create(tp) if world > 1 else ParallelEnv.single()
+        model = GPT2(cfg, dtype=dtype, env=env).to(device)
+        opt = AdamW(model.parameters(), lr=1e-4)
+        reducer = GradReducer(model.parameters(), env.dp_group) \
+            if dp > 1 else None
+        trainer = Trainer(model, opt, grad_accum_steps=grad_accum,
+                          reducer=reducer)
 
+        def run_step():
+            ids = make_ids(micro_size * grad_accum)
+            def bi(i):
+                sl = ids[i * micro_size:(i + 1) * micro_size]
+                return sl[:, :-1].to(device), sl[:, 1:].to(device)
+            return trainer.train_step(bi)
+    else:
+        # pipeline (x dp): stage-major rank layout [pp, dp]
+        from tepdist_amd.models.gpt2 import GPT2Stage, layer_ranges
+        from tepdist_amd.parallel.pp import PipelineEngine
+        stage = rank // dp
+        dp_rank = rank % dp
+        pp_ranks = [s * dp + dp_rank for s in range(pp)]
+        dp_group = None
+        for s in range(pp):
+            ranks = [s * dp + r for r in range(dp)]
+            grp = dist.new_group(ranks)
+            if rank in ranks:
+                dp_group = grp
+        pp_group = None
+        for r in range(dp):
+            ranks = [s * dp + r for s in range(pp)]
+            grp = dist.new_group(ranks)
+            if rank in ranks:
+                pp_group = grp
+        lo, hi = layer_ranges(cfg.n_layer, pp)[stage]
+        mod = GPT2Stage(cfg, lo, hi, stage == 0, stage == pp - 1,
+                        dtype=dtype).to(device)
+        opt = AdamW(mod.parameters(), lr=1e-4)
+        reducer = GradReducer(mod.parameters(), dp_group) if dp > 1 else None
+        m_per = max(local_batch // micro, 1)
+        eng = PipelineEngine(mod, stage, pp, pp_ranks, micro,
+                             act_shape=(m_per, seq, cfg.n_embd),
+                             act_dtype=dtype, device=device,
+                             reducer=reducer, pp_group=pp_group)
+
+        def run_step():
+            ids = make_ids(m_per * micro)
+            def bi(m):
+                sl = ids[m * m_per:(m + 1) * m_per]
+                return sl[:, :-1].to(device), sl[:, 1:].to(device)
+            loss = eng.train_step(bi)
+            opt.step()
+            opt.zero_grad()
+            return loss
+
+    # ---- measure ---------------------------------------------------------
+    for _ in range(args.warmup):
+        loss = run_step()
     barrier_sync()
     t0 = time.perf_counter()
-    loss = 0.0
     for _ in range(args.steps):
-        loss = trainer.train_step(make_batch)
+        loss = run_step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
-
-    # max over ranks
     if world > 1:
         t = torch.tensor([elapsed], dtype=torch.float64,
                          device=device if device.type == "cuda" else "cpu")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    tokens_per_step = args.micro_batch * args.grad_accum * seq * world
+    rows_per_step = (micro_size * grad_accum * dp) if pp == 1 else \
+        (max(local_batch // micro, 1) * micro * dp)
+    tokens_per_step = rows_per_step * seq
     ms_per_step = elapsed / args.steps * 1000.0
     tokens_per_sec = tokens_per_step * args.steps / elapsed
 
@@ -110,14 +204,17 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
-            "final_loss": round(loss, 4),
+            "final_loss": round(float(loss), 4),
             "config": {
                 "model": args.model,
-                "global_batch": args.micro_batch * args.grad_accum * world,
+                "global_batch": rows_per_step,
                 "seq_len": seq,
-                "parallelism": f"dp{world}",
+                "parallelism": f"auto:dp{dp}tp{tp}pp{pp}micro{micro}"
+                               if args.parallel == "auto"
+                               else f"{args.parallel}:dp{dp}tp{tp}pp{pp}",
+                "plan_search_s": round(search_b, 3),
             },
         }))
 
