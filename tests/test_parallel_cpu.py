@@ -163,3 +163,43 @@ def _a2a_worker(rank, world, port):
 @pytest.mark.timeout(300)
 def test_all_to_all_roundtrip():
     _run(_a2a_worker)
+
+
+def _zero_worker(rank, world, port):
+    _init(rank, world, port)
+    from tepdist_amd.parallel import GradReducer
+    from tepdist_amd.train import AdamW, Trainer
+    from tepdist_amd.train.optim import ZeroAdamW
+
+    cfg = GPT2_CONFIGS["gpt2-test"]
+    model = GPT2(cfg, dtype=torch.float32)
+    opt = ZeroAdamW(model.parameters(), lr=1e-3)
+    reducer = GradReducer(model.parameters(), bucket_bytes=1 << 20)
+    trainer = Trainer(model, opt, reducer=reducer)
+
+    ref = GPT2(cfg, dtype=torch.float32)
+    ref_opt = AdamW(ref.parameters(), lr=1e-3)
+
+    g = torch.Generator().manual_seed(31)
+    ids = torch.randint(0, cfg.vocab_size, (2 * world, 17), generator=g)
+    my = ids[rank * 2:(rank + 1) * 2]
+    for _ in range(3):
+        trainer.train_step(lambda i: (my[:, :-1], my[:, 1:]))
+        # reference: full batch, plain AdamW
+        ref_opt.zero_grad()
+        loss = ref(ids[:, :-1], labels=ids[:, 1:])
+        loss.backward()
+        ref_opt.step()
+    # sharded-optimizer training must match full-state training
+    for (n, p), (_, rp) in zip(model.named_parameters(),
+                               ref.named_parameters()):
+        torch.testing.assert_close(p.data, rp.data, rtol=1e-4, atol=1e-5)
+    # optimizer state really is sharded
+    assert opt.state_bytes() < sum(p.numel() * 12 for p in
+                                   model.parameters())
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_zero1_optimizer_matches_full():
+    _run(_zero_worker)
