@@ -29,7 +29,8 @@ class Conv2d(nn.Module):
         super().__init__()
         self.k, self.stride = k, stride
         self.padding = padding if padding is not None else k // 2
-        self.weight = nn.Parameter(torch.empty(cout, cin, k, k, dtype=dtype))
+        w = torch.randn(cout, cin, k, k) * math.sqrt(2.0 / (cin * k * k))
+        self.weight = nn.Parameter(w.to(dtype))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, C, H, W = x.shape
