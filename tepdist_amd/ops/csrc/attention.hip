@@ -1,0 +1,600 @@
+// Fused flash-style causal attention (forward + backward) for gfx950.
+//
+// Forward: one block = 128 q-rows of one (batch, head); 4 waves x 32 rows.
+// Per KV tile of 64: K staged k-contiguous in LDS (ds_read_b128 B-fragments
+// like the GEMM), V staged transposed via the v_perm register transpose,
+// QK^T on mfma_f32_16x16x32_bf16 with Q fragments held in registers across
+// the whole row, online softmax (running max/sum) in fp32 registers with
+// 16-lane shfl_xor row reductions, P routed through a wave-private LDS tile
+// to become the PV A-operand. Saves per-row logsumexp for backward; the
+// S x S score matrix is never materialized in HBM.
+//
+// Backward (flash2-style): one block = 64 kv-rows of one (b,h); recomputes
+// P^T = exp(K Q^T * scale - lse) per q-tile (both operands natural-layout
+// MFMAs), accumulates dV += P^T dO and dK += dS^T Q in registers,
+// dS^T = P^T * (dP^T - delta_q) * scale with dP^T = V dO^T, and scatters
+// dQ partials with fp32 atomics (delta = rowsum(dO*O) precomputed).
+//
+// Replaces the composed scores-GEMM + softmax + PV path (SURVEY.md §2.9:
+// the reference leaves attention to XLA codegen; this is the MI355X-native
+// fused form).
+
+#include <algorithm>
+#include <stdexcept>
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tepdist {
+
+namespace {
+
+constexpr int NT = 256;
+constexpr int QB = 128;   // q rows per block (fwd)
+constexpr int KB = 64;    // kv rows per tile
+
+// LDS addressing for a [rows][C] bf16 image, C in {64,128}: XOR swizzle of
+// the 16B column slot by bit-reversed (row>>1) (see gemm.hip).
+template <int C>
+DEV_INLINE int loff(int row, int col_e) {
+  const int r1 = row >> 1;
+  const int x = ((r1 & 1) << 2) | (r1 & 2) | ((r1 >> 2) & 1);
+  return row * C + (col_e ^ ((x << 3) & (C - 1)));
+}
+
+DEV_INLINE float warp16_max(float v) {
+#pragma unroll
+  for (int d = 1; d < 16; d <<= 1) v = fmaxf(v, __shfl_xor(v, d, 64));
+  return v;
+}
+
+DEV_INLINE float warp16_sum(float v) {
+#pragma unroll
+  for (int d = 1; d < 16; d <<= 1) v += __shfl_xor(v, d, 64);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+template <int D>
+__launch_bounds__(NT) __global__
+void flash_fwd_kernel(const bf16_t* __restrict__ Q,
+                      const bf16_t* __restrict__ K,
+                      const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
+                      float* __restrict__ LSE, int S, float scale,
+                      bool causal) {
+  constexpr int DK = D / 32;   // k-chunks per fragment row
+  constexpr int DF = D / 16;   // output column fragments
+  const int bh = blockIdx.y;
+  const int q0 = blockIdx.x * QB;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wq0 = q0 + wave * 32;
+
+  const bf16_t* qp = Q + (int64_t)bh * S * D;
+  const bf16_t* kp = K + (int64_t)bh * S * D;
+  const bf16_t* vp = V + (int64_t)bh * S * D;
+
+  __shared__ bf16_t smem[KB * D + D * KB + 4 * 32 * KB];
+  bf16_t* sK = smem;                    // [KB][D] k(=D)-contiguous
+  bf16_t* sVT = smem + KB * D;          // [D][KB] k(=kv)-contiguous
+  bf16_t* sP = smem + KB * D + D * KB + wave * 32 * KB;  // [32][KB]
+
+  // Q fragments for this wave's 32 rows, kept in registers for the whole row
+  bf16x8 qf[2][DK];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int kk = 0; kk < DK; ++kk) {
+      const int row = wq0 + mi * 16 + (lane & 15);
+      bf16x8 v8 = {};
+      if (row < S)
+        v8 = *reinterpret_cast<const bf16x8*>(
+            qp + (int64_t)row * D + kk * 32 + 8 * (lane >> 4));
+      qf[mi][kk] = v8;
+    }
+
+  f32x4 acc_o[2][DF] = {};
+  float m_r[2][4], l_r[2][4];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      m_r[mi][e] = -3.0e38f;
+      l_r[mi][e] = 0.f;
+    }
+
+  const int kv_end = causal ? min(S, q0 + QB) : S;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    // --- stage K tile [KB][D] (k-contiguous rows, 16B loads) ---
+    {
+      constexpr int UN = KB * D / 8 / NT;  // units per thread
+#pragma unroll
+      for (int u = 0; u < UN; ++u) {
+        const int idx = threadIdx.x + u * NT;
+        const int row = idx / (D / 8);
+        const int c = (idx % (D / 8)) * 8;
+        bf16x8 v8 = {};
+        if (kv0 + row < S)
+          v8 = *reinterpret_cast<const bf16x8*>(
+              kp + (int64_t)(kv0 + row) * D + c);
+        *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = v8;
+      }
+    }
+    // --- stage V tile transposed -> [D][KB] via 8x2 v_perm slabs ---
+    {
+      constexpr int NSLAB = (KB / 8) * (D / 2);  // 8kv x 2d slabs
+#pragma unroll
+      for (int u = 0; u < (NSLAB + NT - 1) / NT; ++u) {
+        const int idx = threadIdx.x + u * NT;
+        if (idx < NSLAB) {
+          const int f = 2 * (idx % (D / 2));     // d column pair
+          const int kb = idx / (D / 2);          // kv block of 8
+          uint32_t r[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int kv = kv0 + kb * 8 + j;
+            bf16x2 v2 = {};
+            if (kv < S)
+              v2 = *reinterpret_cast<const bf16x2*>(
+                  vp + (int64_t)kv * D + f);
+            r[j] = __builtin_bit_cast(uint32_t, v2);
+          }
+          uint32_t o0[4], o1[4];
+#pragma unroll
+          for (int d2 = 0; d2 < 4; ++d2) {
+            o0[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2],
+                                           0x05040100u);
+            o1[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2],
+                                           0x07060302u);
+          }
+          *reinterpret_cast<uint4*>(sVT + loff<KB>(f, kb * 8)) =
+              make_uint4(o0[0], o0[1], o0[2], o0[3]);
+          *reinterpret_cast<uint4*>(sVT + loff<KB>(f + 1, kb * 8)) =
+              make_uint4(o1[0], o1[1], o1[2], o1[3]);
+        }
+      }
+    }
+    __syncthreads();
+
+    if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
+      // --- S = Q K^T fragments ---
+      f32x4 sf[2][4] = {};
+#pragma unroll
+      for (int kk = 0; kk < DK; ++kk)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+          const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+              sK + loff<D>((lane & 15) + 16 * nf,
+                           8 * (lane >> 4) + 32 * kk));
+#pragma unroll
+          for (int mi = 0; mi < 2; ++mi)
+            sf[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                qf[mi][kk], kf, sf[mi][nf], 0, 0, 0);
+        }
+      // --- scale + mask; online softmax ---
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int qg = wq0 + mi * 16 + 4 * (lane >> 4) + e;
+          float tmax = -3.0e38f;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            const int kg = kv0 + nf * 16 + (lane & 15);
+            float sv = sf[mi][nf][e] * scale;
+            if ((causal && kg > qg) || kg >= S || qg >= S) sv = -3.0e38f;
+            sf[mi][nf][e] = sv;
+            tmax = fmaxf(tmax, sv);
+          }
+          tmax = warp16_max(tmax);
+          const float mn = fmaxf(m_r[mi][e], tmax);
+          const float alpha = (mn <= -1.0e38f) ? 1.f
+                                               : __expf(m_r[mi][e] - mn);
+          m_r[mi][e] = mn;
+          float rs = 0.f;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            const float sv = sf[mi][nf][e];
+            const float p = (sv <= -1.0e38f) ? 0.f : __expf(sv - mn);
+            sf[mi][nf][e] = p;
+            rs += p;
+          }
+          rs = warp16_sum(rs);
+          l_r[mi][e] = l_r[mi][e] * alpha + rs;
+#pragma unroll
+          for (int df = 0; df < DF; ++df) acc_o[mi][df][e] *= alpha;
+          // write P (bf16) to the wave-private LDS tile in C layout
+          const int prow = mi * 16 + 4 * (lane >> 4) + e;
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf)
+            sP[loff<KB>(prow, nf * 16 + (lane & 15))] =
+                f2bf(sf[mi][nf][e]);
+        }
+      }
+      // ensure the wave's sP writes are visible to its own reads
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      // --- O += P V ---
+#pragma unroll
+      for (int kk2 = 0; kk2 < KB / 32; ++kk2) {
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+          const bf16x8 pf = *reinterpret_cast<const bf16x8*>(
+              sP + loff<KB>(mi * 16 + (lane & 15),
+                            8 * (lane >> 4) + 32 * kk2));
+#pragma unroll
+          for (int df = 0; df < DF; ++df) {
+            const bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+                sVT + loff<KB>((lane & 15) + 16 * df,
+                               8 * (lane >> 4) + 32 * kk2));
+            acc_o[mi][df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                pf, vf, acc_o[mi][df], 0, 0, 0);
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // --- epilogue: O /= l, write O and lse ---
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const int qg = wq0 + mi * 16 + 4 * (lane >> 4) + e;
+      if (qg >= S) continue;
+      const float lv = l_r[mi][e];
+      const float inv = (lv > 0.f) ? 1.0f / lv : 0.f;
+      if ((lane & 15) == 0)
+        LSE[(int64_t)bh * S + qg] =
+            (lv > 0.f) ? m_r[mi][e] + __logf(lv) : -3.0e38f;
+#pragma unroll
+      for (int df = 0; df < DF; ++df)
+        O[(int64_t)bh * S * D + (int64_t)qg * D + df * 16 + (lane & 15)] =
+            f2bf(acc_o[mi][df][e] * inv);
+    }
+}
+
+}  // namespace
+
+void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int bh, int S, int D, float scale,
+                        bool causal, hipStream_t stream) {
+  dim3 grid((S + QB - 1) / QB, bh);
+  dim3 block(NT);
+  if (D == 64) {
+    hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, 0, stream,
+                       static_cast<const bf16_t*>(q),
+                       static_cast<const bf16_t*>(k),
+                       static_cast<const bf16_t*>(v),
+                       static_cast<bf16_t*>(o), lse, S, scale, causal);
+  } else if (D == 128) {
+    hipLaunchKernelGGL(flash_fwd_kernel<128>, grid, block, 0, stream,
+                       static_cast<const bf16_t*>(q),
+                       static_cast<const bf16_t*>(k),
+                       static_cast<const bf16_t*>(v),
+                       static_cast<bf16_t*>(o), lse, S, scale, causal);
+  } else {
+    throw std::runtime_error("flash fwd: head dim must be 64 or 128");
+  }
+}
+
+namespace {
+
+// ---------------------------------------------------------------------------
+// backward
+// ---------------------------------------------------------------------------
+
+// delta[bh][s] = rowsum(dO * O) (flash2 preprocess)
+template <int D>
+__launch_bounds__(NT) __global__
+void flash_bwd_delta_kernel(const bf16_t* __restrict__ dO,
+                            const bf16_t* __restrict__ O,
+                            float* __restrict__ delta, int64_t rows) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int64_t nw = (int64_t)gridDim.x * (NT / WAVE);
+  for (int64_t r = (int64_t)blockIdx.x * (NT / WAVE) + wid; r < rows;
+       r += nw) {
+    float s = 0.f;
+    for (int c = lane * 8; c < D; c += WAVE * 8) {
+      const bf16x8 a = *reinterpret_cast<const bf16x8*>(dO + r * D + c);
+      const bf16x8 b = *reinterpret_cast<const bf16x8*>(O + r * D + c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) s += bf2f(a[e]) * bf2f(b[e]);
+    }
+    s = wave_allreduce_sum(s);
+    if (lane == 0) delta[r] = s;
+  }
+}
+
+// cooperative stage of a [T][D] tile as BOTH natural [T][D] and transposed
+// [D][T] images (v_perm 8x2 slabs for the transpose)
+template <int T, int D>
+DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
+                            bf16_t* nat, bf16_t* tr) {
+  constexpr int UN = T * D / 8 / NT;
+#pragma unroll
+  for (int u = 0; u < UN; ++u) {
+    const int idx = threadIdx.x + u * NT;
+    const int row = idx / (D / 8);
+    const int c = (idx % (D / 8)) * 8;
+    bf16x8 v8 = {};
+    if (row0 + row < S)
+      v8 = *reinterpret_cast<const bf16x8*>(src + (int64_t)(row0 + row) * D + c);
+    *reinterpret_cast<bf16x8*>(nat + loff<D>(row, c)) = v8;
+  }
+  constexpr int NSLAB = (T / 8) * (D / 2);
+#pragma unroll
+  for (int u = 0; u < (NSLAB + NT - 1) / NT; ++u) {
+    const int idx = threadIdx.x + u * NT;
+    if (idx < NSLAB) {
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
+      uint32_t r[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int row = row0 + kb * 8 + j;
+        bf16x2 v2 = {};
+        if (row < S)
+          v2 = *reinterpret_cast<const bf16x2*>(src + (int64_t)row * D + f);
+        r[j] = __builtin_bit_cast(uint32_t, v2);
+      }
+      uint32_t o0[4], o1[4];
+#pragma unroll
+      for (int d2 = 0; d2 < 4; ++d2) {
+        o0[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2], 0x05040100u);
+        o1[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2], 0x07060302u);
+      }
+      *reinterpret_cast<uint4*>(tr + loff<T>(f, kb * 8)) =
+          make_uint4(o0[0], o0[1], o0[2], o0[3]);
+      *reinterpret_cast<uint4*>(tr + loff<T>(f + 1, kb * 8)) =
+          make_uint4(o1[0], o1[1], o1[2], o1[3]);
+    }
+  }
+}
+
+// one block = 64 kv rows of one (b,h); 4 waves x 16 kv rows.
+template <int D>
+__launch_bounds__(NT) __global__
+void flash_bwd_kernel(const bf16_t* __restrict__ Q,
+                      const bf16_t* __restrict__ K,
+                      const bf16_t* __restrict__ V,
+                      const bf16_t* __restrict__ dO,
+                      const float* __restrict__ LSE,
+                      const float* __restrict__ DELTA,
+                      float* __restrict__ dQws, bf16_t* __restrict__ dK,
+                      bf16_t* __restrict__ dV, int S, float scale,
+                      bool causal) {
+  constexpr int DK = D / 32;
+  constexpr int DF = D / 16;
+  constexpr int QT = 64;
+  const int bh = blockIdx.y;
+  const int kv0 = blockIdx.x * KB;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wk0 = kv0 + wave * 16;   // this wave's 16 kv rows
+
+  const int64_t base = (int64_t)bh * S * D;
+  const bf16_t* qp = Q + base;
+  const bf16_t* kp = K + base;
+  const bf16_t* vp = V + base;
+  const bf16_t* dop = dO + base;
+
+  __shared__ bf16_t smem[KB * D * 3 + D * KB + QT * D * 2 + D * QT +
+                         QT * KB + 4 * 16 * QT];
+  bf16_t* sKb = smem;                       // [KB][D] natural
+  bf16_t* sKT = sKb + KB * D;               // [D][KB]
+  bf16_t* sVb = sKT + D * KB;               // [KB][D] natural
+  bf16_t* sQ = sVb + KB * D;                // [QT][D] natural
+  bf16_t* sQT = sQ + QT * D;                // [D][QT]
+  bf16_t* sdO = sQT + D * QT;               // [QT][D] natural
+  bf16_t* sdOT = sdO + QT * D;              // [D][QT]
+  bf16_t* sdS = sdOT + D * QT;              // [QT][KB] shared
+  bf16_t* sPT = sdS + QT * KB + wave * 16 * QT;  // [16][QT] wave-private
+  __shared__ float sLSE[QT], sDELTA[QT];
+
+  // stage K, V tiles (fixed for the block)
+  stage_nat_t<KB, D>(kp, kv0, S, sKb, sKT);
+  {
+    constexpr int UN = KB * D / 8 / NT;
+#pragma unroll
+    for (int u = 0; u < UN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      bf16x8 v8 = {};
+      if (kv0 + row < S)
+        v8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)(kv0 + row) * D + c);
+      *reinterpret_cast<bf16x8*>(sVb + loff<D>(row, c)) = v8;
+    }
+  }
+  __syncthreads();
+
+  // A fragments of this wave's K and V rows (constant across q tiles)
+  bf16x8 kf[DK], vf[DK];
+#pragma unroll
+  for (int kk = 0; kk < DK; ++kk) {
+    kf[kk] = *reinterpret_cast<const bf16x8*>(
+        sKb + loff<D>(wave * 16 + (lane & 15), 8 * (lane >> 4) + 32 * kk));
+    vf[kk] = *reinterpret_cast<const bf16x8*>(
+        sVb + loff<D>(wave * 16 + (lane & 15), 8 * (lane >> 4) + 32 * kk));
+  }
+
+  f32x4 acc_dk[DF] = {};
+  f32x4 acc_dv[DF] = {};
+
+  const int q_start = causal ? (kv0 / QT) * QT : 0;
+  for (int q0 = q_start; q0 < S; q0 += QT) {
+    stage_nat_t<QT, D>(qp, q0, S, sQ, sQT);
+    stage_nat_t<QT, D>(dop, q0, S, sdO, sdOT);
+    for (int i = threadIdx.x; i < QT; i += NT) {
+      const int qg = q0 + i;
+      sLSE[i] = (qg < S) ? LSE[(int64_t)bh * S + qg] : -3.0e38f;
+      sDELTA[i] = (qg < S) ? DELTA[(int64_t)bh * S + qg] : 0.f;
+    }
+    __syncthreads();
+
+    // S^T = K Q^T ; dP^T = V dO^T   (both natural-layout B reads)
+    f32x4 st[4] = {};
+    f32x4 dpt[4] = {};
+#pragma unroll
+    for (int kk = 0; kk < DK; ++kk)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+            sQ + loff<D>((lane & 15) + 16 * nf, 8 * (lane >> 4) + 32 * kk));
+        st[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[kk], qb, st[nf],
+                                                         0, 0, 0);
+        const bf16x8 db = *reinterpret_cast<const bf16x8*>(
+            sdO + loff<D>((lane & 15) + 16 * nf, 8 * (lane >> 4) + 32 * kk));
+        dpt[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[kk], db,
+                                                          dpt[nf], 0, 0, 0);
+      }
+
+    // P^T and dS^T (elementwise, C layout: kv row = 4*(lane>>4)+e within
+    // this wave's 16; q col = q0 + nf*16 + (lane&15))
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int qcol = nf * 16 + (lane & 15);
+      const int qg = q0 + qcol;
+      const float lse = sLSE[qcol];
+      const float dl = sDELTA[qcol];
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const int kvg = wk0 + 4 * (lane >> 4) + e;
+        float pt = 0.f;
+        if (qg < S && kvg < S && (!causal || kvg <= qg) && lse > -1.0e38f)
+          pt = __expf(st[nf][e] * scale - lse);
+        st[nf][e] = pt;                                  // now P^T
+        dpt[nf][e] = pt * (dpt[nf][e] - dl) * scale;     // now dS^T
+      }
+    }
+
+    // write P^T to the wave tile, dV += P^T dO (via sdOT)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int e = 0; e < 4; ++e)
+        sPT[loff<QT>(4 * (lane >> 4) + e, nf * 16 + (lane & 15))] =
+            f2bf(st[nf][e]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
+      const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          sPT + loff<QT>(lane & 15, 8 * (lane >> 4) + 32 * kk2));
+#pragma unroll
+      for (int df = 0; df < DF; ++df) {
+        const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
+            sdOT + loff<QT>((lane & 15) + 16 * df,
+                            8 * (lane >> 4) + 32 * kk2));
+        acc_dv[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, dob,
+                                                             acc_dv[df],
+                                                             0, 0, 0);
+      }
+    }
+
+    // write dS^T to the wave tile (reuse) and to the shared [q][kv] image
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int e = 0; e < 4; ++e) {
+        const bf16_t dsv = f2bf(dpt[nf][e]);
+        sPT[loff<QT>(4 * (lane >> 4) + e, nf * 16 + (lane & 15))] = dsv;
+        sdS[loff<KB>(nf * 16 + (lane & 15),
+                     wave * 16 + 4 * (lane >> 4) + e)] = dsv;
+      }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    // dK += dS^T Q (via sQT)
+#pragma unroll
+    for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
+      const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+          sPT + loff<QT>(lane & 15, 8 * (lane >> 4) + 32 * kk2));
+#pragma unroll
+      for (int df = 0; df < DF; ++df) {
+        const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+            sQT + loff<QT>((lane & 15) + 16 * df,
+                           8 * (lane >> 4) + 32 * kk2));
+        acc_dk[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, qb,
+                                                             acc_dk[df],
+                                                             0, 0, 0);
+      }
+    }
+    __syncthreads();  // sdS complete across waves
+
+    // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS @ K (via sKT)
+    {
+      f32x4 acc_dq[DF] = {};
+#pragma unroll
+      for (int kk2 = 0; kk2 < KB / 32; ++kk2) {
+        const bf16x8 dsa = *reinterpret_cast<const bf16x8*>(
+            sdS + loff<KB>(wave * 16 + (lane & 15),
+                           8 * (lane >> 4) + 32 * kk2));
+#pragma unroll
+        for (int df = 0; df < DF; ++df) {
+          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+              sKT + loff<KB>((lane & 15) + 16 * df,
+                             8 * (lane >> 4) + 32 * kk2));
+          acc_dq[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, kb,
+                                                               acc_dq[df],
+                                                               0, 0, 0);
+        }
+      }
+#pragma unroll
+      for (int df = 0; df < DF; ++df)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int qg = q0 + wave * 16 + 4 * (lane >> 4) + e;
+          if (qg < S)
+            atomicAdd(&dQws[base + (int64_t)qg * D + df * 16 + (lane & 15)],
+                      acc_dq[df][e]);
+        }
+    }
+    __syncthreads();  // before restaging sQ/sdO/sdS
+  }
+
+  // write dK, dV (C layout scatter)
+#pragma unroll
+  for (int df = 0; df < DF; ++df)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+      const int kvg = wk0 + 4 * (lane >> 4) + e;
+      if (kvg >= S) continue;
+      const int64_t off = base + (int64_t)kvg * D + df * 16 + (lane & 15);
+      dK[off] = f2bf(acc_dk[df][e]);
+      dV[off] = f2bf(acc_dv[df][e]);
+    }
+}
+
+}  // namespace
+
+void attention_bwd_bf16(const void* q, const void* k, const void* v,
+                        const void* o, const void* dout, const float* lse,
+                        float* delta, float* dq_ws, void* dk, void* dv,
+                        int bh, int S, int D, float scale, bool causal,
+                        hipStream_t stream) {
+  const int64_t rows = (int64_t)bh * S;
+  const int dblocks = (int)std::min<int64_t>((rows + 3) / 4, 2048);
+  dim3 block(NT);
+#define BWD_D(DD)                                                           \
+  do {                                                                      \
+    hipLaunchKernelGGL(flash_bwd_delta_kernel<DD>, dim3(dblocks), block, 0, \
+                       stream, static_cast<const bf16_t*>(dout),            \
+                       static_cast<const bf16_t*>(o), delta, rows);         \
+    dim3 grid((S + KB - 1) / KB, bh);                                       \
+    hipLaunchKernelGGL(flash_bwd_kernel<DD>, grid, block, 0, stream,        \
+                       static_cast<const bf16_t*>(q),                       \
+                       static_cast<const bf16_t*>(k),                       \
+                       static_cast<const bf16_t*>(v),                       \
+                       static_cast<const bf16_t*>(dout), lse, delta, dq_ws, \
+                       static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),  \
+                       S, scale, causal);                                   \
+  } while (0)
+  if (D == 64) BWD_D(64);
+  else if (D == 128) BWD_D(128);
+  else throw std::runtime_error("flash bwd: head dim must be 64 or 128");
+#undef BWD_D
+}
+}  // namespace tepdist

@@ -92,6 +92,34 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
+  m.def("attention_fwd", [](uintptr_t q, uintptr_t k, uintptr_t v,
+                            uintptr_t o, uintptr_t lse, int bh, int seq,
+                            int D, float scale, bool causal,
+                            uintptr_t stream) {
+    attention_fwd_bf16(reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
+                       reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
+                       reinterpret_cast<float*>(lse), bh, seq, D, scale,
+                       causal, S(stream));
+    check_launch();
+  });
+
+  m.def("attention_bwd", [](uintptr_t q, uintptr_t k, uintptr_t v,
+                            uintptr_t o, uintptr_t dout, uintptr_t lse,
+                            uintptr_t delta, uintptr_t dq_ws, uintptr_t dk,
+                            uintptr_t dv, int bh, int seq, int D, float scale,
+                            bool causal, uintptr_t stream) {
+    attention_bwd_bf16(reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
+                       reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
+                       reinterpret_cast<void*>(dout),
+                       reinterpret_cast<const float*>(lse),
+                       reinterpret_cast<float*>(delta),
+                       reinterpret_cast<float*>(dq_ws),
+                       reinterpret_cast<void*>(dk),
+                       reinterpret_cast<void*>(dv), bh, seq, D, scale,
+                       causal, S(stream));
+    check_launch();
+  });
+
   m.def("embedding_fwd", [](uintptr_t ids, uintptr_t table, uintptr_t out,
                             int64_t n_ids, int dim, uintptr_t stream) {
     embedding_fwd_bf16(reinterpret_cast<const int64_t*>(ids),

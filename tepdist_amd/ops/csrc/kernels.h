@@ -46,6 +46,18 @@ void softmax_fwd_bf16(const void* x, void* p, int64_t rows, int cols, int sq,
 void softmax_bwd_bf16(const void* dp, const void* p, void* ds, int64_t rows,
                       int cols, float scale, hipStream_t stream);
 
+// --- Fused flash attention (causal) ---------------------------------------
+// q,k,v,o: [bh, S, D] bf16 contiguous; lse/delta: [bh, S] f32;
+// dq_ws: zeroed f32 [bh, S, D] accumulated with atomics (cast by caller).
+void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
+                        float* lse, int bh, int S, int D, float scale,
+                        bool causal, hipStream_t stream);
+void attention_bwd_bf16(const void* q, const void* k, const void* v,
+                        const void* o, const void* dout, const float* lse,
+                        float* delta, float* dq_ws, void* dk, void* dv,
+                        int bh, int S, int D, float scale, bool causal,
+                        hipStream_t stream);
+
 // --- Embedding -------------------------------------------------------------
 void embedding_fwd_bf16(const int64_t* ids, const void* table, void* out,
                         int64_t n_ids, int dim, hipStream_t stream);

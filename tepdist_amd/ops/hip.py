@@ -9,6 +9,7 @@ runs; see repo policy / driver's native-code check).
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional, Tuple
 
 import torch
@@ -233,22 +234,56 @@ def softmax_bwd(dp: torch.Tensor, p: torch.Tensor,
     return ds
 
 
+_ATTN_IMPL = os.environ.get("TEPDIST_ATTN", "flash")
+
+
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                   causal: bool = True):
-    scale = 1.0 / math.sqrt(q.shape[-1])
-    scores = matmul(q, k.transpose(-1, -2))
-    p = softmax_fwd(scores, scale=scale, causal=causal)
-    out = matmul(p, v)
-    return out, p
+    """Fused flash attention (no S x S matrix in HBM); saves (out, lse)
+    for the fused backward. Falls back to the composed GEMM+softmax path
+    for head dims other than 64/128 or TEPDIST_ATTN=composed."""
+    B, H, S, D = q.shape
+    scale = 1.0 / math.sqrt(D)
+    if _ATTN_IMPL == "composed" or D not in (64, 128):
+        scores = matmul(q, k.transpose(-1, -2))
+        p = softmax_fwd(scores, scale=scale, causal=causal)
+        out = matmul(p, v)
+        return out, (p,)
+    q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+    out = torch.empty_like(q)
+    lse = torch.empty(B * H, S, dtype=torch.float32, device=q.device)
+    ext.attention_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      out.data_ptr(), lse.data_ptr(), B * H, S, D, scale,
+                      causal, _stream())
+    return out, (out, lse)
 
 
-def attention_bwd(dout: torch.Tensor, q, k, v, p, causal: bool = True):
-    scale = 1.0 / math.sqrt(q.shape[-1])
-    dv = matmul(p.transpose(-1, -2), dout)
-    dp = matmul(dout, v.transpose(-1, -2))
-    ds = softmax_bwd(dp, p, scale=scale)
-    dq = matmul(ds, k)
-    dk = matmul(ds.transpose(-1, -2), q)
+def attention_bwd(dout: torch.Tensor, q, k, v, residuals,
+                  causal: bool = True):
+    B, H, S, D = q.shape
+    scale = 1.0 / math.sqrt(D)
+    if len(residuals) == 1:  # composed path
+        (p,) = residuals
+        dv = matmul(p.transpose(-1, -2), dout)
+        dp = matmul(dout, v.transpose(-1, -2))
+        ds = softmax_bwd(dp, p, scale=scale)
+        dq = matmul(ds, k)
+        dk = matmul(ds.transpose(-1, -2), q)
+        return dq, dk, dv
+    out, lse = residuals
+    dout = dout.contiguous()
+    delta = torch.empty(B * H, S, dtype=torch.float32, device=q.device)
+    dq_ws = torch.zeros(B * H * S * D, dtype=torch.float32, device=q.device)
+    dk = torch.empty_like(k)
+    dv = torch.empty_like(v)
+    ext.attention_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      out.data_ptr(), dout.data_ptr(), lse.data_ptr(),
+                      delta.data_ptr(), dq_ws.data_ptr(), dk.data_ptr(),
+                      dv.data_ptr(), B * H, S, D, scale, causal, _stream())
+    dq = torch.empty_like(q)
+    # reuse the split-K reducer as an fp32 -> bf16 cast (nparts=1)
+    ext.splitk_reduce(dq_ws.data_ptr(), dq.data_ptr(), 1, dq_ws.numel(),
+                      _stream())
     return dq, dk, dv
 
 

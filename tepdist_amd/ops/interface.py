@@ -139,16 +139,17 @@ class AttentionFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         be = _backend(q)
-        out, p = be.attention_fwd(q, k, v, causal=causal)
-        ctx.save_for_backward(q, k, v, p)
+        out, residuals = be.attention_fwd(q, k, v, causal=causal)
+        ctx.save_for_backward(q, k, v, *residuals)
         ctx.causal = causal
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, p = ctx.saved_tensors
+        q, k, v, *residuals = ctx.saved_tensors
         be = _backend(dout)
-        dq, dk, dv = be.attention_bwd(dout.contiguous(), q, k, v, p, causal=ctx.causal)
+        dq, dk, dv = be.attention_bwd(dout.contiguous(), q, k, v,
+                                      tuple(residuals), causal=ctx.causal)
         return dq, dk, dv, None
 
 

@@ -145,15 +145,18 @@ def softmax_bwd(dp: torch.Tensor, p: torch.Tensor, scale: float = 1.0) -> torch.
 
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                   causal: bool = True):
-    """q,k,v: [B, H, S, D]. Returns (out, p) with p saved for backward."""
+    """q,k,v: [B, H, S, D]. Returns (out, residuals) with the residuals
+    saved for backward (here: the attention probabilities)."""
     scale = 1.0 / math.sqrt(q.shape[-1])
     scores = matmul(q, k.transpose(-1, -2))
     p = softmax_fwd(scores, scale=scale, causal=causal)
     out = matmul(p, v)
-    return out, p
+    return out, (p,)
 
 
-def attention_bwd(dout: torch.Tensor, q, k, v, p, causal: bool = True):
+def attention_bwd(dout: torch.Tensor, q, k, v, residuals,
+                  causal: bool = True):
+    (p,) = residuals
     scale = 1.0 / math.sqrt(q.shape[-1])
     dv = matmul(p.transpose(-1, -2), dout)
     dp = matmul(dout, v.transpose(-1, -2))
