@@ -277,3 +277,39 @@ def test_tr16_probe_mapping():
         [[l & 3 for _ in range(4)] for l in range(64)], dtype=torch.int32)
     assert torch.equal(pat, expect_pat), f"tr16 pattern mapping:\n{pat}"
     assert torch.equal(uni, expect_uni), f"tr16 uniform mapping:\n{uni}"
+
+
+def test_flash_attention_large_vs_sdpa():
+    """Flash fwd+bwd at benchmark shape (S=1024, D=64) vs fp32 SDPA."""
+    B, H, S, D = 4, 8, 1024, 64
+    q = _randn(B, H, S, D, seed=40)
+    k = _randn(B, H, S, D, seed=41)
+    v = _randn(B, H, S, D, seed=42)
+    out, res = hip.attention_fwd(q, k, v, causal=True)
+    assert len(res) == 2, "flash path should save (out, lse)"
+    dout = _randn(B, H, S, D, seed=43)
+    dq, dk, dv = hip.attention_bwd(dout, q, k, v, res, causal=True)
+    torch.cuda.synchronize()
+
+    q2, k2, v2 = (t.float().requires_grad_() for t in (q, k, v))
+    ref = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2,
+                                                           is_causal=True)
+    ref.backward(dout.float())
+    _assert_close_bf16(out, ref.detach(), rtol=3e-2)
+    _assert_close_bf16(dq, q2.grad, rtol=5e-2, scale=q2.grad.abs().max())
+    _assert_close_bf16(dk, k2.grad, rtol=5e-2, scale=k2.grad.abs().max())
+    _assert_close_bf16(dv, v2.grad, rtol=5e-2, scale=v2.grad.abs().max())
+
+
+def test_flash_attention_ragged_seq():
+    """Non-multiple-of-tile sequence lengths."""
+    B, H, S, D = 2, 2, 200, 64
+    q = _randn(B, H, S, D, seed=50)
+    k = _randn(B, H, S, D, seed=51)
+    v = _randn(B, H, S, D, seed=52)
+    out, res = hip.attention_fwd(q, k, v, causal=True)
+    torch.cuda.synchronize()
+    q2, k2, v2 = (t.float() for t in (q, k, v))
+    ref = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2,
+                                                           is_causal=True)
+    _assert_close_bf16(out, ref, rtol=3e-2)
