@@ -70,14 +70,8 @@ class GPT2Block(nn.Module):
             self.out = RowParallelLinear(4 * d, d, env, bias=True, dtype=dtype)
 
     def _attn(self, qkv: torch.Tensor, B: int, S: int) -> torch.Tensor:
-        Hl, hd = self.n_head_local, self.cfg.n_embd // self.cfg.n_head
-        dl = Hl * hd
-        q, k, v = qkv.split(dl, dim=-1)
-        q = q.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
-        k = k.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
-        v = v.reshape(B, S, Hl, hd).transpose(1, 2).contiguous()
-        a = ops.attention(q, k, v, causal=True)  # [B,Hl,S,hd]
-        return a.transpose(1, 2).reshape(B, S, dl).contiguous()
+        # packed-qkv fused attention: no transpose copies on the GPU path
+        return ops.attention_qkv(qkv, self.n_head_local, causal=True)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, S, d = x.shape

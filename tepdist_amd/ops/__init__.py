@@ -4,6 +4,7 @@ from tepdist_amd.ops.interface import (  # noqa: F401
     layernorm,
     softmax,
     attention,
+    attention_qkv,
     embedding,
     cross_entropy,
     dropout,

@@ -63,19 +63,24 @@ __launch_bounds__(NT) __global__
 void flash_fwd_kernel(const bf16_t* __restrict__ Q,
                       const bf16_t* __restrict__ K,
                       const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
-                      float* __restrict__ LSE, int S, float scale,
-                      bool causal) {
+                      float* __restrict__ LSE, int S, int H, float scale,
+                      bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                      int64_t o_bs, int64_t o_hs, int64_t o_rs) {
   constexpr int DK = D / 32;   // k-chunks per fragment row
   constexpr int DF = D / 16;   // output column fragments
   const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
   const int q0 = blockIdx.x * QB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wq0 = q0 + wave * 32;
 
-  const bf16_t* qp = Q + (int64_t)bh * S * D;
-  const bf16_t* kp = K + (int64_t)bh * S * D;
-  const bf16_t* vp = V + (int64_t)bh * S * D;
+  // q/k/v share the (bs, hs, rs) layout (a packed qkv differs only in the
+  // base pointer); D stays contiguous
+  const bf16_t* qp = Q + (int64_t)b * q_bs + (int64_t)h * q_hs;
+  const bf16_t* kp = K + (int64_t)b * q_bs + (int64_t)h * q_hs;
+  const bf16_t* vp = V + (int64_t)b * q_bs + (int64_t)h * q_hs;
+  bf16_t* op = O + (int64_t)b * o_bs + (int64_t)h * o_hs;
 
   __shared__ bf16_t smem[KB * D + D * KB + 4 * 32 * KB];
   bf16_t* sK = smem;                    // [KB][D] k(=D)-contiguous
@@ -92,7 +97,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       bf16x8 v8 = {};
       if (row < S)
         v8 = *reinterpret_cast<const bf16x8*>(
-            qp + (int64_t)row * D + kk * 32 + 8 * (lane >> 4));
+            qp + (int64_t)row * q_rs + kk * 32 + 8 * (lane >> 4));
       qf[mi][kk] = v8;
     }
 
@@ -119,7 +124,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
         bf16x8 v8 = {};
         if (kv0 + row < S)
           v8 = *reinterpret_cast<const bf16x8*>(
-              kp + (int64_t)(kv0 + row) * D + c);
+              kp + (int64_t)(kv0 + row) * q_rs + c);
         *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = v8;
       }
     }
@@ -139,7 +144,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
             bf16x2 v2 = {};
             if (kv < S)
               v2 = *reinterpret_cast<const bf16x2*>(
-                  vp + (int64_t)kv * D + f);
+                  vp + (int64_t)kv * q_rs + f);
             r[j] = __builtin_bit_cast(uint32_t, v2);
           }
           uint32_t o0[4], o1[4];
@@ -252,7 +257,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
             (lv > 0.f) ? m_r[mi][e] + __logf(lv) : -3.0e38f;
 #pragma unroll
       for (int df = 0; df < DF; ++df)
-        O[(int64_t)bh * S * D + (int64_t)qg * D + df * 16 + (lane & 15)] =
+        op[(int64_t)qg * o_rs + df * 16 + (lane & 15)] =
             f2bf(acc_o[mi][df][e] * inv);
     }
 }
@@ -260,25 +265,23 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
 }  // namespace
 
 void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
-                        float* lse, int bh, int S, int D, float scale,
-                        bool causal, hipStream_t stream) {
-  dim3 grid((S + QB - 1) / QB, bh);
+                        float* lse, int B, int H, int S, int D, float scale,
+                        bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                        int64_t o_bs, int64_t o_hs, int64_t o_rs,
+                        hipStream_t stream) {
+  dim3 grid((S + QB - 1) / QB, B * H);
   dim3 block(NT);
-  if (D == 64) {
-    hipLaunchKernelGGL(flash_fwd_kernel<64>, grid, block, 0, stream,
-                       static_cast<const bf16_t*>(q),
-                       static_cast<const bf16_t*>(k),
-                       static_cast<const bf16_t*>(v),
-                       static_cast<bf16_t*>(o), lse, S, scale, causal);
-  } else if (D == 128) {
-    hipLaunchKernelGGL(flash_fwd_kernel<128>, grid, block, 0, stream,
-                       static_cast<const bf16_t*>(q),
-                       static_cast<const bf16_t*>(k),
-                       static_cast<const bf16_t*>(v),
-                       static_cast<bf16_t*>(o), lse, S, scale, causal);
-  } else {
-    throw std::runtime_error("flash fwd: head dim must be 64 or 128");
-  }
+#define FWD_D(DD)                                                       \
+  hipLaunchKernelGGL(flash_fwd_kernel<DD>, grid, block, 0, stream,      \
+                     static_cast<const bf16_t*>(q),                     \
+                     static_cast<const bf16_t*>(k),                     \
+                     static_cast<const bf16_t*>(v),                     \
+                     static_cast<bf16_t*>(o), lse, S, H, scale, causal, \
+                     q_bs, q_hs, q_rs, o_bs, o_hs, o_rs)
+  if (D == 64) FWD_D(64);
+  else if (D == 128) FWD_D(128);
+  else throw std::runtime_error("flash fwd: head dim must be 64 or 128");
+#undef FWD_D
 }
 
 namespace {
@@ -314,7 +317,7 @@ void flash_bwd_delta_kernel(const bf16_t* __restrict__ dO,
 // [D][T] images (v_perm 8x2 slabs for the transpose)
 template <int T, int D>
 DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
-                            bf16_t* nat, bf16_t* tr) {
+                            int64_t rs, bf16_t* nat, bf16_t* tr) {
   constexpr int UN = T * D / 8 / NT;
 #pragma unroll
   for (int u = 0; u < UN; ++u) {
@@ -323,7 +326,8 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
     const int c = (idx % (D / 8)) * 8;
     bf16x8 v8 = {};
     if (row0 + row < S)
-      v8 = *reinterpret_cast<const bf16x8*>(src + (int64_t)(row0 + row) * D + c);
+      v8 = *reinterpret_cast<const bf16x8*>(
+          src + (int64_t)(row0 + row) * rs + c);
     *reinterpret_cast<bf16x8*>(nat + loff<D>(row, c)) = v8;
   }
   constexpr int NSLAB = (T / 8) * (D / 2);
@@ -339,7 +343,7 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
         const int row = row0 + kb * 8 + j;
         bf16x2 v2 = {};
         if (row < S)
-          v2 = *reinterpret_cast<const bf16x2*>(src + (int64_t)row * D + f);
+          v2 = *reinterpret_cast<const bf16x2*>(src + (int64_t)row * rs + f);
         r[j] = __builtin_bit_cast(uint32_t, v2);
       }
       uint32_t o0[4], o1[4];
@@ -366,22 +370,25 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
                       const float* __restrict__ LSE,
                       const float* __restrict__ DELTA,
                       float* __restrict__ dQws, bf16_t* __restrict__ dK,
-                      bf16_t* __restrict__ dV, int S, float scale,
-                      bool causal) {
+                      bf16_t* __restrict__ dV, int S, int H, float scale,
+                      bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                      int64_t o_bs, int64_t o_hs, int64_t o_rs) {
   constexpr int DK = D / 32;
   constexpr int DF = D / 16;
   constexpr int QT = 64;
   const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
   const int kv0 = blockIdx.x * KB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wk0 = kv0 + wave * 16;   // this wave's 16 kv rows
 
-  const int64_t base = (int64_t)bh * S * D;
-  const bf16_t* qp = Q + base;
-  const bf16_t* kp = K + base;
-  const bf16_t* vp = V + base;
-  const bf16_t* dop = dO + base;
+  const int64_t qoff = (int64_t)b * q_bs + (int64_t)h * q_hs;
+  const int64_t ooff = (int64_t)b * o_bs + (int64_t)h * o_hs;
+  const bf16_t* qp = Q + qoff;
+  const bf16_t* kp = K + qoff;
+  const bf16_t* vp = V + qoff;
+  const bf16_t* dop = dO + ooff;
 
   __shared__ bf16_t smem[KB * D * 3 + D * KB + QT * D * 2 + D * QT +
                          QT * KB + 4 * 16 * QT];
@@ -397,7 +404,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   __shared__ float sLSE[QT], sDELTA[QT];
 
   // stage K, V tiles (fixed for the block)
-  stage_nat_t<KB, D>(kp, kv0, S, sKb, sKT);
+  stage_nat_t<KB, D>(kp, kv0, S, q_rs, sKb, sKT);
   {
     constexpr int UN = KB * D / 8 / NT;
 #pragma unroll
@@ -407,7 +414,8 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       const int c = (idx % (D / 8)) * 8;
       bf16x8 v8 = {};
       if (kv0 + row < S)
-        v8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)(kv0 + row) * D + c);
+        v8 = *reinterpret_cast<const bf16x8*>(
+            vp + (int64_t)(kv0 + row) * q_rs + c);
       *reinterpret_cast<bf16x8*>(sVb + loff<D>(row, c)) = v8;
     }
   }
@@ -428,8 +436,8 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
 
   const int q_start = causal ? (kv0 / QT) * QT : 0;
   for (int q0 = q_start; q0 < S; q0 += QT) {
-    stage_nat_t<QT, D>(qp, q0, S, sQ, sQT);
-    stage_nat_t<QT, D>(dop, q0, S, sdO, sdOT);
+    stage_nat_t<QT, D>(qp, q0, S, q_rs, sQ, sQT);
+    stage_nat_t<QT, D>(dop, q0, S, o_rs, sdO, sdOT);
     for (int i = threadIdx.x; i < QT; i += NT) {
       const int qg = q0 + i;
       sLSE[i] = (qg < S) ? LSE[(int64_t)bh * S + qg] : -3.0e38f;
@@ -548,7 +556,8 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
         for (int e = 0; e < 4; ++e) {
           const int qg = q0 + wave * 16 + 4 * (lane >> 4) + e;
           if (qg < S)
-            atomicAdd(&dQws[base + (int64_t)qg * D + df * 16 + (lane & 15)],
+            atomicAdd(&dQws[(int64_t)bh * S * D + (int64_t)qg * D +
+                            df * 16 + (lane & 15)],
                       acc_dq[df][e]);
         }
     }
@@ -562,35 +571,105 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
     for (int e = 0; e < 4; ++e) {
       const int kvg = wk0 + 4 * (lane >> 4) + e;
       if (kvg >= S) continue;
-      const int64_t off = base + (int64_t)kvg * D + df * 16 + (lane & 15);
+      const int64_t off = qoff + (int64_t)kvg * q_rs + df * 16 + (lane & 15);
       dK[off] = f2bf(acc_dk[df][e]);
       dV[off] = f2bf(acc_dv[df][e]);
     }
 }
 
+
+
+// delta kernel needs contiguous dO/O rows: the host passes per-(b,h)
+// strided views by iterating bh in the kernel via strides
+template <int D>
+__global__ void flash_bwd_delta_strided_kernel(
+    const bf16_t* __restrict__ dO, const bf16_t* __restrict__ O,
+    float* __restrict__ delta, int H, int S, int64_t o_bs, int64_t o_hs,
+    int64_t o_rs) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int64_t rows = (int64_t)gridDim.y * S;  // gridDim.y == B*H
+  const int64_t nw = (int64_t)gridDim.x * (NT / WAVE);
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int64_t off = (int64_t)b * o_bs + (int64_t)h * o_hs;
+  for (int64_t r = (int64_t)blockIdx.x * (NT / WAVE) + wid; r < S; r += nw) {
+    float s = 0.f;
+    for (int c = lane * 8; c < D; c += WAVE * 8) {
+      const bf16x8 a =
+          *reinterpret_cast<const bf16x8*>(dO + off + r * o_rs + c);
+      const bf16x8 bb =
+          *reinterpret_cast<const bf16x8*>(O + off + r * o_rs + c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) s += bf2f(a[e]) * bf2f(bb[e]);
+    }
+    s = wave_allreduce_sum(s);
+    if (lane == 0) delta[(int64_t)bh * S + r] = s;
+  }
+}
+
 }  // namespace
+
+
+namespace {
+__global__ void cast_scatter_kernel(const float* __restrict__ src,
+                                    bf16_t* __restrict__ dst, int H, int S,
+                                    int D, int64_t bs, int64_t hs,
+                                    int64_t rs) {
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int64_t base = (int64_t)b * bs + (int64_t)h * hs;
+  const int64_t n = (int64_t)S * D;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+       i < n; i += stride) {
+    const int64_t row = i / D;
+    const int c = (int)(i - row * D);
+    const f32x4 v = *reinterpret_cast<const f32x4*>(
+        src + (int64_t)bh * n + i);
+    bf16x4 o;
+#pragma unroll
+    for (int e = 0; e < 4; ++e) o[e] = f2bf(v[e]);
+    *reinterpret_cast<bf16x4*>(dst + base + row * rs + c) = o;
+  }
+}
+}  // namespace
+
+void cast_scatter_bf16(const float* src, void* dst, int B, int H, int S,
+                       int D, int64_t bs, int64_t hs, int64_t rs,
+                       hipStream_t stream) {
+  const int blocks = (int)std::min<int64_t>(
+      ((int64_t)S * D / 4 + NT - 1) / NT, 512);
+  hipLaunchKernelGGL(cast_scatter_kernel, dim3(std::max(blocks, 1), B * H),
+                     dim3(NT), 0, stream, src, static_cast<bf16_t*>(dst), H,
+                     S, D, bs, hs, rs);
+}
 
 void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         const void* o, const void* dout, const float* lse,
                         float* delta, float* dq_ws, void* dk, void* dv,
-                        int bh, int S, int D, float scale, bool causal,
+                        int B, int H, int S, int D, float scale, bool causal,
+                        int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                        int64_t o_bs, int64_t o_hs, int64_t o_rs,
                         hipStream_t stream) {
-  const int64_t rows = (int64_t)bh * S;
-  const int dblocks = (int)std::min<int64_t>((rows + 3) / 4, 2048);
   dim3 block(NT);
-#define BWD_D(DD)                                                           \
-  do {                                                                      \
-    hipLaunchKernelGGL(flash_bwd_delta_kernel<DD>, dim3(dblocks), block, 0, \
-                       stream, static_cast<const bf16_t*>(dout),            \
-                       static_cast<const bf16_t*>(o), delta, rows);         \
-    dim3 grid((S + KB - 1) / KB, bh);                                       \
-    hipLaunchKernelGGL(flash_bwd_kernel<DD>, grid, block, 0, stream,        \
-                       static_cast<const bf16_t*>(q),                       \
-                       static_cast<const bf16_t*>(k),                       \
-                       static_cast<const bf16_t*>(v),                       \
-                       static_cast<const bf16_t*>(dout), lse, delta, dq_ws, \
-                       static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),  \
-                       S, scale, causal);                                   \
+#define BWD_D(DD)                                                            \
+  do {                                                                       \
+    dim3 dgrid((S / 4 + (NT / WAVE) - 1) / (NT / WAVE) > 256                 \
+                   ? 256 : (S + 3) / 4, B * H);                              \
+    hipLaunchKernelGGL(flash_bwd_delta_strided_kernel<DD>, dgrid, block, 0,  \
+                       stream, static_cast<const bf16_t*>(dout),             \
+                       static_cast<const bf16_t*>(o), delta, H, S, o_bs,     \
+                       o_hs, o_rs);                                          \
+    dim3 grid((S + KB - 1) / KB, B * H);                                     \
+    hipLaunchKernelGGL(flash_bwd_kernel<DD>, grid, block, 0, stream,         \
+                       static_cast<const bf16_t*>(q),                        \
+                       static_cast<const bf16_t*>(k),                        \
+                       static_cast<const bf16_t*>(v),                        \
+                       static_cast<const bf16_t*>(dout), lse, delta, dq_ws,  \
+                       static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),   \
+                       S, H, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs,    \
+                       o_rs);                                                \
   } while (0)
   if (D == 64) BWD_D(64);
   else if (D == 128) BWD_D(128);

@@ -93,21 +93,25 @@ PYBIND11_MODULE(_tepdist_hip, m) {
   });
 
   m.def("attention_fwd", [](uintptr_t q, uintptr_t k, uintptr_t v,
-                            uintptr_t o, uintptr_t lse, int bh, int seq,
-                            int D, float scale, bool causal,
+                            uintptr_t o, uintptr_t lse, int B, int H,
+                            int seq, int D, float scale, bool causal,
+                            int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                            int64_t o_bs, int64_t o_hs, int64_t o_rs,
                             uintptr_t stream) {
     attention_fwd_bf16(reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
                        reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
-                       reinterpret_cast<float*>(lse), bh, seq, D, scale,
-                       causal, S(stream));
+                       reinterpret_cast<float*>(lse), B, H, seq, D, scale,
+                       causal, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, S(stream));
     check_launch();
   });
 
   m.def("attention_bwd", [](uintptr_t q, uintptr_t k, uintptr_t v,
                             uintptr_t o, uintptr_t dout, uintptr_t lse,
                             uintptr_t delta, uintptr_t dq_ws, uintptr_t dk,
-                            uintptr_t dv, int bh, int seq, int D, float scale,
-                            bool causal, uintptr_t stream) {
+                            uintptr_t dv, int B, int H, int seq, int D,
+                            float scale, bool causal, int64_t q_bs,
+                            int64_t q_hs, int64_t q_rs, int64_t o_bs,
+                            int64_t o_hs, int64_t o_rs, uintptr_t stream) {
     attention_bwd_bf16(reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
                        reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
                        reinterpret_cast<void*>(dout),
@@ -115,8 +119,18 @@ PYBIND11_MODULE(_tepdist_hip, m) {
                        reinterpret_cast<float*>(delta),
                        reinterpret_cast<float*>(dq_ws),
                        reinterpret_cast<void*>(dk),
-                       reinterpret_cast<void*>(dv), bh, seq, D, scale,
-                       causal, S(stream));
+                       reinterpret_cast<void*>(dv), B, H, seq, D, scale,
+                       causal, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs,
+                       S(stream));
+    check_launch();
+  });
+
+  m.def("cast_scatter", [](uintptr_t src, uintptr_t dst, int B, int H,
+                           int seq, int D, int64_t bs, int64_t hs,
+                           int64_t rs, uintptr_t stream) {
+    cast_scatter_bf16(reinterpret_cast<const float*>(src),
+                      reinterpret_cast<void*>(dst), B, H, seq, D, bs, hs,
+                      rs, S(stream));
     check_launch();
   });
 

@@ -47,16 +47,27 @@ void softmax_bwd_bf16(const void* dp, const void* p, void* ds, int64_t rows,
                       int cols, float scale, hipStream_t stream);
 
 // --- Fused flash attention (causal) ---------------------------------------
-// q,k,v,o: [bh, S, D] bf16 contiguous; lse/delta: [bh, S] f32;
-// dq_ws: zeroed f32 [bh, S, D] accumulated with atomics (cast by caller).
+// Strided [B, H, S, D] addressing (D contiguous): q/k/v share
+// (q_bs, q_hs, q_rs) batch/head/row strides — a packed [B,S,3d] qkv is
+// three base pointers with the same strides; o/dout use (o_bs, o_hs, o_rs).
+// lse/delta: [B*H, S] f32; dq_ws: zeroed f32 [B*H, S, D] accumulated with
+// atomics (cast/scattered by the caller).
 void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
-                        float* lse, int bh, int S, int D, float scale,
-                        bool causal, hipStream_t stream);
+                        float* lse, int B, int H, int S, int D, float scale,
+                        bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                        int64_t o_bs, int64_t o_hs, int64_t o_rs,
+                        hipStream_t stream);
 void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         const void* o, const void* dout, const float* lse,
                         float* delta, float* dq_ws, void* dk, void* dv,
-                        int bh, int S, int D, float scale, bool causal,
+                        int B, int H, int S, int D, float scale, bool causal,
+                        int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                        int64_t o_bs, int64_t o_hs, int64_t o_rs,
                         hipStream_t stream);
+
+void cast_scatter_bf16(const float* src, void* dst, int B, int H, int S,
+                       int D, int64_t bs, int64_t hs, int64_t rs,
+                       hipStream_t stream);
 
 // --- Embedding -------------------------------------------------------------
 void embedding_fwd_bf16(const int64_t* ids, const void* table, void* out,

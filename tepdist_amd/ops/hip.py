@@ -252,9 +252,10 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
     out = torch.empty_like(q)
     lse = torch.empty(B * H, S, dtype=torch.float32, device=q.device)
+    sd = (H * S * D, S * D, D)
     ext.attention_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                      out.data_ptr(), lse.data_ptr(), B * H, S, D, scale,
-                      causal, _stream())
+                      out.data_ptr(), lse.data_ptr(), B, H, S, D, scale,
+                      causal, *sd, *sd, _stream())
     return out, (out, lse)
 
 
@@ -276,15 +277,74 @@ def attention_bwd(dout: torch.Tensor, q, k, v, residuals,
     dq_ws = torch.zeros(B * H * S * D, dtype=torch.float32, device=q.device)
     dk = torch.empty_like(k)
     dv = torch.empty_like(v)
+    sd = (H * S * D, S * D, D)
     ext.attention_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       out.data_ptr(), dout.data_ptr(), lse.data_ptr(),
                       delta.data_ptr(), dq_ws.data_ptr(), dk.data_ptr(),
-                      dv.data_ptr(), B * H, S, D, scale, causal, _stream())
+                      dv.data_ptr(), B, H, S, D, scale, causal, *sd, *sd,
+                      _stream())
     dq = torch.empty_like(q)
     # reuse the split-K reducer as an fp32 -> bf16 cast (nparts=1)
     ext.splitk_reduce(dq_ws.data_ptr(), dq.data_ptr(), 1, dq_ws.numel(),
                       _stream())
     return dq, dk, dv
+
+
+def attention_qkv_fwd(qkv: torch.Tensor, heads: int, causal: bool = True):
+    """Packed-projection attention: qkv [B, S, 3*H*D] (q|k|v each [H][D]),
+    out [B, S, H*D] — the flash kernels read/write the packed layout
+    directly (no transpose copies)."""
+    B, S, d3 = qkv.shape
+    d = d3 // 3
+    D = d // heads
+    if D not in (64, 128) or _ATTN_IMPL == "composed":
+        return None  # caller falls back to the split path
+    qkv = qkv.contiguous()
+    scale = 1.0 / math.sqrt(D)
+    out = torch.empty(B, S, d, dtype=qkv.dtype, device=qkv.device)
+    lse = torch.empty(B * heads, S, dtype=torch.float32, device=qkv.device)
+    qs = (S * d3, D, d3)
+    os_ = (S * d, D, d)
+    base = qkv.data_ptr()
+    esz = qkv.element_size()
+    ext.attention_fwd(base, base + d * esz, base + 2 * d * esz,
+                      out.data_ptr(), lse.data_ptr(), B, heads, S, D, scale,
+                      causal, *qs, *os_, _stream())
+    return out, (out, lse)
+
+
+def attention_qkv_bwd(dout: torch.Tensor, qkv: torch.Tensor, heads: int,
+                      residuals, causal: bool = True):
+    B, S, d3 = qkv.shape
+    d = d3 // 3
+    D = d // heads
+    if len(residuals) != 2:  # composed fallback (q, k, v, p)
+        q, k, v, p = residuals
+        dout4 = dout.reshape(B, S, heads, D).transpose(1, 2).contiguous()
+        dq, dk, dv = attention_bwd(dout4, q, k, v, (p,), causal=causal)
+        def back(t):
+            return t.transpose(1, 2).reshape(B, S, d)
+        return torch.cat([back(dq), back(dk), back(dv)], dim=-1)
+    out, lse = residuals
+    scale = 1.0 / math.sqrt(D)
+    dout = dout.contiguous()
+    delta = torch.empty(B * heads, S, dtype=torch.float32, device=qkv.device)
+    dq_ws = torch.zeros(B * heads * S * D, dtype=torch.float32,
+                        device=qkv.device)
+    dqkv = torch.empty_like(qkv)
+    qs = (S * d3, D, d3)
+    os_ = (S * d, D, d)
+    base = qkv.data_ptr()
+    dbase = dqkv.data_ptr()
+    esz = qkv.element_size()
+    ext.attention_bwd(base, base + d * esz, base + 2 * d * esz,
+                      out.data_ptr(), dout.data_ptr(), lse.data_ptr(),
+                      delta.data_ptr(), dq_ws.data_ptr(), dbase + d * esz,
+                      dbase + 2 * d * esz, B, heads, S, D, scale, causal,
+                      *qs, *os_, _stream())
+    # scatter the dense fp32 dQ partials into the packed dqkv
+    ext.cast_scatter(dq_ws.data_ptr(), dbase, B, heads, S, D, *qs, _stream())
+    return dqkv
 
 
 # --------------------------------------------------------------------------

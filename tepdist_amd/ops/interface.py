@@ -157,6 +157,47 @@ def attention(q, k, v, causal: bool = True):
     return AttentionFn.apply(q, k, v, causal)
 
 
+class AttentionQKVFn(torch.autograd.Function):
+    """Attention on the packed qkv projection [B, S, 3*H*D]: the GPU flash
+    kernels consume/produce the packed layout directly (no transposes)."""
+
+    @staticmethod
+    def forward(ctx, qkv, heads, causal):
+        be = _backend(qkv)
+        r = be.attention_qkv_fwd(qkv, heads, causal=causal) \
+            if hasattr(be, "attention_qkv_fwd") else None
+        if r is None:  # hip fallback for unsupported head dims
+            r = ref.attention_qkv_fwd(qkv.cpu(), heads, causal) \
+                if not qkv.is_cuda else _qkv_composed(be, qkv, heads, causal)
+        out, residuals = r
+        ctx.save_for_backward(qkv, *residuals)
+        ctx.heads = heads
+        ctx.causal = causal
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, *residuals = ctx.saved_tensors
+        be = _backend(dout)
+        dqkv = be.attention_qkv_bwd(dout.contiguous(), qkv, ctx.heads,
+                                    tuple(residuals), causal=ctx.causal)
+        return dqkv, None, None
+
+
+def _qkv_composed(be, qkv, heads, causal):
+    B, S, d3 = qkv.shape
+    d = d3 // 3
+    D = d // heads
+    q, k, v = (t.reshape(B, S, heads, D).transpose(1, 2).contiguous()
+               for t in qkv.split(d, dim=-1))
+    out4, res = be.attention_fwd(q, k, v, causal=causal)
+    return out4.transpose(1, 2).reshape(B, S, d), (q, k, v) + res
+
+
+def attention_qkv(qkv, heads: int, causal: bool = True):
+    return AttentionQKVFn.apply(qkv, heads, causal)
+
+
 # --------------------------------------------------------------------------
 
 

@@ -166,6 +166,32 @@ def attention_bwd(dout: torch.Tensor, q, k, v, residuals,
     return dq, dk, dv
 
 
+def attention_qkv_fwd(qkv: torch.Tensor, heads: int, causal: bool = True):
+    """Packed-projection attention (reference): split + transpose, run the
+    composed path, return residuals carrying what backward needs."""
+    B, S, d3 = qkv.shape
+    d = d3 // 3
+    D = d // heads
+    q, k, v = (t.reshape(B, S, heads, D).transpose(1, 2).contiguous()
+               for t in qkv.split(d, dim=-1))
+    out4, (p,) = attention_fwd(q, k, v, causal=causal)
+    out = out4.transpose(1, 2).reshape(B, S, d)
+    return out, (q, k, v, p)
+
+
+def attention_qkv_bwd(dout: torch.Tensor, qkv: torch.Tensor, heads: int,
+                      residuals, causal: bool = True):
+    B, S, d3 = qkv.shape
+    d = d3 // 3
+    D = d // heads
+    q, k, v, p = residuals
+    dout4 = dout.reshape(B, S, heads, D).transpose(1, 2).contiguous()
+    dq, dk, dv = attention_bwd(dout4, q, k, v, (p,), causal=causal)
+    def back(t):
+        return t.transpose(1, 2).reshape(B, S, d)
+    return torch.cat([back(dq), back(dk), back(dv)], dim=-1)
+
+
 # --------------------------------------------------------------------------
 # Embedding
 # --------------------------------------------------------------------------

@@ -313,3 +313,28 @@ def test_flash_attention_ragged_seq():
     ref = torch.nn.functional.scaled_dot_product_attention(q2, k2, v2,
                                                            is_causal=True)
     _assert_close_bf16(out, ref, rtol=3e-2)
+
+
+def test_attention_qkv_packed_matches_split():
+    """Packed-qkv flash path (strided kernels, no transposes) vs the 4-D
+    API and torch SDPA, fwd+bwd."""
+    B, S, H, D = 2, 256, 4, 64
+    d = H * D
+    g = torch.Generator().manual_seed(60)
+    qkv = (torch.randn(B, S, 3 * d, generator=g) * 0.5).to(BF16).cuda()
+    qkv.requires_grad_()
+    out = ops.attention_qkv(qkv, H, causal=True)
+    dout = _randn(B, S, d, seed=61)
+    out.backward(dout)
+    torch.cuda.synchronize()
+
+    qkv2 = qkv.detach().float().requires_grad_()
+    q, k, v = (t.reshape(B, S, H, D).transpose(1, 2)
+               for t in qkv2.split(d, dim=-1))
+    ref = torch.nn.functional.scaled_dot_product_attention(q, k, v,
+                                                           is_causal=True)
+    ref = ref.transpose(1, 2).reshape(B, S, d)
+    ref.backward(dout.float())
+    _assert_close_bf16(out, ref.detach(), rtol=3e-2)
+    _assert_close_bf16(qkv.grad, qkv2.grad, rtol=5e-2,
+                       scale=qkv2.grad.abs().max())
