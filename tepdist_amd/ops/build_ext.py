@@ -25,6 +25,7 @@ ARCH = os.environ.get("TEPDIST_GFX_ARCH", "gfx950")
 SOURCES = [
     "gemm.hip",
     "attention.hip",
+    "transpose.hip",
     "layernorm.hip",
     "softmax.hip",
     "embedding.hip",

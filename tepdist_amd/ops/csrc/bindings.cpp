@@ -43,6 +43,14 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
+  m.def("transpose", [](uintptr_t in, uintptr_t out, int R, int C,
+                        int64_t si, int64_t so, int batch,
+                        uintptr_t stream) {
+    transpose_bf16(reinterpret_cast<void*>(in), reinterpret_cast<void*>(out),
+                   R, C, si, so, batch, S(stream));
+    check_launch();
+  });
+
   m.def("layernorm_fwd", [](uintptr_t x, uintptr_t g, uintptr_t b, uintptr_t y,
                             uintptr_t mean, uintptr_t rstd, int rows, int cols,
                             float eps, uintptr_t stream) {

@@ -26,6 +26,12 @@ void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
 void splitk_reduce(const float* parts, void* out, int nparts, int64_t mn,
                    hipStream_t stream);
 
+// --- Transpose -------------------------------------------------------------
+// out[c][r] = in[r][c] per batch (strides in elements).
+void transpose_bf16(const void* in, void* out, int R, int C,
+                    int64_t stride_in, int64_t stride_out, int batch,
+                    hipStream_t stream);
+
 // --- LayerNorm -------------------------------------------------------------
 void layernorm_fwd_bf16(const void* x, const void* gamma, const void* beta,
                         void* y, float* mean, float* rstd, int rows, int cols,

@@ -1,0 +1,81 @@
+// Tiled bf16 2-D transpose for gfx950: 64x64 tiles through LDS, 16B
+// coalesced loads and stores (the LDS image is written row-major and read
+// column-wise via the v_perm 8x8 register transpose on the way out).
+// Used to canonicalize GEMM operands: dgrad/wgrad become k-contiguous x
+// k-contiguous so the single tuned GEMM schedule serves every case.
+
+#include <algorithm>
+#include <stdexcept>
+
+#include "common.h"
+#include "kernels.h"
+
+namespace tepdist {
+
+namespace {
+
+constexpr int NT = 256;
+constexpr int TS = 64;  // tile size
+
+// out[c][r] = in[r][c]; batched over blockIdx.z.
+__launch_bounds__(NT) __global__
+void transpose_kernel(const bf16_t* __restrict__ in, bf16_t* __restrict__ out,
+                      int R, int C, int64_t stride_in, int64_t stride_out) {
+  __shared__ bf16_t tile[TS * (TS + 8)];  // +16B row pad (b128-aligned)
+  const bf16_t* src = in + blockIdx.z * stride_in;
+  bf16_t* dst = out + blockIdx.z * stride_out;
+  const int r0 = blockIdx.y * TS;
+  const int c0 = blockIdx.x * TS;
+
+  // load [64 rows][64 cols] with 16B vectors: 512 loads / 256 threads
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NT;
+    const int r = idx >> 3;
+    const int c = (idx & 7) * 8;
+    bf16x8 v = {};
+    if (r0 + r < R) {
+      const bf16_t* p = src + (int64_t)(r0 + r) * C + c0 + c;
+      if (c0 + c + 8 <= C) {
+        v = *reinterpret_cast<const bf16x8*>(p);
+      } else {
+        for (int e = 0; e < 8 && c0 + c + e < C; ++e) v[e] = p[e];
+      }
+    }
+    *reinterpret_cast<bf16x8*>(tile + r * (TS + 8) + c) = v;
+  }
+  __syncthreads();
+
+  // store transposed: thread reads a column 8-run via 8 scalar LDS reads
+  // (padded rows -> conflict-light), writes one 16B row of the output
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const int idx = threadIdx.x + u * NT;
+    const int c = idx >> 3;        // output row = input col
+    const int r = (idx & 7) * 8;   // output col run = input rows
+    if (c0 + c < C) {
+      bf16x8 v;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) v[e] = tile[(r + e) * (TS + 8) + c];
+      bf16_t* p = dst + (int64_t)(c0 + c) * R + r0 + r;
+      if (r0 + r + 8 <= R) {
+        *reinterpret_cast<bf16x8*>(p) = v;
+      } else {
+        for (int e = 0; e < 8 && r0 + r + e < R; ++e) p[e] = v[e];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void transpose_bf16(const void* in, void* out, int R, int C,
+                    int64_t stride_in, int64_t stride_out, int batch,
+                    hipStream_t stream) {
+  dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, batch);
+  hipLaunchKernelGGL(transpose_kernel, grid, dim3(NT), 0, stream,
+                     static_cast<const bf16_t*>(in),
+                     static_cast<bf16_t*>(out), R, C, stride_in, stride_out);
+}
+
+}  // namespace tepdist

@@ -93,6 +93,17 @@ def linear_fwd(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
     return y, pre
 
 
+def transpose2d(t: torch.Tensor) -> torch.Tensor:
+    """Materialized bf16 transpose (tiled LDS kernel)."""
+    R, C = t.shape[-2], t.shape[-1]
+    batch = t.numel() // (R * C)
+    t = t.contiguous()
+    out = torch.empty((*t.shape[:-2], C, R), dtype=t.dtype, device=t.device)
+    ext.transpose(t.data_ptr(), out.data_ptr(), R, C, R * C, R * C, batch,
+                  _stream())
+    return out
+
+
 def linear_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
                has_bias: bool, act: str, pre_act: Optional[torch.Tensor]):
     dy = dy.contiguous()
@@ -103,10 +114,15 @@ def linear_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
         ext.gelu_bwd(dy.data_ptr(), pre_act.data_ptr(), dy_eff.data_ptr(),
                      dy.numel(), _stream())
         dy = dy_eff
-    # dx[M,K] = dy[M,N] @ w[N,K]  (contraction N: dy k-inner, w k-outer)
-    dx, _ = _gemm_raw(dy, w, True, False, M, K, N, N, w.stride(0), 0, 0, 1)
-    # dw[N,K] = dy^T[N,M] @ x[M,K]  (contraction M: both k-outer)
-    dw, _ = _gemm_raw(dy, x, False, False, N, K, M, N, K, 0, 0, 1)
+    # canonicalize to KC x KC: one tuned GEMM schedule serves every case,
+    # with cheap materialized transposes (memory-bound, ~2% of GEMM time)
+    wT = transpose2d(w)        # [K, N]
+    dyT = transpose2d(dy)      # [N, M]
+    xT = transpose2d(x)        # [K, M]
+    # dx[M,K] = dy[M,N] @ w[N,K]: A=dy KC (k=N), B=w^T stored [K,N] KC
+    dx, _ = _gemm_raw(dy, wT, True, True, M, K, N, N, N, 0, 0, 1)
+    # dw[N,K] = dy^T[N,M] @ x[M,K]: A=dy^T KC (k=M), B=x^T stored [K,M] KC
+    dw, _ = _gemm_raw(dyT, xT, True, True, N, K, M, M, M, 0, 0, 1)
     db = None
     if has_bias:
         db = torch.empty(N, dtype=BF16, device=dy.device)

@@ -338,3 +338,15 @@ def test_attention_qkv_packed_matches_split():
     _assert_close_bf16(out, ref.detach(), rtol=3e-2)
     _assert_close_bf16(qkv.grad, qkv2.grad, rtol=5e-2,
                        scale=qkv2.grad.abs().max())
+
+
+def test_transpose_kernel():
+    for R, C in ((128, 256), (200, 72), (64, 64), (1000, 1024)):
+        x = _randn(R, C, seed=R + C)
+        y = hip.transpose2d(x)
+        torch.cuda.synchronize()
+        assert torch.equal(y.float(), x.float().t())
+    xb = _randn(6, 96, 160, seed=70)
+    yb = hip.transpose2d(xb)
+    torch.cuda.synchronize()
+    assert torch.equal(yb.float(), xb.float().transpose(-1, -2))
