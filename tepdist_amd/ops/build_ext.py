@@ -24,6 +24,7 @@ ARCH = os.environ.get("TEPDIST_GFX_ARCH", "gfx950")
 
 SOURCES = [
     "gemm.hip",
+    "gemm256.hip",
     "attention.hip",
     "transpose.hip",
     "layernorm.hip",

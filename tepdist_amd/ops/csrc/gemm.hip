@@ -398,6 +398,12 @@ void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
     return;
   }
 
+  if (gemm256_supported(M, N, K, lda, ldb, a_kc, b_kc, epi, split_k)) {
+    gemm256_bf16(A, B, C, c_pre, bias, M, N, K, lda, ldb, ldc, stride_a,
+                 stride_b, stride_c, batch, epi, stream);
+    return;
+  }
+
   dim3 grid(nbx * nby, 1, batch);
 #define GEMM_LAUNCH(AKC, BKC, E)                                            \
   hipLaunchKernelGGL((gemm_kernel<AKC, BKC, E>), grid, block, 0, stream, a, \
