@@ -113,7 +113,40 @@ def bench_membound():
                       "us": round(t * 1e6, 1)}), flush=True)
 
 
+def bench_attention():
+    B, H, S, D = 16, 16, 1024, 64
+    q = torch.randn(B, H, S, D).to(BF16).cuda()
+    k = torch.randn(B, H, S, D).to(BF16).cuda()
+    v = torch.randn(B, H, S, D).to(BF16).cuda()
+    t = timeit(lambda: hip.attention_fwd(q, k, v, True))
+    fl = 4.0 * B * H * S * S * D / 2  # causal
+    print(json.dumps({"kernel": "flash_fwd", "tflops": round(fl / t / 1e12, 1),
+                      "us": round(t * 1e6, 1)}), flush=True)
+    out, res = hip.attention_fwd(q, k, v, True)
+    dout = torch.randn(B, H, S, D).to(BF16).cuda()
+    t = timeit(lambda: hip.attention_bwd(dout, q, k, v, res, True), iters=10)
+    print(json.dumps({"kernel": "flash_bwd",
+                      "tflops": round(2.5 * fl / t / 1e12, 1),
+                      "us": round(t * 1e6, 1)}), flush=True)
+    rows, cols = 32768, 4096
+    dy = torch.randn(rows, cols).to(BF16).cuda()
+    t = timeit(lambda: hip.linear_bwd(dy[:, :1024], torch.randn(rows, 1024).to(BF16).cuda(), torch.randn(1024, 1024).to(BF16).cuda(), True, "none", None), iters=3) if False else None
+    db = torch.empty(cols, dtype=BF16, device="cuda")
+    ws = torch.zeros(cols, dtype=torch.float32, device="cuda")
+    from tepdist_amd.ops import _tepdist_hip as ext
+    import torch as _t
+    def run_bias():
+        ws.zero_()
+        ext.bias_sum(dy.data_ptr(), db.data_ptr(), ws.data_ptr(), rows, cols,
+                     _t.cuda.current_stream().cuda_stream)
+    t = timeit(run_bias)
+    print(json.dumps({"kernel": "bias_sum",
+                      "tb_s": round(rows * cols * 2 / t / 1e12, 2),
+                      "us": round(t * 1e6, 1)}), flush=True)
+
+
 if __name__ == "__main__":
     torch.manual_seed(0)
     bench_gemm()
     bench_membound()
+    bench_attention()

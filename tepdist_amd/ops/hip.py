@@ -174,6 +174,16 @@ def matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 
     a, a_kc, lda, sa = layout(a, contraction_is_last=True)
     b, b_kc, ldb, sb = layout(b, contraction_is_last=False)
+    # canonicalize big single GEMMs to KC x KC (unlocks the 256x256
+    # deep-pipelined kernel; the transpose is memory-bound and cheap)
+    if batch == 1 and M % 256 == 0 and N % 256 == 0 and K % 64 == 0 \
+            and K >= 512:
+        if not a_kc:
+            a = transpose2d(a.reshape(K, M)).reshape(1, M, K)
+            a_kc, lda = True, K
+        if not b_kc:
+            b = transpose2d(b.reshape(K, N)).reshape(1, N, K)
+            b_kc, ldb = True, K
     # b layout: contraction dim is -2; b_kc means the contraction dim has the
     # larger stride... map: for B operand the kernel's b_kc=True expects
     # storage [N,K] (k inner). b's contraction dim is -2; if b.stride(-1)==1
