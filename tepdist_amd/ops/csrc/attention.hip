@@ -111,58 +111,69 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       l_r[mi][e] = 0.f;
     }
 
+  // staging registers (issue-early / write-late split, guide T14):
+  // K: 2 x 16B per thread; V: 8 x 4B per thread (v_perm transpose slabs)
+  constexpr int KUN = KB * D / 8 / NT;
+  bf16x8 krg[KUN];
+  uint32_t vrg[8];
+
+  auto stage_load = [&](int kv0) {
+#pragma unroll
+    for (int u = 0; u < KUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      bf16x8 v8 = {};
+      if (kv0 + row < S)
+        v8 = *reinterpret_cast<const bf16x8*>(
+            kp + (int64_t)(kv0 + row) * q_rs + c);
+      krg[u] = v8;
+    }
+    const int f = 2 * (threadIdx.x % (D / 2));
+    const int kb = threadIdx.x / (D / 2);
+    if (kb < KB / 8) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kv = kv0 + kb * 8 + j;
+        bf16x2 v2 = {};
+        if (kv < S)
+          v2 = *reinterpret_cast<const bf16x2*>(vp + (int64_t)kv * q_rs + f);
+        vrg[j] = __builtin_bit_cast(uint32_t, v2);
+      }
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int u = 0; u < KUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = krg[u];
+    }
+    const int f = 2 * (threadIdx.x % (D / 2));
+    const int kb = threadIdx.x / (D / 2);
+    if (kb < KB / 8) {
+      uint32_t o0[4], o1[4];
+#pragma unroll
+      for (int d2 = 0; d2 < 4; ++d2) {
+        o0[d2] = __builtin_amdgcn_perm(vrg[2 * d2 + 1], vrg[2 * d2],
+                                       0x05040100u);
+        o1[d2] = __builtin_amdgcn_perm(vrg[2 * d2 + 1], vrg[2 * d2],
+                                       0x07060302u);
+      }
+      *reinterpret_cast<uint4*>(sVT + loff<KB>(f, kb * 8)) =
+          make_uint4(o0[0], o0[1], o0[2], o0[3]);
+      *reinterpret_cast<uint4*>(sVT + loff<KB>(f + 1, kb * 8)) =
+          make_uint4(o1[0], o1[1], o1[2], o1[3]);
+    }
+  };
+
   const int kv_end = causal ? min(S, q0 + QB) : S;
+  stage_load(0);
+  stage_write();
+  __syncthreads();
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
-    // --- stage K tile [KB][D] (k-contiguous rows, 16B loads) ---
-    {
-      constexpr int UN = KB * D / 8 / NT;  // units per thread
-#pragma unroll
-      for (int u = 0; u < UN; ++u) {
-        const int idx = threadIdx.x + u * NT;
-        const int row = idx / (D / 8);
-        const int c = (idx % (D / 8)) * 8;
-        bf16x8 v8 = {};
-        if (kv0 + row < S)
-          v8 = *reinterpret_cast<const bf16x8*>(
-              kp + (int64_t)(kv0 + row) * q_rs + c);
-        *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = v8;
-      }
-    }
-    // --- stage V tile transposed -> [D][KB] via 8x2 v_perm slabs ---
-    {
-      constexpr int NSLAB = (KB / 8) * (D / 2);  // 8kv x 2d slabs
-#pragma unroll
-      for (int u = 0; u < (NSLAB + NT - 1) / NT; ++u) {
-        const int idx = threadIdx.x + u * NT;
-        if (idx < NSLAB) {
-          const int f = 2 * (idx % (D / 2));     // d column pair
-          const int kb = idx / (D / 2);          // kv block of 8
-          uint32_t r[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int kv = kv0 + kb * 8 + j;
-            bf16x2 v2 = {};
-            if (kv < S)
-              v2 = *reinterpret_cast<const bf16x2*>(
-                  vp + (int64_t)kv * q_rs + f);
-            r[j] = __builtin_bit_cast(uint32_t, v2);
-          }
-          uint32_t o0[4], o1[4];
-#pragma unroll
-          for (int d2 = 0; d2 < 4; ++d2) {
-            o0[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2],
-                                           0x05040100u);
-            o1[d2] = __builtin_amdgcn_perm(r[2 * d2 + 1], r[2 * d2],
-                                           0x07060302u);
-          }
-          *reinterpret_cast<uint4*>(sVT + loff<KB>(f, kb * 8)) =
-              make_uint4(o0[0], o0[1], o0[2], o0[3]);
-          *reinterpret_cast<uint4*>(sVT + loff<KB>(f + 1, kb * 8)) =
-              make_uint4(o1[0], o1[1], o1[2], o1[3]);
-        }
-      }
-    }
-    __syncthreads();
+    if (kv0 + KB < kv_end) stage_load(kv0 + KB);  // overlap with compute
 
     if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
       // --- S = Q K^T fragments ---
@@ -240,7 +251,11 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
         }
       }
     }
-    __syncthreads();
+    __syncthreads();  // all waves done reading sK/sVT
+    if (kv0 + KB < kv_end) {
+      stage_write();  // overwrite with the pre-loaded next tile
+      __syncthreads();
+    }
   }
 
   // --- epilogue: O /= l, write O and lse ---
@@ -434,16 +449,99 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   f32x4 acc_dk[DF] = {};
   f32x4 acc_dv[DF] = {};
 
-  const int q_start = causal ? (kv0 / QT) * QT : 0;
-  for (int q0 = q_start; q0 < S; q0 += QT) {
-    stage_nat_t<QT, D>(qp, q0, S, q_rs, sQ, sQT);
-    stage_nat_t<QT, D>(dop, q0, S, o_rs, sdO, sdOT);
-    for (int i = threadIdx.x; i < QT; i += NT) {
-      const int qg = q0 + i;
-      sLSE[i] = (qg < S) ? LSE[(int64_t)bh * S + qg] : -3.0e38f;
-      sDELTA[i] = (qg < S) ? DELTA[(int64_t)bh * S + qg] : 0.f;
+  // issue-early / write-late staging registers for the Q and dO tiles
+  constexpr int QUN = QT * D / 8 / NT;
+  bf16x8 qn[QUN], don[QUN];
+  uint32_t qt[8], dot_[8];
+  float lse2[QT / NT + 1], dl2[QT / NT + 1];
+
+  auto tile_load = [&](int q0) {
+#pragma unroll
+    for (int u = 0; u < QUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      bf16x8 a = {}, b2 = {};
+      if (q0 + row < S) {
+        a = *reinterpret_cast<const bf16x8*>(
+            qp + (int64_t)(q0 + row) * q_rs + c);
+        b2 = *reinterpret_cast<const bf16x8*>(
+            dop + (int64_t)(q0 + row) * o_rs + c);
+      }
+      qn[u] = a;
+      don[u] = b2;
     }
-    __syncthreads();
+    const int f = 2 * (threadIdx.x % (D / 2));
+    const int kb = threadIdx.x / (D / 2);
+    if (kb < QT / 8) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int row = q0 + kb * 8 + j;
+        bf16x2 a = {}, b2 = {};
+        if (row < S) {
+          a = *reinterpret_cast<const bf16x2*>(
+              qp + (int64_t)row * q_rs + f);
+          b2 = *reinterpret_cast<const bf16x2*>(
+              dop + (int64_t)row * o_rs + f);
+        }
+        qt[j] = __builtin_bit_cast(uint32_t, a);
+        dot_[j] = __builtin_bit_cast(uint32_t, b2);
+      }
+    }
+    for (int i = threadIdx.x, s2 = 0; i < QT; i += NT, ++s2) {
+      const int qg = q0 + i;
+      lse2[s2] = (qg < S) ? LSE[(int64_t)bh * S + qg] : -3.0e38f;
+      dl2[s2] = (qg < S) ? DELTA[(int64_t)bh * S + qg] : 0.f;
+    }
+  };
+  auto tile_write = [&]() {
+#pragma unroll
+    for (int u = 0; u < QUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(sQ + loff<D>(row, c)) = qn[u];
+      *reinterpret_cast<bf16x8*>(sdO + loff<D>(row, c)) = don[u];
+    }
+    const int f = 2 * (threadIdx.x % (D / 2));
+    const int kb = threadIdx.x / (D / 2);
+    if (kb < QT / 8) {
+      uint32_t o0[4], o1[4];
+#pragma unroll
+      for (int d2 = 0; d2 < 4; ++d2) {
+        o0[d2] = __builtin_amdgcn_perm(qt[2 * d2 + 1], qt[2 * d2],
+                                       0x05040100u);
+        o1[d2] = __builtin_amdgcn_perm(qt[2 * d2 + 1], qt[2 * d2],
+                                       0x07060302u);
+      }
+      *reinterpret_cast<uint4*>(sQT + loff<QT>(f, kb * 8)) =
+          make_uint4(o0[0], o0[1], o0[2], o0[3]);
+      *reinterpret_cast<uint4*>(sQT + loff<QT>(f + 1, kb * 8)) =
+          make_uint4(o1[0], o1[1], o1[2], o1[3]);
+#pragma unroll
+      for (int d2 = 0; d2 < 4; ++d2) {
+        o0[d2] = __builtin_amdgcn_perm(dot_[2 * d2 + 1], dot_[2 * d2],
+                                       0x05040100u);
+        o1[d2] = __builtin_amdgcn_perm(dot_[2 * d2 + 1], dot_[2 * d2],
+                                       0x07060302u);
+      }
+      *reinterpret_cast<uint4*>(sdOT + loff<QT>(f, kb * 8)) =
+          make_uint4(o0[0], o0[1], o0[2], o0[3]);
+      *reinterpret_cast<uint4*>(sdOT + loff<QT>(f + 1, kb * 8)) =
+          make_uint4(o1[0], o1[1], o1[2], o1[3]);
+    }
+    for (int i = threadIdx.x, s2 = 0; i < QT; i += NT, ++s2) {
+      sLSE[i] = lse2[s2];
+      sDELTA[i] = dl2[s2];
+    }
+  };
+
+  const int q_start = causal ? (kv0 / QT) * QT : 0;
+  tile_load(q_start);
+  tile_write();
+  __syncthreads();
+  for (int q0 = q_start; q0 < S; q0 += QT) {
+    if (q0 + QT < S) tile_load(q0 + QT);  // overlap with compute
 
     // S^T = K Q^T ; dP^T = V dO^T   (both natural-layout B reads)
     f32x4 st[4] = {};
@@ -530,7 +628,8 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
                                                              0, 0, 0);
       }
     }
-    __syncthreads();  // sdS complete across waves
+    __syncthreads();  // sdS complete; sQ/sQT/sdO/sdOT reads done
+    if (q0 + QT < S) tile_write();  // overlaps the dQ phase (sdS/sKT only)
 
     // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS @ K (via sKT)
     {
@@ -561,7 +660,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
                       acc_dq[df][e]);
         }
     }
-    __syncthreads();  // before restaging sQ/sdO/sdS
+    __syncthreads();  // next tile's images complete before its reads
   }
 
   // write dK, dV (C layout scatter)
