@@ -10,9 +10,12 @@ the fused AdamW applies them (AG).
 
 from __future__ import annotations
 
+import time
 from typing import Callable, Optional
 
 import torch
+
+from tepdist_amd.config import get_env
 
 from tepdist_amd.train.optim import AdamW
 from tepdist_amd.parallel.dp import GradReducer
@@ -31,6 +34,8 @@ class Trainer:
         """Runs one optimizer step = grad_accum_steps micro-batches.
         batch_iter(i) returns (input_ids, labels) for micro-batch i.
         Returns the mean loss (host float)."""
+        debug = get_env().debug
+        t0 = time.perf_counter() if debug else 0.0
         self.opt.zero_grad()
         if self.reducer is not None:
             self.reducer.reset()
@@ -50,4 +55,10 @@ class Trainer:
         if self.reducer is not None:
             self.reducer.finalize()
         self.opt.step()
+        if debug:
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            print(f"[ExecutePlan Duration] "
+                  f"{(time.perf_counter() - t0) * 1e3:.2f} ms "
+                  f"loss={total / n:.4f}", flush=True)
         return total / n
