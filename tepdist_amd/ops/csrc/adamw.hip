@@ -87,4 +87,78 @@ void adamw_bf16(void* param, float* master, const void* grad_bf16,
   }
 }
 
+namespace {
+
+constexpr int MT_CHUNK = 16384;
+
+// multi-tensor AdamW: one fused launch over all parameters. tabs holds 5
+// pointer tables (param, master, grad(bf16), exp_avg, exp_avg_sq), chunks
+// is [(tensor_idx, chunk_idx)] with MT_CHUNK elements per chunk.
+__launch_bounds__(NT) __global__
+void adamw_mt_kernel(const int64_t* __restrict__ tabs,
+                     const int64_t* __restrict__ numel,
+                     const float* __restrict__ wds,
+                     const int* __restrict__ chunks, int nchunks, int nt,
+                     float lr, float beta1, float beta2, float eps,
+                     float inv_bc1, float inv_bc2) {
+  for (int ci = blockIdx.x; ci < nchunks; ci += gridDim.x) {
+    const int ti = chunks[2 * ci];
+    const int64_t off = (int64_t)chunks[2 * ci + 1] * MT_CHUNK;
+    bf16_t* param = reinterpret_cast<bf16_t*>(tabs[ti]);
+    float* master = reinterpret_cast<float*>(tabs[nt + ti]);
+    const bf16_t* grad = reinterpret_cast<const bf16_t*>(tabs[2 * nt + ti]);
+    float* m = reinterpret_cast<float*>(tabs[3 * nt + ti]);
+    float* v = reinterpret_cast<float*>(tabs[4 * nt + ti]);
+    const float wd = wds[ti];
+    const int64_t end = std::min(off + MT_CHUNK, numel[ti]);
+    for (int64_t i0 = off + (int64_t)threadIdx.x * 4; i0 < end;
+         i0 += NT * 4) {
+      const int cnt = (int)std::min<int64_t>(4, end - i0);
+      if (cnt == 4) {
+        f32x4 mv = *reinterpret_cast<f32x4*>(m + i0);
+        f32x4 vv = *reinterpret_cast<f32x4*>(v + i0);
+        f32x4 ma = *reinterpret_cast<f32x4*>(master + i0);
+        const bf16x4 gv = *reinterpret_cast<const bf16x4*>(grad + i0);
+        bf16x4 pv;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const float g = bf2f(gv[e]);
+          mv[e] = beta1 * mv[e] + (1.f - beta1) * g;
+          vv[e] = beta2 * vv[e] + (1.f - beta2) * g * g;
+          const float denom = sqrtf(vv[e] * inv_bc2) + eps;
+          ma[e] = ma[e] * (1.f - lr * wd) - lr * (mv[e] * inv_bc1) / denom;
+          pv[e] = f2bf(ma[e]);
+        }
+        *reinterpret_cast<f32x4*>(m + i0) = mv;
+        *reinterpret_cast<f32x4*>(v + i0) = vv;
+        *reinterpret_cast<f32x4*>(master + i0) = ma;
+        *reinterpret_cast<bf16x4*>(param + i0) = pv;
+      } else {
+        for (int e = 0; e < cnt; ++e) {
+          const int64_t i = i0 + e;
+          const float g = bf2f(grad[i]);
+          m[i] = beta1 * m[i] + (1.f - beta1) * g;
+          v[i] = beta2 * v[i] + (1.f - beta2) * g * g;
+          const float denom = sqrtf(v[i] * inv_bc2) + eps;
+          master[i] = master[i] * (1.f - lr * wd) -
+                      lr * (m[i] * inv_bc1) / denom;
+          param[i] = f2bf(master[i]);
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
+                   const float* wds, const int* chunks, int nchunks, int nt,
+                   float lr, float beta1, float beta2, float eps, float bc1,
+                   float bc2, hipStream_t stream) {
+  const int blocks = std::min(nchunks, 2048);
+  hipLaunchKernelGGL(adamw_mt_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
+                     stream, tabs, numel, wds, chunks, nchunks, nt, lr,
+                     beta1, beta2, eps, 1.0f / bc1, 1.0f / bc2);
+}
+
 }  // namespace tepdist

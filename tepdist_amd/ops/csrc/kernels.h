@@ -118,6 +118,12 @@ void adamw_bf16(void* param, float* master, const void* grad_bf16,
                 int64_t n, float lr, float beta1, float beta2, float eps,
                 float weight_decay, float bc1, float bc2, hipStream_t stream);
 
+// multi-tensor fused AdamW (pointer tables on device; MT_CHUNK=16384)
+void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
+                   const float* wds, const int* chunks, int nchunks, int nt,
+                   float lr, float beta1, float beta2, float eps, float bc1,
+                   float bc2, hipStream_t stream);
+
 // --- diagnostics -----------------------------------------------------------
 void tr16_probe(float* out_pattern, float* out_uniform, hipStream_t stream);
 

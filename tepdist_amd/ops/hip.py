@@ -467,6 +467,43 @@ def dropout_bwd(dy: torch.Tensor, mask: Optional[torch.Tensor], p: float):
 # AdamW
 # --------------------------------------------------------------------------
 
+class AdamWMT:
+    """Multi-tensor fused AdamW context: static pointer/chunk tables on
+    device; only the gradient pointer table refreshes per step."""
+
+    CHUNK = 16384
+
+    def __init__(self, params, states, wds):
+        dev = params[0].device
+        self.nt = len(params)
+        ptrs = ([p.data_ptr() for p in params] +
+                [s["master"].data_ptr() for s in states] +
+                [0] * self.nt +
+                [s["exp_avg"].data_ptr() for s in states] +
+                [s["exp_avg_sq"].data_ptr() for s in states])
+        self.tabs = torch.tensor(ptrs, dtype=torch.int64, device=dev)
+        self.numel = torch.tensor([p.numel() for p in params],
+                                  dtype=torch.int64, device=dev)
+        self.wds = torch.tensor(wds, dtype=torch.float32, device=dev)
+        chunks = []
+        for ti, p in enumerate(params):
+            for c in range((p.numel() + self.CHUNK - 1) // self.CHUNK):
+                chunks.extend((ti, c))
+        self.chunks = torch.tensor(chunks, dtype=torch.int32, device=dev)
+        self.nchunks = len(chunks) // 2
+        self._gslot = self.tabs[2 * self.nt:3 * self.nt]
+
+    def step(self, grads, lr, beta1, beta2, eps, step_no):
+        gp = torch.tensor([g.data_ptr() for g in grads], dtype=torch.int64)
+        self._gslot.copy_(gp, non_blocking=True)
+        bc1 = 1.0 - beta1 ** step_no
+        bc2 = 1.0 - beta2 ** step_no
+        ext.adamw_mt(self.tabs.data_ptr(), self.numel.data_ptr(),
+                     self.wds.data_ptr(), self.chunks.data_ptr(),
+                     self.nchunks, self.nt, lr, beta1, beta2, eps, bc1, bc2,
+                     _stream())
+
+
 def adamw_step(param_bf16: torch.Tensor, master: torch.Tensor,
                grad: torch.Tensor, exp_avg: torch.Tensor,
                exp_avg_sq: torch.Tensor, lr: float, beta1: float,

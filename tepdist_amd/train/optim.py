@@ -35,9 +35,32 @@ class AdamW:
                 "exp_avg_sq": torch.zeros_like(p, dtype=torch.float32),
             }
 
+    def _try_mt(self):
+        """Multi-tensor fused path (single kernel launch) on GPU with bf16
+        grads for every param."""
+        import torch as _t
+        ps = [p for p in self.params if p.grad is not None]
+        if len(ps) < 4 or not ps[0].is_cuda or \
+                any(p.grad.dtype != _t.bfloat16 or
+                    not p.grad.is_contiguous() for p in ps):
+            return None
+        from tepdist_amd.ops import hip as be
+        key = tuple(id(p) for p in ps)
+        if getattr(self, "_mt_key", None) != key:
+            self._mt_key = key
+            wds = [0.0 if (self.no_decay_1d and p.dim() == 1)
+                   else self.weight_decay for p in ps]
+            self._mt = be.AdamWMT(ps, [self.state[p] for p in ps], wds)
+        return ps
+
     @torch.no_grad()
     def step(self):
         self.step_count += 1
+        ps = self._try_mt()
+        if ps is not None:
+            self._mt.step([p.grad for p in ps], self.lr, self.beta1,
+                          self.beta2, self.eps, self.step_count)
+            return
         for p in self.params:
             if p.grad is None:
                 continue

@@ -350,3 +350,30 @@ def test_transpose_kernel():
     yb = hip.transpose2d(xb)
     torch.cuda.synchronize()
     assert torch.equal(yb.float(), xb.float().transpose(-1, -2))
+
+
+def test_adamw_multi_tensor_matches_single():
+    from tepdist_amd.train import AdamW
+    torch.manual_seed(0)
+    shapes = [(100, 64), (37,), (513, 16), (1000,), (64, 64), (20000,)]
+    params_a = [torch.nn.Parameter(torch.randn(*s).bfloat16().cuda())
+                for s in shapes]
+    params_b = [torch.nn.Parameter(p.detach().clone()) for p in params_a]
+    oa = AdamW(params_a, lr=1e-2)
+    ob = AdamW(params_b, lr=1e-2)
+    for step in range(3):
+        for pa, pb in zip(params_a, params_b):
+            g = torch.randn(pa.shape).bfloat16().cuda()
+            pa.grad = g.clone()
+            pb.grad = g.clone()
+        oa.step()                         # multi-tensor path (>=4 cuda bf16)
+        ob._mt_key = object()             # poison cache
+        saved, ob._try_mt = ob._try_mt, lambda: None  # force per-tensor
+        ob.step()
+        ob._try_mt = saved
+    torch.cuda.synchronize()
+    for pa, pb in zip(params_a, params_b):
+        torch.testing.assert_close(pa.data.float(), pb.data.float())
+        torch.testing.assert_close(oa.state[pa]["master"],
+                                   ob.state[pb]["master"], rtol=1e-6,
+                                   atol=1e-7)
