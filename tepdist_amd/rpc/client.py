@@ -84,12 +84,24 @@ class TepdistClient:
 
 
 class TepdistSession:
-    """Frontend convenience: capture -> plan -> step loop."""
+    """Frontend convenience: capture -> plan -> step loop. Supports the
+    reference client's async step pipelining (NUM_PARALLEL_RPC_STEPS,
+    xla_ops.cc:617-634) and periodic lazy variable fetch
+    (FETCH_RESOURCE_VAR_STEPS)."""
 
     def __init__(self, client: TepdistClient = None):
+        import concurrent.futures as _cf
         self.client = client or TepdistClient()
         self.handle: Optional[int] = None
         self.plan_info: Optional[dict] = None
+        self._n_parallel = int(os.environ.get("NUM_PARALLEL_RPC_STEPS", "1"))
+        self._fetch_every = int(os.environ.get("FETCH_RESOURCE_VAR_STEPS",
+                                               "0"))
+        self._pool = _cf.ThreadPoolExecutor(max_workers=max(
+            self._n_parallel, 1))
+        self._inflight = []
+        self._step_no = 0
+        self.last_vars: Optional[Dict[str, torch.Tensor]] = None
 
     def compile_graph(self, graph, num_devices: int = 1) -> dict:
         self.plan_info = self.client.build_execution_plan(
@@ -98,6 +110,24 @@ class TepdistSession:
         return self.plan_info
 
     def step(self, inputs: Dict[str, torch.Tensor]) -> float:
-        r = self.client.execute_plan(self.handle, inputs)
-        loss = list(r["outputs"].values())[0]
-        return float(loss)
+        """Synchronous when NUM_PARALLEL_RPC_STEPS<=1, else returns the
+        loss of the oldest in-flight step while pipelining new ones."""
+        self._step_no += 1
+        if self._fetch_every and self._step_no % self._fetch_every == 0:
+            self.last_vars = self.client.fetch_resource_vars()
+        if self._n_parallel <= 1:
+            r = self.client.execute_plan(self.handle, inputs)
+            return float(list(r["outputs"].values())[0])
+        self._inflight.append(self._pool.submit(
+            self.client.execute_plan, self.handle, inputs))
+        if len(self._inflight) < self._n_parallel:
+            return float("nan")  # warm-up: no result yet
+        r = self._inflight.pop(0).result()
+        return float(list(r["outputs"].values())[0])
+
+    def drain(self):
+        out = []
+        while self._inflight:
+            r = self._inflight.pop(0).result()
+            out.append(float(list(r["outputs"].values())[0]))
+        return out

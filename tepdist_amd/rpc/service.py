@@ -127,8 +127,18 @@ class TepdistService:
     def execute_plan(self, req: dict) -> dict:
         h = req["handle"]
         plan = self.plans[h]
-        feeds = {k: v.to(self.device) for k, v in req.get("inputs", {}).items()}
         env = get_env()
+        if env.fake_input:
+            # FAKE_INPUT (service_env.h:71): freeze the first step's inputs
+            # to measure pure step time
+            if not hasattr(self, "_frozen_feeds"):
+                self._frozen_feeds = {
+                    k: v.to(self.device)
+                    for k, v in req.get("inputs", {}).items()}
+            feeds = self._frozen_feeds
+        else:
+            feeds = {k: v.to(self.device)
+                     for k, v in req.get("inputs", {}).items()}
         with self._lock:
             t0 = time.time()
             self.step_count += 1
