@@ -59,26 +59,39 @@ void gelu_bwd_kernel(const bf16_t* __restrict__ dy,
 __launch_bounds__(NT) __global__
 void bias_sum_kernel(const bf16_t* __restrict__ dy, float* __restrict__ ws,
                      int64_t rows, int cols) {
-  const int ncg = (cols + 3) / 4;
+  // thread owns an 8-col group; 4-row unroll keeps 4 dwordx4 loads in
+  // flight per thread (latency cover for the strided row walk)
+  const int ncg = (cols + 7) / 8;
   const int cg = blockIdx.x * NT + threadIdx.x;
   if (cg >= ncg) return;
-  const int c0 = cg * 4;
+  const int c0 = cg * 8;
   const int64_t r0 = (rows * blockIdx.y) / gridDim.y;
   const int64_t r1 = (rows * (blockIdx.y + 1)) / gridDim.y;
-  float acc[4] = {};
-  if (c0 + 4 <= cols) {
-    for (int64_t r = r0; r < r1; ++r) {
-      const bf16x4 v = *reinterpret_cast<const bf16x4*>(dy + r * cols + c0);
+  float acc[8] = {};
+  if (c0 + 8 <= cols) {
+    int64_t r = r0;
+    for (; r + 4 <= r1; r += 4) {
+      bf16x8 v[4];
 #pragma unroll
-      for (int e = 0; e < 4; ++e) acc[e] += bf2f(v[e]);
+      for (int u = 0; u < 4; ++u)
+        v[u] = *reinterpret_cast<const bf16x8*>(dy + (r + u) * cols + c0);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+#pragma unroll
+        for (int e = 0; e < 8; ++e) acc[e] += bf2f(v[u][e]);
+    }
+    for (; r < r1; ++r) {
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(dy + r * cols + c0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[e] += bf2f(v[e]);
     }
   } else {
     for (int64_t r = r0; r < r1; ++r)
-      for (int e = 0; e < 4 && c0 + e < cols; ++e)
+      for (int e = 0; e < 8 && c0 + e < cols; ++e)
         acc[e] += bf2f(dy[r * cols + c0 + e]);
   }
 #pragma unroll
-  for (int e = 0; e < 4; ++e)
+  for (int e = 0; e < 8; ++e)
     if (c0 + e < cols) atomicAdd(&ws[c0 + e], acc[e]);
 }
 
@@ -111,7 +124,7 @@ void bias_sum_bf16(const void* dy, void* db_out, float* ws_zeroed,
   bf16_t* db = static_cast<bf16_t*>(db_out);
   float* ws = ws_zeroed;  // caller-zeroed fp32 workspace of `cols`
   const int ysplit = (int)std::min<int64_t>((rows + 31) / 32, 256);
-  const int ncg = (cols + 3) / 4;
+  const int ncg = (cols + 7) / 8;
   dim3 grid((ncg + NT - 1) / NT, std::max(ysplit, 1));
   hipLaunchKernelGGL(bias_sum_kernel, grid, dim3(NT), 0, stream,
                      static_cast<const bf16_t*>(dy), ws, rows, cols);

@@ -384,6 +384,11 @@ void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
   if (split_k > 1) {
     if (batch != 1 || epi != 0)
       throw std::runtime_error("split_k requires batch=1, epi=0");
+    if (gemm256_supported(M, N, K, lda, ldb, a_kc, b_kc, epi, split_k)) {
+      gemm256_bf16(A, B, C, c_pre, bias, M, N, K, lda, ldb, ldc, stride_a,
+                   stride_b, stride_c, batch, epi, split_k, stream);
+      return;
+    }
     const int k_chunk = ((K + split_k - 1) / split_k + BK - 1) / BK * BK;
     dim3 gridk(nbx * nby, 1, split_k);
 #define GEMM_SK(AKC, BKC)                                                   \
@@ -398,9 +403,9 @@ void gemm_bf16(const void* A, const void* B, void* C, void* c_pre,
     return;
   }
 
-  if (gemm256_supported(M, N, K, lda, ldb, a_kc, b_kc, epi, split_k)) {
+  if (gemm256_supported(M, N, K, lda, ldb, a_kc, b_kc, epi, 1)) {
     gemm256_bf16(A, B, C, c_pre, bias, M, N, K, lda, ldb, ldc, stride_a,
-                 stride_b, stride_c, batch, epi, stream);
+                 stride_b, stride_c, batch, epi, 1, stream);
     return;
   }
 

@@ -71,19 +71,27 @@ DEV_INLINE bf16x8 fragq(const bf16_t* slot, int row, int col_e) {
 #define RAW_BAR() __builtin_amdgcn_s_barrier()
 #define VMCNT6() asm volatile("s_waitcnt vmcnt(6)" ::: "memory")
 
-template <int EPI>
+template <int EPI, bool F32OUT = false>
 __launch_bounds__(NTH, 1) __global__
 void gemm256_kernel(const bf16_t* __restrict__ A,
                     const bf16_t* __restrict__ B, void* __restrict__ Cv,
                     bf16_t* __restrict__ Cpre,
                     const bf16_t* __restrict__ bias, int M, int N, int K,
                     int lda, int ldb, int ldc, int64_t strideA,
-                    int64_t strideB, int64_t strideC) {
+                    int64_t strideB, int64_t strideC, int k_chunk) {
   bf16_t* C = static_cast<bf16_t*>(Cv);
-  A += blockIdx.z * strideA;
-  B += blockIdx.z * strideB;
-  C += blockIdx.z * strideC;
-  if (EPI >= 2) Cpre += blockIdx.z * strideC;
+  float* Cf = static_cast<float*>(Cv);
+  int kbeg = 0, kend = K;
+  if (F32OUT) {  // split-K: blockIdx.z = K-chunk, fp32 partial output
+    kbeg = blockIdx.z * k_chunk;
+    kend = min(K, kbeg + k_chunk);
+    Cf += blockIdx.z * strideC;
+  } else {
+    A += blockIdx.z * strideA;
+    B += blockIdx.z * strideB;
+    C += blockIdx.z * strideC;
+    if (EPI >= 2) Cpre += blockIdx.z * strideC;
+  }
 
   // XCD-aware bijective swizzle (guide T1)
   const int nbx = N / BN;
@@ -104,17 +112,19 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
 
   f32x4 acc[8][4] = {};
 
-  const int nk = K / BK;
+  const int nk = (kend - kbeg) / BK;
 
   // A quarter q of tile t covers rows m0 + 64q; B quarter: n0 + 64q.
   auto stage_a = [&](int t, int q2) {  // stage quarters q2 and q2+1
-    stage_q(A, lda, m0 + 64 * q2, t * BK, aslot + ((4 * t + q2) & 7) * SLOT_E);
-    stage_q(A, lda, m0 + 64 * (q2 + 1), t * BK,
+    stage_q(A, lda, m0 + 64 * q2, kbeg + t * BK,
+            aslot + ((4 * t + q2) & 7) * SLOT_E);
+    stage_q(A, lda, m0 + 64 * (q2 + 1), kbeg + t * BK,
             aslot + ((4 * t + q2 + 1) & 7) * SLOT_E);
   };
   auto stage_b = [&](int t, int q2) {
-    stage_q(B, ldb, n0 + 64 * q2, t * BK, bslot + ((4 * t + q2) & 7) * SLOT_E);
-    stage_q(B, ldb, n0 + 64 * (q2 + 1), t * BK,
+    stage_q(B, ldb, n0 + 64 * q2, kbeg + t * BK,
+            bslot + ((4 * t + q2) & 7) * SLOT_E);
+    stage_q(B, ldb, n0 + 64 * (q2 + 1), kbeg + t * BK,
             bslot + ((4 * t + q2 + 1) & 7) * SLOT_E);
   };
 
@@ -210,6 +220,10 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
         float v = acc[mi][p][e];
         if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) v += bv[p];
         const int64_t off = (int64_t)m * ldc + n;
+        if (F32OUT) {
+          Cf[off] = v;
+          continue;
+        }
         if (EPI >= 2) {
           const bf16_t pre = f2bf(v);
           Cpre[off] = pre;
@@ -225,24 +239,37 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
 
 bool gemm256_supported(int M, int N, int K, int lda, int ldb, bool a_kc,
                        bool b_kc, int epi, int split_k) {
-  return a_kc && b_kc && split_k <= 1 && M % BM == 0 && N % BN == 0 &&
-         K % BK == 0 && K >= 2 * BK && (lda & 7) == 0 && (ldb & 7) == 0;
+  if (!(a_kc && b_kc && M % BM == 0 && N % BN == 0 && K % BK == 0 &&
+        (lda & 7) == 0 && (ldb & 7) == 0))
+    return false;
+  if (split_k <= 1) return K >= 2 * BK;
+  const int chunk = (K / split_k + BK - 1) / BK * BK;
+  return epi == 0 && chunk >= 2 * BK && K % chunk != BK;  // every chunk >= 2 tiles
 }
 
 void gemm256_bf16(const void* A, const void* B, void* C, void* c_pre,
                   const void* bias, int M, int N, int K, int lda, int ldb,
                   int ldc, int64_t stride_a, int64_t stride_b,
-                  int64_t stride_c, int batch, int epi, hipStream_t stream) {
-  dim3 grid((N / BN) * (M / BM), 1, batch);
+                  int64_t stride_c, int batch, int epi, int split_k,
+                  hipStream_t stream) {
   dim3 block(NTH);
   const bf16_t* a = static_cast<const bf16_t*>(A);
   const bf16_t* b = static_cast<const bf16_t*>(B);
   bf16_t* cp = static_cast<bf16_t*>(c_pre);
   const bf16_t* bi = static_cast<const bf16_t*>(bias);
+  if (split_k > 1) {
+    const int chunk = (K / split_k + BK - 1) / BK * BK;
+    dim3 gridk((N / BN) * (M / BM), 1, split_k);
+    hipLaunchKernelGGL((gemm256_kernel<0, true>), gridk, block, 0, stream,
+                       a, b, C, cp, bi, M, N, K, lda, ldb, ldc, stride_a,
+                       stride_b, stride_c, chunk);
+    return;
+  }
+  dim3 grid((N / BN) * (M / BM), 1, batch);
 #define G256(E)                                                             \
   hipLaunchKernelGGL((gemm256_kernel<E>), grid, block, 0, stream, a, b, C, \
                      cp, bi, M, N, K, lda, ldb, ldc, stride_a, stride_b,   \
-                     stride_c)
+                     stride_c, 0)
   switch (epi) {
     case 0: G256(0); break;
     case 1: G256(1); break;

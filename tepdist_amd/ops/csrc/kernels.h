@@ -28,7 +28,8 @@ bool gemm256_supported(int M, int N, int K, int lda, int ldb, bool a_kc,
 void gemm256_bf16(const void* A, const void* B, void* C, void* c_pre,
                   const void* bias, int M, int N, int K, int lda, int ldb,
                   int ldc, int64_t stride_a, int64_t stride_b,
-                  int64_t stride_c, int batch, int epi, hipStream_t stream);
+                  int64_t stride_c, int batch, int epi, int split_k,
+                  hipStream_t stream);
 // Reduce split-K fp32 partials [nparts, mn] into bf16 out [mn].
 void splitk_reduce(const float* parts, void* out, int nparts, int64_t mn,
                    hipStream_t stream);

@@ -60,7 +60,10 @@ def _gemm_raw(a: torch.Tensor, b_stored: torch.Tensor, a_kc: bool, b_kc: bool,
         cp = torch.empty_like(c)
 
     # split-K when the (M,N) tile grid underfills the 256-CU chip and K deep
-    blocks = ((M + 127) // 128) * ((N + 127) // 128)
+    if a_kc and b_kc and M % 256 == 0 and N % 256 == 0 and K % 64 == 0:
+        blocks = (M // 256) * (N // 256)
+    else:
+        blocks = ((M + 127) // 128) * ((N + 127) // 128)
     if batch == 1 and epi == 0 and K >= 2048 and blocks < 384:
         split_k = min(8, max(2, (512 + blocks - 1) // blocks))
         parts = torch.empty(split_k, M * N, dtype=torch.float32, device=dev)

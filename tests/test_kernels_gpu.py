@@ -59,6 +59,19 @@ def test_gemm_transposes(ta, tb):
     _assert_close_bf16(c, ref, rtol=3e-2, scale=math.sqrt(K))
 
 
+@pytest.mark.parametrize("M,N,K", [(512, 256, 8192), (256, 256, 4096),
+                                   (768, 512, 16384)])
+def test_gemm_splitk_256(M, N, K):
+    # canonical KCxKC shapes deep enough to trigger the split-K path of the
+    # 256-tile kernel (fp32 partials + reduce)
+    a = _randn(M, K, seed=11)
+    b = _randn(N, K, seed=12)
+    y = hip.matmul(a, b.t())
+    torch.cuda.synchronize()
+    ref = a.float() @ b.float().t()
+    _assert_close_bf16(y, ref, rtol=3e-2, scale=math.sqrt(K))
+
+
 def test_gemm_batched():
     B, M, N, K = 6, 64, 96, 32
     a = _randn(B, M, K, seed=5)
