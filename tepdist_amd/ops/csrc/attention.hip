@@ -42,6 +42,13 @@ DEV_INLINE int loff(int row, int col_e) {
   return row * C + (col_e ^ ((x << 3) & (C - 1)));
 }
 
+// branchless exp: one v_exp_f32, no libcall special-casing. Arguments are
+// always <= 0 here (running-max subtraction / masked to -3e38), so the only
+// edge is underflow to 0 — exactly what the hardware instruction does.
+DEV_INLINE float fast_exp(float x) {
+  return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
+}
+
 DEV_INLINE float warp16_max(float v) {
 #pragma unroll
   for (int d = 1; d < 16; d <<= 1) v = fmaxf(v, __shfl_xor(v, d, 64));
@@ -58,6 +65,19 @@ DEV_INLINE float warp16_sum(float v) {
 // forward
 // ---------------------------------------------------------------------------
 
+// Forward layout (CDNA-native): compute S^T = mfma(K, Q) so the score
+// C-fragment holds q in the lane index (q = lane&15) and kv across lane
+// groups/elements. Softmax row stats then reduce with TWO shfl_xor steps
+// (strides 16, 32) and the running max/sum/alpha are pure per-lane state.
+// P^T is consumed directly from the accumulators as the PV B-fragment via
+// the MFMA k-permutation invariance (a consistent reorder of the k slots of
+// both operands leaves the dot product unchanged): B slot (g, e) holds
+// kv = 32c + 4g + e (e<4, from C-frag 2c) or 32c + 16 + 4g + (e-4) (from
+// C-frag 2c+1), and the A operand (V^T, staged [D][KB] in LDS) is read with
+// the same kv order as two 8B vectors. O accumulates transposed ([d][q],
+// q = lane&15 — the same lane as the softmax state, so the alpha rescale is
+// a per-lane multiply) and is scattered once at the end. No P tile ever
+// touches LDS.
 template <int D>
 __launch_bounds__(NT) __global__
 void flash_fwd_kernel(const bf16_t* __restrict__ Q,
@@ -66,53 +86,46 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
                       float* __restrict__ LSE, int S, int H, float scale,
                       bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
                       int64_t o_bs, int64_t o_hs, int64_t o_rs) {
-  constexpr int DK = D / 32;   // k-chunks per fragment row
-  constexpr int DF = D / 16;   // output column fragments
+  constexpr int DK = D / 32;   // k-chunks per d-contraction fragment
+  constexpr int DF = D / 16;   // d row fragments of O^T
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int q0 = blockIdx.x * QB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wq0 = q0 + wave * 32;
+  const int g = lane >> 4;
+  const int wq0 = q0 + wave * 32;     // this wave's 32 q columns
 
-  // q/k/v share the (bs, hs, rs) layout (a packed qkv differs only in the
-  // base pointer); D stays contiguous
   const bf16_t* qp = Q + (int64_t)b * q_bs + (int64_t)h * q_hs;
   const bf16_t* kp = K + (int64_t)b * q_bs + (int64_t)h * q_hs;
   const bf16_t* vp = V + (int64_t)b * q_bs + (int64_t)h * q_hs;
   bf16_t* op = O + (int64_t)b * o_bs + (int64_t)h * o_hs;
 
-  __shared__ bf16_t smem[KB * D + D * KB + 4 * 32 * KB];
-  bf16_t* sK = smem;                    // [KB][D] k(=D)-contiguous
-  bf16_t* sVT = smem + KB * D;          // [D][KB] k(=kv)-contiguous
-  bf16_t* sP = smem + KB * D + D * KB + wave * 32 * KB;  // [32][KB]
+  __shared__ bf16_t smem[KB * D + D * KB];
+  bf16_t* sK = smem;                    // [KB][D] d-contiguous
+  bf16_t* sVT = smem + KB * D;          // [D][KB] kv-contiguous
 
-  // Q fragments for this wave's 32 rows, kept in registers for the whole row
+  // Q fragments (PV B-layout twin): nf-th 16-q group, kk-th 32-d chunk
   bf16x8 qf[2][DK];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int nf = 0; nf < 2; ++nf)
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
-      const int row = wq0 + mi * 16 + (lane & 15);
+      const int row = wq0 + nf * 16 + (lane & 15);
       bf16x8 v8 = {};
       if (row < S)
         v8 = *reinterpret_cast<const bf16x8*>(
-            qp + (int64_t)row * q_rs + kk * 32 + 8 * (lane >> 4));
-      qf[mi][kk] = v8;
+            qp + (int64_t)row * q_rs + kk * 32 + 8 * g);
+      qf[nf][kk] = v8;
     }
 
-  f32x4 acc_o[2][DF] = {};
-  float m_r[2][4], l_r[2][4];
-#pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
-#pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      m_r[mi][e] = -3.0e38f;
-      l_r[mi][e] = 0.f;
-    }
+  f32x4 acc_o[DF][2] = {};    // O^T: d = 16*df + 4g + e, q = lane&15 (+16nf)
+  float m_r[2], l_r[2];
+  m_r[0] = m_r[1] = -3.0e38f;
+  l_r[0] = l_r[1] = 0.f;
 
-  // staging registers (issue-early / write-late split, guide T14):
-  // K: 2 x 16B per thread; V: 8 x 4B per thread (v_perm transpose slabs)
+  // staging registers (issue-early / write-late split):
+  // K: KUN x 16B per thread; V: 8 x 4B per thread (v_perm transpose slabs)
   constexpr int KUN = KB * D / 8 / NT;
   bf16x8 krg[KUN];
   uint32_t vrg[8];
@@ -176,78 +189,84 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     if (kv0 + KB < kv_end) stage_load(kv0 + KB);  // overlap with compute
 
     if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
-      // --- S = Q K^T fragments ---
-      f32x4 sf[2][4] = {};
+      // --- S^T = K Q^T: kv = 16*mi + 4g + e, q = lane&15 + 16*nf ---
+      f32x4 st[4][2] = {};
 #pragma unroll
       for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
+        for (int mi = 0; mi < 4; ++mi) {
           const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              sK + loff<D>((lane & 15) + 16 * nf,
-                           8 * (lane >> 4) + 32 * kk));
+              sK + loff<D>((lane & 15) + 16 * mi, 8 * g + 32 * kk));
 #pragma unroll
-          for (int mi = 0; mi < 2; ++mi)
-            sf[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                qf[mi][kk], kf, sf[mi][nf], 0, 0, 0);
+          for (int nf = 0; nf < 2; ++nf)
+            st[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kf, qf[nf][kk], st[mi][nf], 0, 0, 0);
         }
-      // --- scale + mask; online softmax ---
+      // --- scale + mask; online softmax (per-lane q state) ---
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi) {
+      for (int nf = 0; nf < 2; ++nf) {
+        const int qg = wq0 + nf * 16 + (lane & 15);
+        float tmax = -3.0e38f;
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const int qg = wq0 + mi * 16 + 4 * (lane >> 4) + e;
-          float tmax = -3.0e38f;
+        for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            const int kg = kv0 + nf * 16 + (lane & 15);
-            float sv = sf[mi][nf][e] * scale;
-            if ((causal && kg > qg) || kg >= S || qg >= S) sv = -3.0e38f;
-            sf[mi][nf][e] = sv;
+          for (int e = 0; e < 4; ++e) {
+            const int kg = kv0 + mi * 16 + 4 * g + e;
+            const bool masked = (causal && kg > qg) || kg >= S || qg >= S;
+            const float sv = masked ? -3.0e38f : st[mi][nf][e] * scale;
+            st[mi][nf][e] = sv;
             tmax = fmaxf(tmax, sv);
           }
-          tmax = warp16_max(tmax);
-          const float mn = fmaxf(m_r[mi][e], tmax);
-          const float alpha = (mn <= -1.0e38f) ? 1.f
-                                               : __expf(m_r[mi][e] - mn);
-          m_r[mi][e] = mn;
-          float rs = 0.f;
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+        const float mn = fmaxf(m_r[nf], tmax);
+        const float alpha = fast_exp(m_r[nf] - mn);  // 0 if mn new, 1 if tied
+        m_r[nf] = mn;
+        float rs = 0.f;
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            const float sv = sf[mi][nf][e];
-            const float p = (sv <= -1.0e38f) ? 0.f : __expf(sv - mn);
-            sf[mi][nf][e] = p;
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            // sv - mn <= 0 always; fully-masked rows give -inf -> 0
+            const float p = fast_exp(st[mi][nf][e] - mn);
+            st[mi][nf][e] = p;
             rs += p;
           }
-          rs = warp16_sum(rs);
-          l_r[mi][e] = l_r[mi][e] * alpha + rs;
+        rs += __shfl_xor(rs, 16, 64);
+        rs += __shfl_xor(rs, 32, 64);
+        l_r[nf] = l_r[nf] * alpha + rs;
 #pragma unroll
-          for (int df = 0; df < DF; ++df) acc_o[mi][df][e] *= alpha;
-          // write P (bf16) to the wave-private LDS tile in C layout
-          const int prow = mi * 16 + 4 * (lane >> 4) + e;
+        for (int df = 0; df < DF; ++df)
 #pragma unroll
-          for (int nf = 0; nf < 4; ++nf)
-            sP[loff<KB>(prow, nf * 16 + (lane & 15))] =
-                f2bf(sf[mi][nf][e]);
-        }
+          for (int e = 0; e < 4; ++e) acc_o[df][nf][e] *= alpha;
       }
-      // ensure the wave's sP writes are visible to its own reads
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      // --- O += P V ---
+      // --- O^T += V^T P^T (P^T direct from accumulators, k-permuted) ---
 #pragma unroll
-      for (int kk2 = 0; kk2 < KB / 32; ++kk2) {
+      for (int c = 0; c < KB / 32; ++c) {
+        bf16x8 pb[2];
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
-          const bf16x8 pf = *reinterpret_cast<const bf16x8*>(
-              sP + loff<KB>(mi * 16 + (lane & 15),
-                            8 * (lane >> 4) + 32 * kk2));
+        for (int nf = 0; nf < 2; ++nf)
 #pragma unroll
-          for (int df = 0; df < DF; ++df) {
-            const bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-                sVT + loff<KB>((lane & 15) + 16 * df,
-                               8 * (lane >> 4) + 32 * kk2));
-            acc_o[mi][df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                pf, vf, acc_o[mi][df], 0, 0, 0);
+          for (int e = 0; e < 4; ++e) {
+            pb[nf][e] = f2bf(st[2 * c][nf][e]);
+            pb[nf][e + 4] = f2bf(st[2 * c + 1][nf][e]);
           }
+#pragma unroll
+        for (int df = 0; df < DF; ++df) {
+          const bf16x4 v0 = *reinterpret_cast<const bf16x4*>(
+              sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 4 * g));
+          const bf16x4 v1 = *reinterpret_cast<const bf16x4*>(
+              sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+          bf16x8 vfr;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            vfr[e] = v0[e];
+            vfr[e + 4] = v1[e];
+          }
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc_o[df][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                vfr, pb[nf], acc_o[df][nf], 0, 0, 0);
         }
       }
     }
@@ -258,23 +277,23 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
-  // --- epilogue: O /= l, write O and lse ---
+  // --- epilogue: O = O^T^T / l, write O and lse ---
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int nf = 0; nf < 2; ++nf) {
+    const int qg = wq0 + nf * 16 + (lane & 15);
+    if (qg >= S) continue;
+    const float lv = l_r[nf];
+    const float inv = (lv > 0.f) ? 1.0f / lv : 0.f;
+    if (g == 0)
+      LSE[(int64_t)bh * S + qg] =
+          (lv > 0.f) ? m_r[nf] + __logf(lv) : -3.0e38f;
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      const int qg = wq0 + mi * 16 + 4 * (lane >> 4) + e;
-      if (qg >= S) continue;
-      const float lv = l_r[mi][e];
-      const float inv = (lv > 0.f) ? 1.0f / lv : 0.f;
-      if ((lane & 15) == 0)
-        LSE[(int64_t)bh * S + qg] =
-            (lv > 0.f) ? m_r[mi][e] + __logf(lv) : -3.0e38f;
+    for (int df = 0; df < DF; ++df)
 #pragma unroll
-      for (int df = 0; df < DF; ++df)
-        op[(int64_t)qg * o_rs + df * 16 + (lane & 15)] =
-            f2bf(acc_o[mi][df][e] * inv);
-    }
+      for (int e = 0; e < 4; ++e)
+        op[(int64_t)qg * o_rs + 16 * df + 4 * g + e] =
+            f2bf(acc_o[df][nf][e] * inv);
+  }
 }
 
 }  // namespace
@@ -375,7 +394,19 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
   }
 }
 
-// one block = 64 kv rows of one (b,h); 4 waves x 16 kv rows.
+// one block = 64 kv rows of one (b,h); wave w owns the 16-kv slice
+// kv = kv0 + 16w + (lane&15) and iterates q tiles of 64.
+//
+// Layout: scores are computed UNtransposed, S = mfma(Q, K), so the
+// C-fragment holds kv in the lane index (kv = lane&15, the wave's slice)
+// and q across groups/elements (q = 16*mi + 4g + e). Then dV^T and dK^T
+// consume P and dS directly from the accumulators as B-fragments via the
+// MFMA k-permutation invariance (see flash_fwd_kernel) against
+// transpose-staged dO^T / Q^T A-operands — no P/dS LDS roundtrip for
+// those two products. Only dQ (contraction over kv, which lives in the
+// lane index) still routes dS through a shared [q][kv] LDS image; its
+// MFMAs read dS naturally and K^T from the block-constant sKT. dK/dV
+// accumulate transposed in registers across all q tiles and scatter once.
 template <int D>
 __launch_bounds__(NT) __global__
 void flash_bwd_kernel(const bf16_t* __restrict__ Q,
@@ -396,7 +427,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   const int kv0 = blockIdx.x * KB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wk0 = kv0 + wave * 16;   // this wave's 16 kv rows
+  const int g = lane >> 4;
 
   const int64_t qoff = (int64_t)b * q_bs + (int64_t)h * q_hs;
   const int64_t ooff = (int64_t)b * o_bs + (int64_t)h * o_hs;
@@ -406,7 +437,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   const bf16_t* dop = dO + ooff;
 
   __shared__ bf16_t smem[KB * D * 3 + D * KB + QT * D * 2 + D * QT +
-                         QT * KB + 4 * 16 * QT];
+                         QT * KB];
   bf16_t* sKb = smem;                       // [KB][D] natural
   bf16_t* sKT = sKb + KB * D;               // [D][KB]
   bf16_t* sVb = sKT + D * KB;               // [KB][D] natural
@@ -415,7 +446,6 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   bf16_t* sdO = sQT + D * QT;               // [QT][D] natural
   bf16_t* sdOT = sdO + QT * D;              // [D][QT]
   bf16_t* sdS = sdOT + D * QT;              // [QT][KB] shared
-  bf16_t* sPT = sdS + QT * KB + wave * 16 * QT;  // [16][QT] wave-private
   __shared__ float sLSE[QT], sDELTA[QT];
 
   // stage K, V tiles (fixed for the block)
@@ -436,18 +466,18 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   }
   __syncthreads();
 
-  // A fragments of this wave's K and V rows (constant across q tiles)
+  // B fragments of this wave's K and V slice (constant across q tiles)
   bf16x8 kf[DK], vf[DK];
 #pragma unroll
   for (int kk = 0; kk < DK; ++kk) {
     kf[kk] = *reinterpret_cast<const bf16x8*>(
-        sKb + loff<D>(wave * 16 + (lane & 15), 8 * (lane >> 4) + 32 * kk));
+        sKb + loff<D>(wave * 16 + (lane & 15), 8 * g + 32 * kk));
     vf[kk] = *reinterpret_cast<const bf16x8*>(
-        sVb + loff<D>(wave * 16 + (lane & 15), 8 * (lane >> 4) + 32 * kk));
+        sVb + loff<D>(wave * 16 + (lane & 15), 8 * g + 32 * kk));
   }
 
-  f32x4 acc_dk[DF] = {};
-  f32x4 acc_dv[DF] = {};
+  f32x4 acc_dkT[DF] = {};   // [d = 16df+4g+e][kv = wave*16 + lane&15]
+  f32x4 acc_dvT[DF] = {};
 
   // issue-early / write-late staging registers for the Q and dO tiles
   constexpr int QUN = QT * D / 8 / NT;
@@ -543,107 +573,101 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   for (int q0 = q_start; q0 < S; q0 += QT) {
     if (q0 + QT < S) tile_load(q0 + QT);  // overlap with compute
 
-    // S^T = K Q^T ; dP^T = V dO^T   (both natural-layout B reads)
+    // S = Q K^T ; dP = dO V^T  — C-frags: q = 16mi+4g+e, kv = lane&15
     f32x4 st[4] = {};
     f32x4 dpt[4] = {};
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
-      for (int nf = 0; nf < 4; ++nf) {
-        const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-            sQ + loff<D>((lane & 15) + 16 * nf, 8 * (lane >> 4) + 32 * kk));
-        st[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf[kk], qb, st[nf],
+      for (int mi = 0; mi < 4; ++mi) {
+        const bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+            sQ + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
+        st[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa, kf[kk], st[mi],
                                                          0, 0, 0);
-        const bf16x8 db = *reinterpret_cast<const bf16x8*>(
-            sdO + loff<D>((lane & 15) + 16 * nf, 8 * (lane >> 4) + 32 * kk));
-        dpt[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf[kk], db,
-                                                          dpt[nf], 0, 0, 0);
+        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+            sdO + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
+        dpt[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, vf[kk],
+                                                          dpt[mi], 0, 0, 0);
       }
 
-    // P^T and dS^T (elementwise, C layout: kv row = 4*(lane>>4)+e within
-    // this wave's 16; q col = q0 + nf*16 + (lane&15))
+    // P and dS (elementwise); kv fixed per lane, q varies per element
+    const int kvg = kv0 + wave * 16 + (lane & 15);
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
-      const int qcol = nf * 16 + (lane & 15);
-      const int qg = q0 + qcol;
-      const float lse = sLSE[qcol];
-      const float dl = sDELTA[qcol];
+    for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
-        const int kvg = wk0 + 4 * (lane >> 4) + e;
-        float pt = 0.f;
-        if (qg < S && kvg < S && (!causal || kvg <= qg) && lse > -1.0e38f)
-          pt = __expf(st[nf][e] * scale - lse);
-        st[nf][e] = pt;                                  // now P^T
-        dpt[nf][e] = pt * (dpt[nf][e] - dl) * scale;     // now dS^T
+        const int qrow = 16 * mi + 4 * g + e;
+        const int qg = q0 + qrow;
+        const float lse = sLSE[qrow];
+        const float dl = sDELTA[qrow];
+        const bool valid = qg < S && kvg < S && (!causal || kvg <= qg) &&
+                           lse > -1.0e38f;
+        const float arg = valid ? st[mi][e] * scale - lse : -3.0e38f;
+        const float pt = fast_exp(arg);                  // select, not branch
+        st[mi][e] = pt;                                  // now P
+        dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale;     // now dS
       }
-    }
 
-    // write P^T to the wave tile, dV += P^T dO (via sdOT)
+    // scatter dS to the shared [q][kv] image early — the stores retire
+    // under the MFMA section below, and the barrier after it publishes them
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
+    for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
       for (int e = 0; e < 4; ++e)
-        sPT[loff<QT>(4 * (lane >> 4) + e, nf * 16 + (lane & 15))] =
-            f2bf(st[nf][e]);
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-#pragma unroll
-    for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
-      const bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-          sPT + loff<QT>(lane & 15, 8 * (lane >> 4) + 32 * kk2));
-#pragma unroll
-      for (int df = 0; df < DF; ++df) {
-        const bf16x8 dob = *reinterpret_cast<const bf16x8*>(
-            sdOT + loff<QT>((lane & 15) + 16 * df,
-                            8 * (lane >> 4) + 32 * kk2));
-        acc_dv[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, dob,
-                                                             acc_dv[df],
-                                                             0, 0, 0);
-      }
-    }
+        sdS[loff<KB>(16 * mi + 4 * g + e, wave * 16 + (lane & 15))] =
+            f2bf(dpt[mi][e]);
 
-    // write dS^T to the wave tile (reuse) and to the shared [q][kv] image
+    // dV^T += dO^T P and dK^T += Q^T dS (both k-permuted B from regs)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf)
+    for (int c = 0; c < QT / 32; ++c) {
+      bf16x8 pb, db;
 #pragma unroll
       for (int e = 0; e < 4; ++e) {
-        const bf16_t dsv = f2bf(dpt[nf][e]);
-        sPT[loff<QT>(4 * (lane >> 4) + e, nf * 16 + (lane & 15))] = dsv;
-        sdS[loff<KB>(nf * 16 + (lane & 15),
-                     wave * 16 + 4 * (lane >> 4) + e)] = dsv;
+        pb[e] = f2bf(st[2 * c][e]);
+        pb[e + 4] = f2bf(st[2 * c + 1][e]);
+        db[e] = f2bf(dpt[2 * c][e]);
+        db[e + 4] = f2bf(dpt[2 * c + 1][e]);
       }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    // dK += dS^T Q (via sQT)
-#pragma unroll
-    for (int kk2 = 0; kk2 < QT / 32; ++kk2) {
-      const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-          sPT + loff<QT>(lane & 15, 8 * (lane >> 4) + 32 * kk2));
 #pragma unroll
       for (int df = 0; df < DF; ++df) {
-        const bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-            sQT + loff<QT>((lane & 15) + 16 * df,
-                           8 * (lane >> 4) + 32 * kk2));
-        acc_dk[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, qb,
-                                                             acc_dk[df],
-                                                             0, 0, 0);
+        const bf16x4 a0 = *reinterpret_cast<const bf16x4*>(
+            sdOT + loff<QT>(16 * df + (lane & 15), 32 * c + 4 * g));
+        const bf16x4 a1 = *reinterpret_cast<const bf16x4*>(
+            sdOT + loff<QT>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+        const bf16x4 q0f = *reinterpret_cast<const bf16x4*>(
+            sQT + loff<QT>(16 * df + (lane & 15), 32 * c + 4 * g));
+        const bf16x4 q1f = *reinterpret_cast<const bf16x4*>(
+            sQT + loff<QT>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+        bf16x8 afr, qfr;
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          afr[e] = a0[e];
+          afr[e + 4] = a1[e];
+          qfr[e] = q0f[e];
+          qfr[e + 4] = q1f[e];
+        }
+        acc_dvT[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afr, pb, acc_dvT[df], 0, 0, 0);
+        acc_dkT[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            qfr, db, acc_dkT[df], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // sdS complete; sQ/sQT/sdO/sdOT reads done
     if (q0 + QT < S) tile_write();  // overlaps the dQ phase (sdS/sKT only)
 
-    // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS @ K (via sKT)
+    // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS K (sdS x sKT)
     {
       f32x4 acc_dq[DF] = {};
 #pragma unroll
       for (int kk2 = 0; kk2 < KB / 32; ++kk2) {
         const bf16x8 dsa = *reinterpret_cast<const bf16x8*>(
-            sdS + loff<KB>(wave * 16 + (lane & 15),
-                           8 * (lane >> 4) + 32 * kk2));
+            sdS + loff<KB>(wave * 16 + (lane & 15), 8 * g + 32 * kk2));
 #pragma unroll
         for (int df = 0; df < DF; ++df) {
           const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-              sKT + loff<KB>((lane & 15) + 16 * df,
-                             8 * (lane >> 4) + 32 * kk2));
+              sKT + loff<KB>((lane & 15) + 16 * df, 8 * g + 32 * kk2));
           acc_dq[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, kb,
                                                                acc_dq[df],
                                                                0, 0, 0);
@@ -653,7 +677,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       for (int df = 0; df < DF; ++df)
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
-          const int qg = q0 + wave * 16 + 4 * (lane >> 4) + e;
+          const int qg = q0 + wave * 16 + 4 * g + e;
           if (qg < S)
             atomicAdd(&dQws[(int64_t)bh * S * D + (int64_t)qg * D +
                             df * 16 + (lane & 15)],
@@ -663,19 +687,25 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
     __syncthreads();  // next tile's images complete before its reads
   }
 
-  // write dK, dV (C layout scatter)
+  // write dK, dV (transposed-accumulator scatter: kv = wave's lane slice)
+  {
+    const int kvg = kv0 + wave * 16 + (lane & 15);
+    if (kvg < S) {
 #pragma unroll
-  for (int df = 0; df < DF; ++df)
+      for (int df = 0; df < DF; ++df) {
+        bf16x4 kv4, vv4;
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-      const int kvg = wk0 + 4 * (lane >> 4) + e;
-      if (kvg >= S) continue;
-      const int64_t off = qoff + (int64_t)kvg * q_rs + df * 16 + (lane & 15);
-      dK[off] = f2bf(acc_dk[df][e]);
-      dV[off] = f2bf(acc_dv[df][e]);
+        for (int e = 0; e < 4; ++e) {
+          kv4[e] = f2bf(acc_dkT[df][e]);
+          vv4[e] = f2bf(acc_dvT[df][e]);
+        }
+        const int64_t off = qoff + (int64_t)kvg * q_rs + 16 * df + 4 * g;
+        *reinterpret_cast<bf16x4*>(dK + off) = kv4;
+        *reinterpret_cast<bf16x4*>(dV + off) = vv4;
+      }
     }
+  }
 }
-
 
 
 // delta kernel needs contiguous dO/O rows: the host passes per-(b,h)
