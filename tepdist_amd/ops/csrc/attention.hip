@@ -191,6 +191,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
       // --- S^T = K Q^T: kv = 16*mi + 4g + e, q = lane&15 + 16*nf ---
       f32x4 st[4][2] = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
@@ -202,21 +203,36 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
             st[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 kf, qf[nf][kk], st[mi][nf], 0, 0, 0);
         }
-      // --- scale + mask; online softmax (per-lane q state) ---
+      __builtin_amdgcn_s_setprio(0);
+      // --- scale + mask; online softmax (per-lane q state). Tiles fully
+      // inside the causal triangle and the sequence skip the per-element
+      // mask selects (wave-uniform branch, no divergence) ---
+      const bool inner = kv0 + KB <= wq0 && kv0 + KB <= S && wq0 + 32 <= S;
 #pragma unroll
       for (int nf = 0; nf < 2; ++nf) {
         const int qg = wq0 + nf * 16 + (lane & 15);
         float tmax = -3.0e38f;
+        if (inner) {
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+          for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            const int kg = kv0 + mi * 16 + 4 * g + e;
-            const bool masked = (causal && kg > qg) || kg >= S || qg >= S;
-            const float sv = masked ? -3.0e38f : st[mi][nf][e] * scale;
-            st[mi][nf][e] = sv;
-            tmax = fmaxf(tmax, sv);
-          }
+            for (int e = 0; e < 4; ++e) {
+              const float sv = st[mi][nf][e] * scale;
+              st[mi][nf][e] = sv;
+              tmax = fmaxf(tmax, sv);
+            }
+        } else {
+#pragma unroll
+          for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              const int kg = kv0 + mi * 16 + 4 * g + e;
+              const bool masked = (causal && kg > qg) || kg >= S || qg >= S;
+              const float sv = masked ? -3.0e38f : st[mi][nf][e] * scale;
+              st[mi][nf][e] = sv;
+              tmax = fmaxf(tmax, sv);
+            }
+        }
         tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
         tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
         const float mn = fmaxf(m_r[nf], tmax);
@@ -241,6 +257,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
           for (int e = 0; e < 4; ++e) acc_o[df][nf][e] *= alpha;
       }
       // --- O^T += V^T P^T (P^T direct from accumulators, k-permuted) ---
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int c = 0; c < KB / 32; ++c) {
         bf16x8 pb[2];
@@ -269,6 +286,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
                 vfr, pb[nf], acc_o[df][nf], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();  // all waves done reading sK/sVT
     if (kv0 + KB < kv_end) {
@@ -590,23 +608,39 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
                                                           dpt[mi], 0, 0, 0);
       }
 
-    // P and dS (elementwise); kv fixed per lane, q varies per element
+    // P and dS (elementwise); kv fixed per lane, q varies per element.
+    // Inner tiles (wave's kv slice fully below the diagonal, rows in
+    // range) skip the per-element mask selects — wave-uniform branch.
     const int kvg = kv0 + wave * 16 + (lane & 15);
+    const bool winner = (!causal || kv0 + wave * 16 + 15 < q0) &&
+                        q0 + QT <= S && kv0 + KB <= S;
+    if (winner) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        const int qrow = 16 * mi + 4 * g + e;
-        const int qg = q0 + qrow;
-        const float lse = sLSE[qrow];
-        const float dl = sDELTA[qrow];
-        const bool valid = qg < S && kvg < S && (!causal || kvg <= qg) &&
-                           lse > -1.0e38f;
-        const float arg = valid ? st[mi][e] * scale - lse : -3.0e38f;
-        const float pt = fast_exp(arg);                  // select, not branch
-        st[mi][e] = pt;                                  // now P
-        dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale;     // now dS
-      }
+        for (int e = 0; e < 4; ++e) {
+          const int qrow = 16 * mi + 4 * g + e;
+          const float pt = fast_exp(st[mi][e] * scale - sLSE[qrow]);
+          st[mi][e] = pt;
+          dpt[mi][e] = pt * (dpt[mi][e] - sDELTA[qrow]) * scale;
+        }
+    } else {
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int qrow = 16 * mi + 4 * g + e;
+          const int qg = q0 + qrow;
+          const float lse = sLSE[qrow];
+          const float dl = sDELTA[qrow];
+          const bool valid = qg < S && kvg < S && (!causal || kvg <= qg) &&
+                             lse > -1.0e38f;
+          const float arg = valid ? st[mi][e] * scale - lse : -3.0e38f;
+          const float pt = fast_exp(arg);              // select, not branch
+          st[mi][e] = pt;                              // now P
+          dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale; // now dS
+        }
+    }
 
     // scatter dS to the shared [q][kv] image early — the stores retire
     // under the MFMA section below, and the barrier after it publishes them
