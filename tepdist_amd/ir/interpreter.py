@@ -14,14 +14,16 @@ import torch
 
 from tepdist_amd import ops
 from tepdist_amd.ir.graph import Graph, Node
+from tepdist_amd.parallel import mappings
 
 
 class GraphInterpreter:
     def __init__(self, graph: Graph, device: str = "cpu",
-                 dtype=torch.float32):
+                 dtype=torch.float32, group=None):
         self.g = graph
         self.device = device
         self.dtype = dtype
+        self.group = group  # process group for reshard collective nodes
 
     def run(self, feeds: Dict[str, torch.Tensor],
             variables: Dict[str, torch.Tensor]) -> Dict[int, torch.Tensor]:
@@ -93,4 +95,26 @@ class GraphInterpreter:
             return ins[0].permute(n.attrs["perm"]).contiguous()
         if n.op == "elementwise":
             return ins[0]
+        if n.op == "scale":
+            return ins[0] * n.attrs.get("scale", 1.0)
+        # reshard collectives inserted by the SpmdTransform (autograd-aware:
+        # backward of the transformed graph is correct by construction)
+        if n.op == "all_reduce":
+            return mappings.reduce_from_group(ins[0], self.group)
+        if n.op == "copy_to":
+            return mappings.copy_to_group(ins[0], self.group)
+        if n.op == "all_gather":
+            return mappings.gather_from_group(ins[0], self.group,
+                                              dim=n.attrs.get("dim", 0))
+        if n.op == "dynamic_slice":
+            return mappings.scatter_to_group(ins[0], self.group,
+                                             dim=n.attrs.get("dim", 0))
+        if n.op == "all_to_all":
+            # split-dim -> split-dim reshard; composed gather+slice keeps
+            # autograd exact (a fused all_to_all_single path would halve the
+            # bytes; the MoE layer uses that form)
+            x = mappings.gather_from_group(ins[0], self.group,
+                                           dim=n.attrs["src_dim"])
+            return mappings.scatter_to_group(x, self.group,
+                                             dim=n.attrs["dst_dim"])
         raise NotImplementedError(f"op {n.op}")
