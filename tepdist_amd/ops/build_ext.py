@@ -57,7 +57,31 @@ def _needs_build() -> bool:
     return any(d.stat().st_mtime > so_mtime for d in deps)
 
 
+RT_CSRC = OPS_DIR.parent / "runtime" / "csrc"
+RT_SO = OPS_DIR.parent / "runtime" / "_tepdist_rt.so"
+CXX = os.environ.get("CXX", "g++")
+
+
+def build_rt(force: bool = False, verbose: bool = True) -> Path:
+    """Builds the pure-C++ native runtime core (scheduler simulator,
+    lifetime/GC, dominance tree) — no HIP dependency, imports on CPU."""
+    src = RT_CSRC / "rt_core.cpp"
+    if not force and RT_SO.exists() and \
+            RT_SO.stat().st_mtime > src.stat().st_mtime:
+        return RT_SO
+    cmd = [CXX, "-O3", "-std=c++17", "-shared", "-fPIC",
+           *_includes(), str(src), "-o", str(RT_SO)]
+    if verbose:
+        print("[build_ext]", " ".join(cmd), flush=True)
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        raise RuntimeError(
+            f"rt_core build failed:\n{r.stdout}\n{r.stderr}")
+    return RT_SO
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
+    build_rt(force, verbose)
     if not force and not _needs_build():
         return OUT_SO
     BUILD.mkdir(exist_ok=True)

@@ -22,6 +22,11 @@ from tepdist_amd.config import get_env
 from tepdist_amd.planner.cost_model import CostModel
 from tepdist_amd.runtime.task_graph import TaskDAG, TaskNode, TaskType
 
+try:  # native runtime core (runtime/csrc/rt_core.cpp); Python fallback below
+    from tepdist_amd.runtime import _tepdist_rt as _rt
+except ImportError:  # pragma: no cover - the build ships the .so in-tree
+    _rt = None
+
 
 @dataclass
 class ScheduleResult:
@@ -52,12 +57,16 @@ class TaskScheduler:
             return t.out_bytes * 3 / (self.cm.hw.hbm_gbps * 1e9) + 5e-6
         return 1e-6
 
-    def schedule(self, sched_cnt: int = None) -> ScheduleResult:
+    def schedule(self, sched_cnt: int = None,
+                 native: bool = None) -> ScheduleResult:
         cnt = sched_cnt or get_env().group_sched_count
+        if native is None:
+            native = _rt is not None
         policies = ["bw_first", "fifo"][:max(cnt, 1)]
         best = None
         for pol in policies:
-            r = self._simulate(pol)
+            r = self._simulate_native(pol) if native and _rt is not None \
+                else self._simulate(pol)
             if r is not None and (best is None or r.makespan < best.makespan):
                 best = r
         assert best is not None, "no feasible schedule"
@@ -153,3 +162,62 @@ class TaskScheduler:
                         push(c)
         makespan = max(finish.values()) if finish else 0.0
         return ScheduleResult(order, makespan, peak, policy)
+
+    # -- native path (C++ rt_core; same semantics as _simulate) -------------
+
+    def _marshal(self):
+        """Flattens the DAG into dense arrays for the native simulator.
+        Task ids may be sparse; idx maps dense<->sparse."""
+        ids = sorted(self.dag.tasks)
+        idx = {tid: i for i, tid in enumerate(ids)}
+        kind, device, micro, dur, out_bytes, release_at = [], [], [], [], [], []
+        p_off, p_ids, c_off, c_ids = [0], [], [0], []
+        for tid in ids:
+            t = self.dag.tasks[tid]
+            kind.append(1 if t.type == TaskType.COMPUTE_FW else
+                        0 if t.type == TaskType.COMPUTE_BW else -1)
+            device.append(t.device)
+            micro.append(t.split.micro)
+            dur.append(self._duration(t))
+            out_bytes.append(t.out_bytes)
+            release_at.append(idx.get(t.release_at, -1)
+                              if t.release_at is not None else -1)
+            p_ids.extend(idx[p] for p in t.parents)
+            p_off.append(len(p_ids))
+            c_ids.extend(idx[c] for c in t.children)
+            c_off.append(len(c_ids))
+        return ids, (kind, device, micro, dur, out_bytes, release_at,
+                     p_off, p_ids, c_off, c_ids)
+
+    def _simulate_native(self, policy: str) -> Optional[ScheduleResult]:
+        ids, arrs = self._marshal()
+        r = _rt.simulate(*arrs, policy == "bw_first", self.micro_limit,
+                         self.mem_cap)
+        if not r.feasible:
+            return None
+        order = {dev: [ids[i] for i in lst] for dev, lst in r.order.items()}
+        return ScheduleResult(order, r.makespan, dict(r.peak), policy)
+
+    def gc_plan(self, order: Dict[int, List[int]]) -> Dict[int, List[int]]:
+        """Buffer-release plan: task id -> producer task ids whose outputs
+        die when it completes (the reference's MakeTaskGraphGCPlan,
+        execution_plan.cc:28-68, via the lifetime tracker). Uses the native
+        core; falls back to a direct Python computation."""
+        ids, arrs = self._marshal()
+        idx = {tid: i for i, tid in enumerate(ids)}
+        if _rt is not None:
+            dense = {d: [idx[t] for t in lst] for d, lst in order.items()}
+            plan = _rt.gc_plan(arrs[6 + 2], arrs[7 + 2], dense)
+            return {ids[k]: [ids[p] for p in v] for k, v in plan.items()}
+        pos = {}
+        for lst in order.values():
+            for i, t in enumerate(lst):
+                pos[t] = i
+        plan: Dict[int, List[int]] = {}
+        for tid, t in self.dag.tasks.items():
+            if not t.children:
+                continue
+            last = max(t.children, key=lambda c: pos.get(c, -1))
+            if pos.get(last, -1) >= 0:
+                plan.setdefault(last, []).append(tid)
+        return plan

@@ -192,3 +192,52 @@ def build_task_dag(num_stages: int, num_micro: int, stage_flops=None,
         dag.edge(ag, out)
         dag.edge(out, sink)
     return dag
+
+
+def idoms(dag: "TaskDAG") -> dict:
+    """Immediate dominators of every task (virtual root = -1): the
+    reference's TaskDAG dominance tree (Cooper-Harvey-Kennedy,
+    task_graph.h:643-717), used by the GC planner. Native C++ core with a
+    Python fallback."""
+    ids = sorted(dag.tasks)
+    idx = {tid: i for i, tid in enumerate(ids)}
+    p_off, p_ids = [0], []
+    for tid in ids:
+        p_ids.extend(idx[p] for p in dag.tasks[tid].parents)
+        p_off.append(len(p_ids))
+    try:
+        from tepdist_amd.runtime import _tepdist_rt as _rt
+        dom = _rt.idom_tree(p_off, p_ids)
+        return {ids[i]: (ids[d] if d >= 0 else -1)
+                for i, d in enumerate(dom)}
+    except ImportError:
+        pass
+    # Python fallback: in a DAG processed in topo order every predecessor
+    # is finalized first, so one Cooper-style pass suffices
+    order = [idx[t.id] for t in dag.topo()]
+    rpo = {u: i for i, u in enumerate(order)}
+    idom = {}
+
+    def intersect(a, b):
+        while a != b:
+            if a == -1 or b == -1:
+                return -1
+            while a != -1 and rpo[a] > rpo[b]:
+                a = idom[a]
+            while b != -1 and a != -1 and rpo[b] > rpo[a]:
+                b = idom[b]
+            if a == -1 or b == -1:
+                return -1
+        return a
+
+    for u in order:
+        preds = [idx[p] for p in dag.tasks[ids[u]].parents]
+        if not preds:
+            idom[u] = -1
+            continue
+        nd = preds[0]
+        for p in preds[1:]:
+            nd = intersect(nd, p)
+        idom[u] = nd
+    return {ids[u]: (ids[d] if d != -1 else -1)
+            for u, d in idom.items()}
