@@ -39,6 +39,7 @@ class ServiceEnv:
     # --- planner mode (reference: auto_parallel.cc:395-409) ---
     rule_mode: bool = field(default_factory=lambda: _env("RULE_MODE", bool, False))
     ignore_annotation: bool = field(default_factory=lambda: _env("IGNORE_ANNOTATION", bool, False))
+    aux_affinity: bool = field(default_factory=lambda: _env("AUX_AFFINITY", bool, True))
     opt_level: int = field(default_factory=lambda: _env("OPT_LEVEL", int, 3))
 
     # --- pipeline / micro-batching (reference: service_env.h) ---

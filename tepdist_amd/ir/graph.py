@@ -71,7 +71,29 @@ class Graph:
     # -- structure ----------------------------------------------------------
 
     def topo(self) -> List[Node]:
-        return [self.nodes[i] for i in sorted(self.nodes)]
+        """Topological order, smallest-id-first among ready nodes (ids are
+        creation order, so graphs built front-to-back keep their layout;
+        passes that append nodes mid-graph — liveness cloning — still
+        serialize correctly)."""
+        import heapq
+        indeg = {i: 0 for i in self.nodes}
+        cons: Dict[int, List[int]] = {i: [] for i in self.nodes}
+        for n in self.nodes.values():
+            for i in n.inputs:
+                indeg[n.id] += 1
+                cons[i].append(n.id)
+        heap = [i for i, d in indeg.items() if d == 0]
+        heapq.heapify(heap)
+        out = []
+        while heap:
+            i = heapq.heappop(heap)
+            out.append(self.nodes[i])
+            for c in cons[i]:
+                indeg[c] -= 1
+                if indeg[c] == 0:
+                    heapq.heappush(heap, c)
+        assert len(out) == len(self.nodes), "cycle in graph"
+        return out
 
     def consumers(self) -> Dict[int, List[int]]:
         cons: Dict[int, List[int]] = {i: [] for i in self.nodes}

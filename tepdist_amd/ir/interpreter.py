@@ -100,7 +100,16 @@ class GraphInterpreter:
         # reshard collectives inserted by the SpmdTransform (autograd-aware:
         # backward of the transformed graph is correct by construction)
         if n.op == "all_reduce":
+            if len(ins) > 1:  # combined bundle (planner/combiner.py): one
+                # flat collective covers all members; bundle_get projects
+                flat = torch.cat([t.reshape(-1) for t in ins])
+                return mappings.reduce_from_group(flat, self.group)
             return mappings.reduce_from_group(ins[0], self.group)
+        if n.op == "bundle_get":
+            offs = n.attrs["offsets"]
+            idx = n.attrs["index"]
+            start = int(sum(offs[:idx]))
+            return ins[0].narrow(0, start, int(offs[idx])).reshape(n.shape)
         if n.op == "copy_to":
             return mappings.copy_to_group(ins[0], self.group)
         if n.op == "all_gather":

@@ -103,10 +103,35 @@ class CostSpmdStrategy:
             prev_out_node = sub[-1]
 
         best_cost, best_specs = min(prev_best.values(), key=lambda v: v[0])
-        # any node not covered (dead code etc.) -> replicated
+        # uncovered nodes (dead code etc.): NeighborVote fallback (the
+        # reference's cost_spmd_strategy.cc:708) then replicated
+        self._neighbor_vote(best_specs)
         for i in self.g.nodes:
             best_specs.setdefault(i, DimStrategy.replicated(self.n))
+        # co-location affinity (InstAffinityMap): aux optimizer vars and
+        # aliased pairs adopt their leader's spec
+        from tepdist_amd.planner.affinity import InstAffinityMap
+        InstAffinityMap(self.g).apply(best_specs)
         return SpmdResult(best_specs, best_cost, used_ilp)
+
+    def _neighbor_vote(self, specs) -> None:
+        """Nodes without a planned spec adopt the majority strategy of
+        their planned neighbors (producers + consumers)."""
+        from collections import Counter
+        for n in self.g.topo():
+            if n.id in specs:
+                continue
+            votes = Counter()
+            for i in list(n.inputs) + self.cons[n.id]:
+                sp = specs.get(i)
+                if sp is not None and not sp.is_glue:
+                    votes[sp] += 1
+            if votes:
+                cand = votes.most_common(1)[0][0]
+                # only adopt if the op supports it exactly
+                from tepdist_amd.planner.rules import back_infer
+                if back_infer(self.g, n, cand, self.n) is not None:
+                    specs[n.id] = cand
 
     # ---------------------------------------------------------------------
 
