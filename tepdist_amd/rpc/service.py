@@ -139,7 +139,9 @@ class TepdistService:
         else:
             feeds = {k: v.to(self.device)
                      for k, v in req.get("inputs", {}).items()}
-        with self._lock:
+        from tepdist_amd.utils.tracing import get_tracer
+        with self._lock, get_tracer().span(
+                f"ExecutePlan/h{h}", args={"step": self.step_count + 1}):
             t0 = time.time()
             self.step_count += 1
             variables = {n: vs.tensor for n, vs in self.vars.items()}

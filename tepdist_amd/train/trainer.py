@@ -41,17 +41,22 @@ class Trainer:
             self.reducer.reset()
         total = 0.0
         n = self.grad_accum_steps
+        sync = get_env().sync_mode  # blocking-sync debug
+        from tepdist_amd.utils.tracing import trace_span
         for i in range(n):
-            input_ids, labels = batch_iter(i)
-            loss = self.model(input_ids, labels=labels)
-            scaled = loss / n
-            if self.reducer is not None and i == n - 1:
-                # overlap the all-reduce of each bucket with the remaining
-                # backward of the LAST micro-batch only (earlier micro-batches
-                # just accumulate locally - the sync-free property)
-                self.reducer.arm()
-            scaled.backward()
-            total += loss.item()
+            with trace_span(f"micro_batch/{i}"):
+                input_ids, labels = batch_iter(i)
+                loss = self.model(input_ids, labels=labels)
+                scaled = loss / n
+                if self.reducer is not None and i == n - 1:
+                    # overlap the all-reduce of each bucket with the
+                    # remaining backward of the LAST micro-batch only
+                    # (earlier micros accumulate locally - sync-free)
+                    self.reducer.arm()
+                scaled.backward()
+                total += loss.item()
+            if sync and torch.cuda.is_available():
+                torch.cuda.synchronize()
         if self.reducer is not None:
             self.reducer.finalize()
         self.opt.step()

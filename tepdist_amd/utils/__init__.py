@@ -1,0 +1,1 @@
+from tepdist_amd.utils.tracing import Tracer, get_tracer, trace_span
