@@ -744,21 +744,29 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
 
 // delta kernel needs contiguous dO/O rows: the host passes per-(b,h)
 // strided views by iterating bh in the kernel via strides
+// each row of D elements is handled by D/8 lanes (one bf16x8 each), so a
+// 64-lane wave covers 64*8/D rows per iteration (8 rows at D=64) — the
+// whole wave's 1 KiB load capacity is used every cycle.
 template <int D>
 __global__ void flash_bwd_delta_strided_kernel(
     const bf16_t* __restrict__ dO, const bf16_t* __restrict__ O,
     float* __restrict__ delta, int H, int S, int64_t o_bs, int64_t o_hs,
     int64_t o_rs) {
+  constexpr int LPR = D / 8;              // lanes per row
+  constexpr int RPW = WAVE / LPR;         // rows per wave per iteration
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int64_t rows = (int64_t)gridDim.y * S;  // gridDim.y == B*H
-  const int64_t nw = (int64_t)gridDim.x * (NT / WAVE);
+  const int sub = lane / LPR;             // row slot within the wave
+  const int c = (lane % LPR) * 8;
+  const int64_t nr = (int64_t)gridDim.x * (NT / WAVE) * RPW;
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int64_t off = (int64_t)b * o_bs + (int64_t)h * o_hs;
-  for (int64_t r = (int64_t)blockIdx.x * (NT / WAVE) + wid; r < S; r += nw) {
+  for (int64_t r0 = ((int64_t)blockIdx.x * (NT / WAVE) + wid) * RPW;
+       r0 < S; r0 += nr) {
+    const int64_t r = r0 + sub;
     float s = 0.f;
-    for (int c = lane * 8; c < D; c += WAVE * 8) {
+    if (r < S) {
       const bf16x8 a =
           *reinterpret_cast<const bf16x8*>(dO + off + r * o_rs + c);
       const bf16x8 bb =
@@ -766,8 +774,10 @@ __global__ void flash_bwd_delta_strided_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e) s += bf2f(a[e]) * bf2f(bb[e]);
     }
-    s = wave_allreduce_sum(s);
-    if (lane == 0) delta[(int64_t)bh * S + r] = s;
+    // reduce across the LPR lanes of each row slot
+#pragma unroll
+    for (int d2 = 1; d2 < LPR; d2 <<= 1) s += __shfl_xor(s, d2, 64);
+    if (r < S && (lane % LPR) == 0) delta[(int64_t)bh * S + r] = s;
   }
 }
 

@@ -207,6 +207,14 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
+  m.def("transpose_gelu_bwd", [](uintptr_t dy, uintptr_t pre, uintptr_t dgt,
+                                 uintptr_t dgn, int R, int C,
+                                 uintptr_t stream) {
+    transpose_gelu_bwd_bf16(reinterpret_cast<void*>(dy),
+                            reinterpret_cast<void*>(pre),
+                            reinterpret_cast<void*>(dgt),
+                            reinterpret_cast<void*>(dgn), R, C, S(stream));
+  });
   m.def("gelu_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t dx, int64_t n,
                        uintptr_t stream) {
     gelu_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),

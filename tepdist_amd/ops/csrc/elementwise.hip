@@ -95,6 +95,49 @@ void bias_sum_kernel(const bf16_t* __restrict__ dy, float* __restrict__ ws,
     if (c0 + e < cols) atomicAdd(&ws[c0 + e], acc[e]);
 }
 
+// streaming variant for cols % 512 == 0 (NV = cols/512 <= 8): each wave
+// reads whole ROWS sequentially (full 1 KiB per instruction, perfect
+// channel interleave) holding NV*8 per-lane column accumulators; waves
+// combine through LDS atomics, one global atomic pass per block.
+template <int NV>
+__launch_bounds__(NT) __global__
+void bias_sum_stream_kernel(const bf16_t* __restrict__ dy,
+                            float* __restrict__ ws, int64_t rows) {
+  const int cols = NV * 512;
+  __shared__ float acc_s[NV * 512];
+  for (int i = threadIdx.x; i < cols; i += NT) acc_s[i] = 0.f;
+  __syncthreads();
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int64_t nw = (int64_t)gridDim.x * (NT / 64);
+  const int64_t w = (int64_t)blockIdx.x * (NT / 64) + wid;
+  const int64_t rb = (rows * w) / nw, re = (rows * (w + 1)) / nw;
+
+  float acc[NV][8] = {};
+  // load ALL of a row's vectors before accumulating (separate destination
+  // registers -> the loads pipeline instead of serializing on vmcnt(0))
+  for (int64_t r = rb; r < re; ++r) {
+    const bf16_t* p = dy + r * cols + lane * 8;
+    bf16x8 v[NV];
+#pragma unroll
+    for (int j = 0; j < NV; ++j)
+      v[j] = *reinterpret_cast<const bf16x8*>(p + j * 512);
+#pragma unroll
+    for (int j = 0; j < NV; ++j)
+#pragma unroll
+      for (int e = 0; e < 8; ++e) acc[j][e] += bf2f(v[j][e]);
+  }
+#pragma unroll
+  for (int j = 0; j < NV; ++j)
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      atomicAdd(&acc_s[j * 512 + lane * 8 + e], acc[j][e]);
+  __syncthreads();
+  for (int i = threadIdx.x; i < cols; i += NT)
+    atomicAdd(&ws[i], acc_s[i]);
+}
+
 __global__ void cast_ws_kernel(const float* __restrict__ ws,
                                bf16_t* __restrict__ db, int cols) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
@@ -123,11 +166,31 @@ void bias_sum_bf16(const void* dy, void* db_out, float* ws_zeroed,
                    int64_t rows, int cols, hipStream_t stream) {
   bf16_t* db = static_cast<bf16_t*>(db_out);
   float* ws = ws_zeroed;  // caller-zeroed fp32 workspace of `cols`
-  const int ysplit = (int)std::min<int64_t>((rows + 31) / 32, 256);
-  const int ncg = (cols + 7) / 8;
-  dim3 grid((ncg + NT - 1) / NT, std::max(ysplit, 1));
-  hipLaunchKernelGGL(bias_sum_kernel, grid, dim3(NT), 0, stream,
-                     static_cast<const bf16_t*>(dy), ws, rows, cols);
+  const int nv = cols / 512;
+  if (cols % 512 == 0 && nv >= 1 && nv <= 8) {
+    const int blocks =
+        (int)std::max<int64_t>(1, std::min<int64_t>((rows + 63) / 64, 256));
+#define BS(NVV)                                                           \
+  hipLaunchKernelGGL(bias_sum_stream_kernel<NVV>, dim3(blocks), dim3(NT), \
+                     0, stream, static_cast<const bf16_t*>(dy), ws, rows)
+    switch (nv) {
+      case 1: BS(1); break;
+      case 2: BS(2); break;
+      case 3: BS(3); break;
+      case 4: BS(4); break;
+      case 5: BS(5); break;
+      case 6: BS(6); break;
+      case 7: BS(7); break;
+      case 8: BS(8); break;
+    }
+#undef BS
+  } else {
+    const int ysplit = (int)std::min<int64_t>((rows + 31) / 32, 256);
+    const int ncg = (cols + 7) / 8;
+    dim3 grid((ncg + NT - 1) / NT, std::max(ysplit, 1));
+    hipLaunchKernelGGL(bias_sum_kernel, grid, dim3(NT), 0, stream,
+                       static_cast<const bf16_t*>(dy), ws, rows, cols);
+  }
   hipLaunchKernelGGL(cast_ws_kernel, dim3((cols + NT - 1) / NT), dim3(NT), 0,
                      stream, ws, db, cols);
 }

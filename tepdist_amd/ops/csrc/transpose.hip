@@ -17,10 +17,18 @@ namespace {
 constexpr int NT = 256;
 constexpr int TS = 64;  // tile size
 
-// out[c][r] = in[r][c]; batched over blockIdx.z.
+// out[c][r] = in[r][c]; batched over blockIdx.z. GELUG additionally
+// multiplies by gelu'(pre) on the way through (the fused gelu-backward:
+// the wgrad path transposes dy anyway, so the elementwise backward rides
+// along instead of a separate full read+write pass) and NAT also writes
+// the post-gelu natural-layout image (the dgrad GEMM operand).
+template <bool GELUG, bool NAT>
 __launch_bounds__(NT) __global__
-void transpose_kernel(const bf16_t* __restrict__ in, bf16_t* __restrict__ out,
-                      int R, int C, int64_t stride_in, int64_t stride_out) {
+void transpose_kernel(const bf16_t* __restrict__ in,
+                      const bf16_t* __restrict__ pre,
+                      bf16_t* __restrict__ out,
+                      bf16_t* __restrict__ out_nat, int R, int C,
+                      int64_t stride_in, int64_t stride_out) {
   __shared__ bf16_t tile[TS * (TS + 8)];  // +16B row pad (b128-aligned)
   const bf16_t* src = in + blockIdx.z * stride_in;
   bf16_t* dst = out + blockIdx.z * stride_out;
@@ -40,6 +48,26 @@ void transpose_kernel(const bf16_t* __restrict__ in, bf16_t* __restrict__ out,
         v = *reinterpret_cast<const bf16x8*>(p);
       } else {
         for (int e = 0; e < 8 && c0 + c + e < C; ++e) v[e] = p[e];
+      }
+      if (GELUG) {
+        const bf16_t* q = pre + (int64_t)(r0 + r) * C + c0 + c;
+        bf16x8 pv = {};
+        if (c0 + c + 8 <= C) {
+          pv = *reinterpret_cast<const bf16x8*>(q);
+        } else {
+          for (int e = 0; e < 8 && c0 + c + e < C; ++e) pv[e] = q[e];
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          v[e] = f2bf(bf2f(v[e]) * gelu_grad_f(bf2f(pv[e])));
+      }
+      if (NAT) {
+        bf16_t* pn = out_nat + (int64_t)(r0 + r) * C + c0 + c;
+        if (c0 + c + 8 <= C) {
+          *reinterpret_cast<bf16x8*>(pn) = v;
+        } else {
+          for (int e = 0; e < 8 && c0 + c + e < C; ++e) pn[e] = v[e];
+        }
       }
     }
     *reinterpret_cast<bf16x8*>(tile + r * (TS + 8) + c) = v;
@@ -73,9 +101,21 @@ void transpose_bf16(const void* in, void* out, int R, int C,
                     int64_t stride_in, int64_t stride_out, int batch,
                     hipStream_t stream) {
   dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, batch);
-  hipLaunchKernelGGL(transpose_kernel, grid, dim3(NT), 0, stream,
-                     static_cast<const bf16_t*>(in),
-                     static_cast<bf16_t*>(out), R, C, stride_in, stride_out);
+  hipLaunchKernelGGL((transpose_kernel<false, false>), grid, dim3(NT), 0,
+                     stream, static_cast<const bf16_t*>(in), nullptr,
+                     static_cast<bf16_t*>(out), nullptr, R, C, stride_in,
+                     stride_out);
+}
+
+void transpose_gelu_bwd_bf16(const void* dy, const void* pre, void* dgelu_t,
+                             void* dgelu_nat, int R, int C,
+                             hipStream_t stream) {
+  dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, 1);
+  hipLaunchKernelGGL((transpose_kernel<true, true>), grid, dim3(NT), 0,
+                     stream, static_cast<const bf16_t*>(dy),
+                     static_cast<const bf16_t*>(pre),
+                     static_cast<bf16_t*>(dgelu_t),
+                     static_cast<bf16_t*>(dgelu_nat), R, C, 0, 0);
 }
 
 }  // namespace tepdist

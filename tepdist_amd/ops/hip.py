@@ -112,15 +112,21 @@ def linear_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     dy = dy.contiguous()
     M, N = dy.shape
     K = w.shape[1]
-    if act == "gelu":
-        dy_eff = torch.empty_like(dy)
-        ext.gelu_bwd(dy.data_ptr(), pre_act.data_ptr(), dy_eff.data_ptr(),
-                     dy.numel(), _stream())
-        dy = dy_eff
     # canonicalize to KC x KC: one tuned GEMM schedule serves every case,
     # with cheap materialized transposes (memory-bound, ~2% of GEMM time)
     wT = transpose2d(w)        # [K, N]
-    dyT = transpose2d(dy)      # [N, M]
+    if act == "gelu":
+        # fused gelu-backward: the wgrad transpose of dy applies gelu'(pre)
+        # in the same pass and emits BOTH layouts (saves the separate
+        # elementwise kernel's full read+write)
+        dy_eff = torch.empty_like(dy)
+        dyT = torch.empty(N, M, dtype=BF16, device=dy.device)
+        ext.transpose_gelu_bwd(dy.data_ptr(), pre_act.data_ptr(),
+                               dyT.data_ptr(), dy_eff.data_ptr(), M, N,
+                               _stream())
+        dy = dy_eff
+    else:
+        dyT = transpose2d(dy)  # [N, M]
     xT = transpose2d(x)        # [K, M]
     # dx[M,K] = dy[M,N] @ w[N,K]: A=dy KC (k=N), B=w^T stored [K,N] KC
     dx, _ = _gemm_raw(dy, wT, True, True, M, K, N, N, N, 0, 0, 1)
