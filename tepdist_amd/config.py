@@ -71,6 +71,7 @@ class ServiceEnv:
     # Blocking-sync debug mode: hipStreamSynchronize after every task
     # (SURVEY.md §5.2 recommends keeping such a kill-switch).
     sync_mode: bool = field(default_factory=lambda: _env("TEPDIST_SYNC_MODE", bool, False))
+    hip_graph: bool = field(default_factory=lambda: _env("TEPDIST_HIP_GRAPH", bool, True))
 
     # --- cluster ---
     cluster_spec: str = field(default_factory=lambda: _env("CLUSTER_SPEC", str, ""))

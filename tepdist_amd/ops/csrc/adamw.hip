@@ -94,13 +94,23 @@ constexpr int MT_CHUNK = 16384;
 // multi-tensor AdamW: one fused launch over all parameters. tabs holds 5
 // pointer tables (param, master, grad(bf16), exp_avg, exp_avg_sq), chunks
 // is [(tensor_idx, chunk_idx)] with MT_CHUNK elements per chunk.
+// hyper: optional device buffer {lr, inv_bc1, inv_bc2, _} read at launch
+// time — lets the kernel live inside a captured hipGraph while the
+// step-dependent bias correction still advances (the host updates the
+// 16-byte buffer before each replay; kernel args would be frozen).
 __launch_bounds__(NT) __global__
 void adamw_mt_kernel(const int64_t* __restrict__ tabs,
                      const int64_t* __restrict__ numel,
                      const float* __restrict__ wds,
                      const int* __restrict__ chunks, int nchunks, int nt,
                      float lr, float beta1, float beta2, float eps,
-                     float inv_bc1, float inv_bc2) {
+                     float inv_bc1, float inv_bc2,
+                     const float* __restrict__ hyper) {
+  if (hyper != nullptr) {
+    lr = hyper[0];
+    inv_bc1 = hyper[1];
+    inv_bc2 = hyper[2];
+  }
   for (int ci = blockIdx.x; ci < nchunks; ci += gridDim.x) {
     const int ti = chunks[2 * ci];
     const int64_t off = (int64_t)chunks[2 * ci + 1] * MT_CHUNK;
@@ -154,11 +164,11 @@ void adamw_mt_kernel(const int64_t* __restrict__ tabs,
 void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
                    const float* wds, const int* chunks, int nchunks, int nt,
                    float lr, float beta1, float beta2, float eps, float bc1,
-                   float bc2, hipStream_t stream) {
+                   float bc2, const float* hyper, hipStream_t stream) {
   const int blocks = std::min(nchunks, 2048);
   hipLaunchKernelGGL(adamw_mt_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
                      stream, tabs, numel, wds, chunks, nchunks, nt, lr,
-                     beta1, beta2, eps, 1.0f / bc1, 1.0f / bc2);
+                     beta1, beta2, eps, 1.0f / bc1, 1.0f / bc2, hyper);
 }
 
 }  // namespace tepdist

@@ -126,7 +126,7 @@ void adamw_bf16(void* param, float* master, const void* grad_bf16,
 void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
                    const float* wds, const int* chunks, int nchunks, int nt,
                    float lr, float beta1, float beta2, float eps, float bc1,
-                   float bc2, hipStream_t stream);
+                   float bc2, const float* hyper, hipStream_t stream);
 
 // --- diagnostics -----------------------------------------------------------
 void tr16_probe(float* out_pattern, float* out_uniform, hipStream_t stream);

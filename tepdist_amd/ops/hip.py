@@ -510,7 +510,29 @@ class AdamWMT:
         ext.adamw_mt(self.tabs.data_ptr(), self.numel.data_ptr(),
                      self.wds.data_ptr(), self.chunks.data_ptr(),
                      self.nchunks, self.nt, lr, beta1, beta2, eps, bc1, bc2,
-                     _stream())
+                     0, _stream())
+
+    # -- hipGraph-capture path: the kernel reads step-dependent scalars
+    # from a persistent device buffer the host updates before each replay
+    # (kernel args would be frozen inside the captured graph) ------------
+
+    def prepare_graph(self, grads):
+        gp = torch.tensor([g.data_ptr() for g in grads], dtype=torch.int64)
+        self._gslot.copy_(gp)
+        self.hyper = torch.zeros(4, dtype=torch.float32,
+                                 device=self.tabs.device)
+
+    def step_graph(self, beta1, beta2, eps):
+        ext.adamw_mt(self.tabs.data_ptr(), self.numel.data_ptr(),
+                     self.wds.data_ptr(), self.chunks.data_ptr(),
+                     self.nchunks, self.nt, 0.0, beta1, beta2, eps, 1.0,
+                     1.0, self.hyper.data_ptr(), _stream())
+
+    def set_hyper(self, lr, beta1, beta2, step_no):
+        vals = torch.tensor([lr, 1.0 / (1.0 - beta1 ** step_no),
+                             1.0 / (1.0 - beta2 ** step_no), 0.0],
+                            dtype=torch.float32)
+        self.hyper.copy_(vals, non_blocking=True)
 
 
 def adamw_step(param_bf16: torch.Tensor, master: torch.Tensor,

@@ -53,8 +53,28 @@ class AdamW:
             self._mt = be.AdamWMT(ps, [self.state[p] for p in ps], wds)
         return ps
 
+    # -- hipGraph capture support (trainer.py): the MT kernel switches to
+    # a device hyper-parameter buffer so the captured launch stays valid
+    # as the step count advances -----------------------------------------
+
+    def prepare_graph(self):
+        ps = self._try_mt()
+        if ps is None:
+            raise RuntimeError("graph capture needs the multi-tensor path")
+        self._graph_ps = ps
+        self._mt.prepare_graph([p.grad for p in ps])
+        self.graph_mode = True
+
+    def refresh_hyper(self):
+        self.step_count += 1
+        self._mt.set_hyper(self.lr, self.beta1, self.beta2, self.step_count)
+
     @torch.no_grad()
     def step(self):
+        if getattr(self, "graph_mode", False):
+            # inside capture (or replayed): hyper buffer carries lr/bc
+            self._mt.step_graph(self.beta1, self.beta2, self.eps)
+            return
         self.step_count += 1
         ps = self._try_mt()
         if ps is not None:
