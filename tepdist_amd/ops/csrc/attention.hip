@@ -49,6 +49,16 @@ DEV_INLINE float fast_exp(float x) {
   return __builtin_amdgcn_exp2f(x * 1.4426950408889634f);
 }
 
+// execution barrier with LDS-visibility only: s_waitcnt lgkmcnt(0) then a
+// raw s_barrier. Unlike __syncthreads there is NO vmcnt(0) drain, so
+// in-flight global prefetch loads and the fire-and-forget dQ atomics keep
+// flowing across tile boundaries instead of serializing at each barrier.
+#define BAR_LDS()                                        \
+  do {                                                   \
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");   \
+    __builtin_amdgcn_s_barrier();                        \
+  } while (0)
+
 DEV_INLINE float warp16_max(float v) {
 #pragma unroll
   for (int d = 1; d < 16; d <<= 1) v = fmaxf(v, __shfl_xor(v, d, 64));
@@ -288,10 +298,10 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();  // all waves done reading sK/sVT
+    BAR_LDS();  // all waves done reading sK/sVT (ds reads retired)
     if (kv0 + KB < kv_end) {
       stage_write();  // overwrite with the pre-loaded next tile
-      __syncthreads();
+      BAR_LDS();
     }
   }
 
@@ -688,7 +698,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       }
     }
     __builtin_amdgcn_s_setprio(0);
-    __syncthreads();  // sdS complete; sQ/sQT/sdO/sdOT reads done
+    BAR_LDS();  // sdS visible; sQ/sQT/sdO/sdOT reads retired
     if (q0 + QT < S) tile_write();  // overlaps the dQ phase (sdS/sKT only)
 
     // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS K (sdS x sKT)
@@ -718,7 +728,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
                       acc_dq[df][e]);
         }
     }
-    __syncthreads();  // next tile's images complete before its reads
+    BAR_LDS();  // next tile's images visible; dQ atomics NOT drained
   }
 
   // write dK, dV (transposed-accumulator scatter: kv = wave's lane slice)

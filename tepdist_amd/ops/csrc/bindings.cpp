@@ -173,12 +173,13 @@ PYBIND11_MODULE(_tepdist_hip, m) {
   });
 
   m.def("cross_entropy_bwd", [](uintptr_t logits, uintptr_t targets,
-                                uintptr_t lse, float dloss_over_n,
+                                uintptr_t lse, uintptr_t dscale,
                                 uintptr_t dlogits, int64_t rows, int cols,
                                 int ignore, uintptr_t stream) {
     cross_entropy_bwd_bf16(reinterpret_cast<void*>(logits),
                            reinterpret_cast<const int64_t*>(targets),
-                           reinterpret_cast<const float*>(lse), dloss_over_n,
+                           reinterpret_cast<const float*>(lse),
+                           reinterpret_cast<const float*>(dscale),
                            reinterpret_cast<void*>(dlogits), rows, cols,
                            ignore, S(stream));
     check_launch();

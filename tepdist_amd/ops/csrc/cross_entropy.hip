@@ -68,9 +68,13 @@ void ce_fwd_kernel(const bf16_t* __restrict__ logits,
 __launch_bounds__(NT) __global__
 void ce_bwd_kernel(const bf16_t* __restrict__ logits,
                    const int64_t* __restrict__ targets,
-                   const float* __restrict__ lse, float dloss_over_n,
+                   const float* __restrict__ lse,
+                   const float* __restrict__ dscale,  // device: dloss / n
                    bf16_t* __restrict__ dlogits, int64_t rows, int cols,
                    int ignore_index) {
+  // read from device memory so the launch is hipGraph-capturable (a host
+  // scalar argument would force a sync to compute it)
+  const float dloss_over_n = *dscale;
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16_t* lr = logits + row * cols;
     bf16_t* dr = dlogits + row * cols;
@@ -121,13 +125,13 @@ void cross_entropy_fwd_bf16(const void* logits, const int64_t* targets,
 }
 
 void cross_entropy_bwd_bf16(const void* logits, const int64_t* targets,
-                            const float* lse, float dloss_over_n,
+                            const float* lse, const float* dscale,
                             void* dlogits, int64_t rows, int cols,
                             int ignore_index, hipStream_t stream) {
   const int blocks = (int)std::min<int64_t>(rows, 2048);
   hipLaunchKernelGGL(ce_bwd_kernel, dim3(blocks), dim3(NT), 0, stream,
                      static_cast<const bf16_t*>(logits), targets, lse,
-                     dloss_over_n, static_cast<bf16_t*>(dlogits), rows, cols,
+                     dscale, static_cast<bf16_t*>(dlogits), rows, cols,
                      ignore_index);
 }
 

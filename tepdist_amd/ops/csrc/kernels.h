@@ -98,7 +98,7 @@ void cross_entropy_fwd_bf16(const void* logits, const int64_t* targets,
                             float* nll, float* lse, int64_t rows, int cols,
                             int ignore_index, hipStream_t stream);
 void cross_entropy_bwd_bf16(const void* logits, const int64_t* targets,
-                            const float* lse, float dloss_over_n,
+                            const float* lse, const float* dscale,
                             void* dlogits, int64_t rows, int cols,
                             int ignore_index, hipStream_t stream);
 

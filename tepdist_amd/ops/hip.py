@@ -439,10 +439,12 @@ def cross_entropy_bwd(dloss: torch.Tensor, logits: torch.Tensor,
     targets = targets.contiguous()
     if targets.dtype != torch.int64:
         targets = targets.long()
-    n = (targets != ignore_index).sum().clamp_min(1).item()
+    # scale stays on device (graph-capturable: no host readback)
+    n = (targets != ignore_index).sum().clamp_min(1)
+    dscale = (dloss.to(logits.device).float() / n).contiguous()
     dlogits = torch.empty_like(logits)
     ext.cross_entropy_bwd(logits.data_ptr(), targets.data_ptr(),
-                          lse.data_ptr(), float(dloss.item()) / n,
+                          lse.data_ptr(), dscale.data_ptr(),
                           dlogits.data_ptr(), rows, cols, ignore_index,
                           _stream())
     return dlogits
