@@ -209,6 +209,54 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
     for (int p = 0; p < 4; ++p)
       bv[p] = bf2f(bias[n0 + 64 * p + wn * 16 + (lane & 15)]);
   }
+  if (!F32OUT) {
+    // Bounce the output through LDS so global stores are whole 128-byte
+    // cachelines: the C-fragment layout is column-strided (per lane, the
+    // four accumulator elements are CONSECUTIVE ROWS of one column), so a
+    // direct scatter is 2-byte stores in 32-byte segments — at K ~ 1k the
+    // C write-out dominates per-block overhead. One 64-column phase at a
+    // time: waves deposit fragments into a swizzled [256][64] image, then
+    // all 512 threads store it row-major (8 lanes = one full row).
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    bf16_t* img = smem;                  // [256][64] post-activation
+    bf16_t* img2 = smem + 256 * BK;      // [256][64] pre-activation
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int cl = wn * 16 + (lane & 15);
+#pragma unroll
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int row = wm * 128 + mi * 16 + (lane >> 4) * 4 + e;
+          float v = acc[mi][p][e];
+          if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) v += bv[p];
+          if (EPI >= 2) {
+            const bf16_t pre = f2bf(v);
+            img2[qoff(row, cl)] = pre;
+            v = gelu_f(bf2f(pre));
+          }
+          img[qoff(row, cl)] = f2bf(v);
+        }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const int sr = threadIdx.x >> 3;         // 64 rows per round
+      const int c8 = (threadIdx.x & 7) * 8;    // 16B chunk within the row
+#pragma unroll
+      for (int rr = 0; rr < BM; rr += 64) {
+        const int row = rr + sr;
+        const int64_t off = (int64_t)(m0 + row) * ldc + n0 + 64 * p + c8;
+        *reinterpret_cast<bf16x8*>(C + off) =
+            *reinterpret_cast<const bf16x8*>(img + qoff(row, c8));
+        if (EPI >= 2)
+          *reinterpret_cast<bf16x8*>(Cpre + off) =
+              *reinterpret_cast<const bf16x8*>(img2 + qoff(row, c8));
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();  // image free for the next phase
+    }
+    return;
+  }
 #pragma unroll
   for (int mi = 0; mi < 8; ++mi) {
 #pragma unroll
@@ -218,18 +266,8 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
       for (int e = 0; e < 4; ++e) {
         const int m = m0 + wm * 128 + mi * 16 + (lane >> 4) * 4 + e;
         float v = acc[mi][p][e];
-        if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) v += bv[p];
         const int64_t off = (int64_t)m * ldc + n;
-        if (F32OUT) {
-          Cf[off] = v;
-          continue;
-        }
-        if (EPI >= 2) {
-          const bf16_t pre = f2bf(v);
-          Cpre[off] = pre;
-          v = gelu_f(bf2f(pre));
-        }
-        C[off] = f2bf(v);
+        Cf[off] = v;
       }
     }
   }
