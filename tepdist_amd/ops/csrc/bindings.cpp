@@ -208,13 +208,21 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
-  m.def("transpose_gelu_bwd", [](uintptr_t dy, uintptr_t pre, uintptr_t dgt,
-                                 uintptr_t dgn, int R, int C,
-                                 uintptr_t stream) {
-    transpose_gelu_bwd_bf16(reinterpret_cast<void*>(dy),
-                            reinterpret_cast<void*>(pre),
-                            reinterpret_cast<void*>(dgt),
-                            reinterpret_cast<void*>(dgn), R, C, S(stream));
+  m.def("cast_ws", [](uintptr_t ws, uintptr_t db, int cols,
+                      uintptr_t stream) {
+    cast_ws_f32_bf16(reinterpret_cast<const float*>(ws),
+                     reinterpret_cast<void*>(db), cols, S(stream));
+    check_launch();
+  });
+  m.def("transpose_dy", [](uintptr_t dy, uintptr_t pre, uintptr_t dgt,
+                           uintptr_t dgn, uintptr_t bias_ws, int R, int C,
+                           uintptr_t stream) {
+    transpose_dy_bf16(reinterpret_cast<void*>(dy),
+                      reinterpret_cast<void*>(pre),
+                      reinterpret_cast<void*>(dgt),
+                      reinterpret_cast<void*>(dgn),
+                      reinterpret_cast<float*>(bias_ws), R, C, S(stream));
+    check_launch();
   });
   m.def("gelu_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t dx, int64_t n,
                        uintptr_t stream) {

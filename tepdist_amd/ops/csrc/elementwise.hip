@@ -162,6 +162,12 @@ void gelu_bwd_bf16(const void* dy, const void* x, void* dx, int64_t n,
                      n);
 }
 
+void cast_ws_f32_bf16(const float* ws, void* db_out, int cols,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(cast_ws_kernel, dim3((cols + NT - 1) / NT), dim3(NT), 0,
+                     stream, ws, static_cast<bf16_t*>(db_out), cols);
+}
+
 void bias_sum_bf16(const void* dy, void* db_out, float* ws_zeroed,
                    int64_t rows, int cols, hipStream_t stream) {
   bf16_t* db = static_cast<bf16_t*>(db_out);

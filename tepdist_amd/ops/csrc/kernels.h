@@ -39,9 +39,11 @@ void splitk_reduce(const float* parts, void* out, int nparts, int64_t mn,
 void transpose_bf16(const void* in, void* out, int R, int C,
                     int64_t stride_in, int64_t stride_out, int batch,
                     hipStream_t stream);
-void transpose_gelu_bwd_bf16(const void* dy, const void* pre, void* dgelu_t,
-                             void* dgelu_nat, int R, int C,
-                             hipStream_t stream);
+void cast_ws_f32_bf16(const float* ws, void* db_out, int cols,
+                      hipStream_t stream);
+void transpose_dy_bf16(const void* dy, const void* pre, void* dy_t,
+                       void* dy_nat, float* bias_ws, int R, int C,
+                       hipStream_t stream);
 
 // --- LayerNorm -------------------------------------------------------------
 void layernorm_fwd_bf16(const void* x, const void* gamma, const void* beta,

@@ -22,18 +22,24 @@ constexpr int TS = 64;  // tile size
 // the wgrad path transposes dy anyway, so the elementwise backward rides
 // along instead of a separate full read+write pass) and NAT also writes
 // the post-gelu natural-layout image (the dgrad GEMM operand).
-template <bool GELUG, bool NAT>
+template <bool GELUG, bool NAT, bool BIAS = false>
 __launch_bounds__(NT) __global__
 void transpose_kernel(const bf16_t* __restrict__ in,
                       const bf16_t* __restrict__ pre,
                       bf16_t* __restrict__ out,
-                      bf16_t* __restrict__ out_nat, int R, int C,
+                      bf16_t* __restrict__ out_nat,
+                      float* __restrict__ bias_ws, int R, int C,
                       int64_t stride_in, int64_t stride_out) {
   __shared__ bf16_t tile[TS * (TS + 8)];  // +16B row pad (b128-aligned)
+  __shared__ float colacc[TS];            // BIAS: per-tile column sums
   const bf16_t* src = in + blockIdx.z * stride_in;
   bf16_t* dst = out + blockIdx.z * stride_out;
   const int r0 = blockIdx.y * TS;
   const int c0 = blockIdx.x * TS;
+  if (BIAS) {
+    if (threadIdx.x < TS) colacc[threadIdx.x] = 0.f;
+    __syncthreads();
+  }
 
   // load [64 rows][64 cols] with 16B vectors: 512 loads / 256 threads
 #pragma unroll
@@ -69,10 +75,19 @@ void transpose_kernel(const bf16_t* __restrict__ in,
           for (int e = 0; e < 8 && c0 + c + e < C; ++e) pn[e] = v[e];
         }
       }
+      if (BIAS) {
+        // column partial sums ride along with the transpose: the
+        // separate full-tensor bias-grad pass disappears
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (c0 + c + e < C) atomicAdd(&colacc[c + e], bf2f(v[e]));
+      }
     }
     *reinterpret_cast<bf16x8*>(tile + r * (TS + 8) + c) = v;
   }
   __syncthreads();
+  if (BIAS && threadIdx.x < TS && c0 + (int)threadIdx.x < C)
+    atomicAdd(&bias_ws[c0 + threadIdx.x], colacc[threadIdx.x]);
 
   // store transposed: thread reads a column 8-run via 8 scalar LDS reads
   // (padded rows -> conflict-light), writes one 16B row of the output
@@ -103,19 +118,31 @@ void transpose_bf16(const void* in, void* out, int R, int C,
   dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, batch);
   hipLaunchKernelGGL((transpose_kernel<false, false>), grid, dim3(NT), 0,
                      stream, static_cast<const bf16_t*>(in), nullptr,
-                     static_cast<bf16_t*>(out), nullptr, R, C, stride_in,
-                     stride_out);
+                     static_cast<bf16_t*>(out), nullptr, nullptr, R, C,
+                     stride_in, stride_out);
 }
 
-void transpose_gelu_bwd_bf16(const void* dy, const void* pre, void* dgelu_t,
-                             void* dgelu_nat, int R, int C,
-                             hipStream_t stream) {
+// dy-transpose variants for linear backward: optional fused gelu'(pre)
+// multiply (+ natural-layout copy) and fused bias column sums.
+void transpose_dy_bf16(const void* dy, const void* pre, void* dy_t,
+                       void* dy_nat, float* bias_ws, int R, int C,
+                       hipStream_t stream) {
   dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, 1);
-  hipLaunchKernelGGL((transpose_kernel<true, true>), grid, dim3(NT), 0,
-                     stream, static_cast<const bf16_t*>(dy),
-                     static_cast<const bf16_t*>(pre),
-                     static_cast<bf16_t*>(dgelu_t),
-                     static_cast<bf16_t*>(dgelu_nat), R, C, 0, 0);
+  const bf16_t* d = static_cast<const bf16_t*>(dy);
+  const bf16_t* p = static_cast<const bf16_t*>(pre);
+  bf16_t* t = static_cast<bf16_t*>(dy_t);
+  bf16_t* n = static_cast<bf16_t*>(dy_nat);
+#define TK(G, NA, B)                                                     \
+  hipLaunchKernelGGL((transpose_kernel<G, NA, B>), grid, dim3(NT), 0,    \
+                     stream, d, p, t, n, bias_ws, R, C, 0, 0)
+  if (pre != nullptr) {
+    if (bias_ws != nullptr) TK(true, true, true);
+    else TK(true, true, false);
+  } else {
+    if (bias_ws != nullptr) TK(false, false, true);
+    else TK(false, false, false);
+  }
+#undef TK
 }
 
 }  // namespace tepdist

@@ -115,18 +115,20 @@ def linear_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     # canonicalize to KC x KC: one tuned GEMM schedule serves every case,
     # with cheap materialized transposes (memory-bound, ~2% of GEMM time)
     wT = transpose2d(w)        # [K, N]
+    # fused gelu-backward: the wgrad transpose of dy applies gelu'(pre)
+    # in the same pass and emits BOTH layouts (saves the separate
+    # elementwise kernel's full read+write). Bias sums stay a separate
+    # streaming pass — folding them into the transpose as LDS atomics
+    # measured ~3x slower transposes (per-element atomic serialization).
+    dyT = torch.empty(N, M, dtype=BF16, device=dy.device)
     if act == "gelu":
-        # fused gelu-backward: the wgrad transpose of dy applies gelu'(pre)
-        # in the same pass and emits BOTH layouts (saves the separate
-        # elementwise kernel's full read+write)
         dy_eff = torch.empty_like(dy)
-        dyT = torch.empty(N, M, dtype=BF16, device=dy.device)
-        ext.transpose_gelu_bwd(dy.data_ptr(), pre_act.data_ptr(),
-                               dyT.data_ptr(), dy_eff.data_ptr(), M, N,
-                               _stream())
+        ext.transpose_dy(dy.data_ptr(), pre_act.data_ptr(), dyT.data_ptr(),
+                         dy_eff.data_ptr(), 0, M, N, _stream())
         dy = dy_eff
     else:
-        dyT = transpose2d(dy)  # [N, M]
+        ext.transpose_dy(dy.data_ptr(), 0, dyT.data_ptr(), 0, 0, M, N,
+                         _stream())
     xT = transpose2d(x)        # [K, M]
     # dx[M,K] = dy[M,N] @ w[N,K]: A=dy KC (k=N), B=w^T stored [K,N] KC
     dx, _ = _gemm_raw(dy, wT, True, True, M, K, N, N, N, 0, 0, 1)
