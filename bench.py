@@ -57,7 +57,7 @@ def main():
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--model", type=str, default="gpt2-345m")
-    ap.add_argument("--micro-batch", type=int, default=32,
+    ap.add_argument("--micro-batch", type=int, default=64,
                     help="per-GPU batch rows per step (weak scaling)")
     ap.add_argument("--seq", type=int, default=1024)
     ap.add_argument("--parallel", type=str, default="auto",

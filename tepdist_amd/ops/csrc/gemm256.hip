@@ -64,6 +64,13 @@ DEV_INLINE void stage_q(const bf16_t* __restrict__ src, int64_t ld, int f0,
   __builtin_amdgcn_global_load_lds((gsrc_t)g, (gdst_t)l, 16, 0, 0);
 }
 
+// f32 image variant of the swizzle (16B slot = 4 floats)
+DEV_INLINE int qoff_f32(int row, int col_e) {
+  const int r1 = row >> 1;
+  const int x = ((r1 & 1) << 2) | (r1 & 2) | ((r1 >> 2) & 1);
+  return row * BK + (col_e ^ ((x << 2) & 63));
+}
+
 DEV_INLINE bf16x8 fragq(const bf16_t* slot, int row, int col_e) {
   return *reinterpret_cast<const bf16x8*>(slot + qoff(row, col_e));
 }
@@ -257,18 +264,34 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
     }
     return;
   }
-#pragma unroll
-  for (int mi = 0; mi < 8; ++mi) {
+  // F32OUT (split-K partials): same LDS bounce with a [256][64] f32 image
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  {
+    float* imgf = reinterpret_cast<float*>(smem);
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
-      const int n = n0 + 64 * p + wn * 16 + (lane & 15);
+      const int cl = wn * 16 + (lane & 15);
 #pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        const int m = m0 + wm * 128 + mi * 16 + (lane >> 4) * 4 + e;
-        float v = acc[mi][p][e];
-        const int64_t off = (int64_t)m * ldc + n;
-        Cf[off] = v;
+      for (int mi = 0; mi < 8; ++mi)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+          const int row = wm * 128 + mi * 16 + (lane >> 4) * 4 + e;
+          imgf[qoff_f32(row, cl)] = acc[mi][p][e];
+        }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const int sr = threadIdx.x >> 4;         // 32 rows per round
+      const int c4 = (threadIdx.x & 15) * 4;   // 16B chunk (4 floats)
+#pragma unroll
+      for (int rr = 0; rr < BM; rr += 32) {
+        const int row = rr + sr;
+        const int64_t off = (int64_t)(m0 + row) * ldc + n0 + 64 * p + c4;
+        *reinterpret_cast<f32x4*>(Cf + off) =
+            *reinterpret_cast<const f32x4*>(imgf + qoff_f32(row, c4));
       }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
     }
   }
 }

@@ -65,7 +65,14 @@ def _gemm_raw(a: torch.Tensor, b_stored: torch.Tensor, a_kc: bool, b_kc: bool,
     else:
         blocks = ((M + 127) // 128) * ((N + 127) // 128)
     if batch == 1 and epi == 0 and K >= 2048 and blocks < 384:
-        split_k = min(8, max(2, (512 + blocks - 1) // blocks))
+        split_k = max(2, (512 + blocks - 1) // blocks)
+        # prefer a whole number of 256-block generations so every CU stays
+        # busy to the end (e.g. 48 blocks: x8 = 1.5 waves -> x16 = 3 full)
+        for cand in range(split_k, min(17, K // 128 + 1)):
+            if (blocks * cand) % 256 == 0:
+                split_k = cand
+                break
+        split_k = max(2, min(split_k, 16, K // 128))
         parts = torch.empty(split_k, M * N, dtype=torch.float32, device=dev)
         ext.gemm(a.data_ptr(), b_stored.data_ptr(), parts.data_ptr(), 0, 0,
                  M, N, K, lda, ldb, N, 0, 0, M * N, 1, a_kc, b_kc, 0,
