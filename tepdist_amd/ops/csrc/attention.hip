@@ -111,9 +111,10 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   const bf16_t* vp = V + (int64_t)b * q_bs + (int64_t)h * q_hs;
   bf16_t* op = O + (int64_t)b * o_bs + (int64_t)h * o_hs;
 
-  __shared__ bf16_t smem[KB * D + D * KB];
-  bf16_t* sK = smem;                    // [KB][D] d-contiguous
-  bf16_t* sVT = smem + KB * D;          // [D][KB] kv-contiguous
+  // double-buffered tiles: compute reads parity (t&1) while the staging
+  // writes parity ^1 — one barrier per tile instead of two
+  __shared__ bf16_t smem[2 * (KB * D + D * KB)];
+  const int BUFSZ = KB * D + D * KB;
 
   // Q fragments (PV B-layout twin): nf-th 16-q group, kk-th 32-d chunk
   bf16x8 qf[2][DK];
@@ -165,7 +166,9 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       }
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int buf) {
+    bf16_t* sK = smem + buf * BUFSZ;
+    bf16_t* sVT = sK + KB * D;
 #pragma unroll
     for (int u = 0; u < KUN; ++u) {
       const int idx = threadIdx.x + u * NT;
@@ -193,9 +196,12 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
 
   const int kv_end = causal ? min(S, q0 + QB) : S;
   stage_load(0);
-  stage_write();
+  stage_write(0);
   __syncthreads();
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+  int t = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB, ++t) {
+    const bf16_t* sK = smem + (t & 1) * BUFSZ;
+    const bf16_t* sVT = sK + KB * D;
     if (kv0 + KB < kv_end) stage_load(kv0 + KB);  // overlap with compute
 
     if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
@@ -298,11 +304,8 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    BAR_LDS();  // all waves done reading sK/sVT (ds reads retired)
-    if (kv0 + KB < kv_end) {
-      stage_write();  // overwrite with the pre-loaded next tile
-      BAR_LDS();
-    }
+    if (kv0 + KB < kv_end) stage_write((t + 1) & 1);  // other buffer: no
+    BAR_LDS();  // wait for readers of this buffer AND the writes above
   }
 
   // --- epilogue: O = O^T^T / l, write O and lse ---
@@ -621,18 +624,25 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
     // P and dS (elementwise); kv fixed per lane, q varies per element.
     // Inner tiles (wave's kv slice fully below the diagonal, rows in
     // range) skip the per-element mask selects — wave-uniform branch.
+    // lse/delta come in as four 16-byte vector reads each (lane g's four
+    // q rows 16*mi+4g.. are consecutive floats), not 16 scalar reads.
     const int kvg = kv0 + wave * 16 + (lane & 15);
     const bool winner = (!causal || kv0 + wave * 16 + 15 < q0) &&
                         q0 + QT <= S && kv0 + KB <= S;
     if (winner) {
+      f32x4 lsev[4], dlv[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        lsev[mi] = *reinterpret_cast<const f32x4*>(sLSE + 16 * mi + 4 * g);
+        dlv[mi] = *reinterpret_cast<const f32x4*>(sDELTA + 16 * mi + 4 * g);
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
-          const int qrow = 16 * mi + 4 * g + e;
-          const float pt = fast_exp(st[mi][e] * scale - sLSE[qrow]);
+          const float pt = fast_exp(st[mi][e] * scale - lsev[mi][e]);
           st[mi][e] = pt;
-          dpt[mi][e] = pt * (dpt[mi][e] - sDELTA[qrow]) * scale;
+          dpt[mi][e] = pt * (dpt[mi][e] - dlv[mi][e]) * scale;
         }
     } else {
 #pragma unroll
