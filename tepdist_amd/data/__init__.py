@@ -1,0 +1,2 @@
+from tepdist_amd.data.synthetic import (SyntheticImages, SyntheticTokens,
+                                        device_prefetcher)
