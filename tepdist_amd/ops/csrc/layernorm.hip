@@ -107,32 +107,48 @@ void ln_bwd_kernel(const bf16_t* __restrict__ dy, const bf16_t* __restrict__ x,
   float dg_acc[NV * 8] = {};
   float db_acc[NV * 8] = {};
 
+  // gamma is row-invariant: load once per wave, not per row
+  bf16x8 gvv[NV];
+#pragma unroll
+  for (int v = 0; v < NV; ++v) {
+    const int c0 = v * WAVE * 8 + lane * 8;
+    bf16x8 gv = {};
+    if (c0 + 8 <= cols) {
+      gv = *reinterpret_cast<const bf16x8*>(g + c0);
+    } else {
+      for (int e = 0; e < 8 && c0 + e < cols; ++e) gv[e] = g[c0 + e];
+    }
+    gvv[v] = gv;
+  }
+
   for (int row = gwave; row < rows; row += nwaves) {
     const bf16_t* dyr = dy + (int64_t)row * cols;
     const bf16_t* xr = x + (int64_t)row * cols;
     const float mu = mean[row], rs = rstd[row];
     float c1 = 0.f, c2 = 0.f;
+    // dy/x stay in registers across both passes (no second global read)
+    bf16x8 dyvv[NV], xvv[NV];
 #pragma unroll
     for (int v = 0; v < NV; ++v) {
       const int c0 = v * WAVE * 8 + lane * 8;
-      bf16x8 dyv = {}, xv = {}, gv = {};
+      bf16x8 dyv = {}, xv = {};
       if (c0 + 8 <= cols) {
         dyv = *reinterpret_cast<const bf16x8*>(dyr + c0);
         xv = *reinterpret_cast<const bf16x8*>(xr + c0);
-        gv = *reinterpret_cast<const bf16x8*>(g + c0);
       } else {
         for (int e = 0; e < 8 && c0 + e < cols; ++e) {
           dyv[e] = dyr[c0 + e];
           xv[e] = xr[c0 + e];
-          gv[e] = g[c0 + e];
         }
       }
+      dyvv[v] = dyv;
+      xvv[v] = xv;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         const int c = c0 + e;
         const float d = bf2f(dyv[e]);
         const float xhat = (c < cols) ? (bf2f(xv[e]) - mu) * rs : 0.f;
-        const float gg = (c < cols) ? d * bf2f(gv[e]) : 0.f;
+        const float gg = (c < cols) ? d * bf2f(gvv[v][e]) : 0.f;
         dg_acc[v * 8 + e] += d * xhat;
         db_acc[v * 8 + e] += d;
         c1 += gg;
@@ -146,25 +162,17 @@ void ln_bwd_kernel(const bf16_t* __restrict__ dy, const bf16_t* __restrict__ x,
     for (int v = 0; v < NV; ++v) {
       const int c0 = v * WAVE * 8 + lane * 8;
       if (c0 >= cols) continue;
-      bf16x8 dyv = {}, xv = {}, gv = {};
-      if (c0 + 8 <= cols) {
-        dyv = *reinterpret_cast<const bf16x8*>(dyr + c0);
-        xv = *reinterpret_cast<const bf16x8*>(xr + c0);
-        gv = *reinterpret_cast<const bf16x8*>(g + c0);
-        bf16x8 dxv;
+      bf16x8 dxv;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const float xhat = (bf2f(xv[e]) - mu) * rs;
-          const float gg = bf2f(dyv[e]) * bf2f(gv[e]);
-          dxv[e] = f2bf((gg - c1 - xhat * c2) * rs);
-        }
+      for (int e = 0; e < 8; ++e) {
+        const float xhat = (bf2f(xvv[v][e]) - mu) * rs;
+        const float gg = bf2f(dyvv[v][e]) * bf2f(gvv[v][e]);
+        dxv[e] = f2bf((gg - c1 - xhat * c2) * rs);
+      }
+      if (c0 + 8 <= cols) {
         *reinterpret_cast<bf16x8*>(dxr + c0) = dxv;
       } else {
-        for (int e = 0; e < 8 && c0 + e < cols; ++e) {
-          const float xhat = (bf2f(xr[c0 + e]) - mu) * rs;
-          const float gg = bf2f(dyr[c0 + e]) * bf2f(g[c0 + e]);
-          dxr[c0 + e] = f2bf((gg - c1 - xhat * c2) * rs);
-        }
+        for (int e = 0; e < 8 && c0 + e < cols; ++e) dxr[c0 + e] = dxv[e];
       }
     }
   }
