@@ -136,10 +136,13 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   l_r[0] = l_r[1] = 0.f;
 
   // staging registers (issue-early / write-late split):
-  // K: KUN x 16B per thread; V: 8 x 4B per thread (v_perm transpose slabs)
+  // K: KUN x 16B per thread; V: 8 x 4B per slab (v_perm transpose slabs;
+  // (KB/8)*(D/2) slabs total — more than one per thread when D = 128)
   constexpr int KUN = KB * D / 8 / NT;
+  constexpr int NSLAB = (KB / 8) * (D / 2);
+  constexpr int SUN = (NSLAB + NT - 1) / NT;
   bf16x8 krg[KUN];
-  uint32_t vrg[8];
+  uint32_t vrg[SUN][8];
 
   auto stage_load = [&](int kv0) {
 #pragma unroll
@@ -153,16 +156,19 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
             kp + (int64_t)(kv0 + row) * q_rs + c);
       krg[u] = v8;
     }
-    const int f = 2 * (threadIdx.x % (D / 2));
-    const int kb = threadIdx.x / (D / 2);
-    if (kb < KB / 8) {
+#pragma unroll
+    for (int u = 0; u < SUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= NSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int kv = kv0 + kb * 8 + j;
         bf16x2 v2 = {};
         if (kv < S)
           v2 = *reinterpret_cast<const bf16x2*>(vp + (int64_t)kv * q_rs + f);
-        vrg[j] = __builtin_bit_cast(uint32_t, v2);
+        vrg[u][j] = __builtin_bit_cast(uint32_t, v2);
       }
     }
   };
@@ -176,15 +182,18 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       const int c = (idx % (D / 8)) * 8;
       *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = krg[u];
     }
-    const int f = 2 * (threadIdx.x % (D / 2));
-    const int kb = threadIdx.x / (D / 2);
-    if (kb < KB / 8) {
+#pragma unroll
+    for (int u = 0; u < SUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= NSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
       uint32_t o0[4], o1[4];
 #pragma unroll
       for (int d2 = 0; d2 < 4; ++d2) {
-        o0[d2] = __builtin_amdgcn_perm(vrg[2 * d2 + 1], vrg[2 * d2],
+        o0[d2] = __builtin_amdgcn_perm(vrg[u][2 * d2 + 1], vrg[u][2 * d2],
                                        0x05040100u);
-        o1[d2] = __builtin_amdgcn_perm(vrg[2 * d2 + 1], vrg[2 * d2],
+        o1[d2] = __builtin_amdgcn_perm(vrg[u][2 * d2 + 1], vrg[u][2 * d2],
                                        0x07060302u);
       }
       *reinterpret_cast<uint4*>(sVT + loff<KB>(f, kb * 8)) =
@@ -512,8 +521,10 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
 
   // issue-early / write-late staging registers for the Q and dO tiles
   constexpr int QUN = QT * D / 8 / NT;
+  constexpr int TSLAB = (QT / 8) * (D / 2);   // v_perm transpose slabs
+  constexpr int TUN = (TSLAB + NT - 1) / NT;
   bf16x8 qn[QUN], don[QUN];
-  uint32_t qt[8], dot_[8];
+  uint32_t qt[TUN][8], dot_[TUN][8];
   float lse2[QT / NT + 1], dl2[QT / NT + 1];
 
   auto tile_load = [&](int q0) {
@@ -532,9 +543,12 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       qn[u] = a;
       don[u] = b2;
     }
-    const int f = 2 * (threadIdx.x % (D / 2));
-    const int kb = threadIdx.x / (D / 2);
-    if (kb < QT / 8) {
+#pragma unroll
+    for (int u = 0; u < TUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= TSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int row = q0 + kb * 8 + j;
@@ -545,8 +559,8 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
           b2 = *reinterpret_cast<const bf16x2*>(
               dop + (int64_t)row * o_rs + f);
         }
-        qt[j] = __builtin_bit_cast(uint32_t, a);
-        dot_[j] = __builtin_bit_cast(uint32_t, b2);
+        qt[u][j] = __builtin_bit_cast(uint32_t, a);
+        dot_[u][j] = __builtin_bit_cast(uint32_t, b2);
       }
     }
     for (int i = threadIdx.x, s2 = 0; i < QT; i += NT, ++s2) {
@@ -564,15 +578,18 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       *reinterpret_cast<bf16x8*>(sQ + loff<D>(row, c)) = qn[u];
       *reinterpret_cast<bf16x8*>(sdO + loff<D>(row, c)) = don[u];
     }
-    const int f = 2 * (threadIdx.x % (D / 2));
-    const int kb = threadIdx.x / (D / 2);
-    if (kb < QT / 8) {
+#pragma unroll
+    for (int u = 0; u < TUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= TSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
       uint32_t o0[4], o1[4];
 #pragma unroll
       for (int d2 = 0; d2 < 4; ++d2) {
-        o0[d2] = __builtin_amdgcn_perm(qt[2 * d2 + 1], qt[2 * d2],
+        o0[d2] = __builtin_amdgcn_perm(qt[u][2 * d2 + 1], qt[u][2 * d2],
                                        0x05040100u);
-        o1[d2] = __builtin_amdgcn_perm(qt[2 * d2 + 1], qt[2 * d2],
+        o1[d2] = __builtin_amdgcn_perm(qt[u][2 * d2 + 1], qt[u][2 * d2],
                                        0x07060302u);
       }
       *reinterpret_cast<uint4*>(sQT + loff<QT>(f, kb * 8)) =
@@ -581,9 +598,9 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
           make_uint4(o1[0], o1[1], o1[2], o1[3]);
 #pragma unroll
       for (int d2 = 0; d2 < 4; ++d2) {
-        o0[d2] = __builtin_amdgcn_perm(dot_[2 * d2 + 1], dot_[2 * d2],
+        o0[d2] = __builtin_amdgcn_perm(dot_[u][2 * d2 + 1], dot_[u][2 * d2],
                                        0x05040100u);
-        o1[d2] = __builtin_amdgcn_perm(dot_[2 * d2 + 1], dot_[2 * d2],
+        o1[d2] = __builtin_amdgcn_perm(dot_[u][2 * d2 + 1], dot_[u][2 * d2],
                                        0x07060302u);
       }
       *reinterpret_cast<uint4*>(sdOT + loff<QT>(f, kb * 8)) =

@@ -292,9 +292,11 @@ def test_tr16_probe_mapping():
     assert torch.equal(uni, expect_uni), f"tr16 uniform mapping:\n{uni}"
 
 
-def test_flash_attention_large_vs_sdpa():
-    """Flash fwd+bwd at benchmark shape (S=1024, D=64) vs fp32 SDPA."""
-    B, H, S, D = 4, 8, 1024, 64
+@pytest.mark.parametrize("D", [64, 128])
+def test_flash_attention_large_vs_sdpa(D):
+    """Flash fwd+bwd at benchmark shape (S=1024) vs fp32 SDPA; D=128
+    covers the wide-head template (llama-class models)."""
+    B, H, S = 4, 8, 1024
     q = _randn(B, H, S, D, seed=40)
     k = _randn(B, H, S, D, seed=41)
     v = _randn(B, H, S, D, seed=42)
