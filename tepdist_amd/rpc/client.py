@@ -91,9 +91,17 @@ class TepdistSession:
 
     def __init__(self, client: TepdistClient = None):
         import concurrent.futures as _cf
+        import json as _json
         self.client = client or TepdistClient()
         self.handle: Optional[int] = None
         self.plan_info: Optional[dict] = None
+        # VARIABLE_MAP_FILE_PATH (reference tf2xla/xla_compiler.cc:957):
+        # preload a variable-name -> global-arg-index map to the server
+        vmap = os.environ.get("VARIABLE_MAP_FILE_PATH")
+        if vmap and os.path.exists(vmap):
+            with open(vmap) as f:
+                self.client._call("TransferVarArgMap",
+                                  {"map": _json.load(f)})
         self._n_parallel = int(os.environ.get("NUM_PARALLEL_RPC_STEPS", "1"))
         self._fetch_every = int(os.environ.get("FETCH_RESOURCE_VAR_STEPS",
                                                "0"))
