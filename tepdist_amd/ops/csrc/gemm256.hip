@@ -93,6 +93,7 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
     kbeg = blockIdx.z * k_chunk;
     kend = min(K, kbeg + k_chunk);
     Cf += blockIdx.z * strideC;
+    if (kend <= kbeg) kbeg = kend = 0;  // dead chunk: no loads, zero out
   } else {
     A += blockIdx.z * strideA;
     B += blockIdx.z * strideB;
@@ -137,20 +138,30 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
 
   // prologue: tile 0 fully + tile 1's A and B q0,q1 (14 units in flight),
   // then complete tile 0 (vmcnt 6) before its reads.
-  stage_a(0, 0);
-  stage_a(0, 2);
-  stage_b(0, 0);
-  stage_b(0, 2);
-  if (nk > 1) {
-    stage_a(1, 0);
-    stage_a(1, 2);
-    stage_b(1, 0);
+  if (nk > 0) {
+    stage_a(0, 0);
+    stage_a(0, 2);
+    stage_b(0, 0);
+    stage_b(0, 2);
+    if (nk > 1) {
+      stage_a(1, 0);
+      stage_a(1, 2);
+      stage_b(1, 0);
+    }
+    VMCNT6();
   }
-  VMCNT6();
   RAW_BAR();
 
   bf16x8 af[8][2];
   for (int t = 0; t < nk; ++t) {
+    if (t + 1 == nk && nk > 1) {
+      // last tile: its B q2/q3 units (staged at (t-1).ph0) are the newest
+      // in flight and the counted vmcnt(6) protocol never covers them —
+      // nothing is staged after them to push them past the watermark.
+      // Drain fully once per block before the final tile's reads.
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      RAW_BAR();
+    }
     // ---- phase 0: read ALL A fragments + B fragment 0; stage B(t+1,q23)
     {
       const bf16_t* as0 = aslot + ((4 * t + 2 * wm) & 7) * SLOT_E;
@@ -305,7 +316,11 @@ bool gemm256_supported(int M, int N, int K, int lda, int ldb, bool a_kc,
     return false;
   if (split_k <= 1) return K >= 2 * BK;
   const int chunk = (K / split_k + BK - 1) / BK * BK;
-  return epi == 0 && chunk >= 2 * BK && K % chunk != BK;  // every chunk >= 2 tiles
+  // every chunk must be >= 2 K-tiles (the pipeline's minimum) and the
+  // chunk count must be exactly split_k (rounding can otherwise leave
+  // trailing chunks with no K range, whose prologue would read past K)
+  return epi == 0 && chunk >= 2 * BK && K % chunk != BK &&
+         (K + chunk - 1) / chunk == split_k;
 }
 
 void gemm256_bf16(const void* A, const void* B, void* C, void* c_pre,

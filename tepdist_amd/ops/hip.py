@@ -73,6 +73,11 @@ def _gemm_raw(a: torch.Tensor, b_stored: torch.Tensor, a_kc: bool, b_kc: bool,
                 split_k = cand
                 break
         split_k = max(2, min(split_k, 16, K // 128))
+        # normalize to the EFFECTIVE chunk count: the kernel rounds the
+        # chunk up to a 64-multiple, which can leave trailing z-blocks
+        # with no K range at all (their prologue would read past K)
+        chunk = (K // split_k + 63) // 64 * 64
+        split_k = max(2, -(-K // chunk))
         parts = torch.empty(split_k, M * N, dtype=torch.float32, device=dev)
         ext.gemm(a.data_ptr(), b_stored.data_ptr(), parts.data_ptr(), 0, 0,
                  M, N, K, lda, ldb, N, 0, 0, M * N, 1, a_kc, b_kc, 0,

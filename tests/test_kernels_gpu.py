@@ -60,7 +60,8 @@ def test_gemm_transposes(ta, tb):
 
 
 @pytest.mark.parametrize("M,N,K", [(512, 256, 8192), (256, 256, 4096),
-                                   (768, 512, 16384)])
+                                   (768, 512, 16384), (3072, 768, 2048),
+                                   (768, 768, 2048), (2304, 768, 2048)])
 def test_gemm_splitk_256(M, N, K):
     # canonical KCxKC shapes deep enough to trigger the split-K path of the
     # 256-tile kernel (fp32 partials + reduce)
