@@ -434,31 +434,30 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
   }
 }
 
-// one block = 64 kv rows of one (b,h); wave w owns the 16-kv slice
-// kv = kv0 + 16w + (lane&15) and iterates q tiles of 64.
+// Backward is TWO kernels, each free of global atomics (a single-kernel
+// flash2 backward accumulates dQ with atomicAdd: 16 read-modify-write
+// passes over an fp32 image dominated the measured time — see
+// profiles/).
 //
-// Layout: scores are computed UNtransposed, S = mfma(Q, K), so the
-// C-fragment holds kv in the lane index (kv = lane&15, the wave's slice)
-// and q across groups/elements (q = 16*mi + 4g + e). Then dV^T and dK^T
+// Kernel A (dK/dV): one block = 64 kv rows of one (b,h); wave w owns the
+// 16-kv slice kv = kv0 + 16w + (lane&15) and iterates q tiles of 64.
+// Scores are computed UNtransposed, S = mfma(Q, K), so dV^T and dK^T
 // consume P and dS directly from the accumulators as B-fragments via the
-// MFMA k-permutation invariance (see flash_fwd_kernel) against
-// transpose-staged dO^T / Q^T A-operands — no P/dS LDS roundtrip for
-// those two products. Only dQ (contraction over kv, which lives in the
-// lane index) still routes dS through a shared [q][kv] LDS image; its
-// MFMAs read dS naturally and K^T from the block-constant sKT. dK/dV
-// accumulate transposed in registers across all q tiles and scatter once.
+// MFMA k-permutation invariance against transpose-staged dO^T / Q^T
+// A-operands — P and dS never touch LDS. dK/dV accumulate transposed in
+// registers across all q tiles and scatter once.
 template <int D>
 __launch_bounds__(NT) __global__
-void flash_bwd_kernel(const bf16_t* __restrict__ Q,
-                      const bf16_t* __restrict__ K,
-                      const bf16_t* __restrict__ V,
-                      const bf16_t* __restrict__ dO,
-                      const float* __restrict__ LSE,
-                      const float* __restrict__ DELTA,
-                      float* __restrict__ dQws, bf16_t* __restrict__ dK,
-                      bf16_t* __restrict__ dV, int S, int H, float scale,
-                      bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
-                      int64_t o_bs, int64_t o_hs, int64_t o_rs) {
+void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
+                         const bf16_t* __restrict__ K,
+                         const bf16_t* __restrict__ V,
+                         const bf16_t* __restrict__ dO,
+                         const float* __restrict__ LSE,
+                         const float* __restrict__ DELTA,
+                         bf16_t* __restrict__ dK, bf16_t* __restrict__ dV,
+                         int S, int H, float scale, bool causal,
+                         int64_t q_bs, int64_t q_hs, int64_t q_rs,
+                         int64_t o_bs, int64_t o_hs, int64_t o_rs) {
   constexpr int DK = D / 32;
   constexpr int DF = D / 16;
   constexpr int QT = 64;
@@ -476,20 +475,16 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
   const bf16_t* vp = V + qoff;
   const bf16_t* dop = dO + ooff;
 
-  __shared__ bf16_t smem[KB * D * 3 + D * KB + QT * D * 2 + D * QT +
-                         QT * KB];
+  __shared__ bf16_t smem[KB * D * 2 + QT * D * 2 + D * QT * 2];
   bf16_t* sKb = smem;                       // [KB][D] natural
-  bf16_t* sKT = sKb + KB * D;               // [D][KB]
-  bf16_t* sVb = sKT + D * KB;               // [KB][D] natural
+  bf16_t* sVb = sKb + KB * D;               // [KB][D] natural
   bf16_t* sQ = sVb + KB * D;                // [QT][D] natural
   bf16_t* sQT = sQ + QT * D;                // [D][QT]
   bf16_t* sdO = sQT + D * QT;               // [QT][D] natural
   bf16_t* sdOT = sdO + QT * D;              // [D][QT]
-  bf16_t* sdS = sdOT + D * QT;              // [QT][KB] shared
   __shared__ float sLSE[QT], sDELTA[QT];
 
-  // stage K, V tiles (fixed for the block)
-  stage_nat_t<KB, D>(kp, kv0, S, q_rs, sKb, sKT);
+  // stage K, V tiles (fixed for the block; natural layout only)
   {
     constexpr int UN = KB * D / 8 / NT;
 #pragma unroll
@@ -497,11 +492,15 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       const int idx = threadIdx.x + u * NT;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
-      bf16x8 v8 = {};
-      if (kv0 + row < S)
-        v8 = *reinterpret_cast<const bf16x8*>(
+      bf16x8 kv8 = {}, vv8 = {};
+      if (kv0 + row < S) {
+        kv8 = *reinterpret_cast<const bf16x8*>(
+            kp + (int64_t)(kv0 + row) * q_rs + c);
+        vv8 = *reinterpret_cast<const bf16x8*>(
             vp + (int64_t)(kv0 + row) * q_rs + c);
-      *reinterpret_cast<bf16x8*>(sVb + loff<D>(row, c)) = v8;
+      }
+      *reinterpret_cast<bf16x8*>(sKb + loff<D>(row, c)) = kv8;
+      *reinterpret_cast<bf16x8*>(sVb + loff<D>(row, c)) = vv8;
     }
   }
   __syncthreads();
@@ -639,10 +638,7 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       }
 
     // P and dS (elementwise); kv fixed per lane, q varies per element.
-    // Inner tiles (wave's kv slice fully below the diagonal, rows in
-    // range) skip the per-element mask selects — wave-uniform branch.
-    // lse/delta come in as four 16-byte vector reads each (lane g's four
-    // q rows 16*mi+4g.. are consecutive floats), not 16 scalar reads.
+    // Inner tiles skip the per-element mask selects (wave-uniform).
     const int kvg = kv0 + wave * 16 + (lane & 15);
     const bool winner = (!causal || kv0 + wave * 16 + 15 < q0) &&
                         q0 + QT <= S && kv0 + KB <= S;
@@ -678,15 +674,6 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
           dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale; // now dS
         }
     }
-
-    // scatter dS to the shared [q][kv] image early — the stores retire
-    // under the MFMA section below, and the barrier after it publishes them
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-      for (int e = 0; e < 4; ++e)
-        sdS[loff<KB>(16 * mi + 4 * g + e, wave * 16 + (lane & 15))] =
-            f2bf(dpt[mi][e]);
 
     // dV^T += dO^T P and dK^T += Q^T dS (both k-permuted B from regs)
     __builtin_amdgcn_s_setprio(1);
@@ -725,37 +712,11 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
       }
     }
     __builtin_amdgcn_s_setprio(0);
-    BAR_LDS();  // sdS visible; sQ/sQT/sdO/sdOT reads retired
-    if (q0 + QT < S) tile_write();  // overlaps the dQ phase (sdS/sKT only)
-
-    // dQ partial: wave w owns q rows [q0+16w, +16): dQ = dS K (sdS x sKT)
-    {
-      f32x4 acc_dq[DF] = {};
-#pragma unroll
-      for (int kk2 = 0; kk2 < KB / 32; ++kk2) {
-        const bf16x8 dsa = *reinterpret_cast<const bf16x8*>(
-            sdS + loff<KB>(wave * 16 + (lane & 15), 8 * g + 32 * kk2));
-#pragma unroll
-        for (int df = 0; df < DF; ++df) {
-          const bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-              sKT + loff<KB>((lane & 15) + 16 * df, 8 * g + 32 * kk2));
-          acc_dq[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, kb,
-                                                               acc_dq[df],
-                                                               0, 0, 0);
-        }
-      }
-#pragma unroll
-      for (int df = 0; df < DF; ++df)
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const int qg = q0 + wave * 16 + 4 * g + e;
-          if (qg < S)
-            atomicAdd(&dQws[(int64_t)bh * S * D + (int64_t)qg * D +
-                            df * 16 + (lane & 15)],
-                      acc_dq[df][e]);
-        }
+    BAR_LDS();  // sQ/sQT/sdO/sdOT reads retired
+    if (q0 + QT < S) {
+      tile_write();
+      BAR_LDS();  // next tile's images visible
     }
-    BAR_LDS();  // next tile's images visible; dQ atomics NOT drained
   }
 
   // write dK, dV (transposed-accumulator scatter: kv = wave's lane slice)
@@ -774,6 +735,254 @@ void flash_bwd_kernel(const bf16_t* __restrict__ Q,
         *reinterpret_cast<bf16x4*>(dK + off) = kv4;
         *reinterpret_cast<bf16x4*>(dV + off) = vv4;
       }
+    }
+  }
+}
+
+// Kernel B (dQ): forward-shaped — one block = 128 q rows, wave w owns 32 q
+// columns, iterates kv tiles; dQ accumulates TRANSPOSED in registers
+// ([d][q], q = lane&15 like the fwd O^T accumulator) with dS^T consumed
+// straight from the accumulators (k-permutation) against a
+// transpose-staged K^T — no LDS scatter, no atomics, bf16 output written
+// directly (strided; serves the packed-qkv layout too).
+template <int D>
+__launch_bounds__(NT) __global__
+void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
+                         const bf16_t* __restrict__ K,
+                         const bf16_t* __restrict__ V,
+                         const bf16_t* __restrict__ dO,
+                         const float* __restrict__ LSE,
+                         const float* __restrict__ DELTA,
+                         bf16_t* __restrict__ dQ, int S, int H, float scale,
+                         bool causal, int64_t q_bs, int64_t q_hs,
+                         int64_t q_rs, int64_t o_bs, int64_t o_hs,
+                         int64_t o_rs) {
+  constexpr int DK = D / 32;
+  constexpr int DF = D / 16;
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int q0 = blockIdx.x * QB;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int g = lane >> 4;
+  const int wq0 = q0 + wave * 32;
+
+  const int64_t qoff = (int64_t)b * q_bs + (int64_t)h * q_hs;
+  const int64_t ooff = (int64_t)b * o_bs + (int64_t)h * o_hs;
+  const bf16_t* qp = Q + qoff;
+  const bf16_t* kp = K + qoff;
+  const bf16_t* vp = V + qoff;
+  const bf16_t* dop = dO + ooff;
+  bf16_t* dqp = dQ + qoff;
+
+  __shared__ bf16_t smem[KB * D * 2 + D * KB];
+  bf16_t* sK = smem;                    // [KB][D] natural
+  bf16_t* sV = smem + KB * D;           // [KB][D] natural
+  bf16_t* sKT = sV + KB * D;            // [D][KB] (dQ^T A-operand)
+
+  // Q and dO B-fragments + per-lane lse/delta (block-constant)
+  bf16x8 qf[2][DK], dof[2][DK];
+  float lse_r[2], dl_r[2];
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    const int row = wq0 + nf * 16 + (lane & 15);
+    lse_r[nf] = (row < S) ? LSE[(int64_t)bh * S + row] : -3.0e38f;
+    dl_r[nf] = (row < S) ? DELTA[(int64_t)bh * S + row] : 0.f;
+#pragma unroll
+    for (int kk = 0; kk < DK; ++kk) {
+      bf16x8 a = {}, b8 = {};
+      if (row < S) {
+        a = *reinterpret_cast<const bf16x8*>(
+            qp + (int64_t)row * q_rs + kk * 32 + 8 * g);
+        b8 = *reinterpret_cast<const bf16x8*>(
+            dop + (int64_t)row * o_rs + kk * 32 + 8 * g);
+      }
+      qf[nf][kk] = a;
+      dof[nf][kk] = b8;
+    }
+  }
+
+  f32x4 acc_dq[DF][2] = {};   // dQ^T: d = 16df+4g+e, q = lane&15 (+16nf)
+
+  // staging registers: K,V natural (16B each) + K^T v_perm slabs
+  constexpr int KUN = KB * D / 8 / NT;
+  constexpr int NSLAB = (KB / 8) * (D / 2);
+  constexpr int SUN = (NSLAB + NT - 1) / NT;
+  bf16x8 krg[KUN], vrg[KUN];
+  uint32_t ktr[SUN][8];
+
+  auto stage_load = [&](int kv0) {
+#pragma unroll
+    for (int u = 0; u < KUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      bf16x8 kv8 = {}, vv8 = {};
+      if (kv0 + row < S) {
+        kv8 = *reinterpret_cast<const bf16x8*>(
+            kp + (int64_t)(kv0 + row) * q_rs + c);
+        vv8 = *reinterpret_cast<const bf16x8*>(
+            vp + (int64_t)(kv0 + row) * q_rs + c);
+      }
+      krg[u] = kv8;
+      vrg[u] = vv8;
+    }
+#pragma unroll
+    for (int u = 0; u < SUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= NSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kv = kv0 + kb * 8 + j;
+        bf16x2 v2 = {};
+        if (kv < S)
+          v2 = *reinterpret_cast<const bf16x2*>(kp + (int64_t)kv * q_rs + f);
+        ktr[u][j] = __builtin_bit_cast(uint32_t, v2);
+      }
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int u = 0; u < KUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      const int row = idx / (D / 8);
+      const int c = (idx % (D / 8)) * 8;
+      *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = krg[u];
+      *reinterpret_cast<bf16x8*>(sV + loff<D>(row, c)) = vrg[u];
+    }
+#pragma unroll
+    for (int u = 0; u < SUN; ++u) {
+      const int idx = threadIdx.x + u * NT;
+      if (idx >= NSLAB) break;
+      const int f = 2 * (idx % (D / 2));
+      const int kb = idx / (D / 2);
+      uint32_t o0[4], o1[4];
+#pragma unroll
+      for (int d2 = 0; d2 < 4; ++d2) {
+        o0[d2] = __builtin_amdgcn_perm(ktr[u][2 * d2 + 1], ktr[u][2 * d2],
+                                       0x05040100u);
+        o1[d2] = __builtin_amdgcn_perm(ktr[u][2 * d2 + 1], ktr[u][2 * d2],
+                                       0x07060302u);
+      }
+      *reinterpret_cast<uint4*>(sKT + loff<KB>(f, kb * 8)) =
+          make_uint4(o0[0], o0[1], o0[2], o0[3]);
+      *reinterpret_cast<uint4*>(sKT + loff<KB>(f + 1, kb * 8)) =
+          make_uint4(o1[0], o1[1], o1[2], o1[3]);
+    }
+  };
+
+  const int kv_end = causal ? min(S, q0 + QB) : S;
+  stage_load(0);
+  stage_write();
+  __syncthreads();
+  for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
+    if (kv0 + KB < kv_end) stage_load(kv0 + KB);
+
+    if (!causal || kv0 <= wq0 + 31) {
+      // per 16-kv block: S^T = K Q^T and dP^T = V dO^T, then dS^T packed
+      // to bf16 immediately (keeps only two f32x4 accumulator pairs live
+      // at a time — register pressure gates a second wave per SIMD)
+      const bool inner = kv0 + KB <= wq0 && kv0 + KB <= S && wq0 + 32 <= S;
+      bf16x4 dsb[4][2];   // dS^T bf16: kv = 16mi+4g+e, q = lane&15+16nf
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        f32x4 st[2] = {};
+        f32x4 dpt[2] = {};
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int kk = 0; kk < DK; ++kk) {
+          const bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+              sK + loff<D>((lane & 15) + 16 * mi, 8 * g + 32 * kk));
+          const bf16x8 va = *reinterpret_cast<const bf16x8*>(
+              sV + loff<D>((lane & 15) + 16 * mi, 8 * g + 32 * kk));
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf) {
+            st[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                ka, qf[nf][kk], st[nf], 0, 0, 0);
+            dpt[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                va, dof[nf][kk], dpt[nf], 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          const int qg = wq0 + nf * 16 + (lane & 15);
+          if (inner) {
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              const float pt = fast_exp(st[nf][e] * scale - lse_r[nf]);
+              dsb[mi][nf][e] =
+                  f2bf(pt * (dpt[nf][e] - dl_r[nf]) * scale);
+            }
+          } else {
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              const int kg = kv0 + 16 * mi + 4 * g + e;
+              const bool valid = qg < S && kg < S &&
+                                 (!causal || kg <= qg) &&
+                                 lse_r[nf] > -1.0e38f;
+              const float arg = valid ? st[nf][e] * scale - lse_r[nf]
+                                      : -3.0e38f;
+              const float pt = fast_exp(arg);
+              dsb[mi][nf][e] =
+                  f2bf(pt * (dpt[nf][e] - dl_r[nf]) * scale);
+            }
+          }
+        }
+      }
+      // dQ^T += K^T dS^T (k-permuted B straight from the accumulators)
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int c = 0; c < KB / 32; ++c) {
+        bf16x8 db[2];
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            db[nf][e] = dsb[2 * c][nf][e];
+            db[nf][e + 4] = dsb[2 * c + 1][nf][e];
+          }
+#pragma unroll
+        for (int df = 0; df < DF; ++df) {
+          const bf16x4 k0 = *reinterpret_cast<const bf16x4*>(
+              sKT + loff<KB>(16 * df + (lane & 15), 32 * c + 4 * g));
+          const bf16x4 k1 = *reinterpret_cast<const bf16x4*>(
+              sKT + loff<KB>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+          bf16x8 kfr;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            kfr[e] = k0[e];
+            kfr[e + 4] = k1[e];
+          }
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc_dq[df][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kfr, db[nf], acc_dq[df][nf], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+    BAR_LDS();
+    if (kv0 + KB < kv_end) {
+      stage_write();
+      BAR_LDS();
+    }
+  }
+
+  // write dQ (transposed-accumulator scatter, bf16, strided layout)
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    const int qg = wq0 + nf * 16 + (lane & 15);
+    if (qg >= S) continue;
+#pragma unroll
+    for (int df = 0; df < DF; ++df) {
+      bf16x4 v4;
+#pragma unroll
+      for (int e = 0; e < 4; ++e) v4[e] = f2bf(acc_dq[df][nf][e]);
+      *reinterpret_cast<bf16x4*>(
+          dqp + (int64_t)qg * q_rs + 16 * df + 4 * g) = v4;
     }
   }
 }
@@ -857,7 +1066,7 @@ void cast_scatter_bf16(const float* src, void* dst, int B, int H, int S,
 
 void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         const void* o, const void* dout, const float* lse,
-                        float* delta, float* dq_ws, void* dk, void* dv,
+                        float* delta, void* dq, void* dk, void* dv,
                         int B, int H, int S, int D, float scale, bool causal,
                         int64_t q_bs, int64_t q_hs, int64_t q_rs,
                         int64_t o_bs, int64_t o_hs, int64_t o_rs,
@@ -871,15 +1080,23 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        stream, static_cast<const bf16_t*>(dout),             \
                        static_cast<const bf16_t*>(o), delta, H, S, o_bs,     \
                        o_hs, o_rs);                                          \
-    dim3 grid((S + KB - 1) / KB, B * H);                                     \
-    hipLaunchKernelGGL(flash_bwd_kernel<DD>, grid, block, 0, stream,         \
+    dim3 kgrid((S + KB - 1) / KB, B * H);                                    \
+    hipLaunchKernelGGL(flash_bwd_kv_kernel<DD>, kgrid, block, 0, stream,     \
                        static_cast<const bf16_t*>(q),                        \
                        static_cast<const bf16_t*>(k),                        \
                        static_cast<const bf16_t*>(v),                        \
-                       static_cast<const bf16_t*>(dout), lse, delta, dq_ws,  \
+                       static_cast<const bf16_t*>(dout), lse, delta,         \
                        static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),   \
                        S, H, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs,    \
                        o_rs);                                                \
+    dim3 qgrid((S + QB - 1) / QB, B * H);                                    \
+    hipLaunchKernelGGL(flash_bwd_dq_kernel<DD>, qgrid, block, 0, stream,     \
+                       static_cast<const bf16_t*>(q),                        \
+                       static_cast<const bf16_t*>(k),                        \
+                       static_cast<const bf16_t*>(v),                        \
+                       static_cast<const bf16_t*>(dout), lse, delta,         \
+                       static_cast<bf16_t*>(dq), S, H, scale, causal, q_bs,  \
+                       q_hs, q_rs, o_bs, o_hs, o_rs);                        \
   } while (0)
   if (D == 64) BWD_D(64);
   else if (D == 128) BWD_D(128);

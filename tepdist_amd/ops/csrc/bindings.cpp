@@ -115,21 +115,18 @@ PYBIND11_MODULE(_tepdist_hip, m) {
 
   m.def("attention_bwd", [](uintptr_t q, uintptr_t k, uintptr_t v,
                             uintptr_t o, uintptr_t dout, uintptr_t lse,
-                            uintptr_t delta, uintptr_t dq_ws, uintptr_t dk,
+                            uintptr_t delta, uintptr_t dq, uintptr_t dk,
                             uintptr_t dv, int B, int H, int seq, int D,
                             float scale, bool causal, int64_t q_bs,
                             int64_t q_hs, int64_t q_rs, int64_t o_bs,
                             int64_t o_hs, int64_t o_rs, uintptr_t stream) {
-    attention_bwd_bf16(reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
-                       reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
-                       reinterpret_cast<void*>(dout),
-                       reinterpret_cast<const float*>(lse),
-                       reinterpret_cast<float*>(delta),
-                       reinterpret_cast<float*>(dq_ws),
-                       reinterpret_cast<void*>(dk),
-                       reinterpret_cast<void*>(dv), B, H, seq, D, scale,
-                       causal, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs,
-                       S(stream));
+    attention_bwd_bf16(
+        reinterpret_cast<void*>(q), reinterpret_cast<void*>(k),
+        reinterpret_cast<void*>(v), reinterpret_cast<void*>(o),
+        reinterpret_cast<void*>(dout), reinterpret_cast<float*>(lse),
+        reinterpret_cast<float*>(delta), reinterpret_cast<void*>(dq),
+        reinterpret_cast<void*>(dk), reinterpret_cast<void*>(dv), B, H, seq,
+        D, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs, S(stream));
     check_launch();
   });
 

@@ -78,7 +78,7 @@ void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
                         hipStream_t stream);
 void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         const void* o, const void* dout, const float* lse,
-                        float* delta, float* dq_ws, void* dk, void* dv,
+                        float* delta, void* dq, void* dk, void* dv,
                         int B, int H, int S, int D, float scale, bool causal,
                         int64_t q_bs, int64_t q_hs, int64_t q_rs,
                         int64_t o_bs, int64_t o_hs, int64_t o_rs,

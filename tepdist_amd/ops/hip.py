@@ -323,18 +323,14 @@ def attention_bwd(dout: torch.Tensor, q, k, v, residuals,
     out, lse = residuals
     dout = dout.contiguous()
     delta = torch.empty(B * H, S, dtype=torch.float32, device=q.device)
-    dq_ws = torch.zeros(B * H * S * D, dtype=torch.float32, device=q.device)
+    dq = torch.empty_like(q)
     dk = torch.empty_like(k)
     dv = torch.empty_like(v)
     sd = (H * S * D, S * D, D)
     ext.attention_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       out.data_ptr(), dout.data_ptr(), lse.data_ptr(),
-                      delta.data_ptr(), dq_ws.data_ptr(), dk.data_ptr(),
+                      delta.data_ptr(), dq.data_ptr(), dk.data_ptr(),
                       dv.data_ptr(), B, H, S, D, scale, causal, *sd, *sd,
-                      _stream())
-    dq = torch.empty_like(q)
-    # reuse the split-K reducer as an fp32 -> bf16 cast (nparts=1)
-    ext.splitk_reduce(dq_ws.data_ptr(), dq.data_ptr(), 1, dq_ws.numel(),
                       _stream())
     return dq, dk, dv
 
@@ -378,21 +374,19 @@ def attention_qkv_bwd(dout: torch.Tensor, qkv: torch.Tensor, heads: int,
     scale = 1.0 / math.sqrt(D)
     dout = dout.contiguous()
     delta = torch.empty(B * heads, S, dtype=torch.float32, device=qkv.device)
-    dq_ws = torch.zeros(B * heads * S * D, dtype=torch.float32,
-                        device=qkv.device)
     dqkv = torch.empty_like(qkv)
     qs = (S * d3, D, d3)
     os_ = (S * d, D, d)
     base = qkv.data_ptr()
     dbase = dqkv.data_ptr()
     esz = qkv.element_size()
+    # the dQ kernel writes the packed strided layout directly (no fp32
+    # workspace, no scatter pass)
     ext.attention_bwd(base, base + d * esz, base + 2 * d * esz,
                       out.data_ptr(), dout.data_ptr(), lse.data_ptr(),
-                      delta.data_ptr(), dq_ws.data_ptr(), dbase + d * esz,
+                      delta.data_ptr(), dbase, dbase + d * esz,
                       dbase + 2 * d * esz, B, heads, S, D, scale, causal,
                       *qs, *os_, _stream())
-    # scatter the dense fp32 dQ partials into the packed dqkv
-    ext.cast_scatter(dq_ws.data_ptr(), dbase, B, heads, S, D, *qs, _stream())
     return dqkv
 
 

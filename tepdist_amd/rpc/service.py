@@ -128,16 +128,23 @@ class TepdistService:
         h = req["handle"]
         plan = self.plans[h]
         env = get_env()
+        def _cast(v):
+            # float feeds follow the service compute dtype (bf16 on GPU):
+            # clients may send fp32 (the smoke examples do)
+            v = v.to(self.device)
+            if v.is_floating_point() and self.device != "cpu":
+                v = v.to(torch.bfloat16)
+            return v
+
         if env.fake_input:
             # FAKE_INPUT (service_env.h:71): freeze the first step's inputs
             # to measure pure step time
             if not hasattr(self, "_frozen_feeds"):
                 self._frozen_feeds = {
-                    k: v.to(self.device)
-                    for k, v in req.get("inputs", {}).items()}
+                    k: _cast(v) for k, v in req.get("inputs", {}).items()}
             feeds = self._frozen_feeds
         else:
-            feeds = {k: v.to(self.device)
+            feeds = {k: _cast(v)
                      for k, v in req.get("inputs", {}).items()}
         from tepdist_amd.utils.tracing import get_tracer
         with self._lock, get_tracer().span(
