@@ -110,11 +110,95 @@ void transpose_kernel(const bf16_t* __restrict__ in,
   }
 }
 
+// v2: 128x128 blocks, XOR-swizzled LDS image, and a v_perm 8x8 register
+// transpose on the way out — every LDS access is a 16-byte vector
+// (conflict-free by the swizzle) and every global write a 16-byte chunk
+// in 128-byte runs, replacing the 8-scalar-LDS-read column walk of the
+// generic kernel. Requires R % 128 == 0 && C % 128 == 0 (all the
+// canonicalization transposes); the generic kernel covers the rest.
+constexpr int T2 = 128;
+
+DEV_INLINE int t2off(int row, int col_e) {  // loff for a [.][128] image
+  const int r1 = row >> 1;
+  const int x = ((r1 & 1) << 2) | (r1 & 2) | ((r1 >> 2) & 1);
+  return row * T2 + (col_e ^ ((x << 3) & 127));
+}
+
+template <bool GELUG, bool NAT>
+__launch_bounds__(NT) __global__
+void transpose2_kernel(const bf16_t* __restrict__ in,
+                       const bf16_t* __restrict__ pre,
+                       bf16_t* __restrict__ out,
+                       bf16_t* __restrict__ out_nat, int R, int C,
+                       int64_t stride_in, int64_t stride_out) {
+  __shared__ bf16_t tile[T2 * T2];
+  const bf16_t* src = in + blockIdx.z * stride_in;
+  bf16_t* dst = out + blockIdx.z * stride_out;
+  const int r0 = blockIdx.y * T2;
+  const int c0 = blockIdx.x * T2;
+
+  // load [128][128]: 2048 16B vectors / 256 threads
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int idx = threadIdx.x + u * NT;
+    const int r = idx >> 4;
+    const int c = (idx & 15) * 8;
+    bf16x8 v =
+        *reinterpret_cast<const bf16x8*>(src + (int64_t)(r0 + r) * C +
+                                         c0 + c);
+    if (GELUG) {
+      const bf16x8 pv = *reinterpret_cast<const bf16x8*>(
+          pre + (int64_t)(r0 + r) * C + c0 + c);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        v[e] = f2bf(bf2f(v[e]) * gelu_grad_f(bf2f(pv[e])));
+    }
+    if (NAT)
+      *reinterpret_cast<bf16x8*>(out_nat + (int64_t)(r0 + r) * C + c0 + c) =
+          v;
+    *reinterpret_cast<bf16x8*>(tile + t2off(r, c)) = v;
+  }
+  __syncthreads();
+
+  // store: thread owns an 8x8 block; cblk is the slow index so that the
+  // 8 lanes of each cblk group write 128 contiguous bytes per output row
+  const int cblk = threadIdx.x >> 4;       // 0..15 (input col block)
+  const int rblk = threadIdx.x & 15;       // 0..15 (input row block)
+  uint32_t rows[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    *reinterpret_cast<bf16x8*>(rows[i]) = *reinterpret_cast<const bf16x8*>(
+        tile + t2off(rblk * 8 + i, cblk * 8));
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    uint32_t lo[4], hi[4];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      lo[p] = __builtin_amdgcn_perm(rows[2 * p + 1][j], rows[2 * p][j],
+                                    0x05040100u);
+      hi[p] = __builtin_amdgcn_perm(rows[2 * p + 1][j], rows[2 * p][j],
+                                    0x07060302u);
+    }
+    bf16_t* d0 = dst + (int64_t)(c0 + cblk * 8 + 2 * j) * R + r0 + rblk * 8;
+    bf16_t* d1 = d0 + R;
+    *reinterpret_cast<uint4*>(d0) = make_uint4(lo[0], lo[1], lo[2], lo[3]);
+    *reinterpret_cast<uint4*>(d1) = make_uint4(hi[0], hi[1], hi[2], hi[3]);
+  }
+}
+
 }  // namespace
 
 void transpose_bf16(const void* in, void* out, int R, int C,
                     int64_t stride_in, int64_t stride_out, int batch,
                     hipStream_t stream) {
+  if (R % T2 == 0 && C % T2 == 0) {
+    dim3 grid(C / T2, R / T2, batch);
+    hipLaunchKernelGGL((transpose2_kernel<false, false>), grid, dim3(NT), 0,
+                       stream, static_cast<const bf16_t*>(in), nullptr,
+                       static_cast<bf16_t*>(out), nullptr, R, C, stride_in,
+                       stride_out);
+    return;
+  }
   dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, batch);
   hipLaunchKernelGGL((transpose_kernel<false, false>), grid, dim3(NT), 0,
                      stream, static_cast<const bf16_t*>(in), nullptr,
@@ -127,11 +211,21 @@ void transpose_bf16(const void* in, void* out, int R, int C,
 void transpose_dy_bf16(const void* dy, const void* pre, void* dy_t,
                        void* dy_nat, float* bias_ws, int R, int C,
                        hipStream_t stream) {
-  dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, 1);
   const bf16_t* d = static_cast<const bf16_t*>(dy);
   const bf16_t* p = static_cast<const bf16_t*>(pre);
   bf16_t* t = static_cast<bf16_t*>(dy_t);
   bf16_t* n = static_cast<bf16_t*>(dy_nat);
+  if (bias_ws == nullptr && R % T2 == 0 && C % T2 == 0) {
+    dim3 g2(C / T2, R / T2, 1);
+    if (pre != nullptr)
+      hipLaunchKernelGGL((transpose2_kernel<true, true>), g2, dim3(NT), 0,
+                         stream, d, p, t, n, R, C, 0, 0);
+    else
+      hipLaunchKernelGGL((transpose2_kernel<false, false>), g2, dim3(NT), 0,
+                         stream, d, p, t, n, R, C, 0, 0);
+    return;
+  }
+  dim3 grid((C + TS - 1) / TS, (R + TS - 1) / TS, 1);
 #define TK(G, NA, B)                                                     \
   hipLaunchKernelGGL((transpose_kernel<G, NA, B>), grid, dim3(NT), 0,    \
                      stream, d, p, t, n, bias_ws, R, C, 0, 0)
