@@ -1,0 +1,103 @@
+"""Llama-family decoder (beyond the reference's model set): RMSNorm,
+rotary position embedding, SwiGLU MLP, no biases, untied head — exercises
+the wide-head (D=128) flash-attention path and the llama op kernels
+(ops/csrc/llama_ops.hip). Same training surface as models/gpt2.py
+(forward(ids, labels) -> loss) so the Trainer / bench / parallel layers
+apply unchanged."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+
+from tepdist_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama-test"
+    n_layer: int = 2
+    n_embd: int = 256
+    n_head: int = 4          # head_dim = n_embd / n_head in {64, 128}
+    n_ctx: int = 512
+    vocab_size: int = 512
+    ffn_mult: int = 2        # intermediate = ffn_mult * n_embd (llama ~2.7)
+    rope_theta: float = 10000.0
+    rms_eps: float = 1e-6
+
+
+LLAMA_CONFIGS = {
+    "llama-test": LlamaConfig(),
+    # ~1.1B-parameter class: 2048 hidden, 16 x 128 heads (wide-head path)
+    "llama-1b": LlamaConfig(name="llama-1b", n_layer=22, n_embd=2048,
+                            n_head=16, n_ctx=2048, vocab_size=32000,
+                            ffn_mult=3),
+}
+
+
+def _param(*shape, std=0.02, dtype=torch.float32):
+    return nn.Parameter(torch.randn(*shape, dtype=dtype) * std)
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.float32):
+        super().__init__()
+        d = cfg.n_embd
+        h = cfg.ffn_mult * d
+        self.cfg = cfg
+        self.head_dim = d // cfg.n_head
+        self.ln1_g = nn.Parameter(torch.ones(d, dtype=dtype))
+        self.ln2_g = nn.Parameter(torch.ones(d, dtype=dtype))
+        self.w_qkv = _param(3 * d, d, dtype=dtype)
+        self.w_o = _param(d, d, dtype=dtype)
+        self.w_gate = _param(h, d, dtype=dtype)
+        self.w_up = _param(h, d, dtype=dtype)
+        self.w_down = _param(d, h, dtype=dtype)
+
+    def forward(self, x, seq_len: int):
+        cfg = self.cfg
+        d = cfg.n_embd
+        h = ops.rmsnorm(x, self.ln1_g, cfg.rms_eps)
+        qkv = ops.linear(h, self.w_qkv)               # [T, 3d]
+        T = qkv.shape[0]
+        nh, hd = cfg.n_head, self.head_dim
+        q, k, v = qkv.split(d, dim=-1)
+        # rotary on q/k per head
+        q = ops.rope(q.reshape(T, nh, hd), seq_len, cfg.rope_theta)
+        k = ops.rope(k.reshape(T, nh, hd), seq_len, cfg.rope_theta)
+        b = T // seq_len
+        def heads(t):
+            return t.reshape(b, seq_len, nh, hd).transpose(1, 2).contiguous()
+        o = ops.attention(heads(q), heads(k),
+                          heads(v.reshape(T, nh, hd)), causal=True)
+        o = o.transpose(1, 2).reshape(T, d).contiguous()
+        x = x + ops.linear(o, self.w_o)
+        hn = ops.rmsnorm(x, self.ln2_g, cfg.rms_eps)
+        gate = ops.linear(hn, self.w_gate)
+        up = ops.linear(hn, self.w_up)
+        x = x + ops.linear(ops.swiglu(gate, up), self.w_down)
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.float32):
+        super().__init__()
+        self.cfg = cfg
+        self.wte = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
+        self.blocks = nn.ModuleList(
+            LlamaBlock(cfg, dtype) for _ in range(cfg.n_layer))
+        self.ln_f_g = nn.Parameter(torch.ones(cfg.n_embd, dtype=dtype))
+        self.lm_head = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
+
+    def forward(self, ids, labels=None):
+        b, s = ids.shape
+        x = ops.embedding(ids.reshape(-1), self.wte)
+        for blk in self.blocks:
+            x = blk(x, s)
+        x = ops.rmsnorm(x, self.ln_f_g, self.cfg.rms_eps)
+        logits = ops.linear(x, self.lm_head)
+        if labels is None:
+            return logits
+        return ops.cross_entropy(logits, labels.reshape(-1))

@@ -34,6 +34,7 @@ SOURCES = [
     "dropout.hip",
     "elementwise.hip",
     "adamw.hip",
+    "llama_ops.hip",
     "debug.hip",
     "bindings.cpp",
 ]

@@ -221,6 +221,48 @@ PYBIND11_MODULE(_tepdist_hip, m) {
                       reinterpret_cast<float*>(bias_ws), R, C, S(stream));
     check_launch();
   });
+  m.def("rmsnorm_fwd", [](uintptr_t x, uintptr_t g, uintptr_t y,
+                          uintptr_t rstd, int64_t rows, int cols, float eps,
+                          uintptr_t stream) {
+    rmsnorm_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(g),
+                     reinterpret_cast<void*>(y),
+                     reinterpret_cast<float*>(rstd), rows, cols, eps,
+                     S(stream));
+    check_launch();
+  });
+  m.def("rmsnorm_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t g,
+                          uintptr_t rstd, uintptr_t dx, uintptr_t dg,
+                          uintptr_t dg_part, int part_rows, int64_t rows,
+                          int cols, uintptr_t stream) {
+    rmsnorm_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),
+                     reinterpret_cast<void*>(g),
+                     reinterpret_cast<const float*>(rstd),
+                     reinterpret_cast<void*>(dx), reinterpret_cast<void*>(dg),
+                     reinterpret_cast<float*>(dg_part), part_rows, rows,
+                     cols, S(stream));
+    check_launch();
+  });
+  m.def("rope", [](uintptr_t x, uintptr_t y, int64_t tokens, int heads,
+                   int D, int seq_len, float theta, bool backward,
+                   uintptr_t stream) {
+    rope_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(y), tokens,
+              heads, D, seq_len, theta, backward, S(stream));
+    check_launch();
+  });
+  m.def("swiglu_fwd", [](uintptr_t a, uintptr_t b, uintptr_t y, int64_t n,
+                         uintptr_t stream) {
+    swiglu_fwd_bf16(reinterpret_cast<void*>(a), reinterpret_cast<void*>(b),
+                    reinterpret_cast<void*>(y), n, S(stream));
+    check_launch();
+  });
+  m.def("swiglu_bwd", [](uintptr_t dy, uintptr_t a, uintptr_t b,
+                         uintptr_t da, uintptr_t db, int64_t n,
+                         uintptr_t stream) {
+    swiglu_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(a),
+                    reinterpret_cast<void*>(b), reinterpret_cast<void*>(da),
+                    reinterpret_cast<void*>(db), n, S(stream));
+    check_launch();
+  });
   m.def("gelu_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t dx, int64_t n,
                        uintptr_t stream) {
     gelu_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),

@@ -133,4 +133,18 @@ void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
 // --- diagnostics -----------------------------------------------------------
 void tr16_probe(float* out_pattern, float* out_uniform, hipStream_t stream);
 
+// --- Llama-family ops ------------------------------------------------------
+void rmsnorm_fwd_bf16(const void* x, const void* g, void* y, float* rstd,
+                      int64_t rows, int cols, float eps, hipStream_t stream);
+void rmsnorm_bwd_bf16(const void* dy, const void* x, const void* g,
+                      const float* rstd, void* dx, void* dgamma,
+                      float* dg_part, int part_rows, int64_t rows, int cols,
+                      hipStream_t stream);
+void rope_bf16(const void* x, void* y, int64_t tokens, int heads, int D,
+               int seq_len, float theta, bool backward, hipStream_t stream);
+void swiglu_fwd_bf16(const void* a, const void* b, void* y, int64_t n,
+                     hipStream_t stream);
+void swiglu_bwd_bf16(const void* dy, const void* a, const void* b, void* da,
+                     void* db, int64_t n, hipStream_t stream);
+
 }  // namespace tepdist

@@ -558,3 +558,64 @@ def adamw_step(param_bf16: torch.Tensor, master: torch.Tensor,
     ext.adamw(param_bf16.data_ptr(), master.data_ptr(), gb, gf,
               exp_avg.data_ptr(), exp_avg_sq.data_ptr(), param_bf16.numel(),
               lr, beta1, beta2, eps, weight_decay, bc1, bc2, _stream())
+
+
+# --------------------------------------------------------------------------
+# Llama-family ops (RMSNorm / RoPE / SwiGLU)
+# --------------------------------------------------------------------------
+
+def rmsnorm_fwd(x: torch.Tensor, gamma: torch.Tensor, eps: float = 1e-6):
+    x = x.contiguous()
+    rows, cols = x.shape
+    y = torch.empty_like(x)
+    rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+    ext.rmsnorm_fwd(x.data_ptr(), gamma.contiguous().data_ptr(), y.data_ptr(),
+                    rstd.data_ptr(), rows, cols, eps, _stream())
+    return y, rstd
+
+
+def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, gamma: torch.Tensor,
+                rstd: torch.Tensor):
+    rows, cols = x.shape
+    dy = dy.contiguous()
+    part_rows = min((rows + 3) // 4 * 4, 2048 * 4)
+    dg_part = torch.zeros(part_rows, cols, dtype=torch.float32,
+                          device=x.device)
+    dx = torch.empty_like(x)
+    dg = torch.empty_like(gamma)
+    ext.rmsnorm_bwd(dy.data_ptr(), x.data_ptr(),
+                    gamma.contiguous().data_ptr(), rstd.data_ptr(),
+                    dx.data_ptr(), dg.data_ptr(), dg_part.data_ptr(),
+                    part_rows, rows, cols, _stream())
+    return dx, dg
+
+
+def rope_fwd(x: torch.Tensor, seq_len: int, theta: float = 10000.0):
+    T, H, D = x.shape
+    y = torch.empty_like(x)
+    ext.rope(x.data_ptr(), y.data_ptr(), T, H, D, seq_len, theta, False,
+             _stream())
+    return y
+
+
+def rope_bwd(dy: torch.Tensor, seq_len: int, theta: float = 10000.0):
+    T, H, D = dy.shape
+    dx = torch.empty_like(dy)
+    ext.rope(dy.data_ptr(), dx.data_ptr(), T, H, D, seq_len, theta, True,
+             _stream())
+    return dx
+
+
+def swiglu_fwd(a: torch.Tensor, b: torch.Tensor):
+    y = torch.empty_like(a)
+    ext.swiglu_fwd(a.data_ptr(), b.data_ptr(), y.data_ptr(), a.numel(),
+                   _stream())
+    return y
+
+
+def swiglu_bwd(dy: torch.Tensor, a: torch.Tensor, b: torch.Tensor):
+    da = torch.empty_like(a)
+    db = torch.empty_like(b)
+    ext.swiglu_bwd(dy.data_ptr(), a.data_ptr(), b.data_ptr(), da.data_ptr(),
+                   db.data_ptr(), a.numel(), _stream())
+    return da, db

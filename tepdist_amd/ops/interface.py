@@ -277,3 +277,71 @@ def adamw_step(param, master, grad, exp_avg, exp_avg_sq, *, lr, beta1=0.9,
     be = _backend(param)
     be.adamw_step(param, master, grad, exp_avg, exp_avg_sq, lr, beta1, beta2,
                   eps, weight_decay, step)
+
+
+# --------------------------------------------------------------------------
+# RMSNorm / RoPE / SwiGLU (llama family)
+# --------------------------------------------------------------------------
+
+
+class RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, eps):
+        be = _backend(x)
+        x2d = x.reshape(-1, x.shape[-1])
+        y, rstd = be.rmsnorm_fwd(x2d, gamma, eps)
+        ctx.save_for_backward(x2d, gamma, rstd)
+        ctx.in_shape = x.shape
+        return y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, gamma, rstd = ctx.saved_tensors
+        be = _backend(dy)
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx, dg = be.rmsnorm_bwd(dy2d, x2d, gamma, rstd)
+        return dx.reshape(ctx.in_shape), dg, None
+
+
+def rmsnorm(x, gamma, eps: float = 1e-6):
+    return RMSNormFn.apply(x, gamma, eps)
+
+
+class RoPEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, seq_len, theta):
+        be = _backend(x)
+        ctx.seq_len, ctx.theta = seq_len, theta
+        return be.rope_fwd(x.contiguous(), seq_len, theta)
+
+    @staticmethod
+    def backward(ctx, dy):
+        be = _backend(dy)
+        return be.rope_bwd(dy.contiguous(), ctx.seq_len, ctx.theta), \
+            None, None
+
+
+def rope(x, seq_len: int, theta: float = 10000.0):
+    """Rotary position embedding over [tokens, heads, head_dim]
+    (rotate-half convention; position = token index % seq_len)."""
+    return RoPEFn.apply(x, seq_len, theta)
+
+
+class SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        be = _backend(a)
+        ctx.save_for_backward(a, b)
+        return be.swiglu_fwd(a.contiguous(), b.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        a, b = ctx.saved_tensors
+        be = _backend(dy)
+        da, db = be.swiglu_bwd(dy.contiguous(), a.contiguous(),
+                               b.contiguous())
+        return da, db
+
+
+def swiglu(a, b):
+    return SwiGLUFn.apply(a, b)

@@ -288,3 +288,68 @@ def adamw_step(param_bf16: torch.Tensor, master: torch.Tensor,
     master.mul_(1 - lr * weight_decay)
     master.addcdiv_(exp_avg / bc1, denom, value=-lr)
     param_bf16.copy_(master.to(param_bf16.dtype))
+
+
+# --------------------------------------------------------------------------
+# RMSNorm / RoPE / SwiGLU (llama-family ops; torch fp32-semantics references)
+# --------------------------------------------------------------------------
+
+def rmsnorm_fwd(x: torch.Tensor, gamma: torch.Tensor, eps: float = 1e-6):
+    xf = x.float()
+    rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    y = (xf * rstd) * gamma.float()
+    return y.to(x.dtype), rstd.squeeze(-1)
+
+
+def rmsnorm_bwd(dy: torch.Tensor, x: torch.Tensor, gamma: torch.Tensor,
+                rstd: torch.Tensor):
+    xf, dyf, gf = x.float(), dy.float(), gamma.float()
+    r = rstd.unsqueeze(-1)
+    xh = xf * r
+    dg = (dyf * xh).sum(0)
+    t = dyf * gf
+    dx = r * (t - xh * (t * xh).mean(-1, keepdim=True))
+    return dx.to(x.dtype), dg.to(gamma.dtype)
+
+
+def rope_fwd(x: torch.Tensor, seq_len: int, theta: float = 10000.0):
+    """x [tokens, heads, D] (tokens = batch*seq flattened, position =
+    token % seq_len); rotate-half convention."""
+    T, H, D = x.shape
+    pos = (torch.arange(T, device=x.device) % seq_len).float()
+    i = torch.arange(D // 2, device=x.device).float()
+    freq = theta ** (-2.0 * i / D)
+    ang = pos[:, None] * freq[None, :]
+    cos, sin = ang.cos()[:, None, :], ang.sin()[:, None, :]
+    xf = x.float()
+    x1, x2 = xf[..., :D // 2], xf[..., D // 2:]
+    y = torch.cat([x1 * cos - x2 * sin, x2 * cos + x1 * sin], dim=-1)
+    return y.to(x.dtype)
+
+
+def rope_bwd(dy: torch.Tensor, seq_len: int, theta: float = 10000.0):
+    """The inverse rotation (transpose of an orthogonal map)."""
+    T, H, D = dy.shape
+    pos = (torch.arange(T, device=dy.device) % seq_len).float()
+    i = torch.arange(D // 2, device=dy.device).float()
+    freq = theta ** (-2.0 * i / D)
+    ang = pos[:, None] * freq[None, :]
+    cos, sin = ang.cos()[:, None, :], ang.sin()[:, None, :]
+    df = dy.float()
+    d1, d2 = df[..., :D // 2], df[..., D // 2:]
+    dx = torch.cat([d1 * cos + d2 * sin, d2 * cos - d1 * sin], dim=-1)
+    return dx.to(dy.dtype)
+
+
+def swiglu_fwd(a: torch.Tensor, b: torch.Tensor):
+    af = a.float()
+    return (af * torch.sigmoid(af) * b.float()).to(a.dtype)
+
+
+def swiglu_bwd(dy: torch.Tensor, a: torch.Tensor, b: torch.Tensor):
+    af, bf, dyf = a.float(), b.float(), dy.float()
+    sig = torch.sigmoid(af)
+    silu = af * sig
+    da = dyf * bf * (sig + silu * (1.0 - sig))
+    db = dyf * silu
+    return da.to(a.dtype), db.to(b.dtype)
