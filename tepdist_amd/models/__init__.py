@@ -7,3 +7,4 @@ from tepdist_amd.models.configs import (  # noqa: F401
     WideResNetConfig,
 )
 from tepdist_amd.models.gpt2 import GPT2, GPT2Block  # noqa: F401
+from tepdist_amd.models.llama import LLAMA_CONFIGS, Llama, LlamaConfig  # noqa: F401
