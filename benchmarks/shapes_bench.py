@@ -23,3 +23,10 @@ for name, (m,n,k) in {
     b = torch.randn(n,k).bfloat16().cuda()
     sec = t(lambda: hip.matmul(a, b.t()))
     print(f"{name:10s} [{m},{n},{k}] {sec*1e6:7.1f}us {2*m*n*k/sec/1e12:7.1f} TF")
+
+# gelu-epilogue shape (fc forward: dual output C + Cpre)
+a = torch.randn(M, 1024).bfloat16().cuda()
+w = torch.randn(4096, 1024).bfloat16().cuda()
+b = torch.zeros(4096).bfloat16().cuda()
+sec = t(lambda: hip.linear_fwd(a, w, b, "gelu"))
+print(f"fc_gelu    [{M},4096,1024] {sec*1e6:7.1f}us {2*M*4096*1024/sec/1e12:7.1f} TF")

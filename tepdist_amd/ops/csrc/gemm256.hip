@@ -237,8 +237,10 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
     // all 512 threads store it row-major (8 lanes = one full row).
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    bf16_t* img = smem;                  // [256][64] post-activation
-    bf16_t* img2 = smem + 256 * BK;      // [256][64] pre-activation
+    // gelu epilogues stage only the PRE-activation image; gelu is applied
+    // at store time on the vector chunks (halves the LDS traffic of the
+    // dual-output fc epilogue)
+    bf16_t* img = smem;                  // [256][64]
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int cl = wn * 16 + (lane & 15);
@@ -249,11 +251,6 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
           const int row = wm * 128 + mi * 16 + (lane >> 4) * 4 + e;
           float v = acc[mi][p][e];
           if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) v += bv[p];
-          if (EPI >= 2) {
-            const bf16_t pre = f2bf(v);
-            img2[qoff(row, cl)] = pre;
-            v = gelu_f(bf2f(pre));
-          }
           img[qoff(row, cl)] = f2bf(v);
         }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
@@ -264,11 +261,13 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
       for (int rr = 0; rr < BM; rr += 64) {
         const int row = rr + sr;
         const int64_t off = (int64_t)(m0 + row) * ldc + n0 + 64 * p + c8;
-        *reinterpret_cast<bf16x8*>(C + off) =
-            *reinterpret_cast<const bf16x8*>(img + qoff(row, c8));
-        if (EPI >= 2)
-          *reinterpret_cast<bf16x8*>(Cpre + off) =
-              *reinterpret_cast<const bf16x8*>(img2 + qoff(row, c8));
+        bf16x8 v8 = *reinterpret_cast<const bf16x8*>(img + qoff(row, c8));
+        if (EPI >= 2) {
+          *reinterpret_cast<bf16x8*>(Cpre + off) = v8;
+#pragma unroll
+          for (int e = 0; e < 8; ++e) v8[e] = f2bf(gelu_f(bf2f(v8[e])));
+        }
+        *reinterpret_cast<bf16x8*>(C + off) = v8;
       }
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();  // image free for the next phase
