@@ -85,6 +85,63 @@ def gpt2_ir(cfg: GPT2Config, batch: int, seq: int) -> Graph:
     return g
 
 
+def llama_ir(cfg, batch: int, seq: int) -> Graph:
+    """Llama-family exporter (models/llama.py): RMSNorm + rotary + SwiGLU
+    decoder in the same flattened (B*S, hidden) form as gpt2_ir, so the
+    auto-parallel planner and SpmdTransform apply unchanged."""
+    g = Graph()
+    V, d, H, L = cfg.vocab_size, cfg.n_embd, cfg.n_head, cfg.n_layer
+    hd = cfg.ffn_mult * d
+    BS = batch * seq
+    ids = g.add_input("input_ids", (BS,), "i64")
+    ids.attrs["batch"] = batch
+    labels = g.add_input("labels", (BS,), "i64")
+    labels.attrs["batch"] = batch
+
+    wte = g.add_param("wte", (V, d))
+    x = g.add("embedding", [ids, wte], (BS, d), attrs={"batch": batch})
+
+    for l in range(L):
+        a = {"batch": batch, "heads": H, "seq": seq}
+        ln1 = g.add_param(f"h{l}.ln1_g", (d,), op_group=l)
+        h = g.add("rmsnorm", [x, ln1], (BS, d), attrs=a, op_group=l)
+        wqkv = g.add_param(f"h{l}.w_qkv", (3 * d, d), op_group=l)
+        qkv = g.add("linear", [h, wqkv], (BS, 3 * d), attrs=a, op_group=l)
+        q = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 0},
+                  op_group=l)
+        k = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 1},
+                  op_group=l)
+        v = g.add("split", [qkv], (BS, d), attrs={**a, "dim": 1, "index": 2},
+                  op_group=l)
+        qr = g.add("rope", [q], (BS, d), attrs=a, op_group=l)
+        kr = g.add("rope", [k], (BS, d), attrs=a, op_group=l)
+        att = g.add("attention", [qr, kr, v], (BS, d), attrs=a, op_group=l)
+        wo = g.add_param(f"h{l}.w_o", (d, d), op_group=l)
+        pr = g.add("linear", [att, wo], (BS, d), attrs=a, op_group=l)
+        x = g.add("add", [x, pr], (BS, d), attrs=a, op_group=l)
+        ln2 = g.add_param(f"h{l}.ln2_g", (d,), op_group=l)
+        h2 = g.add("rmsnorm", [x, ln2], (BS, d), attrs=a, op_group=l)
+        wg = g.add_param(f"h{l}.w_gate", (hd, d), op_group=l)
+        wu = g.add_param(f"h{l}.w_up", (hd, d), op_group=l)
+        gate = g.add("linear", [h2, wg], (BS, hd), attrs=a, op_group=l)
+        up = g.add("linear", [h2, wu], (BS, hd), attrs=a, op_group=l)
+        sw = g.add("swiglu", [gate, up], (BS, hd), attrs=a, op_group=l)
+        wd = g.add_param(f"h{l}.w_down", (d, hd), op_group=l)
+        o = g.add("linear", [sw, wd], (BS, d), attrs=a, op_group=l)
+        x = g.add("add", [x, o], (BS, d), attrs=a, op_group=l)
+
+    lnf = g.add_param("lnf_g", (d,), op_group=L - 1)
+    x = g.add("rmsnorm", [x, lnf], (BS, d), attrs={"batch": batch},
+              op_group=L - 1)
+    head = g.add_param("lm_head", (V, d), op_group=L - 1)
+    logits = g.add("linear", [x, head], (BS, V), attrs={"batch": batch},
+                   op_group=L - 1)
+    loss = g.add("cross_entropy", [logits, labels], (),
+                 attrs={"batch": batch}, op_group=L - 1)
+    g.outputs = [loss.id]
+    return g
+
+
 # --------------------------------------------------------------------------
 # generic fx capture (simple clients)
 # --------------------------------------------------------------------------

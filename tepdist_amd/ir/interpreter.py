@@ -97,6 +97,16 @@ class GraphInterpreter:
             return ins[0]
         if n.op == "scale":
             return ins[0] * n.attrs.get("scale", 1.0)
+        if n.op == "rmsnorm":
+            return ops.rmsnorm(ins[0], ins[1])
+        if n.op == "rope":
+            heads = n.attrs["heads"]
+            T, d = ins[0].shape
+            hd = d // heads
+            y = ops.rope(ins[0].reshape(T, heads, hd), n.attrs["seq"])
+            return y.reshape(T, d)
+        if n.op == "swiglu":
+            return ops.swiglu(ins[0], ins[1])
         # reshard collectives inserted by the SpmdTransform (autograd-aware:
         # backward of the transformed graph is correct by construction)
         if n.op == "all_reduce":

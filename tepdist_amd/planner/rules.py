@@ -96,7 +96,7 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
         out.append(rep())
         return out
 
-    if node.op in ("layernorm", "softmax"):
+    if node.op in ("layernorm", "rmsnorm", "softmax"):
         # reduction over the last dim: split any other dim; aux params rep
         for d in range(len(sh) - 1):
             if (_splittable(sh, d, n) if d > 0 else _split0_ok(node, n)):
@@ -148,7 +148,7 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
         out.append(rep())
         return out
 
-    if node.op in ("add", "mul", "bias_add"):
+    if node.op in ("add", "mul", "bias_add", "swiglu"):
         for d in range(len(sh)):
             if (_splittable(sh, d, n) if d > 0 else _split0_ok(node, n)):
                 ii = []
@@ -165,6 +165,14 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
                         break
                 if ok:
                     out.append(OpStrategy(S(d, n), tuple(ii), f"S{d}"))
+        out.append(rep())
+        return out
+
+    if node.op == "rope":
+        # position depends on the token index: only sequence-preserving
+        # batch splits keep the rotation angles right
+        if _split0_ok(node, n):
+            out.append(OpStrategy(S(0, n), (S(0, n),), "S0"))
         out.append(rep())
         return out
 
