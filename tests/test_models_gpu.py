@@ -112,3 +112,17 @@ def test_llama_gpu_trains():
         opt.step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] - 0.5, losses
+
+
+def test_generation_engine_gpu():
+    from tepdist_amd.inference import Generator
+    from tepdist_amd.models import GPT2, GPT2_CONFIGS
+
+    torch.manual_seed(0)
+    cfg = GPT2_CONFIGS["gpt2-test"]
+    model = GPT2(cfg, dtype=torch.bfloat16).cuda().eval()
+    ids = torch.randint(0, cfg.vocab_size, (2, 12), device="cuda")
+    out = Generator(model).generate(ids, max_new_tokens=8)
+    assert out.shape == (2, 20)
+    assert (out[:, :12] == ids).all()
+    assert (out < cfg.vocab_size).all()
