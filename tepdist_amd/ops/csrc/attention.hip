@@ -1030,39 +1030,6 @@ __global__ void flash_bwd_delta_strided_kernel(
 }  // namespace
 
 
-namespace {
-__global__ void cast_scatter_kernel(const float* __restrict__ src,
-                                    bf16_t* __restrict__ dst, int H, int S,
-                                    int D, int64_t bs, int64_t hs,
-                                    int64_t rs) {
-  const int bh = blockIdx.y;
-  const int b = bh / H, h = bh % H;
-  const int64_t base = (int64_t)b * bs + (int64_t)h * hs;
-  const int64_t n = (int64_t)S * D;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-       i < n; i += stride) {
-    const int64_t row = i / D;
-    const int c = (int)(i - row * D);
-    const f32x4 v = *reinterpret_cast<const f32x4*>(
-        src + (int64_t)bh * n + i);
-    bf16x4 o;
-#pragma unroll
-    for (int e = 0; e < 4; ++e) o[e] = f2bf(v[e]);
-    *reinterpret_cast<bf16x4*>(dst + base + row * rs + c) = o;
-  }
-}
-}  // namespace
-
-void cast_scatter_bf16(const float* src, void* dst, int B, int H, int S,
-                       int D, int64_t bs, int64_t hs, int64_t rs,
-                       hipStream_t stream) {
-  const int blocks = (int)std::min<int64_t>(
-      ((int64_t)S * D / 4 + NT - 1) / NT, 512);
-  hipLaunchKernelGGL(cast_scatter_kernel, dim3(std::max(blocks, 1), B * H),
-                     dim3(NT), 0, stream, src, static_cast<bf16_t*>(dst), H,
-                     S, D, bs, hs, rs);
-}
 
 void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         const void* o, const void* dout, const float* lse,

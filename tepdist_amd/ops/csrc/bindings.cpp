@@ -130,14 +130,6 @@ PYBIND11_MODULE(_tepdist_hip, m) {
     check_launch();
   });
 
-  m.def("cast_scatter", [](uintptr_t src, uintptr_t dst, int B, int H,
-                           int seq, int D, int64_t bs, int64_t hs,
-                           int64_t rs, uintptr_t stream) {
-    cast_scatter_bf16(reinterpret_cast<const float*>(src),
-                      reinterpret_cast<void*>(dst), B, H, seq, D, bs, hs,
-                      rs, S(stream));
-    check_launch();
-  });
 
   m.def("embedding_fwd", [](uintptr_t ids, uintptr_t table, uintptr_t out,
                             int64_t n_ids, int dim, uintptr_t stream) {

@@ -84,9 +84,6 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                         int64_t o_bs, int64_t o_hs, int64_t o_rs,
                         hipStream_t stream);
 
-void cast_scatter_bf16(const float* src, void* dst, int B, int H, int S,
-                       int D, int64_t bs, int64_t hs, int64_t rs,
-                       hipStream_t stream);
 
 // --- Embedding -------------------------------------------------------------
 void embedding_fwd_bf16(const int64_t* ids, const void* table, void* out,
