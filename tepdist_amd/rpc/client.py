@@ -64,11 +64,13 @@ class TepdistClient:
         return self._call("DoRemoteRestore", {"global_step": global_step})
 
     def init_remote_comm(self, master_addr: str, master_port: int, rank: int,
-                         world: int) -> dict:
+                         world: int, join: bool = False) -> dict:
+        """join=True makes the worker enter the process group now (blocks
+        until every rank arrives — fan the call out in parallel)."""
         return self._call("InitRemoteComm",
                           {"master_addr": master_addr,
                            "master_port": master_port, "rank": rank,
-                           "world": world})
+                           "world": world, "join": join})
 
     def transfer_module_and_defctx(self, graph_json: str,
                                    def_tree: str = "") -> dict:

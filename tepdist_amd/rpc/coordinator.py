@@ -50,9 +50,11 @@ class ExecutionCoordinator:
     def dispatch_plan(self, plan: dict = None):
         return self._fanout("dispatch_plan", plan)
 
-    def init_remote_comm(self, master_addr: str, master_port: int):
+    def init_remote_comm(self, master_addr: str, master_port: int,
+                         join: bool = False):
         futs = [self._pool.submit(c.init_remote_comm, master_addr,
-                                  master_port, rank, len(self.clients))
+                                  master_port, rank, len(self.clients),
+                                  join)
                 for rank, c in enumerate(self.clients)]
         return [f.result() for f in futs]
 

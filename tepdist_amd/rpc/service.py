@@ -95,9 +95,14 @@ class TepdistService:
                 "micro_batches": plan.micro_batches,
                 "search_time_s": time.time() - t0}
 
-    def _init_variables(self, g: Graph, init_specs: dict):
+    def _init_variables(self, g: Graph, init_specs: dict,
+                        param_specs: dict = None):
         """Server-side variable creation (RewriteInitializationRemote moved
-        init to the server; DistributedRandomInitializer seeds shards)."""
+        init to the server; DistributedRandomInitializer seeds shards).
+        With param_specs (from a dispatched sharded plan) each worker
+        keeps only its slice — the deterministic global-index generator
+        makes every rank's shard consistent with the full tensor."""
+        rank = getattr(self, "_comm_info", {}).get("rank", 0)
         restore = self.ckpt_opts.get("restore_step")
         if restore is not None:
             want = {name: SliceMeta(tuple(g.nodes[nid].shape))
@@ -108,6 +113,10 @@ class TepdistService:
             if name in self.vars:
                 continue
             node = g.nodes[nid]
+            full_shape = list(node.shape)
+            dim, nsh = (param_specs or {}).get(name, (-1, 1))
+            if dim >= 0:
+                full_shape[dim] *= nsh
             if restore is not None and name in loaded:
                 t = loaded[name].to(dtype)
             elif name in self.host_data:   # client-transferred variable
@@ -115,7 +124,10 @@ class TepdistService:
             else:
                 spec = InitSpec(**init_specs.get(name, {})) \
                     if name in init_specs else _default_spec(name, node.shape)
-                t = init_shard(name, node.shape, spec, dtype=dtype)
+                t = init_shard(name, tuple(full_shape), spec, dtype=dtype)
+            if dim >= 0 and t.shape[dim] == full_shape[dim]:
+                sz = full_shape[dim] // nsh
+                t = t.narrow(dim, rank * sz, sz).contiguous()
             t = t.to(self.device).requires_grad_()
             self.vars[name] = _VarState(
                 t, t.detach().float().clone(),
@@ -204,27 +216,59 @@ class TepdistService:
         return {"ok": True}
 
     def dispatch_plan(self, req: dict) -> dict:
+        """Install a plan on this (slave) worker. When the plan carries
+        the planner's node_specs, the SpmdTransform rewrites the received
+        module to this worker's sharded executable (reshard collectives
+        run over the communicator from init_remote_comm) and variables are
+        initialized as shards — the reference's DispatchPlan +
+        sharded-variable init path."""
         g = getattr(self, "_pending_graph", None)
         if g is None:
             return {"ok": False, "error": "no module transferred"}
-        interp = GraphInterpreter(g, self.device)
+        plan = req.get("plan") or {}
+        param_specs = None
+        exec_graph = g
+        if plan.get("node_specs") and plan.get("nshards", 1) > 1:
+            from tepdist_amd.planner.dist_spec import DimStrategy
+            from tepdist_amd.planner.transform import SpmdTransform
+            specs = {int(k): DimStrategy(*v)
+                     for k, v in plan["node_specs"].items()}
+            res = SpmdTransform(g, specs, int(plan["nshards"])).run()
+            exec_graph = res.graph
+            param_specs = res.param_specs
+        import torch.distributed as dist
+        group = dist.group.WORLD if dist.is_initialized() else None
+        interp = GraphInterpreter(exec_graph, self.device, group=group)
         with self._lock:
             h = self._next_handle
             self._next_handle += 1
-            self.plans[h] = ExecutionPlan(h, g, req.get("plan"), None, interp)
-        self._init_variables(g, {})
+            self.plans[h] = ExecutionPlan(h, exec_graph, plan, None, interp)
+        self._init_variables(exec_graph, {}, param_specs=param_specs)
         return {"ok": True, "handle": h}
 
     def init_remote_comm(self, req: dict) -> dict:
         """Communicator bootstrap: the reference RPCs raw ncclUniqueIds
         (service_rt.cc:310-334); over RCCL we carry the rendezvous
-        (master addr/port + rank/world) for torch.distributed init."""
+        (master addr/port + rank/world) and join the process group here
+        (backend "nccl" = RCCL on GPU, gloo on CPU). Blocks until every
+        worker has joined — the coordinator fans the RPC out in
+        parallel."""
         import os
+
+        import torch.distributed as dist
         os.environ["MASTER_ADDR"] = req.get("master_addr", "127.0.0.1")
         os.environ["MASTER_PORT"] = str(req.get("master_port", 29500))
         self._comm_info = {"rank": req.get("rank", self.task_index),
                            "world": req.get("world", 1)}
-        return {"ok": True}
+        if req.get("join", False) and self._comm_info["world"] > 1 \
+                and not dist.is_initialized():
+            import datetime
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            dist.init_process_group(
+                backend, rank=self._comm_info["rank"],
+                world_size=self._comm_info["world"],
+                timeout=datetime.timedelta(seconds=120))
+        return {"ok": True, "rank": self._comm_info["rank"]}
 
     def execute_remote_plan(self, req: dict) -> dict:
         return self.execute_plan(req)
