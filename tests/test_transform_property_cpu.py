@@ -28,6 +28,9 @@ CASES = [
                   n_ctx=32), 2, 32),
     ("llama", dict(n_layer=2, n_embd=64, n_head=4, vocab_size=128,
                    n_ctx=64, ffn_mult=2), 4, 16),
+    # world-4 case (divisibility/spec-consistency at higher shard counts)
+    ("gpt2", dict(n_layer=1, n_embd=128, n_head=8, vocab_size=256,
+                  n_ctx=32), 8, 16),
 ]
 
 
@@ -93,5 +96,6 @@ def _worker(rank, world, port, case_idx):
 @pytest.mark.parametrize("case_idx", range(len(CASES)))
 @pytest.mark.timeout(600)
 def test_plan_transform_execute_matches_single(case_idx):
+    world = 4 if case_idx == len(CASES) - 1 else 2
     port = torch.randint(20000, 40000, (1,)).item()
-    mp.spawn(_worker, args=(2, port, case_idx), nprocs=2, join=True)
+    mp.spawn(_worker, args=(world, port, case_idx), nprocs=world, join=True)
