@@ -1,2 +1,2 @@
 from tepdist_amd.data.synthetic import (SyntheticImages, SyntheticTokens,
-                                        device_prefetcher)
+                                        device_prefetcher)  # noqa: F401

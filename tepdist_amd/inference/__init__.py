@@ -1,1 +1,1 @@
-from tepdist_amd.inference.engine import Generator
+from tepdist_amd.inference.engine import Generator  # noqa: F401

@@ -15,12 +15,10 @@ batch splits."""
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from tepdist_amd.ir.graph import Graph, Node
-from tepdist_amd.models.configs import GPT2Config, MoEConfig
+from tepdist_amd.models.configs import GPT2Config
 
 
 def gpt2_ir(cfg: GPT2Config, batch: int, seq: int) -> Graph:

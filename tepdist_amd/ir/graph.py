@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import json
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 # ops whose cost is matmul-like (the planner's "compute sensitive" set,
 # reference cost_spmd_strategy.h:42-49 IsComputeSensitive = dot/conv)

@@ -8,7 +8,7 @@ pass and the fused AdamW applying gradients (the AG role)."""
 
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

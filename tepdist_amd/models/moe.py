@@ -13,7 +13,6 @@ xla_sharding.replicate, moe_layers.py:296)."""
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -10,7 +10,6 @@ Every op the planner shards goes through here. Dispatch:
 
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

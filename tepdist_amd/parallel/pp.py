@@ -17,7 +17,6 @@ import torch
 import torch.distributed as dist
 
 from tepdist_amd.config import get_env
-from tepdist_amd.planner.cost_model import CostModel
 from tepdist_amd.runtime.scheduler import TaskScheduler
 from tepdist_amd.runtime.task_graph import TaskType, build_task_dag
 

@@ -10,9 +10,7 @@ vocab-parallel so the largest GEMM and its gradient stay sharded end-to-end.
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -11,7 +11,7 @@ then shard optimizer state identically to the parameter)."""
 
 from __future__ import annotations
 
-from typing import Dict, Iterable, List
+from typing import Dict, List
 
 from tepdist_amd.config import get_env
 from tepdist_amd.ir.graph import Graph

@@ -8,7 +8,7 @@ per device-mesh dim plus the pipeline stage."""
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field, replace
+from dataclasses import dataclass, field
 from typing import List, Optional
 
 GLUE = "glue"          # undecided (planner has not assigned yet)

@@ -8,7 +8,7 @@ the cost search (spmd.py) picks among them."""
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 from tepdist_amd.ir.graph import Graph, Node

@@ -21,11 +21,10 @@ group at execution time.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, Tuple
 
 from tepdist_amd.ir.graph import Graph, Node
-from tepdist_amd.planner.dist_spec import (DimStrategy, GLUE, PARTIAL,
-                                           REPLICATED, SPLIT)
+from tepdist_amd.planner.dist_spec import DimStrategy
 from tepdist_amd.planner.rules import back_infer
 
 

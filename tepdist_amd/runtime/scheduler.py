@@ -15,7 +15,7 @@ The resulting per-device ordered lists drive the executor."""
 from __future__ import annotations
 
 import heapq
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from tepdist_amd.config import get_env

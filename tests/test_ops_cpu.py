@@ -1,9 +1,7 @@
 """CPU tests of the op layer: autograd correctness of every planner-sharded
 op against plain torch (fp32) autograd."""
 
-import math
 
-import pytest
 import torch
 
 from tepdist_amd import ops

@@ -6,7 +6,6 @@ pure HLO->HLO transforms)."""
 import os
 
 import pytest
-import torch
 
 from tepdist_amd.config import ServiceEnv, set_env, get_env
 from tepdist_amd.ir import Graph, gpt2_ir

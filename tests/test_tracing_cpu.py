@@ -2,9 +2,7 @@
 JSON export; dag.dot dump."""
 
 import json
-import os
 
-import torch
 
 from tepdist_amd.runtime.task_graph import build_task_dag
 from tepdist_amd.utils.tracing import Tracer

@@ -8,7 +8,6 @@ node reuses over RCCL)."""
 
 import os
 
-import pytest
 import torch
 import torch.distributed as dist
 import torch.multiprocessing as mp

@@ -14,7 +14,6 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 import torch
 
 from tepdist_amd.ir.graph import Graph
-from tepdist_amd.ir.interpreter import GraphInterpreter
 from tepdist_amd.planner import AutoParallel
 from tepdist_amd.rpc.service import TepdistService
 
