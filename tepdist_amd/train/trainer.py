@@ -30,8 +30,9 @@ class Trainer:
         self.opt = optimizer
         self.grad_accum_steps = grad_accum_steps
         self.reducer = reducer
-        # hipGraph step capture: after two eager warm-up steps (allocator +
-        # MT-optimizer tables settle) the whole step — grad zero, every
+        # hipGraph step capture: after one eager warm-up step (gradients +
+        # MT-optimizer tables exist; graph allocations use their own pool)
+        # the whole step — grad zero, every
         # micro-batch forward+backward, the bucketed all-reduce, the fused
         # optimizer — is captured once and replayed per step, removing the
         # ~2.5k per-step kernel-launch round trips. Kill-switch
@@ -97,7 +98,7 @@ class Trainer:
                       f"loss={loss:.4f} (graph)", flush=True)
             return loss
         if (self.hip_graph and torch.cuda.is_available()
-                and self._eager_steps >= 2):
+                and self._eager_steps >= 1):
             batches = [batch_iter(i) for i in range(self.grad_accum_steps)]
             try:
                 self._try_capture(batches)
