@@ -5,6 +5,8 @@ import os
 
 import pytest
 import torch
+
+from tests.conftest import free_port
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
@@ -78,5 +80,5 @@ def _ep_worker(rank, world, port):
 
 @pytest.mark.timeout(300)
 def test_expert_parallel_matches_single():
-    port = torch.randint(20000, 40000, (1,)).item()
+    port = free_port()
     mp.spawn(_ep_worker, args=(2, port), nprocs=2, join=True)

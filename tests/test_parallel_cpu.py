@@ -8,6 +8,8 @@ import os
 
 import pytest
 import torch
+
+from tests.conftest import free_port
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
@@ -18,7 +20,7 @@ WORLD = 2
 
 
 def _run(fn, world=WORLD):
-    port = torch.randint(20000, 40000, (1,)).item()
+    port = free_port()
     mp.spawn(fn, args=(world, port), nprocs=world, join=True)
 
 

@@ -5,6 +5,8 @@ client), checkpoint RPCs, variable fetch."""
 import pytest
 import torch
 
+from tests.conftest import free_port
+
 from tepdist_amd.ir import gpt2_ir
 from tepdist_amd.models.configs import GPT2_CONFIGS
 from tepdist_amd.rpc.client import TepdistClient, TepdistSession
@@ -13,7 +15,7 @@ from tepdist_amd.rpc.server import serve
 
 @pytest.fixture(scope="module")
 def server(tmp_path_factory):
-    port = torch.randint(21000, 39000, (1,)).item()
+    port = free_port()
     ckpt = str(tmp_path_factory.mktemp("ckpt"))
     srv, svc = serve(port=port, block=False, ckpt_dir=ckpt)
     yield port, svc
@@ -74,7 +76,7 @@ def two_workers(tmp_path_factory):
     servers = []
     workers = []
     for i in range(2):
-        port = torch.randint(21000, 39000, (1,)).item() + i
+        port = free_port()
         ckpt = str(tmp_path_factory.mktemp(f"wck{i}"))
         srv, svc = serve(port=port, task_index=i, block=False, ckpt_dir=ckpt)
         servers.append(srv)
@@ -140,8 +142,8 @@ def test_dispatched_sharded_plan_matches_single(tmp_path):
                   for k, v in plan.node_specs.items()}
 
     ctx = pmp.get_context("spawn")
-    ports = [torch.randint(22000, 39000, (1,)).item() + i for i in range(2)]
-    gloo_port = torch.randint(22000, 39000, (1,)).item() + 7
+    ports = [free_port() for i in range(2)]
+    gloo_port = free_port()
     procs = [ctx.Process(target=_spmd_worker_server,
                          args=(p, str(tmp_path / f"w{i}")), daemon=True)
              for i, p in enumerate(ports)]

@@ -6,6 +6,8 @@ import os
 
 import pytest
 import torch
+
+from tests.conftest import free_port
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
@@ -131,5 +133,5 @@ def _pp_worker(rank, world, port):
 
 @pytest.mark.timeout(300)
 def test_pipeline_engine_matches_single():
-    port = torch.randint(20000, 40000, (1,)).item()
+    port = free_port()
     mp.spawn(_pp_worker, args=(2, port), nprocs=2, join=True)

@@ -9,6 +9,8 @@ node reuses over RCCL)."""
 import os
 
 import torch
+
+from tests.conftest import free_port
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
@@ -22,7 +24,7 @@ R, S, P = (DimStrategy.replicated, DimStrategy.split, DimStrategy.partial)
 
 
 def _run(fn, world=WORLD):
-    port = torch.randint(20000, 40000, (1,)).item()
+    port = free_port()
     mp.spawn(fn, args=(world, port), nprocs=world, join=True)
 
 

@@ -9,6 +9,8 @@ import os
 
 import pytest
 import torch
+
+from tests.conftest import free_port
 import torch.distributed as dist
 import torch.multiprocessing as mp
 
@@ -97,5 +99,5 @@ def _worker(rank, world, port, case_idx):
 @pytest.mark.timeout(600)
 def test_plan_transform_execute_matches_single(case_idx):
     world = 4 if case_idx == len(CASES) - 1 else 2
-    port = torch.randint(20000, 40000, (1,)).item()
+    port = free_port()
     mp.spawn(_worker, args=(world, port, case_idx), nprocs=world, join=True)
