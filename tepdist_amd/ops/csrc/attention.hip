@@ -446,8 +446,11 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
 // MFMA k-permutation invariance against transpose-staged dO^T / Q^T
 // A-operands — P and dS never touch LDS. dK/dV accumulate transposed in
 // registers across all q tiles and scatter once.
+constexpr int NTA = 512;   // 8 waves: one 16-kv slice each (KBA = 128)
+constexpr int KBA = 128;
+
 template <int D>
-__launch_bounds__(NT) __global__
+__launch_bounds__(NTA) __global__
 void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
                          const bf16_t* __restrict__ K,
                          const bf16_t* __restrict__ V,
@@ -463,7 +466,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   constexpr int QT = 64;
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int kv0 = blockIdx.x * KB;
+  const int kv0 = blockIdx.x * KBA;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
@@ -475,10 +478,10 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   const bf16_t* vp = V + qoff;
   const bf16_t* dop = dO + ooff;
 
-  __shared__ bf16_t smem[KB * D * 2 + QT * D * 2 + D * QT * 2];
+  __shared__ bf16_t smem[KBA * D * 2 + QT * D * 2 + D * QT * 2];
   bf16_t* sKb = smem;                       // [KB][D] natural
-  bf16_t* sVb = sKb + KB * D;               // [KB][D] natural
-  bf16_t* sQ = sVb + KB * D;                // [QT][D] natural
+  bf16_t* sVb = sKb + KBA * D;               // [KB][D] natural
+  bf16_t* sQ = sVb + KBA * D;                // [QT][D] natural
   bf16_t* sQT = sQ + QT * D;                // [D][QT]
   bf16_t* sdO = sQT + D * QT;               // [QT][D] natural
   bf16_t* sdOT = sdO + QT * D;              // [D][QT]
@@ -486,10 +489,10 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
 
   // stage K, V tiles (fixed for the block; natural layout only)
   {
-    constexpr int UN = KB * D / 8 / NT;
+    constexpr int UN = KBA * D / 8 / NTA;
 #pragma unroll
     for (int u = 0; u < UN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTA;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       bf16x8 kv8 = {}, vv8 = {};
@@ -519,17 +522,17 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   f32x4 acc_dvT[DF] = {};
 
   // issue-early / write-late staging registers for the Q and dO tiles
-  constexpr int QUN = QT * D / 8 / NT;
+  constexpr int QUN = QT * D / 8 / NTA;
   constexpr int TSLAB = (QT / 8) * (D / 2);   // v_perm transpose slabs
-  constexpr int TUN = (TSLAB + NT - 1) / NT;
+  constexpr int TUN = (TSLAB + NTA - 1) / NTA;
   bf16x8 qn[QUN], don[QUN];
   uint32_t qt[TUN][8], dot_[TUN][8];
-  float lse2[QT / NT + 1], dl2[QT / NT + 1];
+  float lse2[QT / NTA + 1], dl2[QT / NTA + 1];
 
   auto tile_load = [&](int q0) {
 #pragma unroll
     for (int u = 0; u < QUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTA;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       bf16x8 a = {}, b2 = {};
@@ -544,7 +547,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
     }
 #pragma unroll
     for (int u = 0; u < TUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTA;
       if (idx >= TSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -562,7 +565,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
         dot_[u][j] = __builtin_bit_cast(uint32_t, b2);
       }
     }
-    for (int i = threadIdx.x, s2 = 0; i < QT; i += NT, ++s2) {
+    for (int i = threadIdx.x, s2 = 0; i < QT; i += NTA, ++s2) {
       const int qg = q0 + i;
       lse2[s2] = (qg < S) ? LSE[(int64_t)bh * S + qg] : -3.0e38f;
       dl2[s2] = (qg < S) ? DELTA[(int64_t)bh * S + qg] : 0.f;
@@ -571,7 +574,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   auto tile_write = [&]() {
 #pragma unroll
     for (int u = 0; u < QUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTA;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       *reinterpret_cast<bf16x8*>(sQ + loff<D>(row, c)) = qn[u];
@@ -579,7 +582,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
     }
 #pragma unroll
     for (int u = 0; u < TUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTA;
       if (idx >= TSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -607,7 +610,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
       *reinterpret_cast<uint4*>(sdOT + loff<QT>(f + 1, kb * 8)) =
           make_uint4(o1[0], o1[1], o1[2], o1[3]);
     }
-    for (int i = threadIdx.x, s2 = 0; i < QT; i += NT, ++s2) {
+    for (int i = threadIdx.x, s2 = 0; i < QT; i += NTA, ++s2) {
       sLSE[i] = lse2[s2];
       sDELTA[i] = dl2[s2];
     }
@@ -641,7 +644,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
     // Inner tiles skip the per-element mask selects (wave-uniform).
     const int kvg = kv0 + wave * 16 + (lane & 15);
     const bool winner = (!causal || kv0 + wave * 16 + 15 < q0) &&
-                        q0 + QT <= S && kv0 + KB <= S;
+                        q0 + QT <= S && kv0 + KBA <= S;
     if (winner) {
       f32x4 lsev[4], dlv[4];
 #pragma unroll
@@ -745,8 +748,11 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
 // straight from the accumulators (k-permutation) against a
 // transpose-staged K^T — no LDS scatter, no atomics, bf16 output written
 // directly (strided; serves the packed-qkv layout too).
+constexpr int NTB = 512;   // 8 waves x 32 q columns (QBB = 256)
+constexpr int QBB = 256;
+
 template <int D>
-__launch_bounds__(NT) __global__
+__launch_bounds__(NTB) __global__
 void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                          const bf16_t* __restrict__ K,
                          const bf16_t* __restrict__ V,
@@ -761,7 +767,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   constexpr int DF = D / 16;
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QB;
+  const int q0 = blockIdx.x * QBB;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
@@ -805,16 +811,16 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   f32x4 acc_dq[DF][2] = {};   // dQ^T: d = 16df+4g+e, q = lane&15 (+16nf)
 
   // staging registers: K,V natural (16B each) + K^T v_perm slabs
-  constexpr int KUN = KB * D / 8 / NT;
+  constexpr int KUN = KB * D / 8 / NTB;
   constexpr int NSLAB = (KB / 8) * (D / 2);
-  constexpr int SUN = (NSLAB + NT - 1) / NT;
+  constexpr int SUN = (NSLAB + NTB - 1) / NTB;
   bf16x8 krg[KUN], vrg[KUN];
   uint32_t ktr[SUN][8];
 
   auto stage_load = [&](int kv0) {
 #pragma unroll
     for (int u = 0; u < KUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTB;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       bf16x8 kv8 = {}, vv8 = {};
@@ -829,7 +835,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
 #pragma unroll
     for (int u = 0; u < SUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTB;
       if (idx >= NSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -846,7 +852,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   auto stage_write = [&]() {
 #pragma unroll
     for (int u = 0; u < KUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTB;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = krg[u];
@@ -854,7 +860,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
 #pragma unroll
     for (int u = 0; u < SUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTB;
       if (idx >= NSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -873,7 +879,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
   };
 
-  const int kv_end = causal ? min(S, q0 + QB) : S;
+  const int kv_end = causal ? min(S, q0 + QBB) : S;
   stage_load(0);
   stage_write();
   __syncthreads();
@@ -1047,8 +1053,8 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        stream, static_cast<const bf16_t*>(dout),             \
                        static_cast<const bf16_t*>(o), delta, H, S, o_bs,     \
                        o_hs, o_rs);                                          \
-    dim3 kgrid((S + KB - 1) / KB, B * H);                                    \
-    hipLaunchKernelGGL(flash_bwd_kv_kernel<DD>, kgrid, block, 0, stream,     \
+    dim3 kgrid((S + 127) / 128, B * H);                                      \
+    hipLaunchKernelGGL(flash_bwd_kv_kernel<DD>, kgrid, dim3(512), 0, stream, \
                        static_cast<const bf16_t*>(q),                        \
                        static_cast<const bf16_t*>(k),                        \
                        static_cast<const bf16_t*>(v),                        \
@@ -1056,8 +1062,8 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),   \
                        S, H, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs,    \
                        o_rs);                                                \
-    dim3 qgrid((S + QB - 1) / QB, B * H);                                    \
-    hipLaunchKernelGGL(flash_bwd_dq_kernel<DD>, qgrid, block, 0, stream,     \
+    dim3 qgrid((S + QBB - 1) / QBB, B * H);                                  \
+    hipLaunchKernelGGL(flash_bwd_dq_kernel<DD>, qgrid, dim3(NTB), 0, stream, \
                        static_cast<const bf16_t*>(q),                        \
                        static_cast<const bf16_t*>(k),                        \
                        static_cast<const bf16_t*>(v),                        \
