@@ -88,8 +88,11 @@ DEV_INLINE float warp16_sum(float v) {
 // q = lane&15 — the same lane as the softmax state, so the alpha rescale is
 // a per-lane multiply) and is scattered once at the end. No P tile ever
 // touches LDS.
+constexpr int NTF = 512;   // 8 waves x 32 q columns (QBF = 256)
+constexpr int QBF = 256;
+
 template <int D>
-__launch_bounds__(NT) __global__
+__launch_bounds__(NTF) __global__
 void flash_fwd_kernel(const bf16_t* __restrict__ Q,
                       const bf16_t* __restrict__ K,
                       const bf16_t* __restrict__ V, bf16_t* __restrict__ O,
@@ -100,7 +103,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   constexpr int DF = D / 16;   // d row fragments of O^T
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QB;
+  const int q0 = blockIdx.x * QBF;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
@@ -138,16 +141,16 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   // staging registers (issue-early / write-late split):
   // K: KUN x 16B per thread; V: 8 x 4B per slab (v_perm transpose slabs;
   // (KB/8)*(D/2) slabs total — more than one per thread when D = 128)
-  constexpr int KUN = KB * D / 8 / NT;
+  constexpr int KUN = KB * D / 8 / NTF;
   constexpr int NSLAB = (KB / 8) * (D / 2);
-  constexpr int SUN = (NSLAB + NT - 1) / NT;
+  constexpr int SUN = (NSLAB + NTF - 1) / NTF;
   bf16x8 krg[KUN];
   uint32_t vrg[SUN][8];
 
   auto stage_load = [&](int kv0) {
 #pragma unroll
     for (int u = 0; u < KUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTF;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       bf16x8 v8 = {};
@@ -158,7 +161,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     }
 #pragma unroll
     for (int u = 0; u < SUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTF;
       if (idx >= NSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -177,14 +180,14 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     bf16_t* sVT = sK + KB * D;
 #pragma unroll
     for (int u = 0; u < KUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTF;
       const int row = idx / (D / 8);
       const int c = (idx % (D / 8)) * 8;
       *reinterpret_cast<bf16x8*>(sK + loff<D>(row, c)) = krg[u];
     }
 #pragma unroll
     for (int u = 0; u < SUN; ++u) {
-      const int idx = threadIdx.x + u * NT;
+      const int idx = threadIdx.x + u * NTF;
       if (idx >= NSLAB) break;
       const int f = 2 * (idx % (D / 2));
       const int kb = idx / (D / 2);
@@ -203,7 +206,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     }
   };
 
-  const int kv_end = causal ? min(S, q0 + QB) : S;
+  const int kv_end = causal ? min(S, q0 + QBF) : S;
   stage_load(0);
   stage_write(0);
   __syncthreads();
@@ -343,8 +346,8 @@ void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
                         bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
                         int64_t o_bs, int64_t o_hs, int64_t o_rs,
                         hipStream_t stream) {
-  dim3 grid((S + QB - 1) / QB, B * H);
-  dim3 block(NT);
+  dim3 grid((S + QBF - 1) / QBF, B * H);
+  dim3 block(NTF);
 #define FWD_D(DD)                                                       \
   hipLaunchKernelGGL(flash_fwd_kernel<DD>, grid, block, 0, stream,      \
                      static_cast<const bf16_t*>(q),                     \
