@@ -1,23 +1,25 @@
 // Fused flash-style causal attention (forward + backward) for gfx950.
-//
-// Forward: one block = 128 q-rows of one (batch, head); 4 waves x 32 rows.
-// Per KV tile of 64: K staged k-contiguous in LDS (ds_read_b128 B-fragments
-// like the GEMM), V staged transposed via the v_perm register transpose,
-// QK^T on mfma_f32_16x16x32_bf16 with Q fragments held in registers across
-// the whole row, online softmax (running max/sum) in fp32 registers with
-// 16-lane shfl_xor row reductions, P routed through a wave-private LDS tile
-// to become the PV A-operand. Saves per-row logsumexp for backward; the
-// S x S score matrix is never materialized in HBM.
-//
-// Backward (flash2-style): one block = 64 kv-rows of one (b,h); recomputes
-// P^T = exp(K Q^T * scale - lse) per q-tile (both operands natural-layout
-// MFMAs), accumulates dV += P^T dO and dK += dS^T Q in registers,
-// dS^T = P^T * (dP^T - delta_q) * scale with dP^T = V dO^T, and scatters
-// dQ partials with fp32 atomics (delta = rowsum(dO*O) precomputed).
-//
 // Replaces the composed scores-GEMM + softmax + PV path (SURVEY.md §2.9:
-// the reference leaves attention to XLA codegen; this is the MI355X-native
-// fused form).
+// the reference leaves attention to XLA codegen); docs/KERNELS.md has the
+// full design discussion.
+//
+// Forward: one block = 256 q-rows, 8 waves x 32 q columns sharing the
+// double-buffered K/V staging. Scores are computed TRANSPOSED
+// (S^T = mfma(K, Q)) so softmax state is per-lane (two shfl_xor per
+// reduction) and P^T feeds the PV MFMA straight from the accumulators via
+// the k-permutation invariance; O accumulates as O^T and scatters once.
+// Saves per-row logsumexp; the S x S score matrix never touches HBM.
+//
+// Backward: TWO atomics-free kernels (the classic fused flash2 backward
+// spends most of its time on dQ atomicAdd traffic — measured with the
+// section probe, see profiles/):
+//   - dK/dV kernel: block = 128 kv rows (8 waves x 16-kv slices sharing
+//     the Q/dO tile staging); S = mfma(Q, K) untransposed puts kv in the
+//     lane index so dV^T/dK^T consume P/dS from the accumulators
+//     (k-permuted) against v_perm-transposed dO^T/Q^T images.
+//   - dQ kernel: forward-shaped, dQ^T accumulated in registers, bf16
+//     output written directly in the packed-qkv strided layout.
+// delta = rowsum(dO*O) is the flash2 preprocess.
 
 #include <algorithm>
 #include <stdexcept>
