@@ -117,7 +117,35 @@ class AutoParallel:
                     best = cand
             plan = best
         plan.search_time_s = time.time() - t0
+        self._dump(plan)
         return plan
+
+    def _dump(self, plan) -> None:
+        """Debug artifacts when TEPDIST_DUMP_DIR is set (the reference's
+        DumpStrategies + sketch_raw.dot + dag.dot surface,
+        SURVEY.md §5.1)."""
+        d = self.env.dump_dir
+        if not d:
+            return
+        import json
+        import os
+        os.makedirs(d, exist_ok=True)
+        specs = {}
+        for nid, spec in (plan.node_specs or {}).items():
+            specs[str(nid)] = str(spec)
+        with open(os.path.join(d, "strategies.json"), "w") as f:
+            json.dump({"summary": plan.summary(), "mode": plan.mode,
+                       "dp": plan.dp, "tp": plan.tp, "pp": plan.pp,
+                       "micro_batches": plan.micro_batches,
+                       "search_time_s": plan.search_time_s,
+                       "node_specs": specs}, f, indent=1)
+        try:
+            from tepdist_amd.runtime.task_graph import build_task_dag
+            dag = build_task_dag(plan.pp, plan.micro_batches,
+                                 dp_degree=plan.dp)
+            dag.dump_dot(os.path.join(d, "dag.dot"))
+        except Exception:
+            pass
 
     def _single_device_plan(self) -> ParallelPlan:
         specs = {i: DistSpec([DimStrategy.replicated(1)])

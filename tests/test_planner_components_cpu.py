@@ -106,3 +106,24 @@ def test_resolve_forward_backward_apply():
     assert y.id in fwd and loss.id in fwd
     assert gw.id in bwd and y.id not in bwd
     assert resolve_gradients(g) == {"w": gw.id}
+
+
+def test_auto_parallel_dump_artifacts(tmp_path, monkeypatch):
+    import json
+
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2_CONFIGS
+    from tepdist_amd.planner.auto_parallel import AutoParallel
+    import tepdist_amd.config as cfgmod
+
+    monkeypatch.setenv("TEPDIST_DUMP_DIR", str(tmp_path))
+    monkeypatch.setattr(cfgmod, "_GLOBAL_ENV", None)  # drop the env cache
+    g = gpt2_ir(GPT2_CONFIGS["gpt2-test"], batch=4, seq=16)
+    try:
+        AutoParallel(g, 2).run()
+    finally:
+        cfgmod._GLOBAL_ENV = None
+    d = json.load(open(tmp_path / "strategies.json"))
+    assert d["dp"] * d["tp"] * d["pp"] == 2
+    assert d["node_specs"]
+    assert (tmp_path / "dag.dot").exists()
