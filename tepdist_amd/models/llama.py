@@ -13,6 +13,10 @@ import torch
 import torch.nn as nn
 
 from tepdist_amd import ops
+from tepdist_amd.parallel.tp import (ColumnParallelLinear, ParallelEnv,
+                                     RowParallelLinear,
+                                     VocabParallelEmbedding,
+                                     vocab_parallel_cross_entropy)
 
 
 @dataclass
@@ -42,62 +46,109 @@ def _param(*shape, std=0.02, dtype=torch.float32):
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, cfg: LlamaConfig, dtype=torch.float32):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.float32, env=None):
         super().__init__()
+        env = env or ParallelEnv.single()
         d = cfg.n_embd
         h = cfg.ffn_mult * d
+        tp = env.tp_size
+        assert cfg.n_head % tp == 0
         self.cfg = cfg
+        self.env = env
+        self.n_head_local = cfg.n_head // tp
         self.head_dim = d // cfg.n_head
         self.ln1_g = nn.Parameter(torch.ones(d, dtype=dtype))
         self.ln2_g = nn.Parameter(torch.ones(d, dtype=dtype))
-        self.w_qkv = _param(3 * d, d, dtype=dtype)
-        self.w_o = _param(d, d, dtype=dtype)
-        self.w_gate = _param(h, d, dtype=dtype)
-        self.w_up = _param(h, d, dtype=dtype)
-        self.w_down = _param(d, h, dtype=dtype)
+        if tp == 1:
+            self.w_qkv = _param(3 * d, d, dtype=dtype)
+            self.w_o = _param(d, d, dtype=dtype)
+            self.w_gate = _param(h, d, dtype=dtype)
+            self.w_up = _param(h, d, dtype=dtype)
+            self.w_down = _param(d, h, dtype=dtype)
+        else:
+            # Megatron layout: qkv/gate/up column-parallel, o/down
+            # row-parallel (same shard convention as the GPT-2 block)
+            self.qkv = ColumnParallelLinear(d, 3 * d, env, bias=False,
+                                            dtype=dtype)
+            self.o = RowParallelLinear(d, d, env, bias=False, dtype=dtype)
+            self.gate = ColumnParallelLinear(d, h, env, bias=False,
+                                             dtype=dtype)
+            self.up = ColumnParallelLinear(d, h, env, bias=False,
+                                           dtype=dtype)
+            self.down = RowParallelLinear(h, d, env, bias=False, dtype=dtype)
 
-    def forward(self, x, seq_len: int):
+    def _attn(self, qkv, seq_len: int):
         cfg = self.cfg
-        d = cfg.n_embd
-        h = ops.rmsnorm(x, self.ln1_g, cfg.rms_eps)
-        qkv = ops.linear(h, self.w_qkv)               # [T, 3d]
         T = qkv.shape[0]
-        nh, hd = cfg.n_head, self.head_dim
-        q, k, v = qkv.split(d, dim=-1)
-        # rotary on q/k per head
+        nh, hd = self.n_head_local, self.head_dim
+        dloc = nh * hd
+        q, k, v = qkv.split(dloc, dim=-1)
         q = ops.rope(q.reshape(T, nh, hd), seq_len, cfg.rope_theta)
         k = ops.rope(k.reshape(T, nh, hd), seq_len, cfg.rope_theta)
         b = T // seq_len
+
         def heads(t):
             return t.reshape(b, seq_len, nh, hd).transpose(1, 2).contiguous()
         o = ops.attention(heads(q), heads(k),
                           heads(v.reshape(T, nh, hd)), causal=True)
-        o = o.transpose(1, 2).reshape(T, d).contiguous()
-        x = x + ops.linear(o, self.w_o)
-        hn = ops.rmsnorm(x, self.ln2_g, cfg.rms_eps)
-        gate = ops.linear(hn, self.w_gate)
-        up = ops.linear(hn, self.w_up)
-        x = x + ops.linear(ops.swiglu(gate, up), self.w_down)
+        return o.transpose(1, 2).reshape(T, dloc).contiguous()
+
+    def forward(self, x, seq_len: int):
+        cfg = self.cfg
+        hnorm = ops.rmsnorm(x, self.ln1_g, cfg.rms_eps)
+        if self.env.tp_size == 1:
+            o = self._attn(ops.linear(hnorm, self.w_qkv), seq_len)
+            x = x + ops.linear(o, self.w_o)
+            hn = ops.rmsnorm(x, self.ln2_g, cfg.rms_eps)
+            gate = ops.linear(hn, self.w_gate)
+            up = ops.linear(hn, self.w_up)
+            x = x + ops.linear(ops.swiglu(gate, up), self.w_down)
+        else:
+            o = self._attn(self.qkv(hnorm), seq_len)
+            x = x + self.o(o)
+            hn = ops.rmsnorm(x, self.ln2_g, cfg.rms_eps)
+            x = x + self.down(ops.swiglu(self.gate(hn), self.up(hn)))
         return x
 
 
 class Llama(nn.Module):
-    def __init__(self, cfg: LlamaConfig, dtype=torch.float32):
+    def __init__(self, cfg: LlamaConfig, dtype=torch.float32, env=None):
         super().__init__()
         self.cfg = cfg
-        self.wte = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
+        self.env = env or ParallelEnv.single()
+        if self.env.tp_size == 1:
+            self.wte = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
+            self.lm_head = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
+        else:
+            self.wte_mod = VocabParallelEmbedding(cfg.vocab_size,
+                                                  cfg.n_embd, self.env,
+                                                  dtype=dtype)
+            self.head = ColumnParallelLinear(cfg.n_embd, cfg.vocab_size,
+                                             self.env, bias=False,
+                                             dtype=dtype)
         self.blocks = nn.ModuleList(
-            LlamaBlock(cfg, dtype) for _ in range(cfg.n_layer))
+            LlamaBlock(cfg, dtype, self.env) for _ in range(cfg.n_layer))
         self.ln_f_g = nn.Parameter(torch.ones(cfg.n_embd, dtype=dtype))
-        self.lm_head = _param(cfg.vocab_size, cfg.n_embd, dtype=dtype)
 
     def forward(self, ids, labels=None):
         b, s = ids.shape
-        x = ops.embedding(ids.reshape(-1), self.wte)
+        if self.env.tp_size == 1:
+            x = ops.embedding(ids.reshape(-1), self.wte)
+        else:
+            x = self.wte_mod(ids.reshape(-1))
         for blk in self.blocks:
             x = blk(x, s)
         x = ops.rmsnorm(x, self.ln_f_g, self.cfg.rms_eps)
-        logits = ops.linear(x, self.lm_head)
+        if self.env.tp_size == 1:
+            logits = ops.linear(x, self.lm_head)
+            if labels is None:
+                return logits
+            return ops.cross_entropy(logits, labels.reshape(-1))
+        logits_local = self.head(x)
         if labels is None:
-            return logits
-        return ops.cross_entropy(logits, labels.reshape(-1))
+            return logits_local
+        emb = self.env
+        vs = self.head.out_local * emb.tp_rank
+        return vocab_parallel_cross_entropy(
+            logits_local, labels.reshape(-1), vs, self.head.out_local,
+            emb.tp_group)
