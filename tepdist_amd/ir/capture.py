@@ -177,8 +177,6 @@ def from_fx(model: torch.nn.Module, *example_args) -> Graph:
         if n.op == "placeholder":
             env[n] = g.add_input(n.name, shape, dtype)
         elif n.op == "get_attr":
-            t = traced.get_parameter(n.target) if "." not in n.target or \
-                True else None
             try:
                 t = traced.get_parameter(n.target)
             except AttributeError:

@@ -344,3 +344,4 @@ class SwiGLUFn(torch.autograd.Function):
 
 def swiglu(a, b):
     return SwiGLUFn.apply(a, b)
+
