@@ -62,6 +62,8 @@ class PipelineEngine:
         self.order = order or make_1f1b_order(num_stages, micro_batches,
                                               stage)
         self.async_send = get_env().async_send
+        self.async_recv = get_env().async_recv
+        self._posted = None   # (key, work, tensor): one pre-posted irecv
 
     # -- p2p ----------------------------------------------------------------
 
@@ -72,7 +74,32 @@ class PipelineEngine:
         else:
             dist.send(t, self.ranks[to_stage])
 
-    def _recv(self, from_stage: int) -> torch.Tensor:
+    def _recv_key(self, entry):
+        """(from_stage,) the entry will receive from, or None."""
+        kind, m = entry
+        if kind == "fw" and self.stage > 0:
+            return ("fw", m, self.stage - 1)
+        if kind == "bw" and self.stage < self.S - 1:
+            return ("bw", m, self.stage + 1)
+        return None
+
+    def _prepost(self, entry):
+        """ASYNC_RECV: post the next entry's irecv so the transfer runs
+        under the current entry's compute (the reference's dedicated recv
+        stream; kill-switch falls back to blocking recv)."""
+        key = self._recv_key(entry)
+        if key is None or self._posted is not None:
+            return
+        t = torch.empty(self.act_shape, dtype=self.act_dtype,
+                        device=self.device)
+        self._posted = (key, dist.irecv(t, self.ranks[key[2]]), t)
+
+    def _recv(self, from_stage: int, key=None) -> torch.Tensor:
+        if self._posted is not None and self._posted[0] == key:
+            _, work, t = self._posted
+            self._posted = None
+            work.wait()
+            return t
         t = torch.empty(self.act_shape, dtype=self.act_dtype,
                         device=self.device)
         dist.recv(t, self.ranks[from_stage])
@@ -94,13 +121,14 @@ class PipelineEngine:
         if self.reducer is not None:
             self.reducer.reset()
         bw_done = 0
-        for kind, m in self.order:
+        for oi, (kind, m) in enumerate(self.order):
             if kind == "fw":
                 inputs, labels = batch_iter(m)
                 if is_first:
                     x = inputs
                 else:
-                    x = self._recv(self.stage - 1).requires_grad_()
+                    x = self._recv(self.stage - 1,
+                                   ("fw", m, self.stage - 1)).requires_grad_()
                 fw_in[m] = x
                 if is_last:
                     loss = self.mod(x, labels=labels)
@@ -117,12 +145,14 @@ class PipelineEngine:
                 if is_last:
                     (fw_out.pop(m) / self.M).backward()
                 else:
-                    grad = self._recv(self.stage + 1)
+                    grad = self._recv(self.stage + 1, ("bw", m, self.stage + 1))
                     fw_out.pop(m).backward(grad)
                 if not is_first:
                     g = fw_in[m].grad
                     self._send(g, self.stage - 1, pending)
                 fw_in.pop(m)
+            if self.async_recv and oi + 1 < len(self.order):
+                self._prepost(self.order[oi + 1])
         for w, _ in pending:
             w.wait()
         if self.reducer is not None:
