@@ -108,14 +108,31 @@ class AutoParallel:
             plan.mode = "config"
         else:
             best = None
+
+            def _key(p):
+                # feasible plans by duration; infeasible ones by memory so
+                # the caller at least gets the least-oversubscribed layout
+                import math
+                inf = not math.isfinite(p.cost.total_duration)
+                return (inf, p.cost.mem_bytes if inf
+                        else p.cost.total_duration)
+
             for (s, mesh) in _pow2_proposals(self.devs):
                 cand = self._best_over_rounds(s, mesh)
                 if cand is None:
                     continue
-                if best is None or cand.cost.total_duration < \
-                        best.cost.total_duration:
+                if best is None or _key(cand) < _key(best):
                     best = cand
             plan = best
+            import math
+            if plan is not None and not math.isfinite(
+                    plan.cost.total_duration):
+                print(f"[tepdist] WARNING: no plan fits "
+                      f"{self.cm.hw.hbm_bytes >> 30} GiB/device; returning "
+                      f"the minimum-memory layout "
+                      f"(dp{plan.dp} tp{plan.tp} pp{plan.pp}, "
+                      f"{plan.cost.mem_bytes / (1 << 30):.0f} GiB/device)",
+                      flush=True)
         plan.search_time_s = time.time() - t0
         self._dump(plan)
         return plan
@@ -160,13 +177,18 @@ class AutoParallel:
                             cost=cost, def_tree=tree, micro_batches=micro)
 
     def _best_over_rounds(self, stages: int, mesh: int):
+        import math
         best = None
+
+        def _key(p):
+            inf = not math.isfinite(p.cost.total_duration)
+            return (inf, p.cost.mem_bytes if inf else p.cost.total_duration)
+
         for rounds in _mesh_rounds(mesh):
             cand = self._plan_proposal(stages, rounds)
             if cand is None:
                 continue
-            if best is None or cand.cost.total_duration < \
-                    best.cost.total_duration:
+            if best is None or _key(cand) < _key(best):
                 best = cand
         return best
 
@@ -183,7 +205,16 @@ class AutoParallel:
                 valid = [m for m in sf.micro_batches if m <= micro]
                 micro = valid[-1] if valid else 1
 
-        # per-mesh-dim SPMD rounds
+        # per-mesh-dim SPMD rounds. Memory pressure (full optimizer state
+        # would overflow HBM even ZeRO-sharded over this mesh) turns on the
+        # replicated-parameter penalty so rounds prefer weight sharding —
+        # the reference's SplitPlanByMemCost bias.
+        param_bytes = sum(g.bytes_of(g.nodes[i]) for i in g.params.values())
+        mesh = 1
+        for n in rounds:
+            mesh *= n
+        pressure = param_bytes * 8.0 / max(mesh, 1) > \
+            0.6 * self.cm.hw.hbm_bytes
         node_specs: Dict[int, DistSpec] = {
             i: DistSpec([], 0) for i in g.nodes}
         spmd_cost = 0.0
@@ -191,7 +222,8 @@ class AutoParallel:
         for ri, n in enumerate(rounds):
             planner = CostSpmdStrategy(
                 g, n, self.cm,
-                time_limit_s=0.0 if rule else self.env.ilp_time_limit_s)
+                time_limit_s=0.0 if rule else self.env.ilp_time_limit_s,
+                param_mem_penalty=1e-9 if pressure else 0.0)
             res = planner.run()
             spmd_cost += res.cost
             # classify the round: dp if most compute-sensitive flops chose a

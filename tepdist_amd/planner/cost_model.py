@@ -47,6 +47,7 @@ class Cost:
     gpu_efficiency: float = 0.0
     coll_ratio: float = 0.0
     bubble_ratio: float = 0.0
+    mem_bytes: float = 0.0            # peak per-device state+activation
 
     def __lt__(self, other):
         return self.total_duration < other.total_duration
@@ -130,4 +131,4 @@ class Evaluator:
         eff = compute_s / total if total > 0 and total != float("inf") else 0
         return Cost(total_duration=total, gpu_efficiency=eff,
                     coll_ratio=coll_s / max(work, 1e-12),
-                    bubble_ratio=bubble)
+                    bubble_ratio=bubble, mem_bytes=mem_bytes_per_dev)

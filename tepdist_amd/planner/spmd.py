@@ -51,12 +51,18 @@ class SpmdResult:
 
 class CostSpmdStrategy:
     def __init__(self, graph: Graph, nshards: int, cm: CostModel = None,
-                 time_limit_s: float = None):
+                 time_limit_s: float = None,
+                 param_mem_penalty: float = 0.0):
         self.g = graph
         self.n = nshards
         self.cm = cm or CostModel()
         self.time_limit = time_limit_s if time_limit_s is not None \
             else get_env().ilp_time_limit_s
+        # SplitPlanByMemCost's role (reference cost_spmd_strategy.cc:1487):
+        # under memory pressure, an UNsplit parameter pays this many
+        # seconds per byte of replicated weight, steering the ILP toward
+        # weight-sharded (tensor-parallel) strategies
+        self.param_mem_penalty = param_mem_penalty
         self.cons = graph.consumers()
 
     # ---------------------------------------------------------------------
@@ -212,7 +218,11 @@ class CostSpmdStrategy:
 
     def _node_cost(self, n: Node, spec: DimStrategy) -> float:
         shards = self.n if (spec.is_split or spec.is_partial) else 1
-        return self.cm.compute_time(self.g, n, shards)
+        c = self.cm.compute_time(self.g, n, shards)
+        if self.param_mem_penalty > 0 and n.op == "param" \
+                and not spec.is_split:
+            c += self.param_mem_penalty * self.g.bytes_of(n)
+        return c
 
     # ---------------------------------------------------------------------
 

@@ -149,3 +149,19 @@ def test_def_tree_structure():
     s = tree.to_json()
     t2 = type(tree).from_json(s)
     assert len(t2.contexts) == len(tree.contexts)
+
+
+def test_memory_pressure_prefers_weight_sharding():
+    """A model whose optimizer state cannot fit replicated must come back
+    weight-sharded (the SplitPlanByMemCost bias) with an explicit
+    min-memory fallback when nothing fits."""
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2_CONFIGS
+    from tepdist_amd.planner.auto_parallel import AutoParallel
+
+    cfg = GPT2_CONFIGS["gpt2-175b"]
+    g = gpt2_ir(cfg, batch=64, seq=64)
+    plan = AutoParallel(g, 8).run()
+    # dp-only would replicate 2.8 TB of state per device; the planner must
+    # shift the mesh toward tensor sharding
+    assert plan.tp > 1, plan.summary()
