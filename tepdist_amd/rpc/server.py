@@ -69,6 +69,9 @@ def serve(port: int = 2222, task_index: int = 0, block: bool = True,
     server.add_generic_rpc_handlers((_Handler(svc),))
     server.add_insecure_port(f"0.0.0.0:{port}")
     server.start()
+    from tepdist_amd.config import get_env
+    if get_env().debug:
+        print(get_env().dump(), flush=True)   # PrintAllEnvs at startup
     print(f"[tepdist] server listening on :{port} (task {task_index})",
           flush=True)
     if block:
