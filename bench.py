@@ -100,7 +100,12 @@ def main():
     micro_size = max(local_batch // grad_accum, 1)
 
     torch.manual_seed(1234)
-    g = torch.Generator().manual_seed(4321 + rank)
+    # Seed the synthetic-data stream by DATA-PARALLEL rank only:
+    # every rank inside a TP group and every stage of a PP chain must
+    # consume the identical batch stream — tp groups are consecutive ranks
+    # (dp_rank = rank // tp), pp layout is stage-major (dp_rank = rank % dp).
+    data_rank = (rank % dp) if pp > 1 else (rank // tp)
+    g = torch.Generator().manual_seed(4321 + data_rank)
 
     def make_ids(n_rows):
         return torch.randint(0, cfg.vocab_size, (n_rows, seq + 1),

@@ -31,6 +31,21 @@ from tepdist_amd.parallel.tp import (
     VocabParallelEmbedding,
     vocab_parallel_cross_entropy,
 )
+from tepdist_amd.runtime.initializers import InitSpec, init_shard
+
+
+def _draw(name: str, full_shape, std: float, dtype, seed: int,
+          shard_dim: int = -1, shard_index: int = 0, num_shards: int = 1):
+    """Shard-aware weight draw: rank r's shard is bit-identical to slicing
+    the full tensor (counter-based global-index RNG, runtime/initializers).
+    Under tensor parallelism each rank therefore gets a DIFFERENT slice of
+    the same global tensor — heads/neurons are not duplicated across the
+    TP group (identical per-rank seeding silently shrank the
+    effective width by 1/tp)."""
+    return init_shard(name, tuple(full_shape),
+                      InitSpec("random_normal", std=std), global_seed=seed,
+                      shard_dim=shard_dim, shard_index=shard_index,
+                      num_shards=num_shards, dtype=dtype)
 
 
 class GPT2Block(nn.Module):
@@ -112,16 +127,35 @@ class GPT2(nn.Module):
 
     @torch.no_grad()
     def reset_parameters(self, seed: int = 1234):
-        g = torch.Generator().manual_seed(seed)
+        cfg, env = self.cfg, self.env
+        tp, r = env.tp_size, env.tp_rank
+        V, d, H = cfg.padded_vocab, cfg.n_embd, cfg.n_head
+        hd = d // H
         std = 0.02
-        proj_std = std / math.sqrt(2 * self.cfg.n_layer)
-        for name, p in self.named_parameters():
-            if p.dim() == 2:
-                s = proj_std if ("w_proj" in name or "w_out" in name or
-                                 "proj.weight" in name or
-                                 "out.weight" in name) else std
-                p.copy_(torch.randn(p.shape, generator=g) * s)
-        if self.env.tp_size == 1:
+        proj_std = std / math.sqrt(2 * cfg.n_layer)
+        dt = self.wpe.dtype
+        if tp == 1:
+            self.wte.copy_(_draw("wte", (V, d), std, dt, seed))
+        else:
+            self.wte_mod.weight.copy_(_draw("wte", (V, d), std, dt, seed,
+                                            0, r, tp))
+        self.wpe.copy_(_draw("wpe", (cfg.n_ctx, d), std, dt, seed))
+        for i, blk in enumerate(self.blocks):
+            # qkv full tensor is [3d, d] == [3, H, hd*d] flattened; the TP
+            # shard takes this rank's heads (shard_qkv_weight layout)
+            qkv = _draw(f"h{i}.w_qkv", (3, H, hd * d), std, dt, seed,
+                        1, r, tp).reshape(3 * (H // tp) * hd, d)
+            proj = _draw(f"h{i}.w_proj", (d, d), proj_std, dt, seed, 1, r, tp)
+            fc = _draw(f"h{i}.w_fc", (4 * d, d), std, dt, seed, 0, r, tp)
+            out = _draw(f"h{i}.w_out", (d, 4 * d), proj_std, dt, seed,
+                        1, r, tp)
+            if tp == 1:
+                blk.w_qkv.copy_(qkv); blk.w_proj.copy_(proj)
+                blk.w_fc.copy_(fc); blk.w_out.copy_(out)
+            else:
+                blk.qkv.weight.copy_(qkv); blk.proj.weight.copy_(proj)
+                blk.fc.weight.copy_(fc); blk.out.weight.copy_(out)
+        if tp == 1:
             # zero the padded vocab rows so they never win the softmax
             self.wte[self.cfg.vocab_size:].zero_()
         else:
@@ -195,6 +229,7 @@ class GPT2Stage(nn.Module):
         self.cfg = cfg
         self.env = env or ParallelEnv.single()
         self.is_first, self.is_last = is_first, is_last
+        self.layer_start = layer_start
         V, d = cfg.padded_vocab, cfg.n_embd
         if is_first:
             self.wte = nn.Parameter(torch.empty(V, d, dtype=dtype))
@@ -210,18 +245,40 @@ class GPT2Stage(nn.Module):
 
     @torch.no_grad()
     def reset_parameters(self, seed: int = 1234):
-        g = torch.Generator().manual_seed(seed)
+        """Blocks are named by GLOBAL layer index, so a stage's weights are
+        bit-identical to the corresponding layers of the unsplit model
+        (shard-aware under TP, same counter RNG as GPT2)."""
+        cfg, env = self.cfg, self.env
+        tp, r = env.tp_size, env.tp_rank
+        V, d, H = cfg.padded_vocab, cfg.n_embd, cfg.n_head
+        hd = d // H
         std = 0.02
-        proj_std = std / math.sqrt(2 * self.cfg.n_layer)
-        for name, p in self.named_parameters():
-            if p.dim() == 2:
-                s = proj_std if ("w_proj" in name or "w_out" in name or
-                                 "proj.weight" in name or
-                                 "out.weight" in name) else std
-                p.copy_(torch.randn(p.shape, generator=g) * s)
+        proj_std = std / math.sqrt(2 * cfg.n_layer)
+        dt = next(self.parameters()).dtype
+        if self.is_first:
+            self.wte.copy_(_draw("wte", (V, d), std, dt, seed))
+            self.wpe.copy_(_draw("wpe", (cfg.n_ctx, d), std, dt, seed))
+        for li, blk in enumerate(self.blocks):
+            i = self.layer_start + li
+            qkv = _draw(f"h{i}.w_qkv", (3, H, hd * d), std, dt, seed,
+                        1, r, tp).reshape(3 * (H // tp) * hd, d)
+            proj = _draw(f"h{i}.w_proj", (d, d), proj_std, dt, seed, 1, r, tp)
+            fc = _draw(f"h{i}.w_fc", (4 * d, d), std, dt, seed, 0, r, tp)
+            out = _draw(f"h{i}.w_out", (d, 4 * d), proj_std, dt, seed,
+                        1, r, tp)
+            if tp == 1:
+                blk.w_qkv.copy_(qkv); blk.w_proj.copy_(proj)
+                blk.w_fc.copy_(fc); blk.w_out.copy_(out)
+            else:
+                blk.qkv.weight.copy_(qkv); blk.proj.weight.copy_(proj)
+                blk.fc.weight.copy_(fc); blk.out.weight.copy_(out)
         if self.is_first:
             self.wte[self.cfg.vocab_size:].zero_()
         if self.is_last:
+            # untied head (tied form only exists when wte is on this stage):
+            # initialized from the SAME stream as wte so a pipeline split of
+            # the tied model starts from the tied value
+            self.lm_head.copy_(_draw("wte", (V, d), std, dt, seed))
             self.lm_head[self.cfg.vocab_size:].zero_()
 
     def forward(self, x, labels=None):

@@ -129,6 +129,46 @@ class Llama(nn.Module):
         self.blocks = nn.ModuleList(
             LlamaBlock(cfg, dtype, self.env) for _ in range(cfg.n_layer))
         self.ln_f_g = nn.Parameter(torch.ones(cfg.n_embd, dtype=dtype))
+        self.reset_parameters()
+
+    @torch.no_grad()
+    def reset_parameters(self, seed: int = 1234):
+        """Shard-aware init (tp>1 builds left ColumnParallel /
+        RowParallel weights uninitialized): every weight is drawn from the
+        counter-based global-index RNG (runtime/initializers) so each TP
+        rank holds exactly its slice of the same global tensor."""
+        from tepdist_amd.models.gpt2 import _draw
+        cfg, env = self.cfg, self.env
+        tp, r = env.tp_size, env.tp_rank
+        d, H = cfg.n_embd, cfg.n_head
+        hd = d // H
+        h = cfg.ffn_mult * d
+        V = cfg.vocab_size
+        std = 0.02
+        dt = self.ln_f_g.dtype
+        if tp == 1:
+            self.wte.copy_(_draw("wte", (V, d), std, dt, seed))
+            self.lm_head.copy_(_draw("lm_head", (V, d), std, dt, seed))
+        else:
+            self.wte_mod.weight.copy_(
+                _draw("wte", (V, d), std, dt, seed, 0, r, tp))
+            self.head.weight.copy_(
+                _draw("lm_head", (V, d), std, dt, seed, 0, r, tp))
+        for i, blk in enumerate(self.blocks):
+            qkv = _draw(f"h{i}.w_qkv", (3, H, hd * d), std, dt, seed,
+                        1, r, tp).reshape(3 * (H // tp) * hd, d)
+            o = _draw(f"h{i}.w_o", (d, d), std, dt, seed, 1, r, tp)
+            gate = _draw(f"h{i}.w_gate", (h, d), std, dt, seed, 0, r, tp)
+            up = _draw(f"h{i}.w_up", (h, d), std, dt, seed, 0, r, tp)
+            down = _draw(f"h{i}.w_down", (d, h), std, dt, seed, 1, r, tp)
+            if tp == 1:
+                blk.w_qkv.copy_(qkv); blk.w_o.copy_(o)
+                blk.w_gate.copy_(gate); blk.w_up.copy_(up)
+                blk.w_down.copy_(down)
+            else:
+                blk.qkv.weight.copy_(qkv); blk.o.weight.copy_(o)
+                blk.gate.weight.copy_(gate); blk.up.weight.copy_(up)
+                blk.down.weight.copy_(down)
 
     def forward(self, ids, labels=None):
         b, s = ids.shape

@@ -171,9 +171,12 @@ class TepdistService:
             for n, vs in self.vars.items():
                 if vs.tensor.grad is None:
                     continue
+                # norm gains / biases (1-D) are not weight-decayed,
+                # mirroring AdamW.no_decay_1d
+                wd = 0.0 if vs.tensor.dim() <= 1 else 0.01
                 _ops.adamw_step(vs.tensor.data, vs.master, vs.tensor.grad,
                                 vs.exp_avg, vs.exp_avg_sq, lr=self.lr,
-                                step=self.step_count)
+                                weight_decay=wd, step=self.step_count)
                 vs.tensor.grad = None
             dur_ms = (time.time() - t0) * 1e3
             if env.debug:

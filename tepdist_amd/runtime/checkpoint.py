@@ -116,6 +116,10 @@ class CheckpointManager:
             if sm.shard_dim < 0 or sm.num_shards == 1:
                 out[name] = full
             else:
+                if full.shape[sm.shard_dim] % sm.num_shards != 0:
+                    raise ValueError(
+                        f"{name}: dim {sm.shard_dim} of {tuple(full.shape)} "
+                        f"not divisible into {sm.num_shards} shards")
                 n = full.shape[sm.shard_dim] // sm.num_shards
                 out[name] = full.narrow(sm.shard_dim, sm.shard_index * n,
                                         n).contiguous()
@@ -127,10 +131,25 @@ class CheckpointManager:
         if len(pieces) == 1 and metas[0]["shard_dim"] < 0:
             return pieces[0][1]
         dim = metas[0]["shard_dim"]
-        if dim < 0:  # replicated saved by several ranks: take the first
-            return pieces[0][1]
+        if dim < 0:  # replicated saved by several ranks: verify they agree
+            first = pieces[0][1]
+            for m, t in pieces[1:]:
+                if not torch.equal(t, first):
+                    raise ValueError(
+                        f"{metas[0]['name']}: replicated copies from "
+                        f"different ranks disagree")
+            return first
         uniq = {}
         for m, t in pieces:
+            prev = uniq.get(m["shard_index"])
+            if prev is not None and not torch.equal(prev[1], t):
+                raise ValueError(
+                    f"{m['name']}: duplicate shard {m['shard_index']} "
+                    f"copies disagree")
             uniq[m["shard_index"]] = (m, t)
+        if sorted(uniq) != list(range(len(uniq))):
+            raise ValueError(
+                f"{metas[0]['name']}: missing shards "
+                f"(have {sorted(uniq)})")
         ordered = [uniq[i][1] for i in sorted(uniq)]
         return torch.cat(ordered, dim=dim)
