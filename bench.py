@@ -8,10 +8,16 @@ W untimed warmup steps, then exactly K timed steps bracketed by a barrier +
 torch.cuda.synchronize on both sides; elapsed time is MAX over ranks; rank 0
 prints one JSON line with the whole-node aggregate tokens/sec.
 
-Auto-parallel: the model's IR goes through the AutoParallel planner
-(exploration mode), whose chosen dp x tp x pp x micro plan is mapped onto
-process groups (TP sharded layers / DP bucketed all-reduce / 1F1B pipeline).
-Plan-search wall-clock is reported (BASELINE.json: "plan-search sec").
+Auto-parallel (default): the model's IR goes through the AutoParallel
+planner (exploration mode) and the PLANNED GRAPH IS WHAT EXECUTES — the
+plan's per-node DistSpec stacks drive the multi-round SpmdTransform, the
+CommDevManager turns mesh rounds into RCCL process groups, and each rank
+runs its transformed graph through the interpreter over the CDNA4 kernel
+layer (runtime/planned.py; reference: ExecuteRPCPlan runs the planner's
+compiled sub-modules, service_rt.cc:530-671). Plans with pipeline stages
+fall back to the hand 1F1B engine (stage decomposition of the planned
+graph is the remaining gap). `--parallel dp|tp<N>|pp<N>` selects the
+hand-parallelized model classes for A/B comparison.
 
 Metric/config per BASELINE.json: tokens/sec (whole node), GPT-2
 auto-parallel, synthetic data, random-init weights, bf16."""
@@ -33,7 +39,8 @@ from tepdist_amd.train import AdamW, Trainer
 
 def plan_parallelism(cfg, world: int, global_batch: int, seq: int,
                      override: str):
-    """Returns (dp, tp, pp, micro, search_s)."""
+    """Returns (dp, tp, pp, micro, search_s, plan) — plan is the full
+    ParallelPlan for the planned-graph path (None for hand overrides)."""
     if override != "auto":
         dp, tp, pp = world, 1, 1
         if override.startswith("tp"):
@@ -42,13 +49,13 @@ def plan_parallelism(cfg, world: int, global_batch: int, seq: int,
             pp = int(override[2:]); dp = world // pp
         elif override != "dp":
             raise ValueError(override)
-        return dp, tp, pp, 1, 0.0
+        return dp, tp, pp, 1, 0.0, None
     from tepdist_amd.planner import AutoParallel
     t0 = time.time()
     g = gpt2_ir(cfg, batch=global_batch, seq=seq)
     plan = AutoParallel(g, world).run()
     search = time.time() - t0
-    return plan.dp, plan.tp, plan.pp, plan.micro_batches, search
+    return plan.dp, plan.tp, plan.pp, plan.micro_batches, search, plan
 
 
 def main():
@@ -78,22 +85,26 @@ def main():
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     global_batch = args.micro_batch * world
 
-    # ---- plan (deterministic across ranks; rank 0's choice broadcast) ----
+    # ---- plan (rank 0 plans; the FULL plan object is broadcast so every
+    # rank transforms the identical specs — re-planning per rank could
+    # diverge under ILP time limits) ----
     search_s = 0.0
+    plan = None
     if rank == 0:
-        dp, tp, pp, micro, search_s = plan_parallelism(
+        dp, tp, pp, micro, search_s, plan = plan_parallelism(
             cfg, world, global_batch, seq, args.parallel)
     if world > 1:
-        t = torch.tensor([dp, tp, pp, micro] if rank == 0 else [0, 0, 0, 0],
-                         dtype=torch.int64)
-        if device.type == "cuda":
-            t = t.to(device)
-        dist.broadcast(t, 0)
-        dp, tp, pp, micro = (int(v) for v in t.tolist())
+        obj = [plan if rank == 0 else None,
+               (dp, tp, pp, micro) if rank == 0 else None]
+        dist.broadcast_object_list(obj, src=0)
+        plan = obj[0]
+        dp, tp, pp, micro = obj[1]
     if rank == 0:
         print(f"# plan: dp={dp} tp={tp} pp={pp} micro={micro} "
               f"search={search_s:.2f}s", flush=True)
     search_b = search_s if rank == 0 else 0.0
+    planned_path = (args.parallel == "auto" and plan is not None
+                    and pp == 1)
 
     local_batch = max(global_batch // dp, 1)
     grad_accum = max(min(micro, local_batch), 1) if pp == 1 else 1
@@ -120,7 +131,33 @@ def main():
             torch.cuda.synchronize(device)
 
     # ---- build the distributed model per the plan ------------------------
-    if pp == 1:
+    if planned_path:
+        # THE PLANNED GRAPH EXECUTES: multi-round SpmdTransform of the
+        # plan's node specs -> per-rank interpreter over the CDNA4 kernels,
+        # dp rounds synced by a bucketed SUM-mode reducer, driven by the
+        # standard Trainer (GA + fused AdamW + hipGraph step capture)
+        from tepdist_amd.runtime.planned import PlannedModule
+        model = PlannedModule(gpt2_ir(cfg, batch=global_batch, seq=seq),
+                              plan, device=device, dtype=dtype)
+        if rank == 0:
+            print(f"# {model.describe()}", flush=True)
+        opt = AdamW(model.parameters(), lr=1e-4)
+        reducer = model.make_reducer()
+        grad_accum = max(min(micro, global_batch), 1)
+        trainer = Trainer(model, opt, grad_accum_steps=grad_accum,
+                          reducer=reducer)
+        m_rows = max(global_batch // grad_accum, 1)
+        # every rank consumes the IDENTICAL global batch stream; the
+        # planner-inserted dynamic_slice takes this rank's part
+        g = torch.Generator().manual_seed(4321)
+
+        def run_step():
+            ids = make_ids(m_rows * grad_accum)
+            def bi(i):
+                sl = ids[i * m_rows:(i + 1) * m_rows]
+                return sl[:, :-1].to(device), sl[:, 1:].to(device)
+            return trainer.train_step(bi)
+    elif pp == 1:
         from tepdist_amd.parallel.tp import ParallelEnv
         env = ParallelEnv.create(tp) if world > 1 else ParallelEnv.single()
         model = GPT2(cfg, dtype=dtype, env=env).to(device)
@@ -191,8 +228,12 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    rows_per_step = (micro_size * grad_accum * dp) if pp == 1 else \
-        (max(local_batch // micro, 1) * micro * dp)
+    if planned_path:
+        rows_per_step = m_rows * grad_accum          # already global
+    elif pp == 1:
+        rows_per_step = micro_size * grad_accum * dp
+    else:
+        rows_per_step = max(local_batch // micro, 1) * micro * dp
     tokens_per_step = rows_per_step * seq
     ms_per_step = elapsed / args.steps * 1000.0
     tokens_per_sec = tokens_per_step * args.steps / elapsed
@@ -216,9 +257,12 @@ def main():
                 "model": args.model,
                 "global_batch": rows_per_step,
                 "seq_len": seq,
-                "parallelism": f"auto:dp{dp}tp{tp}pp{pp}micro{micro}"
-                               if args.parallel == "auto"
-                               else f"{args.parallel}:dp{dp}tp{tp}pp{pp}",
+                "parallelism": (f"auto-planned-graph:dp{dp}tp{tp}"
+                                f"mesh{plan.mesh_rounds}micro{micro}"
+                                if planned_path else
+                                f"auto:dp{dp}tp{tp}pp{pp}micro{micro}"
+                                if args.parallel == "auto"
+                                else f"{args.parallel}:dp{dp}tp{tp}pp{pp}"),
                 "plan_search_s": round(search_b, 3),
             },
         }))

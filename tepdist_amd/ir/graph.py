@@ -15,7 +15,7 @@ from typing import Dict, List, Tuple
 
 # ops whose cost is matmul-like (the planner's "compute sensitive" set,
 # reference cost_spmd_strategy.h:42-49 IsComputeSensitive = dot/conv)
-COMPUTE_SENSITIVE = {"linear", "matmul", "attention", "conv2d"}
+COMPUTE_SENSITIVE = {"linear", "matmul", "attention", "attention_qkv", "conv2d"}
 
 ELEMENTWISE = {"add", "mul", "gelu", "dropout", "scale", "cast", "bias_add"}
 
@@ -110,7 +110,7 @@ class Graph:
             if k is None:
                 k = self.nodes[n.inputs[0]].shape[-1]
             return 2.0 * n.numel * k
-        if n.op == "attention":
+        if n.op in ("attention", "attention_qkv"):
             if len(n.shape) == 4:
                 b, h, s, d = n.shape
             else:  # flattened (B*S, hidden) form with attrs

@@ -93,6 +93,12 @@ class GraphInterpreter:
                        for t in x)
             o = ops.attention(q, k, v, causal=True)
             return o.transpose(1, 2).reshape(b * s, h * hd).contiguous()
+        if n.op == "attention_qkv":
+            b = n.attrs["batch"]
+            s = n.attrs["seq"]
+            qkv = ins[0].reshape(b, s, -1)
+            o = ops.attention_qkv(qkv, n.attrs["heads"], causal=True)
+            return o.reshape(b * s, -1)
         if n.op == "split":
             dim = n.attrs.get("dim", -1)
             idx = n.attrs.get("index", 0)

@@ -126,6 +126,16 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
         out.append(rep())
         return out
 
+    if node.op == "attention_qkv":
+        # packed [BS, 3d] projection in, [BS, d] out: batch split only (a
+        # head split would need a strided/per-section weight shard — the
+        # reference's stride_on_dim; not expressible as a contiguous
+        # DimStrategy split of the packed projection)
+        if _split0_ok(node, n):
+            out.append(OpStrategy(S(0, n), (S(0, n),), "batch"))
+        out.append(rep())
+        return out
+
     if node.op == "embedding":
         table = ins[1]
         if _split0_ok(node, n):
