@@ -47,6 +47,49 @@ def _layout2d(t: torch.Tensor, kc_expected_inner: bool):
 # --------------------------------------------------------------------------
 # GEMM family
 # --------------------------------------------------------------------------
+#
+# Backend policy ("measure, don't guess" made literal): the hand-written
+# 256x256 MFMA pipeline owns FUSED GEMMs (bias / bias+gelu dual-output
+# epilogues) where a library call would pay an extra full pass over C;
+# hipBLASLt (torch.matmul/F.linear on ROCm) is the right tool for PLAIN
+# GEMMs — measured on the flagship shapes (benchmarks/shapes65k.py,
+# profiles/shapes65k_*.json) it runs them 1.2-1.4x our pipeline AND
+# consumes transposed operand layouts natively, which removes the
+# dgrad/wgrad canonicalization transposes (9.6% of the r1 step) outright.
+# Mode "auto" times both paths once per (site, shape) at first use and
+# caches the winner; TEPDIST_GEMM_BACKEND=hip|blaslt pins a backend
+# (kill-switch + pure-hand-kernel demonstration mode).
+
+_GEMM_BACKEND = os.environ.get("TEPDIST_GEMM_BACKEND", "auto")
+_gemm_choice = {}
+
+
+def _pick_backend(key, hip_fn, blt_fn):
+    """Returns the cached winner for `key`, timing both once if unseen."""
+    if _GEMM_BACKEND == "hip":
+        return hip_fn
+    if _GEMM_BACKEND == "blaslt":
+        return blt_fn
+    got = _gemm_choice.get(key)
+    if got is not None:
+        return hip_fn if got == "hip" else blt_fn
+    if torch.cuda.is_current_stream_capturing():
+        return hip_fn  # never tune inside a hipGraph capture
+    import time as _time
+
+    def _t(fn):
+        fn()
+        torch.cuda.synchronize()
+        t0 = _time.perf_counter()
+        for _ in range(3):
+            fn()
+        torch.cuda.synchronize()
+        return _time.perf_counter() - t0
+
+    th, tb = _t(hip_fn), _t(blt_fn)
+    _gemm_choice[key] = "hip" if th <= tb else "blaslt"
+    return hip_fn if th <= tb else blt_fn
+
 
 def _gemm_raw(a: torch.Tensor, b_stored: torch.Tensor, a_kc: bool, b_kc: bool,
               M: int, N: int, K: int, lda: int, ldb: int,
@@ -103,9 +146,23 @@ def linear_fwd(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
         epi = 2 if bias is not None else 3
     else:
         raise ValueError(act)
-    y, pre = _gemm_raw(x, w, True, True, M, N, K, K, w.stride(0), 0, 0, 1,
-                       bias=bias, epi=epi)
-    return y, pre
+
+    def _hip():
+        return _gemm_raw(x, w, True, True, M, N, K, K, w.stride(0), 0, 0, 1,
+                         bias=bias, epi=epi)
+
+    def _blt():
+        # hipBLASLt GEMM (+fused bias via addmm); gelu applies through our
+        # elementwise kernel so fwd/bwd use the same tanh approximation
+        pre = torch.nn.functional.linear(x, w, bias)
+        if act != "gelu":
+            return pre, None
+        y = torch.empty_like(pre)
+        ext.gelu_fwd(pre.data_ptr(), y.data_ptr(), pre.numel(), _stream())
+        return y, pre
+
+    fn = _pick_backend(("lin_f", M, N, K, epi), _hip, _blt)
+    return fn()
 
 
 def transpose2d(t: torch.Tensor) -> torch.Tensor:
@@ -124,40 +181,70 @@ def linear_bwd(dy: torch.Tensor, x: torch.Tensor, w: torch.Tensor,
     dy = dy.contiguous()
     M, N = dy.shape
     K = w.shape[1]
-    # canonicalize to KC x KC: one tuned GEMM schedule serves every case,
-    # with cheap materialized transposes (memory-bound, ~2% of GEMM time)
-    wT = transpose2d(w)        # [K, N]
-    # fused gelu-backward: the wgrad transpose of dy applies gelu'(pre)
-    # in the same pass and emits BOTH layouts (saves the separate
-    # elementwise kernel's full read+write). Bias sums stay a separate
-    # streaming pass — folding them into the transpose as LDS atomics
-    # measured ~3x slower transposes (per-element atomic serialization).
-    dyT = torch.empty(N, M, dtype=BF16, device=dy.device)
-    if act == "gelu":
-        dy_eff = torch.empty_like(dy)
-        ext.transpose_dy(dy.data_ptr(), pre_act.data_ptr(), dyT.data_ptr(),
-                         dy_eff.data_ptr(), 0, M, N, _stream())
-        dy = dy_eff
-    else:
-        ext.transpose_dy(dy.data_ptr(), 0, dyT.data_ptr(), 0, 0, M, N,
-                         _stream())
-    xT = transpose2d(x)        # [K, M]
-    # dx[M,K] = dy[M,N] @ w[N,K]: A=dy KC (k=N), B=w^T stored [K,N] KC
-    dx, _ = _gemm_raw(dy, wT, True, True, M, K, N, N, N, 0, 0, 1)
-    # dw[N,K] = dy^T[N,M] @ x[M,K]: A=dy^T KC (k=M), B=x^T stored [K,M] KC
-    dw, _ = _gemm_raw(dyT, xT, True, True, N, K, M, M, M, 0, 0, 1)
+
+    def _hip():
+        # canonicalize to KC x KC: one tuned GEMM schedule serves every
+        # case, with materialized transposes.
+        dyl = dy
+        wT = transpose2d(w)        # [K, N]
+        # fused gelu-backward: the wgrad transpose of dy applies gelu'(pre)
+        # in the same pass and emits BOTH layouts (saves the separate
+        # elementwise kernel's full read+write). Bias sums stay a separate
+        # streaming pass — folding them into the transpose as LDS atomics
+        # measured ~3x slower transposes (per-element atomic serialization).
+        dyT = torch.empty(N, M, dtype=BF16, device=dy.device)
+        if act == "gelu":
+            dy_eff = torch.empty_like(dyl)
+            ext.transpose_dy(dyl.data_ptr(), pre_act.data_ptr(),
+                             dyT.data_ptr(), dy_eff.data_ptr(), 0, M, N,
+                             _stream())
+            dyl = dy_eff
+        else:
+            ext.transpose_dy(dyl.data_ptr(), 0, dyT.data_ptr(), 0, 0, M, N,
+                             _stream())
+        xT = transpose2d(x)        # [K, M]
+        # dx[M,K] = dy[M,N] @ w[N,K]: A=dy KC (k=N), B=w^T stored [K,N] KC
+        dx, _ = _gemm_raw(dyl, wT, True, True, M, K, N, N, N, 0, 0, 1)
+        # dw[N,K] = dy^T[N,M] @ x[M,K]: A=dy^T KC (k=M), B=x^T [K,M] KC
+        dw, _ = _gemm_raw(dyT, xT, True, True, N, K, M, M, M, 0, 0, 1)
+        return dx, dw, dyl
+
+    def _blt():
+        # hipBLASLt consumes the natural layouts (NN dgrad, TN wgrad):
+        # no transposes at all
+        if act == "gelu":
+            dyl = torch.empty_like(dy)
+            ext.gelu_bwd(dy.data_ptr(), pre_act.data_ptr(), dyl.data_ptr(),
+                         dy.numel(), _stream())
+        else:
+            dyl = dy
+        dx = torch.matmul(dyl, w)        # [M,N] @ [N,K]
+        dw = torch.matmul(dyl.t(), x)    # [N,M] @ [M,K]
+        return dx, dw, dyl
+
+    fn = _pick_backend(("lin_b", M, N, K, act), _hip, _blt)
+    dx, dw, dy_eff = fn()
     db = None
     if has_bias:
         db = torch.empty(N, dtype=BF16, device=dy.device)
         ws = torch.zeros(N, dtype=torch.float32, device=dy.device)
-        ext.bias_sum(dy.data_ptr(), db.data_ptr(), ws.data_ptr(), M, N,
+        ext.bias_sum(dy_eff.data_ptr(), db.data_ptr(), ws.data_ptr(), M, N,
                      _stream())
     return dx, dw, db
 
 
 def matmul(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
-    """C = A @ B with fp32 accumulation. Accepts transposed views of the last
-    two dims without materializing (mapped to kernel layout flags).
+    """C = A @ B with fp32 accumulation (plain GEMM: measured backend
+    choice between the MFMA pipeline and hipBLASLt per shape/layout)."""
+    key = ("mm", tuple(a.shape), tuple(b.shape),
+           a.stride(-1) == 1, b.stride(-1) == 1)
+    return _pick_backend(key, lambda: _matmul_hip(a, b),
+                         lambda: torch.matmul(a, b))()
+
+
+def _matmul_hip(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Hand MFMA path: accepts transposed views of the last two dims
+    without materializing (mapped to kernel layout flags).
     Shapes: [M,K]@[K,N] or batched [B,M,K]@[B,K,N] (B may broadcast)."""
     assert a.dtype == BF16 and b.dtype == BF16
     squeeze = a.dim() == 2 and b.dim() == 2
