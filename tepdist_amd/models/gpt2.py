@@ -254,7 +254,10 @@ class GPT2Stage(nn.Module):
         hd = d // H
         std = 0.02
         proj_std = std / math.sqrt(2 * cfg.n_layer)
-        dt = next(self.parameters()).dtype
+        params = list(self.parameters())
+        if not params:   # empty middle stage (fewer layers than stages)
+            return
+        dt = params[0].dtype
         if self.is_first:
             self.wte.copy_(_draw("wte", (V, d), std, dt, seed))
             self.wpe.copy_(_draw("wpe", (cfg.n_ctx, d), std, dt, seed))
