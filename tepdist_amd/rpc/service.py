@@ -31,7 +31,7 @@ from tepdist_amd.ir.graph import Graph
 from tepdist_amd.ir.interpreter import GraphInterpreter
 from tepdist_amd.planner.auto_parallel import AutoParallel
 from tepdist_amd.runtime.checkpoint import CheckpointManager, SliceMeta
-from tepdist_amd.runtime.initializers import InitSpec, init_shard
+from tepdist_amd.runtime.initializers import InitSpec, init_shard_multi
 from tepdist_amd.runtime.scheduler import TaskScheduler
 from tepdist_amd.runtime.task_graph import build_task_dag
 from tepdist_amd.train.optim import AdamW
@@ -96,12 +96,13 @@ class TepdistService:
                 "search_time_s": time.time() - t0}
 
     def _init_variables(self, g: Graph, init_specs: dict,
-                        param_specs: dict = None):
+                        param_rounds: dict = None, coords=None):
         """Server-side variable creation (RewriteInitializationRemote moved
         init to the server; DistributedRandomInitializer seeds shards).
-        With param_specs (from a dispatched sharded plan) each worker
-        keeps only its slice — the deterministic global-index generator
-        makes every rank's shard consistent with the full tensor."""
+        With param_rounds (from a dispatched sharded plan: per-round
+        (round, dim, nshards) narrows) each worker keeps only its slice —
+        the counter-based global-index generator makes every rank's shard
+        bit-consistent with the full tensor under any nesting."""
         rank = getattr(self, "_comm_info", {}).get("rank", 0)
         restore = self.ckpt_opts.get("restore_step")
         if restore is not None:
@@ -113,10 +114,13 @@ class TepdistService:
             if name in self.vars:
                 continue
             node = g.nodes[nid]
+            rounds = (param_rounds or {}).get(name, [])
+            splits = [(dim, (coords[r] if coords is not None else rank), n)
+                      for (r, dim, n) in rounds]
+            # full (unsharded) shape: undo the narrows in reverse
             full_shape = list(node.shape)
-            dim, nsh = (param_specs or {}).get(name, (-1, 1))
-            if dim >= 0:
-                full_shape[dim] *= nsh
+            for (_, dim, n) in reversed(rounds):
+                full_shape[dim] *= n
             if restore is not None and name in loaded:
                 t = loaded[name].to(dtype)
             elif name in self.host_data:   # client-transferred variable
@@ -124,10 +128,15 @@ class TepdistService:
             else:
                 spec = InitSpec(**init_specs.get(name, {})) \
                     if name in init_specs else _default_spec(name, node.shape)
-                t = init_shard(name, tuple(full_shape), spec, dtype=dtype)
-            if dim >= 0 and t.shape[dim] == full_shape[dim]:
-                sz = full_shape[dim] // nsh
-                t = t.narrow(dim, rank * sz, sz).contiguous()
+                t = init_shard_multi(name, tuple(full_shape), spec, splits,
+                                     dtype=dtype)
+            if tuple(t.shape) != tuple(node.shape):
+                # full tensor arrived (client transfer / restore): narrow
+                # to this worker's nested shard
+                for (dim, idx, n) in splits:
+                    sz = t.shape[dim] // n
+                    t = t.narrow(dim, idx * sz, sz)
+                t = t.contiguous()
             t = t.to(self.device).requires_grad_()
             self.vars[name] = _VarState(
                 t, t.detach().float().clone(),
@@ -220,33 +229,52 @@ class TepdistService:
 
     def dispatch_plan(self, req: dict) -> dict:
         """Install a plan on this (slave) worker. When the plan carries
-        the planner's node_specs, the SpmdTransform rewrites the received
-        module to this worker's sharded executable (reshard collectives
-        run over the communicator from init_remote_comm) and variables are
-        initialized as shards — the reference's DispatchPlan +
-        sharded-variable init path."""
+        the planner's node_specs — one DimStrategy triple per node, or a
+        LIST of per-round triples plus a "mesh" for hybrid dp x tp plans —
+        the multi-round SpmdTransform rewrites the received module to this
+        worker's sharded executable, a CommDevManager built from the mesh
+        supplies one process group per round ordinal for the reshard
+        collectives, and variables are initialized as (possibly nested)
+        shards — the reference's DispatchPlan + per-group NcclContext +
+        sharded-variable init path (pjrt/nccl_context.h)."""
         g = getattr(self, "_pending_graph", None)
         if g is None:
             return {"ok": False, "error": "no module transferred"}
         plan = req.get("plan") or {}
-        param_specs = None
+        param_rounds = {}
+        coords = None
+        groups = None
         exec_graph = g
-        if plan.get("node_specs") and plan.get("nshards", 1) > 1:
-            from tepdist_amd.planner.dist_spec import DimStrategy
-            from tepdist_amd.planner.transform import SpmdTransform
-            specs = {int(k): DimStrategy(*v)
-                     for k, v in plan["node_specs"].items()}
-            res = SpmdTransform(g, specs, int(plan["nshards"])).run()
+        mesh = plan.get("mesh")
+        if plan.get("node_specs") and (mesh or plan.get("nshards", 1) > 1):
+            from tepdist_amd.planner.dist_spec import DimStrategy, DistSpec
+            from tepdist_amd.planner.transform import multi_round_transform
+            from tepdist_amd.runtime.comm import CommDevManager
+            specs = {}
+            for k, v in plan["node_specs"].items():
+                rounds = v if v and isinstance(v[0], (list, tuple)) else [v]
+                specs[int(k)] = DistSpec([DimStrategy(*r) for r in rounds])
+            if not mesh:
+                mesh = [int(plan.get("nshards", 1))]
+            # service-side execution keeps copy_to wrappers for every round
+            # (exact autograd grads without an external reducer)
+            res = multi_round_transform(g, specs, mesh)
             exec_graph = res.graph
-            param_specs = res.param_specs
+            param_rounds = res.param_rounds
+            cdm = CommDevManager(mesh, pp=int(plan.get("pp", 1)))
+            self.comm_mgr = cdm
+            groups = cdm.groups_dict()
+            _, coords = cdm.coords()
         import torch.distributed as dist
         group = dist.group.WORLD if dist.is_initialized() else None
-        interp = GraphInterpreter(exec_graph, self.device, group=group)
+        interp = GraphInterpreter(exec_graph, self.device, group=group,
+                                  groups=groups)
         with self._lock:
             h = self._next_handle
             self._next_handle += 1
             self.plans[h] = ExecutionPlan(h, exec_graph, plan, None, interp)
-        self._init_variables(exec_graph, {}, param_specs=param_specs)
+        self._init_variables(exec_graph, {}, param_rounds=param_rounds,
+                             coords=coords)
         return {"ok": True, "handle": h}
 
     def init_remote_comm(self, req: dict) -> dict:

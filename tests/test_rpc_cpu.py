@@ -138,7 +138,9 @@ def test_dispatched_sharded_plan_matches_single(tmp_path):
     cfg = GPT2_CONFIGS["gpt2-test"]
     g = gpt2_ir(cfg, batch=4, seq=16)
     plan = CostSpmdStrategy(g, 2).run()
-    node_specs = {str(k): [v.kind, v.partition_dim, v.num_shards]
+    # multi-round wire format: per-node LIST of per-round triples + mesh
+    # (the hybrid-capable DispatchPlan surface; single round here)
+    node_specs = {str(k): [[v.kind, v.partition_dim, v.num_shards]]
                   for k, v in plan.node_specs.items()}
 
     ctx = pmp.get_context("spawn")
@@ -157,7 +159,7 @@ def test_dispatched_sharded_plan_matches_single(tmp_path):
         assert all(r["ok"] for r in
                    coord.transfer_module_and_defctx(g.to_json()))
         coord.init_remote_comm("127.0.0.1", gloo_port, join=True)
-        rs = coord.dispatch_plan({"node_specs": node_specs, "nshards": 2})
+        rs = coord.dispatch_plan({"node_specs": node_specs, "mesh": [2]})
         assert all(r["ok"] for r in rs), rs
         handles = [r["handle"] for r in rs]
         feeds = _batch(cfg, 4, 16, seed=11)
