@@ -103,8 +103,7 @@ def main():
         print(f"# plan: dp={dp} tp={tp} pp={pp} micro={micro} "
               f"search={search_s:.2f}s", flush=True)
     search_b = search_s if rank == 0 else 0.0
-    planned_path = (args.parallel == "auto" and plan is not None
-                    and pp == 1)
+    planned_path = args.parallel == "auto" and plan is not None
 
     local_batch = max(global_batch // dp, 1)
     grad_accum = max(min(micro, local_batch), 1) if pp == 1 else 1
@@ -131,7 +130,7 @@ def main():
             torch.cuda.synchronize(device)
 
     # ---- build the distributed model per the plan ------------------------
-    if planned_path:
+    if planned_path and pp == 1:
         # THE PLANNED GRAPH EXECUTES: multi-round SpmdTransform of the
         # plan's node specs -> per-rank interpreter over the CDNA4 kernels,
         # dp rounds synced by a bucketed SUM-mode reducer, driven by the
@@ -157,6 +156,37 @@ def main():
                 sl = ids[i * m_rows:(i + 1) * m_rows]
                 return sl[:, :-1].to(device), sl[:, 1:].to(device)
             return trainer.train_step(bi)
+    elif planned_path:
+        # planned PIPELINE: generic stage decomposition of the planned
+        # graph + per-stage mesh transform, run by the task-list executor
+        # (scheduled 1F1B order, pre-posted recv queue, gc_plan release)
+        from tepdist_amd.planner.stage_decomposition import decompose_stages
+        from tepdist_amd.runtime.comm import CommDevManager
+        from tepdist_amd.runtime.executor import build_stage_executor
+        from tepdist_amd.runtime.planned import PlannedStageModule
+        m_rows = max(global_batch // micro, 1)   # whole-mesh rows per micro
+        g_micro = gpt2_ir(cfg, batch=m_rows, seq=seq)
+        sp = decompose_stages(g_micro, plan.node_stage, pp)
+        cdm = CommDevManager([n for n in plan.mesh_rounds if n > 1] or [1],
+                             pp=pp)
+        stage, coords = cdm.coords()
+        mod = PlannedStageModule(sp, stage, g_micro, plan=plan,
+                                 device=str(device), dtype=dtype, comm=cdm)
+        opt = AdamW(mod.parameters(), lr=1e-4)
+        reducer = mod.make_reducer()
+        pp_ranks = [cdm.rank_of(s, coords) for s in range(pp)]
+        ex = build_stage_executor(
+            mod, stage, pp, pp_ranks, micro, act_shape=mod.act_shape,
+            act_dtype=dtype, device=device, reducer=reducer, optimizer=opt,
+            pp_group=cdm.pipeline_pair_group(+1 if stage == 0 else -1))
+        g = torch.Generator().manual_seed(4321)
+
+        def run_step():
+            ids = make_ids(m_rows * micro)
+            def bi(m):
+                sl = ids[m * m_rows:(m + 1) * m_rows]
+                return sl[:, :-1].to(device), sl[:, 1:].to(device)
+            return ex.run_step(bi)
     elif pp == 1:
         from tepdist_amd.parallel.tp import ParallelEnv
         env = ParallelEnv.create(tp) if world > 1 else ParallelEnv.single()
@@ -228,8 +258,10 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    if planned_path:
+    if planned_path and pp == 1:
         rows_per_step = m_rows * grad_accum          # already global
+    elif planned_path:
+        rows_per_step = m_rows * micro               # already global
     elif pp == 1:
         rows_per_step = micro_size * grad_accum * dp
     else:
@@ -257,7 +289,7 @@ def main():
                 "model": args.model,
                 "global_batch": rows_per_step,
                 "seq_len": seq,
-                "parallelism": (f"auto-planned-graph:dp{dp}tp{tp}"
+                "parallelism": (f"auto-planned-graph:dp{dp}tp{tp}pp{pp}"
                                 f"mesh{plan.mesh_rounds}micro{micro}"
                                 if planned_path else
                                 f"auto:dp{dp}tp{tp}pp{pp}micro{micro}"

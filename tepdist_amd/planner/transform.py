@@ -85,11 +85,16 @@ _PASS_THROUGH = ("copy_to", "all_reduce", "scale", "gelu", "dropout",
 class SpmdTransform:
     def __init__(self, graph: Graph, node_specs: Dict[int, DimStrategy],
                  nshards: int, round_ordinal: int = 0,
-                 dp_reduce: bool = False):
+                 dp_reduce: bool = False, sharded_outputs=None):
         self.src = graph
         self.specs = node_specs
         self.n = nshards
         self.round = round_ordinal
+        # graph-output node ids whose values STAY SHARDED (pipeline stage
+        # boundaries: the next stage holds the same mesh coordinates and
+        # consumes the local shard); everything else resolves to the full
+        # value (losses)
+        self.sharded_outputs = set(sharded_outputs or ())
         # dp_reduce: this round is a data-parallel round executed with a
         # bucketed gradient reducer — params consumed replicated inside
         # sharded regions are recorded in grad_sync_params instead of being
@@ -141,6 +146,17 @@ class SpmdTransform:
                 new_of[node.id], cur_spec[node.id] = nn, spec
                 continue
             if node.op == "data":
+                if node.attrs.get("stage_boundary") and spec.is_split:
+                    # cross-stage activation: the previous stage (same mesh
+                    # coordinates) already holds the local shard — localize
+                    # the shape, no slice
+                    nn = out.add_input(node.name,
+                                       _local_shape(node.shape, spec),
+                                       node.dtype)
+                    nn.attrs = dict(node.attrs)
+                    self._scale_attrs(nn, spec)
+                    new_of[node.id], cur_spec[node.id] = nn, spec
+                    continue
                 # data arrives global; a dynamic_slice takes the rank's part
                 nn = out.add_input(node.name, node.shape, node.dtype)
                 nn.attrs = dict(node.attrs)
@@ -226,6 +242,9 @@ class SpmdTransform:
         # full value (losses: partial -> all-reduce mean over shards)
         for o in g.outputs:
             x, have = new_of[o], cur_spec[o]
+            if o in self.sharded_outputs:
+                out.outputs.append(x.id)   # stage boundary: stays local
+                continue
             if have.is_partial:
                 # partials (incl. the 1/n-scaled row-split loss) sum to the
                 # global value
@@ -324,8 +343,8 @@ class SpmdTransform:
 def multi_round_transform(graph: Graph,
                           node_specs: Dict[int, DistSpec],
                           mesh: Sequence[int],
-                          dp_rounds: Optional[Sequence[int]] = None
-                          ) -> TransformResult:
+                          dp_rounds: Optional[Sequence[int]] = None,
+                          sharded_outputs=None) -> TransformResult:
     """Applies one SpmdTransform per mesh round (reference: one DoTransform
     per split ordinal, spmd_transform.cc:2155), each over the previous
     round's output graph. `node_specs` is keyed by ORIGINAL graph node ids;
@@ -338,6 +357,7 @@ def multi_round_transform(graph: Graph,
     idmap: Dict[int, int] = {i: i for i in graph.nodes}
     combined = TransformResult(cur)
     combined.id_map = dict(idmap)
+    sharded_out = set(sharded_outputs or ())
     for r, n in enumerate(mesh):
         if n <= 1:
             continue
@@ -349,7 +369,9 @@ def multi_round_transform(graph: Graph,
                 (ds if r == 0 else DimStrategy.glue())
             specs_r[idmap[oid]] = s
         t = SpmdTransform(cur, specs_r, n, round_ordinal=r,
-                          dp_reduce=(r in dp_rounds))
+                          dp_reduce=(r in dp_rounds),
+                          sharded_outputs={idmap[o] for o in sharded_out
+                                           if o in idmap})
         res = t.run()
         # compose id maps and merge param/input specs
         idmap = {oid: res.id_map[mid] for oid, mid in idmap.items()

@@ -144,3 +144,122 @@ class PlannedModule(nn.Module):
                 f"rounds={p.mesh_rounds} dp_flags={p.dp_round_flags} "
                 f"params={len(self.vars)} sharded={nshard} "
                 f"nodes={len(self.exec_graph.nodes)}]")
+
+
+class PlannedStageModule(nn.Module):
+    """One pipeline stage of a planned graph, executable by the task-list
+    executor: the generic StageDecomposition's subgraph (+ optional
+    per-stage multi-round SpmdTransform over the plan's mesh) behind the
+    standard stage surface — stage 0 takes input_ids, inner stages take
+    the boundary activation, the last stage takes (x, labels) and returns
+    the loss. Parameters are drawn from the same counter RNG as the full
+    model, so a stage holds bit-identical weights to the corresponding
+    slice of the unsplit graph (reference StageDecomposition ->
+    per-stage executables, stage_decomposition.cc:718)."""
+
+    def __init__(self, stage_plan, stage: int, graph: Graph, plan=None,
+                 device="cpu", dtype=torch.float32,
+                 comm: Optional[CommDevManager] = None, seed: int = 1234):
+        super().__init__()
+        from tepdist_amd.planner.dist_spec import DistSpec
+        self.stage = stage
+        self.sp = stage_plan
+        self.num_stages = len(stage_plan.stages)
+        self.is_first = stage == 0
+        self.is_last = stage == self.num_stages - 1
+        self.device_ = torch.device(device)
+        sg = stage_plan.stages[stage]
+        self.in_bounds = stage_plan.inputs_of(stage)
+        self.out_bounds = stage_plan.outputs_of(stage)
+        assert len(self.in_bounds) <= 1 and len(self.out_bounds) <= 1, \
+            "task-list executor carries one tensor per stage edge"
+
+        mesh = [n for n in ((plan.mesh_rounds if plan else []) or [])
+                if n > 1]
+        self.comm = comm
+        groups = None
+        param_rounds = {}
+        if mesh and plan is not None:
+            flags = [f for n, f in zip(plan.mesh_rounds,
+                                       plan.dp_round_flags or []) if n > 1]
+            lm = stage_plan.local_maps[stage]
+            specs = {lm[oid]: ds for oid, ds in plan.node_specs.items()
+                     if oid in lm and isinstance(ds, DistSpec)}
+            dp_rounds = [i for i, f in enumerate(flags) if f]
+            sharded = {lm[b.src_node] for b in self.out_bounds
+                       if b.src_node in lm}
+            res = multi_round_transform(sg, specs, mesh,
+                                        dp_rounds=dp_rounds,
+                                        sharded_outputs=sharded)
+            self.transform = res
+            sg = res.graph
+            param_rounds = res.param_rounds
+        else:
+            self.transform = None
+        self.exec_graph = sg
+        coords = comm.coords()[1] if comm is not None else [0] * 8
+
+        self.vars: Dict[str, nn.Parameter] = {}
+        for name, nid in sg.params.items():
+            node = sg.nodes[nid]
+            rounds = param_rounds.get(name, [])
+            full_shape = list(node.shape)
+            for (_, dim, n) in reversed(rounds):
+                full_shape[dim] *= n
+            splits = [(dim, coords[r], n) for (r, dim, n) in rounds]
+            spec = default_init_spec(name, full_shape)
+            t = init_shard_multi(name, tuple(full_shape), spec, splits,
+                                 global_seed=seed, dtype=dtype)
+            p = nn.Parameter(t.to(self.device_))
+            self.register_parameter(name.replace(".", "__"), p)
+            self.vars[name] = p
+        self.interp = GraphInterpreter(
+            sg, str(self.device_), dtype=dtype,
+            groups=comm.groups_dict() if comm is not None else None)
+
+    def make_reducer(self, bucket_bytes: int = 64 << 20):
+        if self.transform is None or not self.transform.grad_sync_params:
+            return None
+        by_set: Dict[tuple, List[nn.Parameter]] = {}
+        for name, rounds in self.transform.grad_sync_params.items():
+            if name in self.vars and rounds:
+                by_set.setdefault(tuple(sorted(set(rounds))),
+                                  []).append(self.vars[name])
+        reducers = [GradReducer(params, self.comm.rounds_group(rs),
+                                bucket_bytes=bucket_bytes, average=False)
+                    for rs, params in sorted(by_set.items())]
+        return MultiReducer(reducers) if reducers else None
+
+    @property
+    def act_shape(self):
+        """LOCAL shape of this stage's INCOMING boundary tensor (None for
+        stage 0): read from the (possibly mesh-transformed) stage graph."""
+        if not self.in_bounds:
+            return self.out_act_shape  # stage 0 still RECEIVES bw grads
+        name = self.in_bounds[0].input_name
+        for nid in self.exec_graph.inputs:
+            if self.exec_graph.nodes[nid].name == name:
+                return tuple(self.exec_graph.nodes[nid].shape)
+        return tuple(self.in_bounds[0].shape)
+
+    @property
+    def out_act_shape(self):
+        """LOCAL shape of the OUTGOING boundary tensor (whose gradient
+        this stage receives back)."""
+        if not self.out_bounds:
+            return None
+        oid = self.exec_graph.outputs[self.out_bounds[0].src_output_idx]
+        return tuple(self.exec_graph.nodes[oid].shape)
+
+    def forward(self, x, labels=None):
+        feeds = {}
+        if self.is_first:
+            feeds["input_ids"] = x.reshape(-1)
+        else:
+            feeds[self.in_bounds[0].input_name] = x
+        if labels is not None and self.is_last:
+            feeds["labels"] = labels.reshape(-1)
+        outs = list(self.interp.run(feeds, self.vars).values())
+        if self.is_last:
+            return outs[0]          # the loss
+        return outs[self.out_bounds[0].src_output_idx]

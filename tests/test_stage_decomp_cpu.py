@@ -1,0 +1,176 @@
+"""Generic stage decomposition (planner/stage_decomposition.py): any
+planned graph splits into per-stage subgraphs with inferred boundaries,
+and the task-list executor runs them as a pipeline that matches the
+single-device loss and gradients — no model-specific stage class
+(reference StageDecomposition, stage_decomposition.cc:718)."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from tests.conftest import free_port
+
+
+def _graph(batch=2, seq=16):
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2Config
+    cfg = GPT2Config(name="t", n_layer=4, n_embd=64, n_head=4,
+                     vocab_size=128, n_ctx=64)
+    return gpt2_ir(cfg, batch=batch, seq=seq), cfg
+
+
+def test_decompose_boundaries():
+    from tepdist_amd.planner.pipeline import GraphSketch
+    from tepdist_amd.planner.stage_decomposition import decompose_stages
+    g, cfg = _graph()
+    sp = GraphSketch(g).stage_plan(2)
+    plan = decompose_stages(g, sp.node_stage, 2)
+    assert len(plan.stages) == 2
+    # exactly one activation crosses the single stage edge
+    assert len(plan.boundaries) == 1
+    b = plan.boundaries[0]
+    assert b.src_stage == 0 and b.dst_stage == 1
+    assert b.shape == (2 * 16, cfg.n_embd)
+    # params are disjoint between stages (each lives where consumed)
+    p0 = set(plan.stages[0].params)
+    p1 = set(plan.stages[1].params)
+    assert "wte" in p0
+    assert p0 and p1
+    # every original param appears somewhere
+    assert set(g.params) <= (p0 | p1)
+
+
+def test_llama_decomposes_too():
+    from tepdist_amd.ir.capture import llama_ir
+    from tepdist_amd.models.llama import LlamaConfig
+    from tepdist_amd.planner.pipeline import GraphSketch
+    from tepdist_amd.planner.stage_decomposition import decompose_stages
+    cfg = LlamaConfig(name="t", n_layer=4, n_embd=64, n_head=4,
+                      vocab_size=128, n_ctx=64, ffn_mult=2)
+    g = llama_ir(cfg, batch=2, seq=16)
+    sp = GraphSketch(g).stage_plan(2)
+    plan = decompose_stages(g, sp.node_stage, 2)
+    assert len(plan.boundaries) == 1
+
+
+def _pp_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from tepdist_amd.planner.pipeline import GraphSketch
+        from tepdist_amd.planner.stage_decomposition import decompose_stages
+        from tepdist_amd.runtime.comm import CommDevManager
+        from tepdist_amd.runtime.executor import build_stage_executor
+        from tepdist_amd.runtime.planned import (PlannedModule,
+                                                 PlannedStageModule)
+
+        M = 2
+        g, cfg = _graph(batch=2, seq=16)   # graph at MICRO batch size
+        sk = GraphSketch(g).stage_plan(world)
+        plan = decompose_stages(g, sk.node_stage, world)
+        mod = PlannedStageModule(plan, rank, g)
+        ex = build_stage_executor(
+            mod, rank, world, list(range(world)), M,
+            act_shape=mod.act_shape, act_dtype=torch.float32,
+            device="cpu")
+
+        gen = torch.Generator().manual_seed(9)
+        ids = torch.randint(0, 128, (2 * M, 17), generator=gen)
+
+        def batch_iter(m):
+            sl = ids[m * 2:(m + 1) * 2]
+            return sl[:, :-1], sl[:, 1:]
+
+        loss = ex.run_step(batch_iter)
+
+        # single-device reference: same weights (counter RNG), same data
+        from tepdist_amd.planner.auto_parallel import AutoParallel
+        ref_mod = PlannedModule(g, AutoParallel(g, 1).run(),
+                                comm=CommDevManager([1], 1, 0, 1))
+        ref = sum(ref_mod(*batch_iter(m)) for m in range(M)) / M
+        ref.backward()
+        assert abs(loss - float(ref)) < 1e-4, (rank, loss, float(ref))
+        # stage grads match the reference's for this stage's params.
+        # params consumed on BOTH stages (the tied wte: embedding on
+        # stage 0, logits on stage 1) become untied by decomposition —
+        # each stage holds a partial grad; skip them (GPT2Stage has the
+        # same documented semantics)
+        shared = set(plan.stages[0].params) & set(plan.stages[1].params)
+        checked = 0
+        for name, p in mod.vars.items():
+            if name in shared:
+                continue
+            rg = ref_mod.vars[name].grad
+            if p.grad is not None and rg is not None:
+                # executor scales each micro loss by 1/M; the reference
+                # backward runs on the mean — same gradient scale
+                torch.testing.assert_close(p.grad, rg, rtol=1e-3,
+                                           atol=1e-5)
+                checked += 1
+        assert checked >= 3, checked
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_planned_pipeline_matches_single():
+    port = free_port()
+    mp.spawn(_pp_worker, args=(2, port), nprocs=2, join=True)
+
+
+def _hybrid_worker(rank, world, port):
+    """pp2 x dp2: stage-major layout, stage = rank // 2; each stage's
+    subgraph is dp-transformed over mesh [2] (multi-round SpmdTransform on
+    the DECOMPOSED graph) and the task-list executor runs the pipeline."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from tepdist_amd.planner.auto_parallel import AutoParallel
+        from tepdist_amd.planner.stage_decomposition import decompose_stages
+        from tepdist_amd.runtime.comm import CommDevManager
+        from tepdist_amd.runtime.executor import build_stage_executor
+        from tepdist_amd.runtime.planned import (PlannedModule,
+                                                 PlannedStageModule)
+
+        M = 2
+        g, cfg = _graph(batch=4, seq=16)   # GLOBAL micro batch (2/dp rank)
+        ap = AutoParallel(g, world)
+        plan = ap._plan_proposal(2, [2])   # 2 stages x mesh [2]
+        assert plan.mesh_rounds == [2]
+        sp = decompose_stages(g, plan.node_stage, 2)
+        cdm = CommDevManager(plan.mesh_rounds, pp=2)
+        stage, coords = cdm.coords()
+        mod = PlannedStageModule(sp, stage, g, plan=plan, comm=cdm)
+        reducer = mod.make_reducer()
+        pp_ranks = [cdm.rank_of(s, coords) for s in range(2)]
+        ex = build_stage_executor(
+            mod, stage, 2, pp_ranks, M, act_shape=mod.act_shape,
+            act_dtype=torch.float32, device="cpu", reducer=reducer,
+            pp_group=cdm.pipeline_pair_group(+1 if stage == 0 else -1))
+        gen = torch.Generator().manual_seed(9)
+        ids = torch.randint(0, 128, (4 * M, 17), generator=gen)
+
+        def batch_iter(m):
+            sl = ids[m * 4:(m + 1) * 4]
+            return sl[:, :-1], sl[:, 1:]
+
+        loss = ex.run_step(batch_iter)
+
+        from tepdist_amd.runtime.comm import CommDevManager as CDM
+        ref_mod = PlannedModule(g, AutoParallel(g, 1).run(),
+                                comm=CDM([1], 1, 0, 1))
+        ref = sum(ref_mod(*batch_iter(m)) for m in range(M)) / M
+        assert abs(loss - float(ref)) < 1e-4, (rank, loss, float(ref))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_planned_pipeline_x_dp_hybrid():
+    port = free_port()
+    mp.spawn(_hybrid_worker, args=(4, port), nprocs=4, join=True)
