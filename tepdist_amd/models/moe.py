@@ -87,6 +87,11 @@ class MoELayer(nn.Module):
         self.aux_loss = torch.zeros(())
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """STATIC-SHAPE GShard dispatch (hipGraph-capture-safe): every
+        step exchanges exactly E x capacity padded slots — no .item() /
+        .tolist() host syncs, no data-dependent tensor shapes. Dropped
+        tokens (over capacity) contribute zero; padded slots compute
+        garbage that is never read (their combine weights are zero)."""
         B, S, d = x.shape
         T = B * S
         xt = x.reshape(T, d)
@@ -106,83 +111,58 @@ class MoELayer(nn.Module):
         P = gates.float().mean(0)
         self.aux_loss = self.E * (f * P).sum()
 
-        capacity = max(int(self.cf * T * self.k / self.E), 4)
+        C = max(int(self.cf * T * self.k / self.E), 4)   # static capacity
 
-        # flatten (token, slot) assignments, capacity-drop per expert
         flat_e = topi.reshape(-1)                        # [T*k]
         flat_w = topv.reshape(-1)
         flat_t = torch.arange(T, device=x.device).repeat_interleave(self.k)
+        # position within expert (stable sort -> rank - expert offset)
         order = torch.argsort(flat_e, stable=True)
-        fe, fw, ft = flat_e[order], flat_w[order], flat_t[order]
-        # position within expert
-        ones = torch.ones_like(fe)
         counts = torch.zeros(self.E, dtype=torch.long,
-                             device=x.device).scatter_add_(0, fe, ones)
+                             device=x.device).scatter_add_(
+            0, flat_e, torch.ones_like(flat_e))
         offs = torch.cumsum(counts, 0) - counts
-        pos = torch.arange(fe.numel(), device=x.device) - offs[fe]
-        keep = pos < capacity
-        fe, fw, ft = fe[keep], fw[keep], ft[keep]
+        r = torch.arange(flat_e.numel(), device=x.device)
+        pos = torch.empty_like(r)
+        pos[order] = r - offs[flat_e[order]]
+        keep = pos < C
+        slot = flat_e * C + pos
+        slot_safe = torch.where(keep, slot, torch.zeros_like(slot))
 
-        # gather tokens in expert order, exchange across EP ranks
-        send = xt[ft]                                    # [n_send, d]
-        kept_counts = torch.zeros(self.E, dtype=torch.long,
-                                  device=x.device).scatter_add_(
-            0, fe, torch.ones_like(fe))
-        per_rank = kept_counts.reshape(self.ep_size, self.e_local).sum(-1)
-        in_splits = per_rank.tolist()
+        # dispatch: D[E*C, d]; dropped entries accumulate ZERO into slot 0
+        contrib = xt[flat_t] * keep.unsqueeze(-1).to(xt.dtype)
+        D = torch.zeros(self.E * C, d, dtype=xt.dtype, device=x.device)
+        D = D.index_put((slot_safe,), contrib, accumulate=True)
+
         if self.ep_size > 1:
-            ex = torch.tensor(in_splits, device=x.device)
-            all_splits = [torch.zeros_like(ex) for _ in range(self.ep_size)]
-            dist.all_gather(all_splits, ex, group=self.ep_group)
-            out_splits = [int(s[self.ep_rank].item()) for s in all_splits]
-            recv = _AllToAllVar.apply(send, out_splits, in_splits,
-                                      self.ep_group)
-            # exchange per-expert counts for exact segmentation
-            pe = kept_counts.reshape(self.ep_size, self.e_local).contiguous()
-            pe_all = torch.empty_like(pe)
-            dist.all_to_all_single(pe_all, pe, group=self.ep_group)
-            seg = pe_all  # [src_rank, local_expert]
+            from tepdist_amd.parallel.mappings import all_to_all
+            Dx = all_to_all(D, self.ep_group)            # equal splits
+            recv = Dx.reshape(self.ep_size, self.e_local, C, d)
         else:
-            recv = send
-            seg = kept_counts.reshape(1, self.E)[:, self.e_start:
-                                                 self.e_start + self.e_local]
+            recv = D.reshape(1, self.E, C, d)[:, self.e_start:
+                                              self.e_start + self.e_local]
 
-        # process: received tokens are grouped rank-major, expert-minor;
-        # rebuild expert-contiguous batches (out-of-place index_add keeps
-        # autograd through the expert FFNs)
-        cursor = 0
-        segs = []
-        for r in range(seg.shape[0]):
-            for e in range(self.e_local):
-                c = int(seg[r, e].item())
-                segs.append((e, cursor, c))
-                cursor += c
-        sels, ys = [], []
+        # local experts over padded batches [src_ranks * C, d]
+        ys = []
         for e in range(self.e_local):
-            idxs = [torch.arange(st, st + c, device=x.device)
-                    for (ee, st, c) in segs if ee == e and c > 0]
-            if not idxs:
-                continue
-            sel = torch.cat(idxs)
-            h = ops.linear(recv[sel], self.w1[e], self.b1[e], act="gelu")
-            y = ops.linear(h, self.w2[e], self.b2[e])
-            sels.append(sel)
-            ys.append(y)
-        if sels:
-            outs = torch.zeros_like(recv).index_add(
-                0, torch.cat(sels), torch.cat(ys).to(recv.dtype))
-        else:
-            outs = torch.zeros_like(recv)
+            xe = recv[:, e].reshape(-1, d).contiguous()
+            h = ops.linear(xe, self.w1[e], self.b1[e], act="gelu")
+            ys.append(ops.linear(h, self.w2[e], self.b2[e]))
+        Y = torch.stack(ys, dim=0)                       # [e_local, src*C, d]
+        Y = Y.reshape(self.e_local, -1, C, d).transpose(0, 1)  # [src, e_l, C, d]
 
         if self.ep_size > 1:
-            back = _AllToAllVar.apply(outs, in_splits, out_splits,
-                                      self.ep_group)
+            from tepdist_amd.parallel.mappings import all_to_all
+            back = all_to_all(Y.reshape(self.ep_size, -1, d).reshape(
+                self.ep_size * self.e_local * C, d).contiguous(),
+                self.ep_group)
         else:
-            back = outs
+            back = Y.reshape(self.E * C, d)
 
-        # combine: scatter back to tokens with gate weights
-        out = torch.zeros_like(xt).index_add(
-            0, ft, back * fw.unsqueeze(-1).to(back.dtype))
+        # combine: gather each (token, slot)'s expert output, weight, sum
+        gathered = back[slot_safe] *             (flat_w * keep.to(flat_w.dtype)).unsqueeze(-1).to(back.dtype)
+        out = torch.zeros_like(xt).index_add(0, flat_t,
+                                             gathered.to(xt.dtype))
         return out.reshape(B, S, d)
 
 
@@ -234,11 +214,45 @@ class GPTMoE(nn.Module):
         self.reset_parameters()
 
     @torch.no_grad()
+    @torch.no_grad()
     def reset_parameters(self, seed: int = 1234):
-        g = torch.Generator().manual_seed(seed)
-        for name, p in self.named_parameters():
-            if p.dim() >= 2:
-                p.copy_(torch.randn(p.shape, generator=g) * 0.02)
+        """Shard-aware (counter RNG): EP ranks draw DIFFERENT experts —
+        their w1/w2 shards are slices of one global [E, ...] tensor (the
+        per-rank-identical seeding would duplicate experts across the EP
+        group, the ADVICE r1 bug class); TP shards follow the GPT-2
+        convention via models.gpt2._draw."""
+        from tepdist_amd.models.gpt2 import _draw
+        cfg, env = self.cfg, self.env
+        tp, r = env.tp_size, env.tp_rank
+        V, d, H = cfg.padded_vocab, cfg.n_embd, cfg.n_head
+        hd = d // H
+        std = 0.02
+        dt = self.wte.dtype
+        self.wte.copy_(_draw("wte", (V, d), std, dt, seed))
+        self.wpe.copy_(_draw("wpe", (cfg.n_ctx, d), std, dt, seed))
+        for i, blk in enumerate(self.blocks):
+            b = blk.inner
+            qkv = _draw(f"h{i}.w_qkv", (3, H, hd * d), std, dt, seed,
+                        1, r, tp).reshape(3 * (H // tp) * hd, d)
+            proj = _draw(f"h{i}.w_proj", (d, d), std, dt, seed, 1, r, tp)
+            fc = _draw(f"h{i}.w_fc", (4 * d, d), std, dt, seed, 0, r, tp)
+            out = _draw(f"h{i}.w_out", (d, 4 * d), std, dt, seed, 1, r, tp)
+            if tp == 1:
+                b.w_qkv.copy_(qkv); b.w_proj.copy_(proj)
+                b.w_fc.copy_(fc); b.w_out.copy_(out)
+            else:
+                b.qkv.weight.copy_(qkv); b.proj.weight.copy_(proj)
+                b.fc.weight.copy_(fc); b.out.weight.copy_(out)
+            m = blk.moe
+            if m is not None:
+                # gate replicated (reference pins it replicated); experts
+                # sharded on the GLOBAL expert dim across the EP group
+                m.w_gate.copy_(_draw(f"h{i}.moe.gate", (m.E, d), std, dt,
+                                     seed))
+                m.w1.copy_(_draw(f"h{i}.moe.w1", (m.E, 4 * d, d), std, dt,
+                                 seed, 0, m.ep_rank, m.ep_size))
+                m.w2.copy_(_draw(f"h{i}.moe.w2", (m.E, d, 4 * d), std, dt,
+                                 seed, 0, m.ep_rank, m.ep_size))
         self.wte[self.cfg.vocab_size:].zero_()
 
     def forward(self, input_ids, labels=None):

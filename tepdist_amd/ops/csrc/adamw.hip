@@ -92,8 +92,10 @@ namespace {
 constexpr int MT_CHUNK = 16384;
 
 // multi-tensor AdamW: one fused launch over all parameters. tabs holds 5
-// pointer tables (param, master, grad(bf16), exp_avg, exp_avg_sq), chunks
-// is [(tensor_idx, chunk_idx)] with MT_CHUNK elements per chunk.
+// pointer tables (param, master, grad, exp_avg, exp_avg_sq), chunks
+// is [(tensor_idx, chunk_idx)] with MT_CHUNK elements per chunk; ptypes[t]
+// selects the per-tensor storage type (0 = bf16 param+grad, 1 = fp32
+// param+grad — e.g. Wide-ResNet's fp32 batch-norm affines).
 // hyper: optional device buffer {lr, inv_bc1, inv_bc2, _} read at launch
 // time — lets the kernel live inside a captured hipGraph while the
 // step-dependent bias correction still advances (the host updates the
@@ -102,6 +104,7 @@ __launch_bounds__(NT) __global__
 void adamw_mt_kernel(const int64_t* __restrict__ tabs,
                      const int64_t* __restrict__ numel,
                      const float* __restrict__ wds,
+                     const unsigned char* __restrict__ ptypes,
                      const int* __restrict__ chunks, int nchunks, int nt,
                      float lr, float beta1, float beta2, float eps,
                      float inv_bc1, float inv_bc2,
@@ -115,12 +118,29 @@ void adamw_mt_kernel(const int64_t* __restrict__ tabs,
     const int ti = chunks[2 * ci];
     const int64_t off = (int64_t)chunks[2 * ci + 1] * MT_CHUNK;
     bf16_t* param = reinterpret_cast<bf16_t*>(tabs[ti]);
+    float* paramf = reinterpret_cast<float*>(tabs[ti]);
     float* master = reinterpret_cast<float*>(tabs[nt + ti]);
     const bf16_t* grad = reinterpret_cast<const bf16_t*>(tabs[2 * nt + ti]);
+    const float* gradf = reinterpret_cast<const float*>(tabs[2 * nt + ti]);
     float* m = reinterpret_cast<float*>(tabs[3 * nt + ti]);
     float* v = reinterpret_cast<float*>(tabs[4 * nt + ti]);
     const float wd = wds[ti];
+    const bool f32p = ptypes != nullptr && ptypes[ti] == 1;
     const int64_t end = std::min(off + MT_CHUNK, numel[ti]);
+    if (f32p) {
+      // fp32 param + fp32 grad (no bf16 image): scalar loop, these
+      // tensors are tiny (norm affines)
+      for (int64_t i = off + threadIdx.x; i < end; i += NT) {
+        const float g = gradf[i];
+        m[i] = beta1 * m[i] + (1.f - beta1) * g;
+        v[i] = beta2 * v[i] + (1.f - beta2) * g * g;
+        const float denom = sqrtf(v[i] * inv_bc2) + eps;
+        master[i] = master[i] * (1.f - lr * wd) -
+                    lr * (m[i] * inv_bc1) / denom;
+        paramf[i] = master[i];
+      }
+      continue;
+    }
     for (int64_t i0 = off + (int64_t)threadIdx.x * 4; i0 < end;
          i0 += NT * 4) {
       const int cnt = (int)std::min<int64_t>(4, end - i0);
@@ -162,13 +182,14 @@ void adamw_mt_kernel(const int64_t* __restrict__ tabs,
 }  // namespace
 
 void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
-                   const float* wds, const int* chunks, int nchunks, int nt,
+                   const float* wds, const unsigned char* ptypes,
+                   const int* chunks, int nchunks, int nt,
                    float lr, float beta1, float beta2, float eps, float bc1,
                    float bc2, const float* hyper, hipStream_t stream) {
   const int blocks = std::min(nchunks, 2048);
   hipLaunchKernelGGL(adamw_mt_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
-                     stream, tabs, numel, wds, chunks, nchunks, nt, lr,
-                     beta1, beta2, eps, 1.0f / bc1, 1.0f / bc2, hyper);
+                     stream, tabs, numel, wds, ptypes, chunks, nchunks, nt,
+                     lr, beta1, beta2, eps, 1.0f / bc1, 1.0f / bc2, hyper);
 }
 
 }  // namespace tepdist

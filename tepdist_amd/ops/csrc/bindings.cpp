@@ -288,12 +288,14 @@ PYBIND11_MODULE(_tepdist_hip, m) {
   });
 
   m.def("adamw_mt", [](uintptr_t tabs, uintptr_t numel, uintptr_t wds,
-                       uintptr_t chunks, int nchunks, int nt, float lr,
-                       float b1, float b2, float eps, float bc1, float bc2,
-                       uintptr_t hyper, uintptr_t stream) {
+                       uintptr_t ptypes, uintptr_t chunks, int nchunks,
+                       int nt, float lr, float b1, float b2, float eps,
+                       float bc1, float bc2, uintptr_t hyper,
+                       uintptr_t stream) {
     adamw_mt_bf16(reinterpret_cast<const int64_t*>(tabs),
                   reinterpret_cast<const int64_t*>(numel),
                   reinterpret_cast<const float*>(wds),
+                  reinterpret_cast<const unsigned char*>(ptypes),
                   reinterpret_cast<const int*>(chunks), nchunks, nt, lr, b1,
                   b2, eps, bc1, bc2,
                   reinterpret_cast<const float*>(hyper), S(stream));

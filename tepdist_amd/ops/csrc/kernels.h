@@ -123,7 +123,8 @@ void adamw_bf16(void* param, float* master, const void* grad_bf16,
 
 // multi-tensor fused AdamW (pointer tables on device; MT_CHUNK=16384)
 void adamw_mt_bf16(const int64_t* tabs, const int64_t* numel,
-                   const float* wds, const int* chunks, int nchunks, int nt,
+                   const float* wds, const unsigned char* ptypes,
+                   const int* chunks, int nchunks, int nt,
                    float lr, float beta1, float beta2, float eps, float bc1,
                    float bc2, const float* hyper, hipStream_t stream);
 

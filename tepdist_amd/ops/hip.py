@@ -582,6 +582,11 @@ class AdamWMT:
     def __init__(self, params, states, wds):
         dev = params[0].device
         self.nt = len(params)
+        # per-tensor storage type: 0 = bf16 param+grad, 1 = fp32 (e.g.
+        # Wide-ResNet batch-norm affines)
+        self.ptypes = torch.tensor(
+            [1 if p.dtype == torch.float32 else 0 for p in params],
+            dtype=torch.uint8, device=dev)
         ptrs = ([p.data_ptr() for p in params] +
                 [s["master"].data_ptr() for s in states] +
                 [0] * self.nt +
@@ -605,7 +610,8 @@ class AdamWMT:
         bc1 = 1.0 - beta1 ** step_no
         bc2 = 1.0 - beta2 ** step_no
         ext.adamw_mt(self.tabs.data_ptr(), self.numel.data_ptr(),
-                     self.wds.data_ptr(), self.chunks.data_ptr(),
+                     self.wds.data_ptr(), self.ptypes.data_ptr(),
+                     self.chunks.data_ptr(),
                      self.nchunks, self.nt, lr, beta1, beta2, eps, bc1, bc2,
                      0, _stream())
 
@@ -621,7 +627,8 @@ class AdamWMT:
 
     def step_graph(self, beta1, beta2, eps):
         ext.adamw_mt(self.tabs.data_ptr(), self.numel.data_ptr(),
-                     self.wds.data_ptr(), self.chunks.data_ptr(),
+                     self.wds.data_ptr(), self.ptypes.data_ptr(),
+                     self.chunks.data_ptr(),
                      self.nchunks, self.nt, 0.0, beta1, beta2, eps, 1.0,
                      1.0, self.hyper.data_ptr(), _stream())
 

@@ -40,9 +40,15 @@ class AdamW:
         grads for every param."""
         import torch as _t
         ps = [p for p in self.params if p.grad is not None]
-        if len(ps) < 4 or not ps[0].is_cuda or \
-                any(p.grad.dtype != _t.bfloat16 or
-                    not p.grad.is_contiguous() for p in ps):
+
+        def _ok(p):
+            # bf16 param+grad or fp32 param+grad (Wide-ResNet batch-norm
+            # affines), contiguous — mixed sets run in ONE mt launch via
+            # the per-tensor ptypes table
+            return p.dtype in (_t.bfloat16, _t.float32) and \
+                p.grad.dtype == p.dtype and p.grad.is_contiguous()
+
+        if len(ps) < 4 or not ps[0].is_cuda or not all(_ok(p) for p in ps):
             return None
         from tepdist_amd.ops import hip as be
         key = tuple(id(p) for p in ps)
