@@ -269,4 +269,12 @@ class GPTMoE(nn.Module):
                                  labels.reshape(-1), ignore_index=-1)
         aux = sum(blk.moe.aux_loss for blk in self.blocks
                   if blk.moe is not None)
-        return loss + self.aux_weight * aux
+        loss = loss + self.aux_weight * aux
+        # drop the layers' live references to this step's autograd graph:
+        # a retained aux_loss keeps last step's AccumulateGrad nodes alive
+        # on the default stream, which breaks (segfaults) hipGraph capture
+        # of the next step
+        for blk in self.blocks:
+            if blk.moe is not None:
+                blk.moe.aux_loss = blk.moe.aux_loss.detach()
+        return loss
