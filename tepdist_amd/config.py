@@ -123,3 +123,10 @@ def get_env() -> ServiceEnv:
 def set_env(env: ServiceEnv) -> None:
     global _GLOBAL_ENV
     _GLOBAL_ENV = env
+
+
+def reset_env() -> None:
+    """Drops the cached ServiceEnv so the next get_env() re-reads the
+    process environment (tests toggling kill-switches)."""
+    global _GLOBAL_ENV
+    _GLOBAL_ENV = None

@@ -227,10 +227,16 @@ class AutoParallel:
         dp = tp = 1
         dp_flags = []
         for ri, n in enumerate(rounds):
-            planner = CostSpmdStrategy(
-                g, n, self.cm,
-                time_limit_s=0.0 if rule else self.env.ilp_time_limit_s,
-                param_mem_penalty=1e-9 if pressure else 0.0)
+            if rule:
+                # RULE_MODE: the one-pass annotation/rule inference engine
+                # (reference AnnotFastSpmdStrategy), not the cost search
+                from tepdist_amd.planner.fast_spmd import FastSpmdStrategy
+                planner = FastSpmdStrategy(g, n, self.cm)
+            else:
+                planner = CostSpmdStrategy(
+                    g, n, self.cm,
+                    time_limit_s=self.env.ilp_time_limit_s,
+                    param_mem_penalty=1e-9 if pressure else 0.0)
             res = planner.run()
             spmd_cost += res.cost
             # classify the round: dp if most compute-sensitive flops chose a

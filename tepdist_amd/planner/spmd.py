@@ -64,6 +64,13 @@ class CostSpmdStrategy:
         # weight-sharded (tensor-parallel) strategies
         self.param_mem_penalty = param_mem_penalty
         self.cons = graph.consumers()
+        # user annotations (the reference's ExtractUserSplit,
+        # cost_spmd_strategy.cc:588-680; IGNORE_ANNOTATION drops them)
+        if get_env().ignore_annotation:
+            self.pins = {}
+        else:
+            from tepdist_amd.ir.sharding import collect_pins
+            self.pins = collect_pins(graph, nshards)
 
     # ---------------------------------------------------------------------
 
@@ -114,6 +121,9 @@ class CostSpmdStrategy:
         self._neighbor_vote(best_specs)
         for i in self.g.nodes:
             best_specs.setdefault(i, DimStrategy.replicated(self.n))
+        # user annotations are a CONTRACT: enforce them on the result
+        # (SpmdTransform reshards any producer/consumer mismatch)
+        best_specs.update(self.pins)
         # co-location affinity (InstAffinityMap): aux optimizer vars and
         # aliased pairs adopt their leader's spec
         from tepdist_amd.planner.affinity import InstAffinityMap
@@ -177,7 +187,15 @@ class CostSpmdStrategy:
     def _populate_cone_strategies(self, cones: List[Cone]):
         for cone in cones:
             root = self.g.nodes[cone.root]
-            for st in op_strategies(self.g, root, self.n):
+            pin = self.pins.get(cone.root)
+            cands = op_strategies(self.g, root, self.n)
+            if pin is not None:
+                # annotated root: only strategies producing the pinned
+                # spec survive (fall back to all if none match — the
+                # final enforcement pass still applies the pin)
+                matched = [st for st in cands if st.out == pin]
+                cands = matched or cands
+            for st in cands:
                 specs, in_demand, cost = self._grow(cone, st)
                 cone.strategies.append(specs)
                 cone.strat_in.append(in_demand)
