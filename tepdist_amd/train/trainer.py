@@ -37,8 +37,20 @@ class Trainer:
         # optimizer — is captured once and replayed per step, removing the
         # ~2.5k per-step kernel-launch round trips. Kill-switch
         # TEPDIST_HIP_GRAPH=0; any capture failure falls back to eager.
-        self.hip_graph = get_env().hip_graph if hip_graph is None \
-            else hip_graph
+        if hip_graph is None:
+            import os as _os
+
+            import torch.distributed as _dist
+            multi = _dist.is_initialized() and _dist.get_world_size() > 1
+            if multi and "TEPDIST_HIP_GRAPH" not in _os.environ:
+                # multi-rank default: eager. Capturing RCCL collectives
+                # inside a hipGraph is unverified on this pool (1-GPU
+                # leases); a capture segfault would kill the whole job.
+                # TEPDIST_HIP_GRAPH=1 opts in explicitly.
+                hip_graph = False
+            else:
+                hip_graph = get_env().hip_graph
+        self.hip_graph = hip_graph
         self._graph = None
         self._static = None
         self._graph_losses = None
