@@ -22,6 +22,7 @@
 // delta = rowsum(dO*O) is the flash2 preprocess.
 
 #include <algorithm>
+#include <cstdlib>
 #include <stdexcept>
 
 #include "common.h"
@@ -454,7 +455,7 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
 constexpr int NTA = 512;   // 8 waves: one 16-kv slice each (KBA = 128)
 constexpr int KBA = 128;
 
-template <int D>
+template <int D, int PROBE = 0>
 __launch_bounds__(NTA) __global__
 void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
                          const bf16_t* __restrict__ K,
@@ -697,14 +698,18 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
       }
 #pragma unroll
       for (int df = 0; df < DF; ++df) {
+        // PROBE=1: broadcast row (conflict-free, WRONG numerics) — an
+        // upper-bound timing probe for the bank-conflict cost of these
+        // reads (benchmarks/prof_attn.py probe mode)
+        const int fr = PROBE ? 0 : (lane & 15);
         const bf16x4 a0 = *reinterpret_cast<const bf16x4*>(
-            sdOT + loff<QT>(16 * df + (lane & 15), 32 * c + 4 * g));
+            sdOT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
         const bf16x4 a1 = *reinterpret_cast<const bf16x4*>(
-            sdOT + loff<QT>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+            sdOT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
         const bf16x4 q0f = *reinterpret_cast<const bf16x4*>(
-            sQT + loff<QT>(16 * df + (lane & 15), 32 * c + 4 * g));
+            sQT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
         const bf16x4 q1f = *reinterpret_cast<const bf16x4*>(
-            sQT + loff<QT>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+            sQT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
         bf16x8 afr, qfr;
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
@@ -1059,6 +1064,17 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        static_cast<const bf16_t*>(o), delta, H, S, o_bs,     \
                        o_hs, o_rs);                                          \
     dim3 kgrid((S + 127) / 128, B * H);                                      \
+    static const bool kv_probe = std::getenv("TEPDIST_FLASH_PROBE");         \
+    if (kv_probe)                                                            \
+      hipLaunchKernelGGL((flash_bwd_kv_kernel<DD, 1>), kgrid, dim3(512), 0,  \
+                         stream, static_cast<const bf16_t*>(q),              \
+                         static_cast<const bf16_t*>(k),                      \
+                         static_cast<const bf16_t*>(v),                      \
+                         static_cast<const bf16_t*>(dout), lse, delta,       \
+                         static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv), \
+                         S, H, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs,  \
+                         o_rs);                                              \
+    else                                                                     \
     hipLaunchKernelGGL(flash_bwd_kv_kernel<DD>, kgrid, dim3(512), 0, stream, \
                        static_cast<const bf16_t*>(q),                        \
                        static_cast<const bf16_t*>(k),                        \
