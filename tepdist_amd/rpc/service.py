@@ -287,10 +287,25 @@ class TepdistService:
         import os
 
         import torch.distributed as dist
+        rank = int(req.get("rank", self.task_index))
+        world = int(req.get("world", 1))
+        # request validation (blind trust in request-supplied
+        # ranks): coordinates must be coherent, and a second bootstrap
+        # with DIFFERENT coordinates while a group is live is an error —
+        # silently proceeding would pair mismatched communicators
+        if world < 1 or not (0 <= rank < world):
+            return {"ok": False,
+                    "error": f"bad comm coords rank={rank} world={world}"}
+        prev = getattr(self, "_comm_info", None)
+        if dist.is_initialized() and prev is not None and \
+                (prev["rank"] != rank or prev["world"] != world):
+            return {"ok": False,
+                    "error": f"communicator already initialized as "
+                             f"{prev}; refusing re-init as "
+                             f"rank={rank}/world={world}"}
         os.environ["MASTER_ADDR"] = req.get("master_addr", "127.0.0.1")
         os.environ["MASTER_PORT"] = str(req.get("master_port", 29500))
-        self._comm_info = {"rank": req.get("rank", self.task_index),
-                           "world": req.get("world", 1)}
+        self._comm_info = {"rank": rank, "world": world}
         if req.get("join", False) and self._comm_info["world"] > 1 \
                 and not dist.is_initialized():
             import datetime
