@@ -68,6 +68,21 @@ class TaskListExecutor:
         self.is_last = stage == len(pp_ranks) - 1
         self._n_bw = sum(1 for t in self.list
                          if t.type == TaskType.COMPUTE_BW)
+        # recv-buffer pool (reference ExecutionState's per-(device, type)
+        # recv pools, execution_state.h:43-135): buffers cycle instead of
+        # re-allocating every step
+        self._buf_pool: List[torch.Tensor] = []
+
+    def _get_buf(self) -> torch.Tensor:
+        if self._buf_pool:
+            return self._buf_pool.pop()
+        return torch.empty(self.act_shape, dtype=self.act_dtype,
+                           device=self.device)
+
+    def _put_buf(self, t: torch.Tensor):
+        if t.shape == torch.Size(self.act_shape) and \
+                t.dtype == self.act_dtype and len(self._buf_pool) < 8:
+            self._buf_pool.append(t)
 
     # -- helpers ----------------------------------------------------------
 
@@ -85,8 +100,7 @@ class TaskListExecutor:
             t = self.list[i]
             if t.type == TaskType.RECV and t.id not in posted \
                     and not hasattr(t, "_done"):
-                buf = torch.empty(self.act_shape, dtype=self.act_dtype,
-                                  device=self.device)
+                buf = self._get_buf()
                 posted[t.id] = (dist.irecv(buf, self._peer_rank(t)), buf)
                 outstanding += 1
             i += 1
@@ -164,8 +178,7 @@ class TaskListExecutor:
                     work, buf = posted.pop(t.id)
                     work.wait()
                 else:
-                    buf = torch.empty(self.act_shape, dtype=self.act_dtype,
-                                      device=self.device)
+                    buf = self._get_buf()
                     dist.recv(buf, self._peer_rank(t))
                 store[t.id] = buf
                 t._done = True
@@ -185,7 +198,11 @@ class TaskListExecutor:
 
             # gc_plan consumer: buffers whose last reader was this task
             for dead in self.gc.get(t.id, ()):
-                store.pop(dead, None)
+                freed = store.pop(dead, None)
+                if freed is not None and \
+                        self.dag.tasks[dead].type == TaskType.RECV and \
+                        not freed.requires_grad and freed.grad_fn is None:
+                    self._put_buf(freed.detach())
             self.peak_store = max(self.peak_store, len(store))
             self._pump_recvs(idx + 1, posted)
             if self.sync_mode and self.device.type == "cuda":
