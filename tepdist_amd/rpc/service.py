@@ -265,6 +265,16 @@ class TepdistService:
             self.comm_mgr = cdm
             groups = cdm.groups_dict()
             _, coords = cdm.coords()
+        # rebuild the runtime plan when the master shipped its scheduled
+        # task lists (the reference's ComputeTask protos -> slave
+        # BuildDistributedPlanRPC + BuildLocalPlan, service_rt.cc:417-465)
+        local_plan = None
+        if plan.get("task_dag"):
+            from tepdist_amd.runtime.task_graph import TaskDAG
+            rdag = TaskDAG.from_wire(plan["task_dag"])
+            order = {int(k): [int(t) for t in v]
+                     for k, v in (plan.get("sched_order") or {}).items()}
+            local_plan = (rdag, order)
         import torch.distributed as dist
         group = dist.group.WORLD if dist.is_initialized() else None
         interp = GraphInterpreter(exec_graph, self.device, group=group,
@@ -272,7 +282,8 @@ class TepdistService:
         with self._lock:
             h = self._next_handle
             self._next_handle += 1
-            self.plans[h] = ExecutionPlan(h, exec_graph, plan, None, interp)
+            self.plans[h] = ExecutionPlan(h, exec_graph, plan, local_plan,
+                                          interp)
         self._init_variables(exec_graph, {}, param_rounds=param_rounds,
                              coords=coords)
         return {"ok": True, "handle": h}

@@ -101,6 +101,41 @@ class TaskDAG:
         assert len(order) == len(self.tasks), "cycle in TaskDAG"
         return order
 
+    # -- wire form (the reference serializes TaskNodes as ComputeTask
+    # protos for DispatchPlan, xla.proto:491-508) -------------------------
+
+    def to_wire(self) -> dict:
+        """JSON-able form of the full DAG: one record per task with type,
+        split address, device, def id, schedule index, edges and peer."""
+        return {
+            "source": self.source, "sink": self.sink,
+            "tasks": [{
+                "id": t.id, "type": t.type.value,
+                "micro": t.split.micro, "stage": t.split.stage,
+                "shard": t.split.shard, "device": t.device,
+                "def_id": t.def_id, "sched_idx": t.sched_idx,
+                "parents": list(t.parents), "children": list(t.children),
+                "flops": t.flops, "out_bytes": t.out_bytes,
+                "release_at": t.release_at, "peer": t.peer,
+            } for t in self.tasks.values()],
+        }
+
+    @staticmethod
+    def from_wire(d: dict) -> "TaskDAG":
+        dag = TaskDAG()
+        for r in d["tasks"]:
+            t = TaskNode(r["id"], TaskType(r["type"]),
+                         SplitId(r["micro"], r["stage"], r["shard"]),
+                         r["device"], r["def_id"],
+                         list(r["parents"]), list(r["children"]),
+                         r["sched_idx"], r["flops"], r["out_bytes"],
+                         r["release_at"], r["peer"])
+            dag.tasks[t.id] = t
+            dag._next = max(dag._next, t.id + 1)
+        dag.source = d.get("source")
+        dag.sink = d.get("sink")
+        return dag
+
     def dump_dot(self, path: str):
         """Graphviz dump (the reference's TaskDAG::Dump dag.dot,
         task_graph.cc:377)."""

@@ -159,7 +159,16 @@ def test_dispatched_sharded_plan_matches_single(tmp_path):
         assert all(r["ok"] for r in
                    coord.transfer_module_and_defctx(g.to_json()))
         coord.init_remote_comm("127.0.0.1", gloo_port, join=True)
-        rs = coord.dispatch_plan({"node_specs": node_specs, "mesh": [2]})
+        # ship the scheduled runtime task lists too (the reference's
+        # ComputeTask protos): workers rebuild the LocalPlan from them
+        from tepdist_amd.runtime.scheduler import TaskScheduler
+        from tepdist_amd.runtime.task_graph import build_task_dag
+        dag = build_task_dag(1, 1)
+        sched = TaskScheduler(dag, mem_cap_bytes=float("inf")).schedule()
+        rs = coord.dispatch_plan({"node_specs": node_specs, "mesh": [2],
+                                  "task_dag": dag.to_wire(),
+                                  "sched_order": {str(k): v for k, v
+                                                  in sched.order.items()}})
         assert all(r["ok"] for r in rs), rs
         handles = [r["handle"] for r in rs]
         feeds = _batch(cfg, 4, 16, seed=11)

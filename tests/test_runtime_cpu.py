@@ -184,3 +184,25 @@ def _exec_worker(rank, world, port, results):
 def test_task_list_executor_gc_and_recv_queue():
     port = free_port()
     mp.spawn(_exec_worker, args=(4, port, None), nprocs=4, join=True)
+
+
+def test_task_dag_wire_roundtrip():
+    """TaskDAG wire form (the reference's ComputeTask serialization for
+    DispatchPlan, xla.proto:491-508): to_wire/from_wire preserves types,
+    split addresses, edges, peers and schedule indices."""
+    from tepdist_amd.runtime.task_graph import TaskDAG, build_task_dag
+    from tepdist_amd.runtime.scheduler import TaskScheduler
+    import json
+    dag = build_task_dag(2, 4, act_bytes_per_micro=1.0)
+    TaskScheduler(dag, mem_cap_bytes=float("inf")).schedule()
+    wire = json.loads(json.dumps(dag.to_wire()))   # through JSON
+    back = TaskDAG.from_wire(wire)
+    assert set(back.tasks) == set(dag.tasks)
+    for tid, t in dag.tasks.items():
+        b = back.tasks[tid]
+        assert (b.type, b.split.micro, b.split.stage, b.device,
+                b.sched_idx, b.peer) == \
+            (t.type, t.split.micro, t.split.stage, t.device,
+             t.sched_idx, t.peer)
+        assert b.parents == t.parents and b.children == t.children
+    assert back.topo()  # acyclic, complete
