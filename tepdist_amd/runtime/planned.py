@@ -125,6 +125,25 @@ class PlannedModule(nn.Module):
                                         average=False))
         return MultiReducer(reducers) if reducers else None
 
+    def dist_param(self, name: str):
+        """This parameter as a DistTensor (DAPPLEBuffer equivalent): local
+        shard + slicing metadata; .to_full() gathers the global tensor."""
+        from tepdist_amd.runtime.dist_tensor import DistTensor
+        return DistTensor(self.vars[name].detach(),
+                          self.transform.param_rounds.get(name, []),
+                          self.comm, name)
+
+    def save_checkpoint(self, ckpt, step: int):
+        """Sharded save through the CheckpointManager: every rank writes
+        its shards with slice metadata (reference CheckpointUtil +
+        VariableSpecsMgr)."""
+        shards = {}
+        for name in self.vars:
+            t, sm = self.dist_param(name).slice_meta()
+            shards[name] = (t, sm)
+        ckpt.save(step, shards, rank=self.comm.rank,
+                  world=self.comm.world)
+
     def forward(self, input_ids: torch.Tensor,
                 labels: Optional[torch.Tensor] = None):
         """input_ids/labels are the GLOBAL (whole-mesh) batch in [B, S] or
