@@ -452,10 +452,13 @@ DEV_INLINE void stage_nat_t(const bf16_t* __restrict__ src, int row0, int S,
 // MFMA k-permutation invariance against transpose-staged dO^T / Q^T
 // A-operands — P and dS never touch LDS. dK/dV accumulate transposed in
 // registers across all q tiles and scatter once.
-constexpr int NTA = 512;   // 8 waves: one 16-kv slice each (KBA = 128)
-constexpr int KBA = 128;
+constexpr int NTA = 512;   // 8 waves; each wave owns one 16-kv slice per
+                           // 128-kv slab (KBA/128 slabs per block: the q/dO
+                           // tile staging+A-fragment reads amortize over
+                           // every slab — the backward's staging VALU was
+                           // the measured bound, profiles/flash_bwd_pmc_r2)
 
-template <int D, int PROBE = 0>
+template <int D, int PROBE = 0, int KBA = (D == 64 ? 256 : 128)>
 __launch_bounds__(NTA) __global__
 void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
                          const bf16_t* __restrict__ K,
@@ -470,6 +473,7 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   constexpr int DK = D / 32;
   constexpr int DF = D / 16;
   constexpr int QT = 64;
+  constexpr int NSL = KBA / 128;   // kv slabs per block
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int kv0 = blockIdx.x * KBA;
@@ -514,18 +518,11 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   }
   __syncthreads();
 
-  // B fragments of this wave's K and V slice (constant across q tiles)
-  bf16x8 kf[DK], vf[DK];
-#pragma unroll
-  for (int kk = 0; kk < DK; ++kk) {
-    kf[kk] = *reinterpret_cast<const bf16x8*>(
-        sKb + loff<D>(wave * 16 + (lane & 15), 8 * g + 32 * kk));
-    vf[kk] = *reinterpret_cast<const bf16x8*>(
-        sVb + loff<D>(wave * 16 + (lane & 15), 8 * g + 32 * kk));
-  }
-
-  f32x4 acc_dkT[DF] = {};   // [d = 16df+4g+e][kv = wave*16 + lane&15]
-  f32x4 acc_dvT[DF] = {};
+  // K/V B-fragments are read from LDS per use (register-caching both
+  // slabs' fragments pushed the kernel to 256 VGPR + scratch spills; the
+  // b128 re-reads are cheaper than spilling)
+  f32x4 acc_dkT[NSL][DF] = {};  // [d = 16df+4g+e][kv = slab + wave*16+l15]
+  f32x4 acc_dvT[NSL][DF] = {};
 
   // issue-early / write-late staging registers for the Q and dO tiles
   constexpr int QUN = QT * D / 8 / NTA;
@@ -629,102 +626,113 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   for (int q0 = q_start; q0 < S; q0 += QT) {
     if (q0 + QT < S) tile_load(q0 + QT);  // overlap with compute
 
-    // S = Q K^T ; dP = dO V^T  — C-frags: q = 16mi+4g+e, kv = lane&15
-    f32x4 st[4] = {};
-    f32x4 dpt[4] = {};
 #pragma unroll
-    for (int kk = 0; kk < DK; ++kk)
+    for (int sl = 0; sl < NSL; ++sl) {
+      // S = Q K^T ; dP = dO V^T  — C-frags: q = 16mi+4g+e, kv = lane&15
+      f32x4 st[4] = {};
+      f32x4 dpt[4] = {};
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        const bf16x8 qa = *reinterpret_cast<const bf16x8*>(
-            sQ + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
-        st[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa, kf[kk], st[mi],
-                                                         0, 0, 0);
-        const bf16x8 da = *reinterpret_cast<const bf16x8*>(
-            sdO + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
-        dpt[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, vf[kk],
-                                                          dpt[mi], 0, 0, 0);
+      for (int kk = 0; kk < DK; ++kk) {
+        const bf16x8 kfr = *reinterpret_cast<const bf16x8*>(
+            sKb + loff<D>(sl * 128 + wave * 16 + (lane & 15),
+                          8 * g + 32 * kk));
+        const bf16x8 vfr = *reinterpret_cast<const bf16x8*>(
+            sVb + loff<D>(sl * 128 + wave * 16 + (lane & 15),
+                          8 * g + 32 * kk));
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const bf16x8 qa = *reinterpret_cast<const bf16x8*>(
+              sQ + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
+          st[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qa, kfr,
+                                                           st[mi], 0, 0, 0);
+          const bf16x8 da = *reinterpret_cast<const bf16x8*>(
+              sdO + loff<D>(16 * mi + (lane & 15), 8 * g + 32 * kk));
+          dpt[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              da, vfr, dpt[mi], 0, 0, 0);
+        }
       }
 
-    // P and dS (elementwise); kv fixed per lane, q varies per element.
-    // Inner tiles skip the per-element mask selects (wave-uniform).
-    const int kvg = kv0 + wave * 16 + (lane & 15);
-    const bool winner = (!causal || kv0 + wave * 16 + 15 < q0) &&
-                        q0 + QT <= S && kv0 + KBA <= S;
-    if (winner) {
-      f32x4 lsev[4], dlv[4];
+      // P and dS (elementwise); kv fixed per lane, q varies per element.
+      // Inner tiles skip the per-element mask selects (wave-uniform).
+      const int kvb = kv0 + sl * 128;
+      const int kvg = kvb + wave * 16 + (lane & 15);
+      const bool winner = (!causal || kvb + wave * 16 + 15 < q0) &&
+                          q0 + QT <= S && kvb + 128 <= S;
+      if (winner) {
+        f32x4 lsev[4], dlv[4];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        lsev[mi] = *reinterpret_cast<const f32x4*>(sLSE + 16 * mi + 4 * g);
-        dlv[mi] = *reinterpret_cast<const f32x4*>(sDELTA + 16 * mi + 4 * g);
+        for (int mi = 0; mi < 4; ++mi) {
+          lsev[mi] = *reinterpret_cast<const f32x4*>(sLSE + 16 * mi + 4 * g);
+          dlv[mi] = *reinterpret_cast<const f32x4*>(sDELTA + 16 * mi + 4 * g);
+        }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const float pt = fast_exp(st[mi][e] * scale - lsev[mi][e]);
+            st[mi][e] = pt;
+            dpt[mi][e] = pt * (dpt[mi][e] - dlv[mi][e]) * scale;
+          }
+      } else {
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            const int qrow = 16 * mi + 4 * g + e;
+            const int qg = q0 + qrow;
+            const float lse = sLSE[qrow];
+            const float dl = sDELTA[qrow];
+            const bool valid = qg < S && kvg < S && (!causal || kvg <= qg) &&
+                               lse > -1.0e38f;
+            const float arg = valid ? st[mi][e] * scale - lse : -3.0e38f;
+            const float pt = fast_exp(arg);              // select, not branch
+            st[mi][e] = pt;                              // now P
+            dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale; // now dS
+          }
       }
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const float pt = fast_exp(st[mi][e] * scale - lsev[mi][e]);
-          st[mi][e] = pt;
-          dpt[mi][e] = pt * (dpt[mi][e] - dlv[mi][e]) * scale;
-        }
-    } else {
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-          const int qrow = 16 * mi + 4 * g + e;
-          const int qg = q0 + qrow;
-          const float lse = sLSE[qrow];
-          const float dl = sDELTA[qrow];
-          const bool valid = qg < S && kvg < S && (!causal || kvg <= qg) &&
-                             lse > -1.0e38f;
-          const float arg = valid ? st[mi][e] * scale - lse : -3.0e38f;
-          const float pt = fast_exp(arg);              // select, not branch
-          st[mi][e] = pt;                              // now P
-          dpt[mi][e] = pt * (dpt[mi][e] - dl) * scale; // now dS
-        }
-    }
 
-    // dV^T += dO^T P and dK^T += Q^T dS (both k-permuted B from regs)
-    __builtin_amdgcn_s_setprio(1);
+      // dV^T += dO^T P and dK^T += Q^T dS (both k-permuted B from regs)
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int c = 0; c < QT / 32; ++c) {
-      bf16x8 pb, db;
-#pragma unroll
-      for (int e = 0; e < 4; ++e) {
-        pb[e] = f2bf(st[2 * c][e]);
-        pb[e + 4] = f2bf(st[2 * c + 1][e]);
-        db[e] = f2bf(dpt[2 * c][e]);
-        db[e + 4] = f2bf(dpt[2 * c + 1][e]);
-      }
-#pragma unroll
-      for (int df = 0; df < DF; ++df) {
-        // PROBE=1: broadcast row (conflict-free, WRONG numerics) — an
-        // upper-bound timing probe for the bank-conflict cost of these
-        // reads (benchmarks/prof_attn.py probe mode)
-        const int fr = PROBE ? 0 : (lane & 15);
-        const bf16x4 a0 = *reinterpret_cast<const bf16x4*>(
-            sdOT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
-        const bf16x4 a1 = *reinterpret_cast<const bf16x4*>(
-            sdOT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
-        const bf16x4 q0f = *reinterpret_cast<const bf16x4*>(
-            sQT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
-        const bf16x4 q1f = *reinterpret_cast<const bf16x4*>(
-            sQT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
-        bf16x8 afr, qfr;
+      for (int c = 0; c < QT / 32; ++c) {
+        bf16x8 pb, db;
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
-          afr[e] = a0[e];
-          afr[e + 4] = a1[e];
-          qfr[e] = q0f[e];
-          qfr[e + 4] = q1f[e];
+          pb[e] = f2bf(st[2 * c][e]);
+          pb[e + 4] = f2bf(st[2 * c + 1][e]);
+          db[e] = f2bf(dpt[2 * c][e]);
+          db[e + 4] = f2bf(dpt[2 * c + 1][e]);
         }
-        acc_dvT[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afr, pb, acc_dvT[df], 0, 0, 0);
-        acc_dkT[df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            qfr, db, acc_dkT[df], 0, 0, 0);
+#pragma unroll
+        for (int df = 0; df < DF; ++df) {
+          // PROBE=1: broadcast row (conflict-free, WRONG numerics) — an
+          // upper-bound timing probe for the bank-conflict cost of these
+          // reads (benchmarks/prof_attn.py probe mode)
+          const int fr = PROBE ? 0 : (lane & 15);
+          const bf16x4 a0 = *reinterpret_cast<const bf16x4*>(
+              sdOT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
+          const bf16x4 a1 = *reinterpret_cast<const bf16x4*>(
+              sdOT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
+          const bf16x4 q0f = *reinterpret_cast<const bf16x4*>(
+              sQT + loff<QT>(16 * df + fr, 32 * c + 4 * g));
+          const bf16x4 q1f = *reinterpret_cast<const bf16x4*>(
+              sQT + loff<QT>(16 * df + fr, 32 * c + 16 + 4 * g));
+          bf16x8 afr, qfr;
+#pragma unroll
+          for (int e = 0; e < 4; ++e) {
+            afr[e] = a0[e];
+            afr[e + 4] = a1[e];
+            qfr[e] = q0f[e];
+            qfr[e + 4] = q1f[e];
+          }
+          acc_dvT[sl][df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr, pb, acc_dvT[sl][df], 0, 0, 0);
+          acc_dkT[sl][df] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              qfr, db, acc_dkT[sl][df], 0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
     BAR_LDS();  // sQ/sQT/sdO/sdOT reads retired
     if (q0 + QT < S) {
       tile_write();
@@ -733,16 +741,17 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
   }
 
   // write dK, dV (transposed-accumulator scatter: kv = wave's lane slice)
-  {
-    const int kvg = kv0 + wave * 16 + (lane & 15);
+#pragma unroll
+  for (int sl = 0; sl < NSL; ++sl) {
+    const int kvg = kv0 + sl * 128 + wave * 16 + (lane & 15);
     if (kvg < S) {
 #pragma unroll
       for (int df = 0; df < DF; ++df) {
         bf16x4 kv4, vv4;
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
-          kv4[e] = f2bf(acc_dkT[df][e]);
-          vv4[e] = f2bf(acc_dvT[df][e]);
+          kv4[e] = f2bf(acc_dkT[sl][df][e]);
+          vv4[e] = f2bf(acc_dvT[sl][df][e]);
         }
         const int64_t off = qoff + (int64_t)kvg * q_rs + 16 * df + 4 * g;
         *reinterpret_cast<bf16x4*>(dK + off) = kv4;
@@ -1063,7 +1072,8 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        stream, static_cast<const bf16_t*>(dout),             \
                        static_cast<const bf16_t*>(o), delta, H, S, o_bs,     \
                        o_hs, o_rs);                                          \
-    dim3 kgrid((S + 127) / 128, B * H);                                      \
+    dim3 kgrid((S + (DD == 64 ? 255 : 127)) / (DD == 64 ? 256 : 128),       \
+               B * H);                                                       \
     static const bool kv_probe = std::getenv("TEPDIST_FLASH_PROBE");         \
     if (kv_probe)                                                            \
       hipLaunchKernelGGL((flash_bwd_kv_kernel<DD, 1>), kgrid, dim3(512), 0,  \
