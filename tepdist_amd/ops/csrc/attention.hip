@@ -91,10 +91,16 @@ DEV_INLINE float warp16_sum(float v) {
 // q = lane&15 — the same lane as the softmax state, so the alpha rescale is
 // a per-lane multiply) and is scattered once at the end. No P tile ever
 // touches LDS.
-constexpr int NTF = 512;   // 8 waves x 32 q columns (QBF = 256)
-constexpr int QBF = 256;
+constexpr int NTF = 512;   // 8 waves x 64 q columns (QBF = 512): the
+                           // K/V staging volume per block is fixed by S,
+                           // so doubling the q rows it feeds halves the
+                           // staging cost per unit of MFMA work (same
+                           // lever as the backward's 256-kv blocks); the
+                           // 64 q are processed as two sequential 16-q
+                           // pairs so the score accumulators stay [4][2]
+constexpr int QBF = 512;   // D=64; D=128 uses 256 (register budget)
 
-template <int D>
+template <int D, int NQ = (D == 64 ? 4 : 2)>  // 16-q groups per wave
 __launch_bounds__(NTF) __global__
 void flash_fwd_kernel(const bf16_t* __restrict__ Q,
                       const bf16_t* __restrict__ K,
@@ -106,11 +112,11 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   constexpr int DF = D / 16;   // d row fragments of O^T
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QBF;
+  const int q0 = blockIdx.x * (128 * NQ);   // 8 waves x 16*NQ q rows
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
-  const int wq0 = q0 + wave * 32;     // this wave's 32 q columns
+  const int wq0 = q0 + wave * 16 * NQ;   // this wave's q columns
 
   const bf16_t* qp = Q + (int64_t)b * q_bs + (int64_t)h * q_hs;
   const bf16_t* kp = K + (int64_t)b * q_bs + (int64_t)h * q_hs;
@@ -123,9 +129,9 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
   const int BUFSZ = KB * D + D * KB;
 
   // Q fragments (PV B-layout twin): nf-th 16-q group, kk-th 32-d chunk
-  bf16x8 qf[2][DK];
+  bf16x8 qf[NQ][DK];
 #pragma unroll
-  for (int nf = 0; nf < 2; ++nf)
+  for (int nf = 0; nf < NQ; ++nf)
 #pragma unroll
     for (int kk = 0; kk < DK; ++kk) {
       const int row = wq0 + nf * 16 + (lane & 15);
@@ -136,10 +142,13 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
       qf[nf][kk] = v8;
     }
 
-  f32x4 acc_o[DF][2] = {};    // O^T: d = 16*df + 4g + e, q = lane&15 (+16nf)
-  float m_r[2], l_r[2];
-  m_r[0] = m_r[1] = -3.0e38f;
-  l_r[0] = l_r[1] = 0.f;
+  f32x4 acc_o[DF][NQ] = {};   // O^T: d = 16*df + 4g + e, q = lane&15 (+16nf)
+  float m_r[NQ], l_r[NQ];
+#pragma unroll
+  for (int i = 0; i < NQ; ++i) {
+    m_r[i] = -3.0e38f;
+    l_r[i] = 0.f;
+  }
 
   // staging registers (issue-early / write-late split):
   // K: KUN x 16B per thread; V: 8 x 4B per slab (v_perm transpose slabs;
@@ -209,7 +218,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     }
   };
 
-  const int kv_end = causal ? min(S, q0 + QBF) : S;
+  const int kv_end = causal ? min(S, q0 + 128 * NQ) : S;
   stage_load(0);
   stage_write(0);
   __syncthreads();
@@ -219,105 +228,113 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
     const bf16_t* sVT = sK + KB * D;
     if (kv0 + KB < kv_end) stage_load(kv0 + KB);  // overlap with compute
 
-    if (!causal || kv0 <= wq0 + 31) {  // wave has unmasked work
-      // --- S^T = K Q^T: kv = 16*mi + 4g + e, q = lane&15 + 16*nf ---
-      f32x4 st[4][2] = {};
-      __builtin_amdgcn_s_setprio(1);
+    if (!causal || kv0 <= wq0 + 16 * NQ - 1) {  // wave has unmasked work
 #pragma unroll
-      for (int kk = 0; kk < DK; ++kk)
+      for (int hp = 0; hp < NQ / 2; ++hp) {  // sequential 32-q pairs
+        const int pq0 = wq0 + 32 * hp;
+        if (causal && kv0 > pq0 + 31) continue;  // pair fully masked
+        // --- S^T = K Q^T: kv = 16*mi + 4g + e, q = lane&15 + 16*nf ---
+        f32x4 st[4][2] = {};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-          const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-              sK + loff<D>((lane & 15) + 16 * mi, 8 * g + 32 * kk));
+        for (int kk = 0; kk < DK; ++kk)
 #pragma unroll
-          for (int nf = 0; nf < 2; ++nf)
-            st[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                kf, qf[nf][kk], st[mi][nf], 0, 0, 0);
-        }
-      __builtin_amdgcn_s_setprio(0);
-      // --- scale + mask; online softmax (per-lane q state). Tiles fully
-      // inside the causal triangle and the sequence skip the per-element
-      // mask selects (wave-uniform branch, no divergence) ---
-      const bool inner = kv0 + KB <= wq0 && kv0 + KB <= S && wq0 + 32 <= S;
+          for (int mi = 0; mi < 4; ++mi) {
+            const bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+                sK + loff<D>((lane & 15) + 16 * mi, 8 * g + 32 * kk));
 #pragma unroll
-      for (int nf = 0; nf < 2; ++nf) {
-        const int qg = wq0 + nf * 16 + (lane & 15);
-        float tmax = -3.0e38f;
-        if (inner) {
+            for (int nf = 0; nf < 2; ++nf)
+              st[mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  kf, qf[2 * hp + nf][kk], st[mi][nf], 0, 0, 0);
+          }
+        __builtin_amdgcn_s_setprio(0);
+        // --- scale + mask; online softmax (per-lane q state). Tiles
+        // fully inside the causal triangle and the sequence skip the
+        // per-element mask selects (wave-uniform branch) ---
+        const bool inner = kv0 + KB <= pq0 && kv0 + KB <= S &&
+                           pq0 + 32 <= S;
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf) {
+          const int nfg = 2 * hp + nf;
+          const int qg = pq0 + nf * 16 + (lane & 15);
+          float tmax = -3.0e38f;
+          if (inner) {
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+              for (int e = 0; e < 4; ++e) {
+                const float sv = st[mi][nf][e] * scale;
+                st[mi][nf][e] = sv;
+                tmax = fmaxf(tmax, sv);
+              }
+          } else {
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+              for (int e = 0; e < 4; ++e) {
+                const int kg = kv0 + mi * 16 + 4 * g + e;
+                const bool masked = (causal && kg > qg) || kg >= S ||
+                                    qg >= S;
+                const float sv = masked ? -3.0e38f : st[mi][nf][e] * scale;
+                st[mi][nf][e] = sv;
+                tmax = fmaxf(tmax, sv);
+              }
+          }
+          tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+          tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+          const float mn = fmaxf(m_r[nfg], tmax);
+          const float alpha = fast_exp(m_r[nfg] - mn);
+          m_r[nfg] = mn;
+          float rs = 0.f;
 #pragma unroll
           for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
             for (int e = 0; e < 4; ++e) {
-              const float sv = st[mi][nf][e] * scale;
-              st[mi][nf][e] = sv;
-              tmax = fmaxf(tmax, sv);
+              // sv - mn <= 0 always; fully-masked rows give -inf -> 0
+              const float p = fast_exp(st[mi][nf][e] - mn);
+              st[mi][nf][e] = p;
+              rs += p;
             }
-        } else {
+          rs += __shfl_xor(rs, 16, 64);
+          rs += __shfl_xor(rs, 32, 64);
+          l_r[nfg] = l_r[nfg] * alpha + rs;
 #pragma unroll
-          for (int mi = 0; mi < 4; ++mi)
+          for (int df = 0; df < DF; ++df)
 #pragma unroll
-            for (int e = 0; e < 4; ++e) {
-              const int kg = kv0 + mi * 16 + 4 * g + e;
-              const bool masked = (causal && kg > qg) || kg >= S || qg >= S;
-              const float sv = masked ? -3.0e38f : st[mi][nf][e] * scale;
-              st[mi][nf][e] = sv;
-              tmax = fmaxf(tmax, sv);
-            }
+            for (int e = 0; e < 4; ++e) acc_o[df][nfg][e] *= alpha;
         }
-        tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
-        tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
-        const float mn = fmaxf(m_r[nf], tmax);
-        const float alpha = fast_exp(m_r[nf] - mn);  // 0 if mn new, 1 if tied
-        m_r[nf] = mn;
-        float rs = 0.f;
+        // --- O^T += V^T P^T (P^T direct from accumulators) ---
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            // sv - mn <= 0 always; fully-masked rows give -inf -> 0
-            const float p = fast_exp(st[mi][nf][e] - mn);
-            st[mi][nf][e] = p;
-            rs += p;
-          }
-        rs += __shfl_xor(rs, 16, 64);
-        rs += __shfl_xor(rs, 32, 64);
-        l_r[nf] = l_r[nf] * alpha + rs;
-#pragma unroll
-        for (int df = 0; df < DF; ++df)
-#pragma unroll
-          for (int e = 0; e < 4; ++e) acc_o[df][nf][e] *= alpha;
-      }
-      // --- O^T += V^T P^T (P^T direct from accumulators, k-permuted) ---
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int c = 0; c < KB / 32; ++c) {
-        bf16x8 pb[2];
-#pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
-#pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            pb[nf][e] = f2bf(st[2 * c][nf][e]);
-            pb[nf][e + 4] = f2bf(st[2 * c + 1][nf][e]);
-          }
-#pragma unroll
-        for (int df = 0; df < DF; ++df) {
-          const bf16x4 v0 = *reinterpret_cast<const bf16x4*>(
-              sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 4 * g));
-          const bf16x4 v1 = *reinterpret_cast<const bf16x4*>(
-              sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
-          bf16x8 vfr;
-#pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            vfr[e] = v0[e];
-            vfr[e + 4] = v1[e];
-          }
+        for (int c = 0; c < KB / 32; ++c) {
+          bf16x8 pb[2];
 #pragma unroll
           for (int nf = 0; nf < 2; ++nf)
-            acc_o[df][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                vfr, pb[nf], acc_o[df][nf], 0, 0, 0);
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              pb[nf][e] = f2bf(st[2 * c][nf][e]);
+              pb[nf][e + 4] = f2bf(st[2 * c + 1][nf][e]);
+            }
+#pragma unroll
+          for (int df = 0; df < DF; ++df) {
+            const bf16x4 v0 = *reinterpret_cast<const bf16x4*>(
+                sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 4 * g));
+            const bf16x4 v1 = *reinterpret_cast<const bf16x4*>(
+                sVT + loff<KB>(16 * df + (lane & 15), 32 * c + 16 + 4 * g));
+            bf16x8 vfr;
+#pragma unroll
+            for (int e = 0; e < 4; ++e) {
+              vfr[e] = v0[e];
+              vfr[e + 4] = v1[e];
+            }
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+              acc_o[df][2 * hp + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  vfr, pb[nf], acc_o[df][2 * hp + nf], 0, 0, 0);
+          }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
-      __builtin_amdgcn_s_setprio(0);
     }
     if (kv0 + KB < kv_end) stage_write((t + 1) & 1);  // other buffer: no
     BAR_LDS();  // wait for readers of this buffer AND the writes above
@@ -325,7 +342,7 @@ void flash_fwd_kernel(const bf16_t* __restrict__ Q,
 
   // --- epilogue: O = O^T^T / l, write O and lse ---
 #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
+  for (int nf = 0; nf < NQ; ++nf) {
     const int qg = wq0 + nf * 16 + (lane & 15);
     if (qg >= S) continue;
     const float lv = l_r[nf];
@@ -349,15 +366,18 @@ void attention_fwd_bf16(const void* q, const void* k, const void* v, void* o,
                         bool causal, int64_t q_bs, int64_t q_hs, int64_t q_rs,
                         int64_t o_bs, int64_t o_hs, int64_t o_rs,
                         hipStream_t stream) {
-  dim3 grid((S + QBF - 1) / QBF, B * H);
   dim3 block(NTF);
-#define FWD_D(DD)                                                       \
-  hipLaunchKernelGGL(flash_fwd_kernel<DD>, grid, block, 0, stream,      \
-                     static_cast<const bf16_t*>(q),                     \
-                     static_cast<const bf16_t*>(k),                     \
-                     static_cast<const bf16_t*>(v),                     \
-                     static_cast<bf16_t*>(o), lse, S, H, scale, causal, \
-                     q_bs, q_hs, q_rs, o_bs, o_hs, o_rs)
+#define FWD_D(DD)                                                        \
+  do {                                                                   \
+    dim3 grid((S + (DD == 64 ? QBF : 256) - 1) / (DD == 64 ? QBF : 256), \
+              B * H);                                                    \
+    hipLaunchKernelGGL(flash_fwd_kernel<DD>, grid, block, 0, stream,     \
+                       static_cast<const bf16_t*>(q),                    \
+                       static_cast<const bf16_t*>(k),                    \
+                       static_cast<const bf16_t*>(v),                    \
+                       static_cast<bf16_t*>(o), lse, S, H, scale,        \
+                       causal, q_bs, q_hs, q_rs, o_bs, o_hs, o_rs);      \
+  } while (0)
   if (D == 64) FWD_D(64);
   else if (D == 128) FWD_D(128);
   else throw std::runtime_error("flash fwd: head dim must be 64 or 128");
