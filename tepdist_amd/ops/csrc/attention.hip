@@ -787,10 +787,13 @@ void flash_bwd_kv_kernel(const bf16_t* __restrict__ Q,
 // straight from the accumulators (k-permutation) against a
 // transpose-staged K^T — no LDS scatter, no atomics, bf16 output written
 // directly (strided; serves the packed-qkv layout too).
-constexpr int NTB = 512;   // 8 waves x 32 q columns (QBB = 256)
-constexpr int QBB = 256;
+constexpr int NTB = 512;   // 8 waves; D=64 blocks take 512 q (two
+                           // sequential 32-q pairs per wave — the same
+                           // staging-amortization as fwd/bwd-kv), D=128
+                           // keeps 256 (register budget)
+constexpr int QBB = 256;   // D=128 block size; D=64 uses 128*NQ
 
-template <int D>
+template <int D, int NQ = (D == 64 ? 4 : 2)>
 __launch_bounds__(NTB) __global__
 void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                          const bf16_t* __restrict__ K,
@@ -806,11 +809,11 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   constexpr int DF = D / 16;
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
-  const int q0 = blockIdx.x * QBB;
+  const int q0 = blockIdx.x * (128 * NQ);
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int g = lane >> 4;
-  const int wq0 = q0 + wave * 32;
+  const int wq0 = q0 + wave * 16 * NQ;
 
   const int64_t qoff = (int64_t)b * q_bs + (int64_t)h * q_hs;
   const int64_t ooff = (int64_t)b * o_bs + (int64_t)h * o_hs;
@@ -826,10 +829,10 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   bf16_t* sKT = sV + KB * D;            // [D][KB] (dQ^T A-operand)
 
   // Q and dO B-fragments + per-lane lse/delta (block-constant)
-  bf16x8 qf[2][DK], dof[2][DK];
-  float lse_r[2], dl_r[2];
+  bf16x8 qf[NQ][DK], dof[NQ][DK];
+  float lse_r[NQ], dl_r[NQ];
 #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
+  for (int nf = 0; nf < NQ; ++nf) {
     const int row = wq0 + nf * 16 + (lane & 15);
     lse_r[nf] = (row < S) ? LSE[(int64_t)bh * S + row] : -3.0e38f;
     dl_r[nf] = (row < S) ? DELTA[(int64_t)bh * S + row] : 0.f;
@@ -847,7 +850,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
-  f32x4 acc_dq[DF][2] = {};   // dQ^T: d = 16df+4g+e, q = lane&15 (+16nf)
+  f32x4 acc_dq[DF][NQ] = {};  // dQ^T: d = 16df+4g+e, q = lane&15 (+16nf)
 
   // staging registers: K,V natural (16B each) + K^T v_perm slabs
   constexpr int KUN = KB * D / 8 / NTB;
@@ -918,18 +921,22 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
   };
 
-  const int kv_end = causal ? min(S, q0 + QBB) : S;
+  const int kv_end = causal ? min(S, q0 + 128 * NQ) : S;
   stage_load(0);
   stage_write();
   __syncthreads();
   for (int kv0 = 0; kv0 < kv_end; kv0 += KB) {
     if (kv0 + KB < kv_end) stage_load(kv0 + KB);
 
-    if (!causal || kv0 <= wq0 + 31) {
+    if (!causal || kv0 <= wq0 + 16 * NQ - 1) {
+#pragma unroll
+     for (int hp = 0; hp < NQ / 2; ++hp) {
+      const int pq0 = wq0 + 32 * hp;
+      if (causal && kv0 > pq0 + 31) continue;
       // per 16-kv block: S^T = K Q^T and dP^T = V dO^T, then dS^T packed
       // to bf16 immediately (keeps only two f32x4 accumulator pairs live
       // at a time — register pressure gates a second wave per SIMD)
-      const bool inner = kv0 + KB <= wq0 && kv0 + KB <= S && wq0 + 32 <= S;
+      const bool inner = kv0 + KB <= pq0 && kv0 + KB <= S && pq0 + 32 <= S;
       bf16x4 dsb[4][2];   // dS^T bf16: kv = 16mi+4g+e, q = lane&15+16nf
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
@@ -945,21 +952,22 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
           for (int nf = 0; nf < 2; ++nf) {
             st[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                ka, qf[nf][kk], st[nf], 0, 0, 0);
+                ka, qf[2 * hp + nf][kk], st[nf], 0, 0, 0);
             dpt[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                va, dof[nf][kk], dpt[nf], 0, 0, 0);
+                va, dof[2 * hp + nf][kk], dpt[nf], 0, 0, 0);
           }
         }
         __builtin_amdgcn_s_setprio(0);
 #pragma unroll
         for (int nf = 0; nf < 2; ++nf) {
-          const int qg = wq0 + nf * 16 + (lane & 15);
+          const int nfg = 2 * hp + nf;
+          const int qg = pq0 + nf * 16 + (lane & 15);
           if (inner) {
 #pragma unroll
             for (int e = 0; e < 4; ++e) {
-              const float pt = fast_exp(st[nf][e] * scale - lse_r[nf]);
+              const float pt = fast_exp(st[nf][e] * scale - lse_r[nfg]);
               dsb[mi][nf][e] =
-                  f2bf(pt * (dpt[nf][e] - dl_r[nf]) * scale);
+                  f2bf(pt * (dpt[nf][e] - dl_r[nfg]) * scale);
             }
           } else {
 #pragma unroll
@@ -967,12 +975,12 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
               const int kg = kv0 + 16 * mi + 4 * g + e;
               const bool valid = qg < S && kg < S &&
                                  (!causal || kg <= qg) &&
-                                 lse_r[nf] > -1.0e38f;
-              const float arg = valid ? st[nf][e] * scale - lse_r[nf]
+                                 lse_r[nfg] > -1.0e38f;
+              const float arg = valid ? st[nf][e] * scale - lse_r[nfg]
                                       : -3.0e38f;
               const float pt = fast_exp(arg);
               dsb[mi][nf][e] =
-                  f2bf(pt * (dpt[nf][e] - dl_r[nf]) * scale);
+                  f2bf(pt * (dpt[nf][e] - dl_r[nfg]) * scale);
             }
           }
         }
@@ -1003,11 +1011,12 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
           }
 #pragma unroll
           for (int nf = 0; nf < 2; ++nf)
-            acc_dq[df][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                kfr, db[nf], acc_dq[df][nf], 0, 0, 0);
+            acc_dq[df][2 * hp + nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kfr, db[nf], acc_dq[df][2 * hp + nf], 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
+     }
     }
     BAR_LDS();
     if (kv0 + KB < kv_end) {
@@ -1018,7 +1027,7 @@ void flash_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 
   // write dQ (transposed-accumulator scatter, bf16, strided layout)
 #pragma unroll
-  for (int nf = 0; nf < 2; ++nf) {
+  for (int nf = 0; nf < NQ; ++nf) {
     const int qg = wq0 + nf * 16 + (lane & 15);
     if (qg >= S) continue;
 #pragma unroll
@@ -1113,7 +1122,8 @@ void attention_bwd_bf16(const void* q, const void* k, const void* v,
                        static_cast<bf16_t*>(dk), static_cast<bf16_t*>(dv),   \
                        S, H, scale, causal, q_bs, q_hs, q_rs, o_bs, o_hs,    \
                        o_rs);                                                \
-    dim3 qgrid((S + QBB - 1) / QBB, B * H);                                  \
+    dim3 qgrid((S + (DD == 64 ? 511 : 255)) / (DD == 64 ? 512 : 256),       \
+               B * H);                                                       \
     hipLaunchKernelGGL(flash_bwd_dq_kernel<DD>, qgrid, dim3(NTB), 0, stream, \
                        static_cast<const bf16_t*>(q),                        \
                        static_cast<const bf16_t*>(k),                        \
