@@ -20,11 +20,13 @@ void gelu_fwd_kernel(const bf16_t* __restrict__ x, bf16_t* __restrict__ y,
   for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
        i0 < n; i0 += stride) {
     if (i0 + 8 <= n) {
-      const bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i0);
+      // streaming, never re-read: bypass L2 (nontemporal)
+      const bf16x8 xv = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16x8*>(x + i0));
       bf16x8 yv;
 #pragma unroll
       for (int e = 0; e < 8; ++e) yv[e] = f2bf(gelu_f(bf2f(xv[e])));
-      *reinterpret_cast<bf16x8*>(y + i0) = yv;
+      __builtin_nontemporal_store(yv, reinterpret_cast<bf16x8*>(y + i0));
     } else {
       for (int64_t i = i0; i < n; ++i) y[i] = f2bf(gelu_f(bf2f(x[i])));
     }
@@ -39,13 +41,15 @@ void gelu_bwd_kernel(const bf16_t* __restrict__ dy,
   for (int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
        i0 < n; i0 += stride) {
     if (i0 + 8 <= n) {
-      const bf16x8 dv = *reinterpret_cast<const bf16x8*>(dy + i0);
-      const bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i0);
+      const bf16x8 dv = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16x8*>(dy + i0));
+      const bf16x8 xv = __builtin_nontemporal_load(
+          reinterpret_cast<const bf16x8*>(x + i0));
       bf16x8 o;
 #pragma unroll
       for (int e = 0; e < 8; ++e)
         o[e] = f2bf(bf2f(dv[e]) * gelu_grad_f(bf2f(xv[e])));
-      *reinterpret_cast<bf16x8*>(dx + i0) = o;
+      __builtin_nontemporal_store(o, reinterpret_cast<bf16x8*>(dx + i0));
     } else {
       for (int64_t i = i0; i < n; ++i)
         dx[i] = f2bf(bf2f(dy[i]) * gelu_grad_f(bf2f(x[i])));
@@ -147,7 +151,7 @@ __global__ void cast_ws_kernel(const float* __restrict__ ws,
 }  // namespace
 
 void gelu_fwd_bf16(const void* x, void* y, int64_t n, hipStream_t stream) {
-  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 2048);
+  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 4096);
   hipLaunchKernelGGL(gelu_fwd_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
                      stream, static_cast<const bf16_t*>(x),
                      static_cast<bf16_t*>(y), n);
@@ -155,7 +159,7 @@ void gelu_fwd_bf16(const void* x, void* y, int64_t n, hipStream_t stream) {
 
 void gelu_bwd_bf16(const void* dy, const void* x, void* dx, int64_t n,
                    hipStream_t stream) {
-  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 2048);
+  const int blocks = (int)std::min<int64_t>((n / 8 + NT - 1) / NT, 4096);
   hipLaunchKernelGGL(gelu_bwd_kernel, dim3(std::max(blocks, 1)), dim3(NT), 0,
                      stream, static_cast<const bf16_t*>(dy),
                      static_cast<const bf16_t*>(x), static_cast<bf16_t*>(dx),
