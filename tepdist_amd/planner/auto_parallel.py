@@ -27,7 +27,7 @@ from typing import Dict, List, Optional, Tuple
 
 from tepdist_amd.config import get_env
 from tepdist_amd.ir.graph import COMPUTE_SENSITIVE, Graph
-from tepdist_amd.planner.cost_model import Cost, CostModel, Evaluator
+from tepdist_amd.planner.cost_model import Cost, CostModel
 from tepdist_amd.planner.def_context import DefContextTree, build_def_tree
 from tepdist_amd.planner.dist_spec import DimStrategy, DistSpec
 from tepdist_amd.planner.pipeline import GraphSketch
