@@ -4,20 +4,19 @@ Re-implements the reference AutoParallel pass structure
 (auto_parallel.cc:395-409) on our IR:
 
   - three modes: EXPLORATION (default; enumerate power-of-2 device-split
-    proposals stages x mesh, plan each, pick min Evaluator cost), CONFIG
-    (NUM_STAGES / NUM_MICRO_BATCHES pinned via env), RULE (one-pass greedy
-    inference, no ILP);
+    proposals stages x mesh, plan each, pick min cost), CONFIG
+    (NUM_STAGES / NUM_MICRO_BATCHES pinned via env), RULE (the one-pass
+    FastSpmdStrategy inference engine, no ILP);
   - per proposal: SyncFree micro-batch analysis -> per-mesh-dim
     CostSpmdStrategy rounds -> GraphSketch ILP stage cut -> ZeRO memory
-    decision -> analytic Evaluator;
-  - output: ParallelPlan with per-node DistSpec stacks, per-node stage,
-    the DefContext tree, and the degrees (dp/tp/pp/micro) the executor
-    maps onto process groups.
-
-Mesh-round stacking note: round r>0 plans on the original shapes with
-costs already divided by the chosen spec's shard count; the second-order
-interaction between rounds is approximated rather than re-deriving shapes
-(the reference re-runs SpmdTransform between rounds)."""
+    decision -> planner/evaluate.py, which APPLIES the proposal
+    (multi-round transform per stage subgraph) and prices the transformed
+    result — collective bytes from the inserted nodes, pipeline bubble
+    from a schedule simulation (the reference evaluates the
+    actually-transformed module, auto_parallel.cc:236-324);
+  - output: ParallelPlan with per-node DistSpec stacks + mesh_rounds /
+    dp_round_flags (what multi_round_transform and the executors consume),
+    per-node stage, and the DefContext tree."""
 
 from __future__ import annotations
 
