@@ -5,7 +5,7 @@ from TF):
   - from_fx: generic torch.fx symbolic trace + shape propagation for simple
     models whose forward is built from tepdist_amd.ops calls (the
     smoke_testing MLP class of clients);
-  - gpt2_ir / moe_ir: explicit exporters for the benchmark model families
+  - gpt2_ir / llama_ir: explicit exporters for the benchmark model families
     (the client KNOWS its graph; op_group tags = layer ids, which is what
     the reference's dapple_scope provides).
 
@@ -197,4 +197,90 @@ def from_fx(model: torch.nn.Module, *example_args) -> Graph:
             args = n.args[0]
             outs = args if isinstance(args, (tuple, list)) else [args]
             g.outputs = [env[a].id for a in outs if isinstance(a, fx.Node)]
+    return g
+
+
+def moe_ir(cfg, batch: int, seq: int,
+           capacity_factor: float = None) -> Graph:
+    """GPT-MoE exporter (models/moe.py semantics; reference
+    examples/gpt_moe): GShard-style MoE FFN layers expressed STATICALLY —
+    `moe_dispatch` builds the group-blocked [E, capacity, d] tensor,
+    the experts run as batched matmuls over the expert dim, and
+    `moe_combine` returns tokens. Expert parallelism is then a planner
+    RESHARD: dispatch is capacity-split (each token shard fills its
+    capacity block), the expert matmuls want expert-dim splits, and the
+    SpmdTransform inserts the all-to-all on the mismatch (the reference's
+    kDAPPLEAllToAll path, SURVEY.md §2.7 EP). Auxiliary load-balancing
+    loss is a training-loop concern and not part of the planned graph."""
+    from tepdist_amd.models.configs import MoEConfig  # noqa: F401
+    g = Graph()
+    V, d, H, L = cfg.padded_vocab, cfg.n_embd, cfg.n_head, cfg.n_layer
+    E, K = cfg.num_experts, cfg.top_k
+    cf = capacity_factor if capacity_factor is not None \
+        else cfg.capacity_factor
+    BS = batch * seq
+    C = max(int(cf * BS * K / E), 4)
+    ids = g.add_input("input_ids", (BS,), "i64")
+    ids.attrs["batch"] = batch
+    labels = g.add_input("labels", (BS,), "i64")
+    labels.attrs["batch"] = batch
+    wte = g.add_param("wte", (V, d))
+    wpe = g.add_param("wpe", (cfg.n_ctx, d))
+    pos = g.add("data", [], (BS,), "i64", attrs={"batch": batch}, name="pos")
+    xe = g.add("embedding", [ids, wte], (BS, d), attrs={"batch": batch})
+    xp = g.add("embedding", [pos, wpe], (BS, d), attrs={"batch": batch})
+    x = g.add("add", [xe, xp], (BS, d), attrs={"batch": batch})
+
+    for l in range(L):
+        a = {"batch": batch, "heads": H, "seq": seq}
+        ln1g = g.add_param(f"h{l}.ln1_g", (d,), op_group=l)
+        ln1b = g.add_param(f"h{l}.ln1_b", (d,), op_group=l)
+        h = g.add("layernorm", [x, ln1g, ln1b], (BS, d), attrs=a, op_group=l)
+        wqkv = g.add_param(f"h{l}.w_qkv", (3 * d, d), op_group=l)
+        bqkv = g.add_param(f"h{l}.b_qkv", (3 * d,), op_group=l)
+        qkv = g.add("linear", [h, wqkv, bqkv], (BS, 3 * d), attrs=a,
+                    op_group=l)
+        att = g.add("attention_qkv", [qkv], (BS, d), attrs=a, op_group=l)
+        wproj = g.add_param(f"h{l}.w_proj", (d, d), op_group=l)
+        bproj = g.add_param(f"h{l}.b_proj", (d,), op_group=l)
+        pr = g.add("linear", [att, wproj, bproj], (BS, d), attrs=a,
+                   op_group=l)
+        x = g.add("add", [x, pr], (BS, d), attrs=a, op_group=l)
+        ln2g = g.add_param(f"h{l}.ln2_g", (d,), op_group=l)
+        ln2b = g.add_param(f"h{l}.ln2_b", (d,), op_group=l)
+        h2 = g.add("layernorm", [x, ln2g, ln2b], (BS, d), attrs=a,
+                   op_group=l)
+        if (l + 1) % cfg.moe_every == 0:
+            # MoE FFN: gate -> dispatch -> expert batched matmuls -> combine
+            wgate = g.add_param(f"h{l}.moe_gate", (E, d), op_group=l)
+            glog = g.add("linear", [h2, wgate], (BS, E), attrs=a, op_group=l)
+            disp = g.add("moe_dispatch", [h2, glog], (E, C, d),
+                         attrs={"batch": batch, "k": K}, op_group=l)
+            w1 = g.add_param(f"h{l}.moe_w1", (E, d, 4 * d), op_group=l)
+            hh = g.add("matmul", [disp, w1], (E, C, 4 * d), op_group=l)
+            hg = g.add("gelu", [hh], (E, C, 4 * d), op_group=l)
+            w2 = g.add_param(f"h{l}.moe_w2", (E, 4 * d, d), op_group=l)
+            ye = g.add("matmul", [hg, w2], (E, C, d), op_group=l)
+            o = g.add("moe_combine", [ye, h2, glog], (BS, d),
+                      attrs={"batch": batch, "k": K}, op_group=l)
+        else:
+            wfc = g.add_param(f"h{l}.w_fc", (4 * d, d), op_group=l)
+            bfc = g.add_param(f"h{l}.b_fc", (4 * d,), op_group=l)
+            f = g.add("linear", [h2, wfc, bfc], (BS, 4 * d),
+                      attrs={**a, "act": "gelu"}, op_group=l)
+            wout = g.add_param(f"h{l}.w_out", (d, 4 * d), op_group=l)
+            bout = g.add_param(f"h{l}.b_out", (d,), op_group=l)
+            o = g.add("linear", [f, wout, bout], (BS, d), attrs=a,
+                      op_group=l)
+        x = g.add("add", [x, o], (BS, d), attrs=a, op_group=l)
+
+    lnfg = g.add_param("lnf_g", (d,), op_group=L - 1)
+    lnfb = g.add_param("lnf_b", (d,), op_group=L - 1)
+    x = g.add("layernorm", [x, lnfg, lnfb], (BS, d),
+              attrs={"batch": batch}, op_group=L - 1)
+    logits = g.add("linear", [x, wte], (BS, V), attrs={"batch": batch},
+                   op_group=L - 1)
+    loss = g.add("cross_entropy", [logits, labels], (),
+                 attrs={"batch": batch}, op_group=L - 1)
+    g.outputs = [loss.id]
     return g

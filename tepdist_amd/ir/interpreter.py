@@ -99,6 +99,18 @@ class GraphInterpreter:
             qkv = ins[0].reshape(b, s, -1)
             o = ops.attention_qkv(qkv, n.attrs["heads"], causal=True)
             return o.reshape(b * s, -1)
+        if n.op == "moe_dispatch":
+            # GShard static dispatch -> [E, C_local, d] (C_local from the
+            # node's possibly-sharded shape: each token shard fills its
+            # own capacity block — what makes EP a planner reshard)
+            from tepdist_amd.models.moe import static_dispatch
+            gates = torch.softmax(ins[1].float(), -1)
+            D, _ = static_dispatch(ins[0], gates, n.attrs["k"], n.shape[1])
+            return D
+        if n.op == "moe_combine":
+            from tepdist_amd.models.moe import static_combine
+            gates = torch.softmax(ins[2].float(), -1)
+            return static_combine(ins[0], ins[1], gates, n.attrs["k"])
         if n.op == "split":
             dim = n.attrs.get("dim", -1)
             idx = n.attrs.get("index", 0)

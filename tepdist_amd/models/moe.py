@@ -278,3 +278,48 @@ class GPTMoE(nn.Module):
             if blk.moe is not None:
                 blk.moe.aux_loss = blk.moe.aux_loss.detach()
         return loss
+
+
+# -- functional static-capacity dispatch/combine (the IR ops' semantics) ----
+
+def static_dispatch(x: torch.Tensor, gates: torch.Tensor, k: int,
+                    capacity: int):
+    """GShard dispatch with BLOCKED capacity: x [T, d], gates [T, E] ->
+    D [E, C, d]. Deterministic in (x, gates); autograd flows through the
+    gather/scatter and the combine weights. Shared by MoELayer and the
+    planner IR's moe_dispatch op (capacity is per token GROUP, so a
+    token-sharded dispatch is exactly the capacity-dim narrow of the
+    grouped tensor — what makes expert parallelism a planner reshard)."""
+    T, d = x.shape
+    E = gates.shape[1]
+    topv, topi = torch.topk(gates.float(), k, dim=-1)
+    topv = topv / topv.sum(-1, keepdim=True).clamp_min(1e-9)
+    flat_e = topi.reshape(-1)
+    flat_t = torch.arange(T, device=x.device).repeat_interleave(k)
+    order = torch.argsort(flat_e, stable=True)
+    counts = torch.zeros(E, dtype=torch.long, device=x.device).scatter_add_(
+        0, flat_e, torch.ones_like(flat_e))
+    offs = torch.cumsum(counts, 0) - counts
+    r = torch.arange(flat_e.numel(), device=x.device)
+    pos = torch.empty_like(r)
+    pos[order] = r - offs[flat_e[order]]
+    keep = pos < capacity
+    slot = flat_e * capacity + pos
+    slot_safe = torch.where(keep, slot, torch.zeros_like(slot))
+    contrib = x[flat_t] * keep.unsqueeze(-1).to(x.dtype)
+    D = torch.zeros(E * capacity, d, dtype=x.dtype, device=x.device)
+    D = D.index_put((slot_safe,), contrib, accumulate=True)
+    return D.reshape(E, capacity, d), (flat_t, slot_safe, keep,
+                                       topv.reshape(-1))
+
+
+def static_combine(y: torch.Tensor, x: torch.Tensor, gates: torch.Tensor,
+                   k: int):
+    """Inverse of static_dispatch: y [E, C, d] expert outputs -> [T, d]
+    (recomputes the deterministic slot map from gates)."""
+    E, C, d = y.shape
+    _, (flat_t, slot_safe, keep, flat_w) = static_dispatch(x, gates, k, C)
+    back = y.reshape(E * C, d)
+    gathered = back[slot_safe] * \
+        (flat_w * keep.to(flat_w.dtype)).unsqueeze(-1).to(back.dtype)
+    return torch.zeros_like(x).index_add(0, flat_t, gathered.to(x.dtype))

@@ -126,6 +126,26 @@ def op_strategies(g: Graph, node: Node, n: int) -> List[OpStrategy]:
         out.append(rep())
         return out
 
+    if node.op == "moe_dispatch":
+        # [T,d] x [T,E] -> [E, C, d] group-blocked capacity: a token
+        # shard fills ITS capacity block, so the capacity dim (1) splits
+        # with the token inputs; the downstream expert matmuls want
+        # expert-dim (0) splits — the planner reshard between them is the
+        # MoE all-to-all (reference kDAPPLEAllToAll, SURVEY §2.7 EP)
+        if node.shape[1] % n == 0 and _split0_ok(g.nodes[node.inputs[0]], n):
+            out.append(OpStrategy(S(1, n), (S(0, n), S(0, n)), "ep"))
+        out.append(rep())
+        return out
+
+    if node.op == "moe_combine":
+        # [E,C,d] x [T,d] x [T,E] -> [T,d]: token-split output consumes
+        # the capacity-split expert results for ITS token shard
+        if _split0_ok(node, n) and g.nodes[node.inputs[0]].shape[1] % n == 0:
+            out.append(OpStrategy(S(0, n), (S(1, n), S(0, n), S(0, n)),
+                                  "ep"))
+        out.append(rep())
+        return out
+
     if node.op == "attention_qkv":
         # packed [BS, 3d] projection in, [BS, d] out: batch split only (a
         # head split would need a strided/per-section weight shard — the

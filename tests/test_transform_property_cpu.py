@@ -33,6 +33,9 @@ CASES = [
     # world-4 case (divisibility/spec-consistency at higher shard counts)
     ("gpt2", dict(n_layer=1, n_embd=128, n_head=8, vocab_size=256,
                   n_ctx=32), 8, 16),
+    # MoE graph: the cost search must produce an executable plan over the
+    # moe_dispatch/combine + batched-expert-matmul op family
+    ("moe", dict(), 4, 16),
 ]
 
 
@@ -40,6 +43,12 @@ def _build(kind, cfg_kw, batch, seq):
     if kind == "gpt2":
         cfg = GPT2Config(name="t", **cfg_kw)
         return gpt2_ir(cfg, batch=batch, seq=seq), cfg.padded_vocab
+    if kind == "moe":
+        from tepdist_amd.ir.capture import moe_ir
+        from tepdist_amd.models.configs import MOE_CONFIGS
+        cfg = MOE_CONFIGS["gpt-moe-test"]
+        return moe_ir(cfg, batch=batch, seq=seq,
+                      capacity_factor=4.0), cfg.vocab_size
     cfg = LlamaConfig(name="t", **cfg_kw)
     return llama_ir(cfg, batch=batch, seq=seq), cfg.vocab_size
 
