@@ -284,3 +284,61 @@ def moe_ir(cfg, batch: int, seq: int,
                  attrs={"batch": batch}, op_group=L - 1)
     g.outputs = [loss.id]
     return g
+
+
+def wrn_ir(cfg, batch: int) -> Graph:
+    """Wide-ResNet PLANNER graph (models/wide_resnet.py structure;
+    reference examples/wide_resnet): conv2d/add/linear/cross_entropy
+    nodes with real shapes and flops so AutoParallel can search the
+    image path (batch vs out-channel splits, stage cuts, memory). The
+    execution engine for WRN remains the module path (conv-as-GEMM
+    kernels + Trainer); this graph is the planning surface."""
+    from tepdist_amd.models.wide_resnet import _DEPTH
+    g = Graph()
+    blocks = _DEPTH[cfg.n_layer]
+    base = 64 * cfg.width_factor
+    r = cfg.image_size // 4          # stem 7x7/2 + maxpool/2
+    x = g.add_input("images", (batch, 3, cfg.image_size, cfg.image_size))
+    x.attrs["batch"] = batch
+    labels = g.add_input("labels", (batch,), "i64")
+    w = g.add_param("stem.w", (base, 3, 7, 7))
+    x = g.add("conv2d", [x, w], (batch, base, r, r),
+              attrs={"batch": batch, "kernel": (7, 7)})
+    cin = base
+    for si, nb in enumerate(blocks):
+        planes = base * (2 ** si)
+        cout = planes * 4
+        stride = 1 if si == 0 else 2
+        if si > 0:
+            r //= 2
+        for b in range(nb):
+            grp = si * 10 + b
+            a = {"batch": batch}
+            w1 = g.add_param(f"s{si}.b{b}.w1", (planes, cin, 1, 1),
+                             op_group=grp)
+            h = g.add("conv2d", [x, w1], (batch, planes, r, r),
+                      attrs={**a, "kernel": (1, 1)}, op_group=grp)
+            w2 = g.add_param(f"s{si}.b{b}.w2", (planes, planes, 3, 3),
+                             op_group=grp)
+            h = g.add("conv2d", [h, w2], (batch, planes, r, r),
+                      attrs={**a, "kernel": (3, 3)}, op_group=grp)
+            w3 = g.add_param(f"s{si}.b{b}.w3", (cout, planes, 1, 1),
+                             op_group=grp)
+            h = g.add("conv2d", [h, w3], (batch, cout, r, r),
+                      attrs={**a, "kernel": (1, 1)}, op_group=grp)
+            if cin != cout:
+                wd = g.add_param(f"s{si}.b{b}.down", (cout, cin, 1, 1),
+                                 op_group=grp)
+                x = g.add("conv2d", [x, wd], (batch, cout, r, r),
+                          attrs={**a, "kernel": (1, 1)}, op_group=grp)
+            x = g.add("add", [x, h], (batch, cout, r, r), attrs=a,
+                      op_group=grp)
+            cin = cout
+    x = g.add("elementwise", [x], (batch, cin), attrs={"batch": batch})
+    wfc = g.add_param("fc.w", (cfg.num_classes, cin))
+    logits = g.add("linear", [x, wfc], (batch, cfg.num_classes),
+                   attrs={"batch": batch})
+    loss = g.add("cross_entropy", [logits, labels], (),
+                 attrs={"batch": batch})
+    g.outputs = [loss.id]
+    return g
