@@ -140,7 +140,8 @@ def main():
                               plan, device=device, dtype=dtype)
         if rank == 0:
             print(f"# {model.describe()}", flush=True)
-        opt = AdamW(model.parameters(), lr=1e-4)
+        opt = model.make_optimizer(lr=1e-4)   # ZeroAdamW when the plan's
+        # ZeRO decision shards optimizer state over the dp group
         reducer = model.make_reducer()
         grad_accum = max(min(micro, global_batch), 1)
         trainer = Trainer(model, opt, grad_accum_steps=grad_accum,

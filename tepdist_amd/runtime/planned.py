@@ -106,6 +106,23 @@ class PlannedModule(nn.Module):
                                        dtype=dtype,
                                        groups=self.comm.groups_dict())
 
+    def make_optimizer(self, lr: float = 1e-4, **kw):
+        """The plan's optimizer: when the ZeRO decision shards optimizer
+        state (planner/zero.py SplitPlanByMemCost role) and this module
+        has a dp round, fp32 masters + moments shard over the dp group
+        (ZeroAdamW); otherwise the fused AdamW."""
+        from tepdist_amd.train.optim import AdamW, ZeroAdamW
+        zp = getattr(self.plan, "zero", None)
+        dp_rounds = sorted({r for rounds in
+                            self.transform.grad_sync_params.values()
+                            for r in rounds})
+        if zp is not None and getattr(zp, "shard_optimizer", False) \
+                and dp_rounds:
+            grp = self.comm.rounds_group(dp_rounds)
+            if grp is not None or self.comm.world > 1:
+                return ZeroAdamW(self.parameters(), group=grp, lr=lr, **kw)
+        return AdamW(self.parameters(), lr=lr, **kw)
+
     def make_reducer(self, bucket_bytes: int = 64 << 20):
         """One bucketed SUM-mode reducer per distinct round-SET: a param
         whose gradient must be summed over several dp rounds gets a single

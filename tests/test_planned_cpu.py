@@ -196,3 +196,51 @@ def _hybrid_worker(rank, world, port):
 def test_planned_dp_x_tp_hybrid():
     port = free_port()
     mp.spawn(_hybrid_worker, args=(4, port), nprocs=4, join=True)
+
+
+def _zero_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from tepdist_amd.planner.auto_parallel import AutoParallel
+        from tepdist_amd.runtime.planned import PlannedModule
+        from tepdist_amd.train.optim import ZeroAdamW
+        g, cfg = _build_case()
+        plan = AutoParallel(g, world)._plan_proposal(1, [world])
+        plan.zero.shard_optimizer = True    # force the ZeRO decision
+        m = PlannedModule(g, plan)
+        opt = m.make_optimizer(lr=1e-3)
+        assert isinstance(opt, ZeroAdamW)
+        # each rank owns a disjoint subset; union covers everything
+        owned = torch.tensor([sum(p.numel() for p in opt._owned)])
+        tot = torch.tensor([sum(p.numel() for p in m.parameters())])
+        all_owned = [torch.zeros_like(owned) for _ in range(world)]
+        dist.all_gather(all_owned, owned)
+        assert sum(int(t) for t in all_owned) == int(tot)
+        reducer = m.make_reducer()
+        ids, labels = _feeds(128, 64)
+        for _ in range(2):
+            opt.zero_grad()
+            if reducer:
+                reducer.reset(); reducer.arm()
+            loss = m(ids, labels)
+            loss.backward()
+            if reducer:
+                reducer.finalize()
+            opt.step()
+        # after ZeRO broadcast all ranks hold identical params
+        w = m.vars["wte"].detach()
+        ws = [torch.zeros_like(w) for _ in range(world)]
+        dist.all_gather(ws, w)
+        for t in ws[1:]:
+            assert torch.equal(t, ws[0])
+        assert torch.isfinite(loss).all()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_planned_zero_optimizer():
+    port = free_port()
+    mp.spawn(_zero_worker, args=(2, port), nprocs=2, join=True)
