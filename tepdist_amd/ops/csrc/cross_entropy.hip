@@ -1,7 +1,9 @@
 // Fused cross entropy (log-softmax + NLL) over large vocab rows for gfx950.
-// fwd: one 256-thread block per row computes max + logsumexp (two passes,
-// second pass L2-resident), writes per-row nll and lse.
-// bwd: elementwise d_logits = (softmax - onehot) * dloss/n, vectorized.
+// fwd: one 256-thread block per row computes logsumexp in a SINGLE online
+// pass (flash-style running max/sum rescale — the two-pass version
+// re-read each 100 KB row and only partly hit L2), writes nll and lse.
+// bwd: elementwise d_logits = (softmax - onehot) * dloss/n, vectorized;
+// nontemporal streams (nothing is re-read).
 
 #include <algorithm>
 #include <stdexcept>
@@ -23,38 +25,40 @@ void ce_fwd_kernel(const bf16_t* __restrict__ logits,
   __shared__ float scratch[NT / WAVE];
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const bf16_t* lr = logits + row * cols;
+    // online logsumexp: one pass, per-thread running (max, sum)
     float m = -3.0e38f;
+    float s = 0.f;
     for (int c0 = threadIdx.x * 8; c0 < cols; c0 += NT * 8) {
-      bf16x8 v = {};
+      bf16x8 v;
       if (c0 + 8 <= cols) {
-        v = *reinterpret_cast<const bf16x8*>(lr + c0);
+        v = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8*>(lr + c0));
       } else {
         for (int e = 0; e < 8 && c0 + e < cols; ++e) v[e] = lr[c0 + e];
         for (int e = cols - c0; e < 8; ++e) v[e] = f2bf(-3.0e38f);
       }
+      float cm = -3.0e38f;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) m = fmaxf(m, bf2f(v[e]));
+      for (int e = 0; e < 8; ++e) cm = fmaxf(cm, bf2f(v[e]));
+      const float mn = fmaxf(m, cm);
+      float cs = 0.f;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) cs += __expf(bf2f(v[e]) - mn);
+      s = s * __expf(m - mn) + cs;
+      m = mn;
     }
-    m = block_allreduce(m, scratch, NT / WAVE,
-                        [](float a, float b) { return fmaxf(a, b); },
-                        -3.0e38f);
+    // combine per-thread (m, s) pairs: block max, then rescaled sums
+    const float bm = block_allreduce(m, scratch, NT / WAVE,
+                                     [](float a, float b) {
+                                       return fmaxf(a, b);
+                                     }, -3.0e38f);
     __syncthreads();
-    float s = 0.f;
-    for (int c0 = threadIdx.x * 8; c0 < cols; c0 += NT * 8) {
-      bf16x8 v = {};
-      if (c0 + 8 <= cols) {
-        v = *reinterpret_cast<const bf16x8*>(lr + c0);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) s += __expf(bf2f(v[e]) - m);
-      } else {
-        for (int e = 0; e < 8 && c0 + e < cols; ++e)
-          s += __expf(bf2f(lr[c0 + e]) - m);
-      }
-    }
+    s = (m > -3.0e38f) ? s * __expf(m - bm) : 0.f;
     s = block_allreduce(s, scratch, NT / WAVE,
                         [](float a, float b) { return a + b; }, 0.f);
+    const float m2 = bm;
     if (threadIdx.x == 0) {
-      const float lse = m + __logf(s);
+      const float lse = m2 + __logf(s);
       lse_out[row] = lse;
       const int64_t t = targets[row];
       // t < 0: either the ignore index or a "no target in this vocab
@@ -94,14 +98,15 @@ void ce_bwd_kernel(const bf16_t* __restrict__ logits,
     for (int c0 = threadIdx.x * 8; c0 < cols; c0 += NT * 8) {
       bf16x8 v = {}, o;
       if (c0 + 8 <= cols) {
-        v = *reinterpret_cast<const bf16x8*>(lr + c0);
+        v = __builtin_nontemporal_load(
+            reinterpret_cast<const bf16x8*>(lr + c0));
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
           float p = __expf(bf2f(v[e]) - l);
           if (c0 + e == t) p -= 1.f;
           o[e] = f2bf(p * dloss_over_n);
         }
-        *reinterpret_cast<bf16x8*>(dr + c0) = o;
+        __builtin_nontemporal_store(o, reinterpret_cast<bf16x8*>(dr + c0));
       } else {
         for (int e = 0; e < 8 && c0 + e < cols; ++e) {
           float p = __expf(bf2f(lr[c0 + e]) - l);
