@@ -121,9 +121,7 @@ class GraphInterpreter:
         if n.op == "mul":
             return ins[0] * ins[1]
         if n.op == "gelu":
-            from tepdist_amd.ops.interface import _backend
-            return _backend(ins[0]).gelu_fwd(ins[0]) if ins[0].is_cuda \
-                else torch.nn.functional.gelu(ins[0], approximate="tanh")
+            return ops.gelu(ins[0])
         if n.op == "dropout":
             return ops.dropout(ins[0], n.attrs.get("p", 0.0))
         if n.op == "cross_entropy":

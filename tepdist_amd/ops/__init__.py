@@ -1,4 +1,5 @@
 from tepdist_amd.ops.interface import (  # noqa: F401
+    gelu,
     linear,
     matmul,
     layernorm,

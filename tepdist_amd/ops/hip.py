@@ -169,6 +169,21 @@ def linear_fwd(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
     return fn()
 
 
+def gelu_fwd(x: torch.Tensor) -> torch.Tensor:
+    x = x.contiguous()
+    y = torch.empty_like(x)
+    ext.gelu_fwd(x.data_ptr(), y.data_ptr(), x.numel(), _stream())
+    return y
+
+
+def gelu_bwd(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+    dy, x = dy.contiguous(), x.contiguous()
+    dx = torch.empty_like(dy)
+    ext.gelu_bwd(dy.data_ptr(), x.data_ptr(), dx.data_ptr(), dy.numel(),
+                 _stream())
+    return dx
+
+
 def transpose2d(t: torch.Tensor) -> torch.Tensor:
     """Materialized bf16 transpose (tiled LDS kernel)."""
     R, C = t.shape[-2], t.shape[-1]

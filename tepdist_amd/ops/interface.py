@@ -200,6 +200,24 @@ def attention_qkv(qkv, heads: int, causal: bool = True):
 # --------------------------------------------------------------------------
 
 
+class GeluFn(torch.autograd.Function):
+    """Standalone tanh-gelu (the IR's `gelu` op; linear fuses its own)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        return _backend(x).gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return _backend(x).gelu_bwd(dy.contiguous(), x)
+
+
+def gelu(x):
+    return GeluFn.apply(x)
+
+
 class EmbeddingFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, ids, table):
