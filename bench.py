@@ -148,11 +148,20 @@ def main():
                           reducer=reducer)
         m_rows = max(global_batch // grad_accum, 1)
         # every rank consumes the IDENTICAL global batch stream; the
-        # planner-inserted dynamic_slice takes this rank's part
-        g = torch.Generator().manual_seed(4321)
+        # planner-inserted dynamic_slice takes this rank's part. Generate
+        # ON DEVICE (same seed + same GPU model => identical Philox
+        # sequences on every rank): at N=8 the global batch is ~4 MB of
+        # ids per step — CPU generation + pageable upload inside the
+        # timed loop would tax weak scaling.
+        if device.type == "cuda":
+            gdev = torch.Generator(device=device).manual_seed(4321)
+        else:
+            gdev = torch.Generator().manual_seed(4321)
 
         def run_step():
-            ids = make_ids(m_rows * grad_accum)
+            ids = torch.randint(0, cfg.vocab_size,
+                                (m_rows * grad_accum, seq + 1),
+                                generator=gdev, device=gdev.device)
             def bi(i):
                 sl = ids[i * m_rows:(i + 1) * m_rows]
                 return sl[:, :-1].to(device), sl[:, 1:].to(device)
