@@ -189,6 +189,9 @@ def main():
             mod, stage, pp, pp_ranks, micro, act_shape=mod.act_shape,
             act_dtype=dtype, device=device, reducer=reducer, optimizer=opt,
             pp_group=cdm.pipeline_pair_group(+1 if stage == 0 else -1))
+        # rank-IDENTICAL stream: the stage graphs dp-slice the global
+        # batch in-graph (dynamic_slice over the mesh round), so every
+        # rank must feed the same ids
         g = torch.Generator().manual_seed(4321)
 
         def run_step():
