@@ -55,16 +55,19 @@ def main():
 
     plan = None
     if rank == 0:
-        plan = AutoParallel(g, world).run()
+        ap = AutoParallel(g, world)
+        plan = ap.run()
+        if plan.pp > 1:
+            # this minimal script shows the flat-mesh module; bench.py's
+            # auto path runs pipeline plans through the stage-decomposed
+            # task-list executor
+            plan = ap._best_over_rounds(1, world)
         print(plan.summary(), flush=True)
     if world > 1:
         import torch.distributed as dist
         obj = [plan]
         dist.broadcast_object_list(obj, src=0)
         plan = obj[0]
-    if plan.pp > 1:
-        raise SystemExit("pipeline plan: see bench.py's stage-decomposed "
-                         "executor path for the full treatment")
 
     model = PlannedModule(g, plan, device=device, dtype=dtype)
     opt = model.make_optimizer(lr=1e-4)
