@@ -123,6 +123,8 @@ class GradReducer:
                     n = p.numel()
                     p.grad.reshape(-1).copy_(b.flat[off:off + n])
                     off += n
+                b.work = None   # idempotent finalize (AR task + AG task
+                # may both call it in the executor)
 
 
 def init_distributed(backend: Optional[str] = None) -> tuple:
