@@ -180,3 +180,32 @@ def test_moe_ir_graph_gpu():
     assert torch.isfinite(loss)
     assert vars_["h0.moe_w1"].grad is not None
     assert torch.isfinite(vars_["h0.moe_w1"].grad).all()
+
+
+@pytest.mark.gpu
+def test_ring_attention_flash_gpu():
+    """Ring attention's per-block flash path (world 1 degenerates to one
+    diagonal block): bf16 flash kernels vs the fp32 composed reference."""
+    from tepdist_amd.parallel.ring_attention import ring_attention
+    torch.manual_seed(0)
+    B, H, S, D = 2, 4, 256, 64
+    q = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = ring_attention(q, k, v, causal=True)
+    out.float().sum().backward()
+    q2, k2, v2 = (t.detach().float().clone().requires_grad_()
+                  for t in (q, k, v))
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q2, k2, v2, is_causal=True)
+    ref.sum().backward()
+    torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(q.grad.float(), q2.grad, rtol=5e-2,
+                               atol=5e-2)
+    torch.testing.assert_close(k.grad.float(), k2.grad, rtol=5e-2,
+                               atol=5e-2)
+    torch.testing.assert_close(v.grad.float(), v2.grad, rtol=5e-2,
+                               atol=5e-2)
