@@ -24,9 +24,14 @@ partial gradients, so dQ accumulates locally while (dK, dV) accumulate
 in buffers that travel WITH their K/V around the ring and arrive home
 after n steps.
 
-Note: contiguous blocks load-imbalance causal work (rank r computes r+1
-blocks); zigzag block assignment fixes that and is left as the next
-step."""
+Causal load balance: contiguous blocks make rank r compute r+1 blocks
+(the last rank bounds wall-clock). `zigzag=True` instead gives each rank
+chunks (r, 2W-1-r) of 2W global chunks; every step then computes exactly
+2 chunk-pair equivalents of work on every rank (the j==rank step is the
+two diagonals plus one full pair; j<rank is both q chunks against the
+LOW kv chunk; j>rank is the HIGH q chunk against both kv chunks), so
+causal work is identical across ranks. `zigzag_shard`/`zigzag_unshard`
+convert between a full sequence and the zigzag layout."""
 
 from __future__ import annotations
 
@@ -92,6 +97,36 @@ def _blk_bwd(dout, q, k, v, out_g, lse_g, delta, causal):
     dk = (ds.transpose(-1, -2) @ q.float()).to(q.dtype)
     dv = (p.transpose(-1, -2) @ dout.float()).to(q.dtype)
     return dq, dk, dv
+
+
+def _zz_pairs(rank, j, world):
+    """Visible (q_chunk, kv_chunk, causal) triples for the zigzag layout
+    when rank `rank` holds the kv blocks of rank `j`. Local chunk 0 is
+    global chunk `owner`, chunk 1 is global chunk 2*world-1-owner; chunk
+    visibility follows global causal order (see module docstring)."""
+    if j == rank:
+        return ((0, 0, True), (1, 0, False), (1, 1, True))
+    if j < rank:
+        return ((0, 0, False), (1, 0, False))
+    return ((1, 0, False), (1, 1, False))
+
+
+def zigzag_shard(x, world, dim=2):
+    """Full sequence -> list of per-rank zigzag shards: rank r gets
+    chunks (r, 2*world-1-r) of 2*world, concatenated along `dim`."""
+    chunks = x.chunk(2 * world, dim=dim)
+    return [torch.cat([chunks[r], chunks[2 * world - 1 - r]], dim=dim)
+            for r in range(world)]
+
+
+def zigzag_unshard(shards, dim=2):
+    """Inverse of zigzag_shard: per-rank shards -> full sequence."""
+    world = len(shards)
+    out = [None] * (2 * world)
+    for r, s in enumerate(shards):
+        a, b = s.chunk(2, dim=dim)
+        out[r], out[2 * world - 1 - r] = a, b
+    return torch.cat(out, dim=dim)
 
 
 class _RingAttention(torch.autograd.Function):
@@ -166,8 +201,93 @@ class _RingAttention(torch.autograd.Function):
         return dq, dk_acc, dv_acc, None, None
 
 
-def ring_attention(q, k, v, group=None, causal: bool = True):
-    """Context-parallel attention: q/k/v are this rank's contiguous
-    sequence block [B, H, S_local, D]; returns this rank's output block.
-    Exact (up to dtype rounding) vs full-sequence attention."""
+class _ZigzagRingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, group):
+        world = dist.get_world_size(group) if dist.is_initialized() else 1
+        if world == 1:
+            out, lse = _blk_fwd(q, k, v, True)
+            ctx.save_for_backward(q, k, v, out, lse)
+            ctx.group, ctx.world = group, 1
+            return out
+        rank = dist.get_rank(group)
+        ranks = dist.get_process_group_ranks(group) if group is not None \
+            else list(range(world))
+        nxt, prv = ranks[(rank + 1) % world], ranks[(rank - 1) % world]
+        c = q.shape[2] // 2
+        out = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        lse = torch.full(q.shape[:3], float("-inf"),
+                         dtype=torch.float32, device=q.device)
+        kj, vj = k, v
+        for step in range(world):
+            j = (rank - step) % world
+            if step + 1 < world:
+                nk, nv = _ring_sendrecv([kj, vj], prv, nxt, group)
+            for qa, kb, caus in _zz_pairs(rank, j, world):
+                o_p, lse_p = _blk_fwd(q.narrow(2, qa * c, c),
+                                      kj.narrow(2, kb * c, c),
+                                      vj.narrow(2, kb * c, c), caus)
+                sl = out.narrow(2, qa * c, c)
+                ll = lse.narrow(2, qa * c, c)
+                lse_n = torch.logaddexp(ll, lse_p)
+                sl.mul_(torch.exp(ll - lse_n).unsqueeze(-1))
+                sl.add_(o_p.float() *
+                        torch.exp(lse_p - lse_n).unsqueeze(-1))
+                ll.copy_(lse_n)
+            if step + 1 < world:
+                kj, vj = nk, nv
+        out = out.to(q.dtype)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.group, ctx.world = group, world
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        group, world = ctx.group, ctx.world
+        if world == 1:
+            dq, dk, dv = _blk_bwd(dout.contiguous(), q, k, v, out, lse,
+                                  (dout.float() * out.float()).sum(-1),
+                                  True)
+            return dq, dk, dv, None
+        rank = dist.get_rank(group)
+        ranks = dist.get_process_group_ranks(group) if group is not None \
+            else list(range(world))
+        nxt, prv = ranks[(rank + 1) % world], ranks[(rank - 1) % world]
+        c = q.shape[2] // 2
+        delta = (dout.float() * out.float()).sum(-1)
+        dout = dout.contiguous()
+        dq = torch.zeros_like(q)
+        kj, vj = k, v
+        dk_acc = torch.zeros_like(k)
+        dv_acc = torch.zeros_like(v)
+        for step in range(world):
+            j = (rank - step) % world
+            for qa, kb, caus in _zz_pairs(rank, j, world):
+                dq_p, dk_p, dv_p = _blk_bwd(
+                    dout.narrow(2, qa * c, c).contiguous(),
+                    q.narrow(2, qa * c, c).contiguous(),
+                    kj.narrow(2, kb * c, c).contiguous(),
+                    vj.narrow(2, kb * c, c).contiguous(),
+                    out.narrow(2, qa * c, c).contiguous(),
+                    lse.narrow(2, qa * c, c).contiguous(),
+                    delta.narrow(2, qa * c, c).contiguous(), caus)
+                dq.narrow(2, qa * c, c).add_(dq_p)
+                dk_acc.narrow(2, kb * c, c).add_(dk_p)
+                dv_acc.narrow(2, kb * c, c).add_(dv_p)
+            kj, vj, dk_acc, dv_acc = _ring_sendrecv(
+                [kj, vj, dk_acc, dv_acc], prv, nxt, group)
+        return dq, dk_acc, dv_acc, None
+
+
+def ring_attention(q, k, v, group=None, causal: bool = True,
+                   zigzag: bool = False):
+    """Context-parallel attention: q/k/v are this rank's sequence block
+    [B, H, S_local, D]; returns this rank's output block. Exact (up to
+    dtype rounding) vs full-sequence attention. zigzag=True uses the
+    causal load-balanced chunk layout (shards from `zigzag_shard`)."""
+    if zigzag:
+        if not causal:
+            raise ValueError("zigzag layout is causal-only")
+        return _ZigzagRingAttention.apply(q, k, v, group)
     return _RingAttention.apply(q, k, v, group, causal)
