@@ -50,6 +50,13 @@ class GradReducer:
         self.group = process_group
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.params = [p for p in params if p.requires_grad]
+        if comm_dtype is None and os.environ.get("FP16_COMM", "").lower() \
+                in ("1", "true"):
+            # reference ServiceEnv FP16_COMM: communicate
+            # gradients in 16 bit. bf16 here — fp16's range underflows
+            # grads; bf16 grads are already 16-bit so this only changes
+            # fp32-grad params (e.g. batch-norm affines)
+            comm_dtype = torch.bfloat16
         self.comm_dtype = comm_dtype
         self._armed = False
         self._build_buckets(bucket_bytes)

@@ -205,3 +205,35 @@ def _zero_worker(rank, world, port):
 @pytest.mark.timeout(300)
 def test_zero1_optimizer_matches_full():
     _run(_zero_worker)
+
+
+def _fp16_comm_worker(rank, world, port):
+    import os as _os
+    _os.environ["MASTER_ADDR"] = "127.0.0.1"
+    _os.environ["MASTER_PORT"] = str(port)
+    _os.environ["FP16_COMM"] = "true"
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from tepdist_amd.parallel.dp import GradReducer
+        torch.manual_seed(rank)
+        m = torch.nn.Linear(8, 8)  # fp32 params -> bf16 comm under the flag
+        red = GradReducer(m.parameters(), bucket_bytes=1 << 10)
+        assert red.comm_dtype == torch.bfloat16
+        red.reset()
+        red.arm()
+        m(torch.randn(4, 8)).sum().backward()
+        red.finalize()
+        # grads are averaged across ranks (through the bf16 wire)
+        ref = m.weight.grad.clone()
+        dist.all_reduce(ref)
+        torch.testing.assert_close(m.weight.grad, ref / world, rtol=1e-2,
+                                   atol=1e-2)
+    finally:
+        del _os.environ["FP16_COMM"]
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_fp16_comm_flag():
+    port = free_port()
+    mp.spawn(_fp16_comm_worker, args=(2, port), nprocs=2, join=True)
