@@ -85,6 +85,9 @@ class GPT2Block(nn.Module):
             self.out = RowParallelLinear(4 * d, d, env, bias=True, dtype=dtype)
 
     def _attn(self, qkv: torch.Tensor, B: int, S: int) -> torch.Tensor:
+        if getattr(self, "attn_impl", None) is not None:
+            # override hook (context parallelism swaps in ring attention)
+            return self.attn_impl(qkv)
         # packed-qkv fused attention: no transpose copies on the GPU path
         return ops.attention_qkv(qkv, self.n_head_local, causal=True)
 
@@ -165,12 +168,16 @@ class GPT2(nn.Module):
                 emb.weight[lo:].zero_()
 
     def forward(self, input_ids: torch.Tensor,
-                labels: Optional[torch.Tensor] = None):
+                labels: Optional[torch.Tensor] = None,
+                pos: Optional[torch.Tensor] = None):
         """input_ids [B, S]; labels [B, S] (next-token ids, -1 = ignore).
         Returns loss (scalar f32) if labels given, else logits [B,S,V]
-        ([B,S,V/tp] local shard under tensor parallelism)."""
+        ([B,S,V/tp] local shard under tensor parallelism). `pos` overrides
+        position ids (context parallelism feeds the shard's GLOBAL
+        positions)."""
         B, S = input_ids.shape
-        pos = torch.arange(S, device=input_ids.device)
+        if pos is None:
+            pos = torch.arange(S, device=input_ids.device)
         if self.env.tp_size == 1:
             x = ops.embedding(input_ids, self.wte) + ops.embedding(pos, self.wpe)
         else:
