@@ -8,6 +8,7 @@ pass and the fused AdamW applying gradients (the AG role)."""
 
 from __future__ import annotations
 
+import os
 from typing import Dict
 
 import torch
@@ -17,10 +18,41 @@ from tepdist_amd.ir.graph import Graph, Node
 from tepdist_amd.parallel import mappings
 
 
+def _fuse_mlp(g: Graph) -> None:
+    """Backend peephole (GPU): linear(act=gelu) -> linear, single
+    consumer, both biased, becomes ONE `mlp` node so the gelu rides the
+    GEMM epilogues (ops.mlp: hipBLASLt GELU_AUX_BIAS / DGELU_BGRAD). Runs
+    AFTER planning/transform — a mesh round that resharded between the
+    two linears leaves a collective in between and the pattern simply
+    does not match."""
+    cons: Dict[int, int] = {}
+    for n in g.nodes.values():
+        for i in n.inputs:
+            cons[i] = cons.get(i, 0) + 1
+    outs = set(g.outputs)
+    for n in list(g.nodes.values()):
+        if (n.op != "linear" or n.attrs.get("act", "none") != "none"
+                or len(n.inputs) != 3):
+            continue
+        p = g.nodes.get(n.inputs[0])
+        if (p is None or p.op != "linear"
+                or p.attrs.get("act") != "gelu" or len(p.inputs) != 3
+                or cons.get(p.id, 0) != 1 or p.id in outs):
+            continue
+        n.op = "mlp"
+        n.inputs = [p.inputs[0], p.inputs[1], p.inputs[2], n.inputs[1],
+                    n.inputs[2]]
+        n.attrs.pop("act", None)
+        del g.nodes[p.id]
+
+
 class GraphInterpreter:
     def __init__(self, graph: Graph, device: str = "cpu",
                  dtype=torch.float32, group=None, groups=None):
         self.g = graph
+        if str(device).startswith("cuda") and \
+                os.environ.get("TEPDIST_MLP_FUSE", "1") != "0":
+            _fuse_mlp(self.g)
         self.device = device
         self.dtype = dtype
         self.group = group  # process group for reshard collective nodes
@@ -78,6 +110,8 @@ class GraphInterpreter:
                               act=n.attrs.get("act", "none"))
         if n.op == "matmul":
             return ops.matmul(ins[0], ins[1])
+        if n.op == "mlp":
+            return ops.mlp(ins[0], ins[1], ins[2], ins[3], ins[4])
         if n.op == "layernorm":
             return ops.layernorm(ins[0], ins[1], ins[2])
         if n.op == "softmax":

@@ -99,8 +99,8 @@ class GPT2Block(nn.Module):
             a = self._attn(qkv, B, S)
             x = x + ops.linear(a, self.w_proj, self.b_proj)
             h = ops.layernorm(x, self.ln2_g, self.ln2_b, self.cfg.ln_eps)
-            h = ops.linear(h, self.w_fc, self.b_fc, act="gelu")
-            x = x + ops.linear(h, self.w_out, self.b_out)
+            # fused MLP: the gelu rides the GEMM epilogues on GPU
+            x = x + ops.mlp(h, self.w_fc, self.b_fc, self.w_out, self.b_out)
         else:
             qkv = self.qkv(h)                     # [B,S,3*d/tp]
             a = self._attn(qkv, B, S)

@@ -1,6 +1,7 @@
 from tepdist_amd.ops.interface import (  # noqa: F401
     gelu,
     linear,
+    mlp,
     matmul,
     layernorm,
     softmax,

@@ -81,8 +81,30 @@ def build_rt(force: bool = False, verbose: bool = True) -> Path:
     return RT_SO
 
 
+BLT_SO = OPS_DIR / "_tepdist_blt.so"
+
+
+def build_blt(force: bool = False, verbose: bool = True) -> Path:
+    """Builds the hipBLASLt fused-epilogue driver (host-only C++, links
+    libhipblaslt)."""
+    src = CSRC / "blaslt.cpp"
+    if not force and BLT_SO.exists() and \
+            BLT_SO.stat().st_mtime > src.stat().st_mtime:
+        return BLT_SO
+    cmd = [HIPCC, "-O3", "-std=c++17", "-shared", "-fPIC",
+           *_includes(), str(src), "-L", "/opt/rocm/lib", "-lhipblaslt",
+           "-o", str(BLT_SO)]
+    if verbose:
+        print("[build_ext]", " ".join(cmd), flush=True)
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        raise RuntimeError(f"blaslt build failed:\n{r.stdout}\n{r.stderr}")
+    return BLT_SO
+
+
 def build(force: bool = False, verbose: bool = True) -> Path:
     build_rt(force, verbose)
+    build_blt(force, verbose)
     if not force and not _needs_build():
         return OUT_SO
     BUILD.mkdir(exist_ok=True)

@@ -184,6 +184,109 @@ def gelu_bwd(dy: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     return dx
 
 
+# --------------------------------------------------------------------------
+# Fused transformer MLP (hipBLASLt fused epilogues)
+# --------------------------------------------------------------------------
+#
+# The gelu between the two MLP GEMMs costs two full HBM round-trips as
+# standalone kernels (read+write of the [T, 4d] tensor fwd AND bwd). The
+# hipBLASLt epilogues fold them into the producing GEMMs: GELU_AUX_BIAS on
+# the c_fc forward (pre-gelu saved as the aux buffer the backward needs
+# anyway) and DGELU_BGRAD on the proj dgrad (gelu' applied to the GEMM
+# result + the fc bias gradient reduced in the same pass). Fused vs
+# composed is measured per shape at first use like every other GEMM site;
+# TEPDIST_MLP_FUSED=0|1 pins.
+
+try:
+    from tepdist_amd.ops import _tepdist_blt as _blt
+except Exception:                      # extension not built
+    _blt = None
+
+_MLP_FUSED = os.environ.get("TEPDIST_MLP_FUSED", "auto")
+
+
+def _blt_mlp_gelu(x: torch.Tensor, w1: torch.Tensor, b1: torch.Tensor):
+    M, K = x.shape
+    N = w1.shape[0]
+    y = torch.empty(M, N, dtype=BF16, device=x.device)
+    aux = torch.empty(M, N, dtype=BF16, device=x.device)
+    _blt.blt_fused(0, M, N, K, w1.data_ptr(), x.data_ptr(), y.data_ptr(),
+                   b1.data_ptr(), aux.data_ptr(), _stream())
+    return y, aux
+
+
+def _blt_mlp_dgelu(dy: torch.Tensor, w2: torch.Tensor, pre: torch.Tensor):
+    M, N = dy.shape
+    K = w2.shape[1]
+    dh = torch.empty(M, K, dtype=BF16, device=dy.device)
+    db1 = torch.empty(K, dtype=BF16, device=dy.device)
+    _blt.blt_fused(1, M, N, K, w2.data_ptr(), dy.data_ptr(), dh.data_ptr(),
+                   db1.data_ptr(), pre.data_ptr(), _stream())
+    return dh, db1
+
+
+def mlp_fwd(x, w1, b1, w2, b2):
+    """y = gelu(x@w1^T+b1) @ w2^T + b2. Returns (y, (h, pre)): the
+    post-gelu activation and the pre-gelu aux for backward."""
+    x = x.contiguous()
+    res = {}
+
+    def _fused():
+        h, pre = _blt_mlp_gelu(x, w1, b1)
+        res["h"], res["pre"] = h, pre
+        return linear_fwd(h, w2, b2, "none")[0]
+
+    def _composed():
+        h, pre = linear_fwd(x, w1, b1, "gelu")
+        res["h"], res["pre"] = h, pre
+        return linear_fwd(h, w2, b2, "none")[0]
+
+    if _blt is None or _MLP_FUSED == "0":
+        y = _composed()
+    elif _MLP_FUSED == "1":
+        y = _fused()
+    else:
+        y = _pick_backend(("mlp_f", x.shape[0], w1.shape[0], x.shape[1]),
+                          _fused, _composed)()
+    return y, (res["h"], res["pre"])
+
+
+def mlp_bwd(dy, x, w1, w2, h, pre):
+    """Backward of mlp_fwd. Returns (dx, dw1, db1, dw2, db2)."""
+    dy = dy.contiguous()
+    out = {}
+
+    def _fused():
+        out["dh"], out["db1"] = _blt_mlp_dgelu(dy, w2, pre)
+
+    def _composed():
+        dhp = torch.matmul(dy, w2)
+        dh = torch.empty_like(dhp)
+        ext.gelu_bwd(dhp.data_ptr(), pre.data_ptr(), dh.data_ptr(),
+                     dhp.numel(), _stream())
+        db1 = torch.empty(w2.shape[1], dtype=BF16, device=dy.device)
+        ws = torch.zeros(w2.shape[1], dtype=torch.float32, device=dy.device)
+        ext.bias_sum(dh.data_ptr(), db1.data_ptr(), ws.data_ptr(),
+                     dh.shape[0], dh.shape[1], _stream())
+        out["dh"], out["db1"] = dh, db1
+
+    if _blt is None or _MLP_FUSED == "0":
+        _composed()
+    elif _MLP_FUSED == "1":
+        _fused()
+    else:
+        _pick_backend(("mlp_b", dy.shape[0], dy.shape[1], w2.shape[1]),
+                      _fused, _composed)()
+    dh = out["dh"]
+    dw2 = torch.matmul(dy.t(), h)
+    db2 = torch.empty(dy.shape[1], dtype=BF16, device=dy.device)
+    ws2 = torch.zeros(dy.shape[1], dtype=torch.float32, device=dy.device)
+    ext.bias_sum(dy.data_ptr(), db2.data_ptr(), ws2.data_ptr(),
+                 dy.shape[0], dy.shape[1], _stream())
+    dx, dw1, _ = linear_bwd(dh, x, w1, False, "none", None)
+    return dx, dw1, out["db1"], dw2, db2
+
+
 def transpose2d(t: torch.Tensor) -> torch.Tensor:
     """Materialized bf16 transpose (tiled LDS kernel)."""
     R, C = t.shape[-2], t.shape[-1]

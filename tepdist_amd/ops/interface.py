@@ -56,6 +56,38 @@ def linear(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor] = None
     return LinearFn.apply(x, w, bias, act)
 
 
+class MlpFn(torch.autograd.Function):
+    """Fused transformer MLP: gelu epilogues ride the producing GEMMs
+    (hip.mlp_fwd/mlp_bwd — hipBLASLt GELU_AUX_BIAS / DGELU_BGRAD, measured
+    per shape against the composed path)."""
+
+    @staticmethod
+    def forward(ctx, x, w1, b1, w2, b2):
+        be = _backend(x)
+        x2d = x.reshape(-1, x.shape[-1]).contiguous()
+        y, (h, pre) = be.mlp_fwd(x2d, w1, b1, w2, b2)
+        ctx.save_for_backward(x2d, w1, w2, h, pre)
+        ctx.in_shape = x.shape
+        return y.reshape(*x.shape[:-1], w2.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2d, w1, w2, h, pre = ctx.saved_tensors
+        be = _backend(dy)
+        dy2d = dy.reshape(-1, dy.shape[-1]).contiguous()
+        dx, dw1, db1, dw2, db2 = be.mlp_bwd(dy2d, x2d, w1, w2, h, pre)
+        return dx.reshape(ctx.in_shape), dw1, db1, dw2, db2
+
+
+def mlp(x, w1, b1, w2, b2):
+    """Transformer MLP y = gelu(x@w1^T+b1)@w2^T+b2 (both biases
+    required). GPU runs the fused-epilogue path; CPU composes the
+    reference linears (same autograd surface)."""
+    if x.is_cuda:
+        return MlpFn.apply(x, w1, b1, w2, b2)
+    return linear(linear(x, w1, b1, act="gelu"), w2, b2)
+
+
 class MatmulFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, a, b):

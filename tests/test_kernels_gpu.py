@@ -393,3 +393,43 @@ def test_adamw_multi_tensor_matches_single():
         torch.testing.assert_close(oa.state[pa]["master"],
                                    ob.state[pb]["master"], rtol=1e-6,
                                    atol=1e-7)
+
+
+@pytest.mark.gpu
+def test_mlp_fused_epilogue():
+    """hipBLASLt fused-epilogue MLP (GELU_AUX_BIAS fwd, DGELU_BGRAD
+    dgrad) vs the fp32 tanh-gelu reference: values and all five grads."""
+    from tepdist_amd.ops import hip
+    if hip._blt is None:
+        pytest.skip("blaslt extension not built")
+    torch.manual_seed(0)
+    T, d = 1024, 256
+    mk = lambda *s: (torch.randn(*s, device="cuda", dtype=torch.bfloat16)
+                     * 0.05).requires_grad_()
+    x, w1, b1 = mk(T, d), mk(4 * d, d), mk(4 * d)
+    w2, b2 = mk(d, 4 * d), mk(d)
+    dy = torch.randn(T, d, device="cuda", dtype=torch.bfloat16)
+
+    old = hip._MLP_FUSED
+    hip._MLP_FUSED = "1"
+    try:
+        from tepdist_amd import ops
+        y = ops.mlp(x, w1, b1, w2, b2)
+        y.backward(dy)
+    finally:
+        hip._MLP_FUSED = old
+
+    xf, w1f, b1f, w2f, b2f = (t.detach().float().clone().requires_grad_()
+                              for t in (x, w1, b1, w2, b2))
+    h = torch.nn.functional.gelu(xf @ w1f.t() + b1f, approximate="tanh")
+    ref = h @ w2f.t() + b2f
+    ref.backward(dy.float())
+
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-2)
+    for got, want, name in ((x.grad, xf.grad, "dx"),
+                            (w1.grad, w1f.grad, "dw1"),
+                            (b1.grad, b1f.grad, "db1"),
+                            (w2.grad, w2f.grad, "dw2"),
+                            (b2.grad, b2f.grad, "db2")):
+        torch.testing.assert_close(got.float(), want, rtol=5e-2,
+                                   atol=5e-2, msg=name)
