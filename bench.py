@@ -81,7 +81,11 @@ def main():
             else torch.device("cpu")
 
     cfg = GPT2_CONFIGS[args.model]
-    seq = min(args.seq, cfg.n_ctx)
+    if args.seq > cfg.n_ctx:
+        # grow the position table instead of silently clamping --seq
+        import dataclasses
+        cfg = dataclasses.replace(cfg, n_ctx=args.seq)
+    seq = args.seq
     dtype = torch.bfloat16 if device.type == "cuda" else torch.float32
     global_batch = args.micro_batch * world
 
