@@ -175,9 +175,102 @@ void blt_fused(int mode, long M, long N, long K, uintptr_t A, uintptr_t B,
       p.ws_size, reinterpret_cast<hipStream_t>(stream)));
 }
 
+// Support probe: which epilogue / bias-dtype / aux-dtype combinations
+// does the library's heuristic actually offer algorithms for on this
+// device + shape? Returns {label: n_algos}.
+pybind11::dict probe_epilogues(long M, long N, long K) {
+  pybind11::dict out;
+  struct Case {
+    const char* label;
+    hipblasLtEpilogue_t epi;
+    int bias_t;  // hipDataType or -1 to skip attr
+    int aux_t;   // hipDataType or -1 to skip attr
+    bool need_aux;
+  };
+  const Case cases[] = {
+      {"default", HIPBLASLT_EPILOGUE_DEFAULT, -1, -1, false},
+      {"bias_bf16", HIPBLASLT_EPILOGUE_BIAS, HIP_R_16BF, -1, false},
+      {"bias_f32", HIPBLASLT_EPILOGUE_BIAS, HIP_R_32F, -1, false},
+      {"gelu", HIPBLASLT_EPILOGUE_GELU, -1, -1, false},
+      {"gelu_bias_bf16", HIPBLASLT_EPILOGUE_GELU_BIAS, HIP_R_16BF, -1,
+       false},
+      {"gelu_aux", HIPBLASLT_EPILOGUE_GELU_AUX, -1, -1, true},
+      {"gelu_aux_bias_bf16", HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, HIP_R_16BF,
+       -1, true},
+      {"gelu_aux_bias_f32", HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, HIP_R_32F,
+       -1, true},
+      {"gelu_aux_bias_bf16_auxf32", HIPBLASLT_EPILOGUE_GELU_AUX_BIAS,
+       HIP_R_16BF, HIP_R_32F, true},
+      {"dgelu", HIPBLASLT_EPILOGUE_DGELU, -1, -1, true},
+      {"dgelu_bgrad_bf16", HIPBLASLT_EPILOGUE_DGELU_BGRAD, HIP_R_16BF, -1,
+       true},
+      {"dgelu_bgrad_f32", HIPBLASLT_EPILOGUE_DGELU_BGRAD, HIP_R_32F, -1,
+       true},
+      {"bgradb_bf16", HIPBLASLT_EPILOGUE_BGRADB, HIP_R_16BF, -1, false},
+  };
+  for (const Case& c : cases) {
+    hipblasLtMatmulDesc_t desc;
+    BLT_CHECK(hipblasLtMatmulDescCreate(&desc, HIPBLAS_COMPUTE_32F,
+                                        HIP_R_32F));
+    int32_t opT = HIPBLAS_OP_T, opN = HIPBLAS_OP_N;
+    hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSA,
+                                    &opT, sizeof(opT));
+    hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_TRANSB,
+                                    &opN, sizeof(opN));
+    hipblasLtEpilogue_t e = c.epi;
+    hipblasLtMatmulDescSetAttribute(desc, HIPBLASLT_MATMUL_DESC_EPILOGUE,
+                                    &e, sizeof(e));
+    if (c.bias_t >= 0) {
+      int32_t bt = c.bias_t;
+      hipblasLtMatmulDescSetAttribute(
+          desc, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &bt, sizeof(bt));
+      void* dummy = workspace();
+      hipblasLtMatmulDescSetAttribute(
+          desc, HIPBLASLT_MATMUL_DESC_BIAS_POINTER, &dummy, sizeof(dummy));
+    }
+    if (c.need_aux) {
+      int64_t ld = N;
+      void* dummy = workspace();
+      hipblasLtMatmulDescSetAttribute(
+          desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &dummy,
+          sizeof(dummy));
+      hipblasLtMatmulDescSetAttribute(
+          desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &ld, sizeof(ld));
+      if (c.aux_t >= 0) {
+        int32_t at = c.aux_t;
+        hipblasLtMatmulDescSetAttribute(
+            desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE, &at,
+            sizeof(at));
+      }
+    }
+    hipblasLtMatrixLayout_t la, lb, ld_;
+    BLT_CHECK(hipblasLtMatrixLayoutCreate(&la, HIP_R_16BF, K, N, K));
+    BLT_CHECK(hipblasLtMatrixLayoutCreate(&lb, HIP_R_16BF, K, M, K));
+    BLT_CHECK(hipblasLtMatrixLayoutCreate(&ld_, HIP_R_16BF, N, M, N));
+    hipblasLtMatmulPreference_t pref;
+    BLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
+    size_t ws = kMaxWorkspace;
+    hipblasLtMatmulPreferenceSetAttribute(
+        pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &ws, sizeof(ws));
+    hipblasLtMatmulHeuristicResult_t res[8];
+    int nres = 0;
+    hipblasStatus_t st = hipblasLtMatmulAlgoGetHeuristic(
+        handle(), desc, la, lb, ld_, ld_, pref, 8, res, &nres);
+    out[c.label] =
+        (st == HIPBLAS_STATUS_SUCCESS) ? nres : -int(st);
+    hipblasLtMatmulPreferenceDestroy(pref);
+    hipblasLtMatrixLayoutDestroy(la);
+    hipblasLtMatrixLayoutDestroy(lb);
+    hipblasLtMatrixLayoutDestroy(ld_);
+    hipblasLtMatmulDescDestroy(desc);
+  }
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_tepdist_blt, m) {
   m.doc() = "hipBLASLt fused-epilogue GEMMs (GELU_AUX_BIAS / DGELU_BGRAD)";
   m.def("blt_fused", &blt_fused, "fused-epilogue matmul");
+  m.def("probe_epilogues", &probe_epilogues, "heuristic support probe");
 }

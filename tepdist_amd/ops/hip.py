@@ -225,9 +225,18 @@ def _blt_mlp_dgelu(dy: torch.Tensor, w2: torch.Tensor, pre: torch.Tensor):
     return dh, db1
 
 
+_blt_ok = True    # flipped off permanently on the first library refusal
+                  # (e.g. "no algo for fused epilogue" on this shape set)
+
+
+def _blt_usable() -> bool:
+    return _blt is not None and _blt_ok and _MLP_FUSED != "0"
+
+
 def mlp_fwd(x, w1, b1, w2, b2):
     """y = gelu(x@w1^T+b1) @ w2^T + b2. Returns (y, (h, pre)): the
     post-gelu activation and the pre-gelu aux for backward."""
+    global _blt_ok
     x = x.contiguous()
     res = {}
 
@@ -241,18 +250,23 @@ def mlp_fwd(x, w1, b1, w2, b2):
         res["h"], res["pre"] = h, pre
         return linear_fwd(h, w2, b2, "none")[0]
 
-    if _blt is None or _MLP_FUSED == "0":
+    if not _blt_usable():
         y = _composed()
     elif _MLP_FUSED == "1":
         y = _fused()
     else:
-        y = _pick_backend(("mlp_f", x.shape[0], w1.shape[0], x.shape[1]),
-                          _fused, _composed)()
+        try:
+            y = _pick_backend(("mlp_f", x.shape[0], w1.shape[0],
+                               x.shape[1]), _fused, _composed)()
+        except RuntimeError:
+            _blt_ok = False
+            y = _composed()
     return y, (res["h"], res["pre"])
 
 
 def mlp_bwd(dy, x, w1, w2, h, pre):
     """Backward of mlp_fwd. Returns (dx, dw1, db1, dw2, db2)."""
+    global _blt_ok
     dy = dy.contiguous()
     out = {}
 
@@ -270,13 +284,17 @@ def mlp_bwd(dy, x, w1, w2, h, pre):
                      dh.shape[0], dh.shape[1], _stream())
         out["dh"], out["db1"] = dh, db1
 
-    if _blt is None or _MLP_FUSED == "0":
+    if not _blt_usable():
         _composed()
     elif _MLP_FUSED == "1":
         _fused()
     else:
-        _pick_backend(("mlp_b", dy.shape[0], dy.shape[1], w2.shape[1]),
-                      _fused, _composed)()
+        try:
+            _pick_backend(("mlp_b", dy.shape[0], dy.shape[1],
+                           w2.shape[1]), _fused, _composed)()
+        except RuntimeError:
+            _blt_ok = False
+            _composed()
     dh = out["dh"]
     dw2 = torch.matmul(dy.t(), h)
     db2 = torch.empty(dy.shape[1], dtype=BF16, device=dy.device)
