@@ -402,6 +402,12 @@ def test_mlp_fused_epilogue():
     from tepdist_amd.ops import hip
     if hip._blt is None:
         pytest.skip("blaslt extension not built")
+    if hip._blt.probe_epilogues(256, 256, 256)["gelu_aux_bias_bf16"] <= 0:
+        # measured: hipBLASLt 1.2.7 offers no algos for ANY aux epilogue
+        # on gfx950 (profiles/blaslt_epilogue_probe.md) — ops.mlp then
+        # permanently falls back to the composed path, which the rest of
+        # the GPU suite covers
+        pytest.skip("hipblaslt lacks aux epilogues on this stack")
     torch.manual_seed(0)
     T, d = 1024, 256
     mk = lambda *s: (torch.randn(*s, device="cuda", dtype=torch.bfloat16)
