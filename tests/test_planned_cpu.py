@@ -244,3 +244,49 @@ def _zero_worker(rank, world, port):
 def test_planned_zero_optimizer():
     port = free_port()
     mp.spawn(_zero_worker, args=(2, port), nprocs=2, join=True)
+
+
+def test_fuse_mlp_respects_collectives():
+    """The MLP peephole only fuses DIRECTLY adjacent linear(gelu)->linear
+    pairs: a planner-inserted collective between them (TP reshard) must
+    block the fusion; a clean pair must fuse."""
+    import dataclasses
+
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.ir.interpreter import _fuse_mlp
+    from tepdist_amd.models.configs import GPT2_CONFIGS
+    from tepdist_amd.planner.dist_spec import DimStrategy
+    from tepdist_amd.planner.transform import SpmdTransform
+
+    cfg = dataclasses.replace(GPT2_CONFIGS["gpt2-117m"], n_layer=1,
+                              n_embd=64, n_head=2, vocab_size=128)
+    g = gpt2_ir(cfg, batch=2, seq=8)
+    n_mlp_in = sum(1 for n in g.nodes.values()
+                   if n.op == "linear" and n.attrs.get("act") == "gelu")
+    assert n_mlp_in == 1
+    _fuse_mlp(g)
+    assert sum(1 for n in g.nodes.values() if n.op == "mlp") == 1
+
+    # Megatron TP over the MLP: w_fc column-split / w_out row-split puts
+    # an all_reduce after the second linear but keeps fc->out adjacent
+    # (fusable); a batch-split plan slices activations BETWEEN them only
+    # if a reshard lands there — build one explicitly: split fc's output
+    g2 = gpt2_ir(cfg, batch=2, seq=8)
+    specs = {}
+    for n in g2.nodes.values():
+        specs[n.id] = DimStrategy.replicated(2)
+    # shard the gelu-linear's OUTPUT differently from the consumer's
+    # expectation so the transform inserts a collective between them
+    fc = next(n for n in g2.nodes.values()
+              if n.op == "linear" and n.attrs.get("act") == "gelu")
+    specs[fc.id] = DimStrategy.split(0, 2)
+    res = SpmdTransform(g2, specs, 2).run()
+    tg = res.graph
+    fused_before = sum(1 for n in tg.nodes.values() if n.op == "mlp")
+    _fuse_mlp(tg)
+    fc_t = [n for n in tg.nodes.values()
+            if n.op == "linear" and n.attrs.get("act") == "gelu"]
+    mlps = sum(1 for n in tg.nodes.values() if n.op == "mlp") - fused_before
+    # the collective (all_gather of the split output) sits between the
+    # pair -> the gelu linear survives unfused
+    assert fc_t and mlps == 0
