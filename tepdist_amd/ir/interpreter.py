@@ -55,6 +55,24 @@ class GraphInterpreter:
             _fuse_mlp(self.g)
         self.device = device
         self.dtype = dtype
+        # residual-add + layernorm pairs fused at eval time (GPU): the add
+        # returns the sum as usual AND stashes the fused ln output for the
+        # consumer node (one kernel, one HBM pass)
+        self._addln = {}
+        self._ln_cache = {}
+        if str(device).startswith("cuda") and \
+                os.environ.get("TEPDIST_ADDLN_FUSE", "1") != "0":
+            for n in self.g.nodes.values():
+                if n.op != "layernorm" or len(n.inputs) != 3:
+                    continue
+                a = self.g.nodes.get(n.inputs[0])
+                g1 = self.g.nodes.get(n.inputs[1])
+                b1 = self.g.nodes.get(n.inputs[2])
+                if (a is not None and a.op == "add" and len(a.inputs) == 2
+                        and g1 is not None and g1.op == "param"
+                        and b1 is not None and b1.op == "param"
+                        and a.id not in self._addln):
+                    self._addln[a.id] = n.id
         self.group = group  # process group for reshard collective nodes
         # mesh-round ordinal -> process group (multi-round transforms tag
         # every collective with attrs["mesh_round"]; CommDevManager builds
@@ -113,6 +131,8 @@ class GraphInterpreter:
         if n.op == "mlp":
             return ops.mlp(ins[0], ins[1], ins[2], ins[3], ins[4])
         if n.op == "layernorm":
+            if n.id in self._ln_cache:
+                return self._ln_cache.pop(n.id)
             return ops.layernorm(ins[0], ins[1], ins[2])
         if n.op == "softmax":
             return ops.softmax(ins[0], scale=n.attrs.get("scale", 1.0),
@@ -151,6 +171,16 @@ class GraphInterpreter:
             size = n.shape[dim if dim >= 0 else len(n.shape) - 1]
             return ins[0].narrow(dim, idx * size, size)
         if n.op == "add":
+            ln_id = self._addln.get(n.id)
+            if (ln_id is not None and ins[0].is_cuda
+                    and ins[0].dtype == torch.bfloat16
+                    and ins[0].shape == ins[1].shape):
+                ln = self.g.nodes[ln_id]
+                gamma = variables[self.g.nodes[ln.inputs[1]].name]
+                beta = variables[self.g.nodes[ln.inputs[2]].name]
+                s, y = ops.add_layernorm(ins[0], ins[1], gamma, beta)
+                self._ln_cache[ln_id] = y
+                return s
             return ins[0] + ins[1]
         if n.op == "mul":
             return ins[0] * ins[1]

@@ -97,9 +97,10 @@ class GPT2Block(nn.Module):
         if self.env.tp_size == 1:
             qkv = ops.linear(h, self.w_qkv, self.b_qkv)
             a = self._attn(qkv, B, S)
-            x = x + ops.linear(a, self.w_proj, self.b_proj)
-            h = ops.layernorm(x, self.ln2_g, self.ln2_b, self.cfg.ln_eps)
-            # fused MLP: the gelu rides the GEMM epilogues on GPU
+            # fused residual+ln (one kernel) and fused MLP on GPU
+            x, h = ops.add_layernorm(
+                x, ops.linear(a, self.w_proj, self.b_proj),
+                self.ln2_g, self.ln2_b, self.cfg.ln_eps)
             x = x + ops.mlp(h, self.w_fc, self.b_fc, self.w_out, self.b_out)
         else:
             qkv = self.qkv(h)                     # [B,S,3*d/tp]

@@ -4,6 +4,7 @@ from tepdist_amd.ops.interface import (  # noqa: F401
     mlp,
     matmul,
     layernorm,
+    add_layernorm,
     softmax,
     attention,
     attention_qkv,

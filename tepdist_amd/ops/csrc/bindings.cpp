@@ -53,20 +53,27 @@ PYBIND11_MODULE(_tepdist_hip, m) {
 
   m.def("layernorm_fwd", [](uintptr_t x, uintptr_t g, uintptr_t b, uintptr_t y,
                             uintptr_t mean, uintptr_t rstd, int rows, int cols,
-                            float eps, uintptr_t stream) {
+                            float eps, uintptr_t stream, uintptr_t res,
+                            uintptr_t sum_out) {
     layernorm_fwd_bf16(reinterpret_cast<void*>(x), reinterpret_cast<void*>(g),
                        reinterpret_cast<void*>(b), reinterpret_cast<void*>(y),
                        reinterpret_cast<float*>(mean),
                        reinterpret_cast<float*>(rstd), rows, cols, eps,
-                       S(stream));
+                       S(stream), reinterpret_cast<void*>(res),
+                       reinterpret_cast<void*>(sum_out));
     check_launch();
-  });
+  }, pybind11::arg("x"), pybind11::arg("g"), pybind11::arg("b"),
+     pybind11::arg("y"), pybind11::arg("mean"), pybind11::arg("rstd"),
+     pybind11::arg("rows"), pybind11::arg("cols"), pybind11::arg("eps"),
+     pybind11::arg("stream"), pybind11::arg("res") = 0,
+     pybind11::arg("sum_out") = 0);
 
   m.def("layernorm_bwd", [](uintptr_t dy, uintptr_t x, uintptr_t g,
                             uintptr_t mean, uintptr_t rstd, uintptr_t dx,
                             uintptr_t dg_part, uintptr_t db_part,
                             uintptr_t dgamma, uintptr_t dbeta, int rows,
-                            int cols, int part_rows, uintptr_t stream) {
+                            int cols, int part_rows, uintptr_t stream,
+                            uintptr_t dsum) {
     layernorm_bwd_bf16(reinterpret_cast<void*>(dy), reinterpret_cast<void*>(x),
                        reinterpret_cast<void*>(g),
                        reinterpret_cast<float*>(mean),
@@ -74,7 +81,7 @@ PYBIND11_MODULE(_tepdist_hip, m) {
                        reinterpret_cast<void*>(dx),
                        reinterpret_cast<float*>(dg_part),
                        reinterpret_cast<float*>(db_part), rows, cols,
-                       part_rows, S(stream));
+                       part_rows, S(stream), reinterpret_cast<void*>(dsum));
     check_launch();
     layernorm_bwd_reduce(reinterpret_cast<float*>(dg_part),
                          reinterpret_cast<float*>(db_part),
@@ -82,7 +89,12 @@ PYBIND11_MODULE(_tepdist_hip, m) {
                          reinterpret_cast<void*>(dbeta), part_rows, cols,
                          S(stream));
     check_launch();
-  });
+  }, pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("g"),
+     pybind11::arg("mean"), pybind11::arg("rstd"), pybind11::arg("dx"),
+     pybind11::arg("dg_part"), pybind11::arg("db_part"),
+     pybind11::arg("dgamma"), pybind11::arg("dbeta"), pybind11::arg("rows"),
+     pybind11::arg("cols"), pybind11::arg("part_rows"),
+     pybind11::arg("stream"), pybind11::arg("dsum") = 0);
 
   m.def("softmax_fwd", [](uintptr_t x, uintptr_t p, int64_t rows, int cols,
                           int sq, float scale, bool causal, uintptr_t stream) {

@@ -46,13 +46,18 @@ void transpose_dy_bf16(const void* dy, const void* pre, void* dy_t,
                        hipStream_t stream);
 
 // --- LayerNorm -------------------------------------------------------------
+// res/sum_out non-null = fused residual add: sum_out := x + res (rounded
+// once to bf16), statistics and y over the sum. dsum non-null in bwd adds
+// the straight-through gradient into dx in the same pass.
 void layernorm_fwd_bf16(const void* x, const void* gamma, const void* beta,
                         void* y, float* mean, float* rstd, int rows, int cols,
-                        float eps, hipStream_t stream);
+                        float eps, hipStream_t stream,
+                        const void* res = nullptr, void* sum_out = nullptr);
 void layernorm_bwd_bf16(const void* dy, const void* x, const void* gamma,
                         const float* mean, const float* rstd, void* dx,
                         float* dgamma_part, float* dbeta_part, int rows,
-                        int cols, int part_rows, hipStream_t stream);
+                        int cols, int part_rows, hipStream_t stream,
+                        const void* dsum = nullptr);
 void layernorm_bwd_reduce(const float* dgamma_part, const float* dbeta_part,
                           void* dgamma, void* dbeta, int part_rows, int cols,
                           hipStream_t stream);

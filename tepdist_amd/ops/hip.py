@@ -484,6 +484,45 @@ def layernorm_bwd(dy: torch.Tensor, x: torch.Tensor, gamma: torch.Tensor,
     return dx, dgamma, dbeta
 
 
+def add_layernorm_fwd(x, res, gamma, beta, eps: float = 1e-5):
+    """Fused residual add + layernorm: s = x + res (bf16-rounded once),
+    y = ln(s). The standalone add kernel's write+read of `s` disappears.
+    Returns (s, y, mean, rstd)."""
+    x, res = x.contiguous(), res.contiguous()
+    rows, cols = x.shape
+    s = torch.empty_like(x)
+    y = torch.empty_like(x)
+    mean = torch.empty(rows, dtype=torch.float32, device=x.device)
+    rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
+    ext.layernorm_fwd(x.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
+                      y.data_ptr(), mean.data_ptr(), rstd.data_ptr(), rows,
+                      cols, eps, _stream(), res.data_ptr(), s.data_ptr())
+    return s, y, mean, rstd
+
+
+def add_layernorm_bwd(dy, dsum, s, gamma, mean, rstd):
+    """Backward of add_layernorm: dx = dsum + d(ln)/ds in one pass
+    (the autograd grad-accumulation add disappears). dsum may be None
+    (sum unused elsewhere). Returns (dx, dgamma, dbeta); dres == dx."""
+    dy = dy.contiguous()
+    dsum = dsum.contiguous() if dsum is not None else None
+    rows, cols = s.shape
+    part = min(LN_PART_ROWS, max(4, (rows + 3) // 4 * 4))
+    part = (part // 4) * 4
+    dx = torch.empty_like(s)
+    dg_part = torch.empty(part, cols, dtype=torch.float32, device=s.device)
+    db_part = torch.empty(part, cols, dtype=torch.float32, device=s.device)
+    dgamma = torch.empty_like(gamma)
+    dbeta = torch.empty_like(gamma)
+    ext.layernorm_bwd(dy.data_ptr(), s.data_ptr(), gamma.data_ptr(),
+                      mean.data_ptr(), rstd.data_ptr(), dx.data_ptr(),
+                      dg_part.data_ptr(), db_part.data_ptr(),
+                      dgamma.data_ptr(), dbeta.data_ptr(), rows, cols, part,
+                      _stream(),
+                      0 if dsum is None else dsum.data_ptr())
+    return dx, dgamma, dbeta
+
+
 # --------------------------------------------------------------------------
 # Softmax / attention
 # --------------------------------------------------------------------------

@@ -138,6 +138,47 @@ def layernorm(x, gamma, beta, eps: float = 1e-5):
 # --------------------------------------------------------------------------
 
 
+class AddLayerNormFn(torch.autograd.Function):
+    """Fused residual add + layernorm: one kernel produces BOTH the
+    residual sum (bf16-rounded once, bit-identical to a separate add) and
+    the normalized output; backward folds the straight-through gradient
+    into the layernorm dx pass (kills the autograd grad-sum add)."""
+
+    @staticmethod
+    def forward(ctx, x, res, gamma, beta, eps):
+        ctx.set_materialize_grads(False)
+        be = _backend(x)
+        x2d = x.reshape(-1, x.shape[-1])
+        r2d = res.reshape(-1, res.shape[-1])
+        s, y, mean, rstd = be.add_layernorm_fwd(x2d, r2d, gamma, beta, eps)
+        ctx.save_for_backward(s, gamma, mean, rstd)
+        ctx.in_shape = x.shape
+        return s.reshape(x.shape), y.reshape(x.shape)
+
+    @staticmethod
+    def backward(ctx, dsum, dy):
+        s, gamma, mean, rstd = ctx.saved_tensors
+        be = _backend(dy if dy is not None else dsum)
+        if dy is None:
+            # normalized output unused: grads flow through the sum only
+            d = dsum.reshape(ctx.in_shape)
+            return d, d, None, None, None
+        dy2d = dy.reshape(-1, dy.shape[-1])
+        ds2d = None if dsum is None else dsum.reshape(-1, dsum.shape[-1])
+        dx, dg, db = be.add_layernorm_bwd(dy2d, ds2d, s, gamma, mean, rstd)
+        dx = dx.reshape(ctx.in_shape)
+        return dx, dx, dg, db, None
+
+
+def add_layernorm(x, res, gamma, beta, eps: float = 1e-5):
+    """(s, y) = (x + res, layernorm(x + res)). Fused on GPU; composed on
+    CPU with identical autograd semantics."""
+    if x.is_cuda:
+        return AddLayerNormFn.apply(x, res, gamma, beta, eps)
+    s = x + res
+    return s, layernorm(s, gamma, beta, eps)
+
+
 class SoftmaxFn(torch.autograd.Function):
     """Fused scale + optional causal mask + softmax over the last dim."""
 

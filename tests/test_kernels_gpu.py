@@ -439,3 +439,56 @@ def test_mlp_fused_epilogue():
                             (b2.grad, b2f.grad, "db2")):
         torch.testing.assert_close(got.float(), want, rtol=5e-2,
                                    atol=5e-2, msg=name)
+
+
+@pytest.mark.gpu
+def test_add_layernorm_fused():
+    """Fused residual add + layernorm vs fp32 composed reference:
+    forward sum + normed output, and x/res/gamma/beta grads with the sum
+    consumed by a second path (exercises the fused dsum accumulation)."""
+    from tepdist_amd import ops
+    torch.manual_seed(0)
+    R, C = 512, 1024
+    x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    r = torch.randn(R, C, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    g = (1 + 0.1 * torch.randn(C, device="cuda")).bfloat16() \
+        .requires_grad_()
+    b = (0.1 * torch.randn(C, device="cuda")).bfloat16().requires_grad_()
+    s, y = ops.add_layernorm(x, r, g, b)
+    (y.float().sum() + (s.float() ** 2).sum()).backward()
+
+    xf, rf, gf, bf = (t.detach().float().clone().requires_grad_()
+                      for t in (x, r, g, b))
+    sf = xf + rf
+    yf = torch.nn.functional.layer_norm(sf, (C,), gf, bf, 1e-5)
+    (yf.sum() + (sf ** 2).sum()).backward()
+
+    torch.testing.assert_close(s.float(), sf, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(y.float(), yf, rtol=2e-2, atol=3e-2)
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=5e-2,
+                               atol=2e-1)
+    torch.testing.assert_close(r.grad.float(), rf.grad, rtol=5e-2,
+                               atol=2e-1)
+    torch.testing.assert_close(g.grad.float(), gf.grad, rtol=5e-2,
+                               atol=5e-1)
+    torch.testing.assert_close(b.grad.float(), bf.grad, rtol=5e-2,
+                               atol=5e-1)
+
+
+@pytest.mark.gpu
+def test_add_layernorm_sum_only_grad():
+    """When the normalized output is unused, backward routes the sum
+    gradient straight through (the None-dy branch)."""
+    from tepdist_amd import ops
+    x = torch.randn(64, 128, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    r = torch.randn(64, 128, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    g = torch.ones(128, device="cuda", dtype=torch.bfloat16)
+    b = torch.zeros(128, device="cuda", dtype=torch.bfloat16)
+    s, _y = ops.add_layernorm(x, r, g, b)
+    s.float().sum().backward()
+    torch.testing.assert_close(x.grad.float(),
+                               torch.ones(64, 128, device="cuda"))
