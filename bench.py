@@ -192,7 +192,7 @@ def main():
         ex = build_stage_executor(
             mod, stage, pp, pp_ranks, micro, act_shape=mod.act_shape,
             act_dtype=dtype, device=device, reducer=reducer, optimizer=opt,
-            pp_group=cdm.pipeline_pair_group(+1 if stage == 0 else -1))
+            pp_group=cdm.pipeline_column_group())
         # rank-IDENTICAL stream: the stage graphs dp-slice the global
         # batch in-graph (dynamic_slice over the mesh round), so every
         # rank must feed the same ids

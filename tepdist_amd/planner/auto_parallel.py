@@ -257,6 +257,25 @@ class AutoParallel:
             for nid, sp in res.node_specs.items():
                 node_specs[nid].set_round(ri, sp)
 
+        # per-micro mesh feasibility: the runtime mesh-transforms EACH
+        # micro-batch's graph, so the SAMPLE batch per micro must still
+        # divide by the batch-splitting (dp) rounds — attention keeps
+        # sequences whole (rules._split0_ok). Clamp micro to the largest
+        # count that keeps every micro splittable.
+        bsz = max((n.attrs.get("batch", 0) for n in g.nodes.values()),
+                  default=0)
+        dpprod = 1
+        for n_r, f in zip(rounds, dp_flags):
+            if f:
+                dpprod *= n_r
+        if bsz and dpprod > 1:
+            for m in range(micro, 0, -1):
+                if bsz % m == 0 and (bsz // m) % dpprod == 0:
+                    micro = m
+                    break
+            else:
+                return None   # no feasible micro count for this mesh
+
         # pipeline stage cut
         sk = GraphSketch(g)
         sp = sk.stage_plan(stages)

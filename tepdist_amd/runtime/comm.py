@@ -195,6 +195,21 @@ class CommDevManager:
                     out = g
         return out
 
+    def pipeline_column_group(self):
+        """Group of ALL pipeline stages at this rank's mesh coordinate
+        (the per-dp-slice pipeline column) — for whole-pipeline
+        collectives like the final loss broadcast. A pair group is NOT a
+        substitute at pp>=3: broadcast from the last stage needs every
+        stage in one group."""
+        out = None
+        for i in range(self.mesh_size):
+            coords = self._unflatten(i)
+            ranks = [self.rank_of(s, coords) for s in range(self.pp)]
+            g = self._group(ranks)
+            if self.rank in ranks:
+                out = g
+        return out
+
     def describe(self) -> str:
         stage, coords = self.coords()
         return (f"CommDevManager[world={self.world} pp={self.pp} "
