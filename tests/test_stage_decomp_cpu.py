@@ -174,3 +174,73 @@ def _hybrid_worker(rank, world, port):
 def test_planned_pipeline_x_dp_hybrid():
     port = free_port()
     mp.spawn(_hybrid_worker, args=(4, port), nprocs=4, join=True)
+
+
+def _pp4_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from tepdist_amd.planner.auto_parallel import AutoParallel
+        from tepdist_amd.planner.pipeline import GraphSketch
+        from tepdist_amd.planner.stage_decomposition import decompose_stages
+        from tepdist_amd.runtime.comm import CommDevManager
+        from tepdist_amd.runtime.executor import build_stage_executor
+        from tepdist_amd.runtime.planned import (PlannedModule,
+                                                 PlannedStageModule)
+        M = 4
+        g, cfg = _graph(batch=2, seq=16)
+        sk = GraphSketch(g).stage_plan(world)
+        plan = decompose_stages(g, sk.node_stage, world)
+        cdm = CommDevManager([1], pp=world)
+        stage, coords = cdm.coords()
+        mod = PlannedStageModule(plan, stage, g)
+        # the bench wiring: pp ranks from the CommDevManager, the final
+        # loss broadcast over the FULL pipeline column group (a 2-rank
+        # pair group deadlocked/failed here at pp>=3 — regression)
+        ex = build_stage_executor(
+            mod, stage, world,
+            [cdm.rank_of(s, coords) for s in range(world)], M,
+            act_shape=mod.act_shape, act_dtype=torch.float32,
+            device="cpu", pp_group=cdm.pipeline_column_group())
+        gen = torch.Generator().manual_seed(9)
+        ids = torch.randint(0, 128, (2 * M, 17), generator=gen)
+
+        def batch_iter(m):
+            sl = ids[m * 2:(m + 1) * 2]
+            return sl[:, :-1], sl[:, 1:]
+
+        loss = ex.run_step(batch_iter)
+        ref_mod = PlannedModule(g, AutoParallel(g, 1).run(),
+                                comm=CommDevManager([1], 1, 0, 1))
+        ref = sum(ref_mod(*batch_iter(m)) for m in range(M)) / M
+        assert abs(loss - float(ref)) < 1e-4, (rank, loss, float(ref))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_planned_pipeline_pp4():
+    port = free_port()
+    mp.spawn(_pp4_worker, args=(4, port), nprocs=4, join=True)
+
+
+def test_micro_count_respects_mesh_feasibility():
+    """Planner regression: micro count must keep every micro-batch's
+    sample count divisible by the batch-splitting rounds (a dp2 x pp4
+    plan at batch 8 once chose micro=8 -> per-micro batch 1, which the
+    mesh transform rejects at run time)."""
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2Config
+    from tepdist_amd.planner.auto_parallel import AutoParallel
+    cfg = GPT2Config(name="t", n_layer=4, n_embd=64, n_head=4,
+                     vocab_size=128, n_ctx=256)
+    g = gpt2_ir(cfg, batch=8, seq=128)
+    plan = AutoParallel(g, 8).run()
+    dpprod = 1
+    for n, f in zip(plan.mesh_rounds, plan.dp_round_flags):
+        if f:
+            dpprod *= n
+    per_micro = 8 // plan.micro_batches
+    assert 8 % plan.micro_batches == 0
+    assert per_micro % max(dpprod, 1) == 0, plan.summary()
