@@ -271,18 +271,24 @@ def mlp_bwd(dy, x, w1, w2, h, pre):
     out = {}
 
     def _fused():
-        out["dh"], out["db1"] = _blt_mlp_dgelu(dy, w2, pre)
+        # proj dgrad with the DGELU_BGRAD epilogue: one library call
+        # yields dgelu'd dh AND the fc bias grad
+        dh, db1 = _blt_mlp_dgelu(dy, w2, pre)
+        dw2 = torch.matmul(dy.t(), h)
+        db2 = torch.empty(dy.shape[1], dtype=BF16, device=dy.device)
+        ws = torch.zeros(dy.shape[1], dtype=torch.float32,
+                         device=dy.device)
+        ext.bias_sum(dy.data_ptr(), db2.data_ptr(), ws.data_ptr(),
+                     dy.shape[0], dy.shape[1], _stream())
+        dx, dw1, _ = linear_bwd(dh, x, w1, False, "none", None)
+        out["r"] = (dx, dw1, db1, dw2, db2)
 
     def _composed():
-        dhp = torch.matmul(dy, w2)
-        dh = torch.empty_like(dhp)
-        ext.gelu_bwd(dhp.data_ptr(), pre.data_ptr(), dh.data_ptr(),
-                     dhp.numel(), _stream())
-        db1 = torch.empty(w2.shape[1], dtype=BF16, device=dy.device)
-        ws = torch.zeros(w2.shape[1], dtype=torch.float32, device=dy.device)
-        ext.bias_sum(dh.data_ptr(), db1.data_ptr(), ws.data_ptr(),
-                     dh.shape[0], dh.shape[1], _stream())
-        out["dh"], out["db1"] = dh, db1
+        # exactly the unfused two-linear backward: per-shape autotuned,
+        # the hand path keeps its fused gelu'-in-transpose wgrad trick
+        dh_post, dw2, db2 = linear_bwd(dy, h, w2, True, "none", None)
+        dx, dw1, db1 = linear_bwd(dh_post, x, w1, True, "gelu", pre)
+        out["r"] = (dx, dw1, db1, dw2, db2)
 
     if not _blt_usable():
         _composed()
@@ -295,14 +301,7 @@ def mlp_bwd(dy, x, w1, w2, h, pre):
         except RuntimeError:
             _blt_ok = False
             _composed()
-    dh = out["dh"]
-    dw2 = torch.matmul(dy.t(), h)
-    db2 = torch.empty(dy.shape[1], dtype=BF16, device=dy.device)
-    ws2 = torch.zeros(dy.shape[1], dtype=torch.float32, device=dy.device)
-    ext.bias_sum(dy.data_ptr(), db2.data_ptr(), ws2.data_ptr(),
-                 dy.shape[0], dy.shape[1], _stream())
-    dx, dw1, _ = linear_bwd(dh, x, w1, False, "none", None)
-    return dx, dw1, out["db1"], dw2, db2
+    return out["r"]
 
 
 def transpose2d(t: torch.Tensor) -> torch.Tensor:
