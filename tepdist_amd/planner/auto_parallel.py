@@ -264,6 +264,11 @@ class AutoParallel:
         # count that keeps every micro splittable.
         bsz = max((n.attrs.get("batch", 0) for n in g.nodes.values()),
                   default=0)
+        if not bsz and g.inputs:
+            # image graphs carry no flattened-token `batch` attr: the
+            # sample count IS the input's dim 0
+            shp = g.nodes[g.inputs[0]].shape
+            bsz = shp[0] if shp else 0
         dpprod = 1
         for n_r, f in zip(rounds, dp_flags):
             if f:
