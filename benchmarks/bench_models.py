@@ -56,9 +56,25 @@ def bench_moe(batch=8, seq=1024):
                       "final_loss": round(float(loss), 4)}), flush=True)
 
 
+def bench_llama(name="llama-1b", batch=8, seq=2048):
+    from tepdist_amd.models.llama import LLAMA_CONFIGS, Llama
+    cfg = LLAMA_CONFIGS[name]
+    m = Llama(cfg, dtype=torch.bfloat16).cuda()
+    seq = min(seq, cfg.n_ctx)
+    ids = torch.randint(0, cfg.vocab_size, (batch, seq + 1)).cuda()
+    dt, loss = measure(m, lambda i: (ids[:, :-1], ids[:, 1:]))
+    print(json.dumps({"bench": "llama", "config": cfg.name,
+                      "batch": batch, "seq": seq, "dtype": "bf16",
+                      "ms_per_step": round(dt * 1e3, 2),
+                      "tokens_per_s": round(batch * seq / dt, 1),
+                      "final_loss": round(float(loss), 4)}), flush=True)
+
+
 if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     if which in ("all", "wrn"):
         bench_wrn()
     if which in ("all", "moe"):
         bench_moe()
+    if which in ("all", "llama"):
+        bench_llama()
