@@ -37,7 +37,11 @@ constexpr int NTH = 512;  // 8 waves
 constexpr int QROWS = 64;          // stage-unit rows
 constexpr int SLOT_E = QROWS * BK; // elements per slot (8 KiB)
 
-enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3 };
+enum { EPI_NONE = 0, EPI_BIAS = 1, EPI_BIAS_GELU = 2, EPI_GELU = 3,
+       // dgelu: C = (A@B) * gelu'(aux) — the Cpre pointer is the INPUT
+       // pre-activation saved by the forward (the fusion hipBLASLt has
+       // no algorithms for on gfx950, profiles/blaslt_epilogue_probe.md)
+       EPI_DGELU = 4 };
 
 typedef __attribute__((address_space(1))) const void* gsrc_t;
 typedef __attribute__((address_space(3))) void* gdst_t;
@@ -262,7 +266,12 @@ void gemm256_kernel(const bf16_t* __restrict__ A,
         const int row = rr + sr;
         const int64_t off = (int64_t)(m0 + row) * ldc + n0 + 64 * p + c8;
         bf16x8 v8 = *reinterpret_cast<const bf16x8*>(img + qoff(row, c8));
-        if (EPI >= 2) {
+        if (EPI == EPI_DGELU) {
+          const bf16x8 pv = *reinterpret_cast<const bf16x8*>(Cpre + off);
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            v8[e] = f2bf(bf2f(v8[e]) * gelu_grad_f(bf2f(pv[e])));
+        } else if (EPI >= 2) {
           *reinterpret_cast<bf16x8*>(Cpre + off) = v8;
 #pragma unroll
           for (int e = 0; e < 8; ++e) v8[e] = f2bf(gelu_f(bf2f(v8[e])));
@@ -350,6 +359,7 @@ void gemm256_bf16(const void* A, const void* B, void* C, void* c_pre,
     case 1: G256(1); break;
     case 2: G256(2); break;
     case 3: G256(3); break;
+    case 4: G256(4); break;
     default: throw std::runtime_error("bad epi");
   }
 #undef G256

@@ -290,17 +290,49 @@ def mlp_bwd(dy, x, w1, w2, h, pre):
         dx, dw1, db1 = linear_bwd(dh_post, x, w1, True, "gelu", pre)
         out["r"] = (dx, dw1, db1, dw2, db2)
 
+    def _hand_dgelu():
+        # the fusion the library cannot express on this stack: proj dgrad
+        # on the 256 schedule with a dgelu epilogue reading the saved
+        # pre-activation — dh = (dy @ w2) * gelu'(pre) in ONE kernel, the
+        # standalone gelu_bwd pass disappears
+        M, N2 = dy.shape
+        K2 = w2.shape[1]
+        w2T = transpose2d(w2)                       # [4d, d] KC
+        dh, _ = _gemm_raw(dy, w2T, True, True, M, K2, N2, N2, N2, 0, 0, 1,
+                          epi=4, out_pre=pre)
+        db1 = torch.empty(K2, dtype=BF16, device=dy.device)
+        ws = torch.zeros(K2, dtype=torch.float32, device=dy.device)
+        ext.bias_sum(dh.data_ptr(), db1.data_ptr(), ws.data_ptr(), M, K2,
+                     _stream())
+        dw2 = torch.matmul(dy.t(), h)
+        db2 = torch.empty(N2, dtype=BF16, device=dy.device)
+        ws2 = torch.zeros(N2, dtype=torch.float32, device=dy.device)
+        ext.bias_sum(dy.data_ptr(), db2.data_ptr(), ws2.data_ptr(), M, N2,
+                     _stream())
+        dx, dw1, _ = linear_bwd(dh, x, w1, False, "none", None)
+        out["r"] = (dx, dw1, db1, dw2, db2)
+
+    def _base():
+        # hand-dgelu vs composed, measured per shape (256-schedule shapes
+        # only; TEPDIST_GEMM_BACKEND pins propagate through _pick_backend)
+        if (dy.shape[0] % 256 == 0 and w2.shape[1] % 256 == 0
+                and dy.shape[1] % 64 == 0 and pre.is_contiguous()):
+            _pick_backend(("mlp_bh", dy.shape[0], dy.shape[1],
+                           w2.shape[1]), _hand_dgelu, _composed)()
+        else:
+            _composed()
+
     if not _blt_usable():
-        _composed()
+        _base()
     elif _MLP_FUSED == "1":
         _fused()
     else:
         try:
             _pick_backend(("mlp_b", dy.shape[0], dy.shape[1],
-                           w2.shape[1]), _fused, _composed)()
+                           w2.shape[1]), _fused, _base)()
         except RuntimeError:
             _blt_ok = False
-            _composed()
+            _base()
     return out["r"]
 
 

@@ -492,3 +492,54 @@ def test_add_layernorm_sum_only_grad():
     s.float().sum().backward()
     torch.testing.assert_close(x.grad.float(),
                                torch.ones(64, 128, device="cuda"))
+
+
+@pytest.mark.gpu
+def test_gemm_dgelu_epilogue():
+    """256-schedule GEMM with the dgelu epilogue: C = (A@B) * gelu'(aux)
+    vs fp32 autograd through tanh-gelu."""
+    M, N, K = 512, 256, 64
+    a = _randn(M, K, seed=31)
+    b = _randn(N, K, seed=32)
+    pre = _randn(M, N, seed=33)
+    dh, _ = hip._gemm_raw(a, b, True, True, M, N, K, K, K, 0, 0, 1,
+                          epi=4, out_pre=pre)
+    torch.cuda.synchronize()
+    up = a.float() @ b.float().t()
+    p32 = pre.float().requires_grad_()
+    torch.nn.functional.gelu(p32, approximate="tanh").backward(up)
+    _assert_close_bf16(dh, p32.grad, rtol=3e-2, scale=math.sqrt(K))
+
+
+@pytest.mark.gpu
+def test_mlp_hand_dgelu_backward():
+    """ops.mlp backward with the hand dgelu-epilogue dgrad (pinned via
+    TEPDIST_GEMM_BACKEND=hip) vs the fp32 reference."""
+    torch.manual_seed(0)
+    T, d = 512, 256
+    mk = lambda *s: (torch.randn(*s, device="cuda", dtype=torch.bfloat16)
+                     * 0.05).requires_grad_()
+    x, w1, b1 = mk(T, d), mk(4 * d, d), mk(4 * d)
+    w2, b2 = mk(d, 4 * d), mk(d)
+    dy = torch.randn(T, d, device="cuda", dtype=torch.bfloat16)
+    old = hip._GEMM_BACKEND
+    hip._GEMM_BACKEND = "hip"
+    try:
+        from tepdist_amd import ops
+        y = ops.mlp(x, w1, b1, w2, b2)
+        y.backward(dy)
+    finally:
+        hip._GEMM_BACKEND = old
+    xf, w1f, b1f, w2f, b2f = (t.detach().float().clone().requires_grad_()
+                              for t in (x, w1, b1, w2, b2))
+    h = torch.nn.functional.gelu(xf @ w1f.t() + b1f, approximate="tanh")
+    ref = h @ w2f.t() + b2f
+    ref.backward(dy.float())
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-2)
+    for got, want, name in ((x.grad, xf.grad, "dx"),
+                            (w1.grad, w1f.grad, "dw1"),
+                            (b1.grad, b1f.grad, "db1"),
+                            (w2.grad, w2f.grad, "dw2"),
+                            (b2.grad, b2f.grad, "db2")):
+        torch.testing.assert_close(got.float(), want, rtol=5e-2,
+                                   atol=5e-2, msg=name)
