@@ -316,7 +316,8 @@ def mlp_bwd(dy, x, w1, w2, h, pre):
         # hand-dgelu vs composed, measured per shape (256-schedule shapes
         # only; TEPDIST_GEMM_BACKEND pins propagate through _pick_backend)
         if (dy.shape[0] % 256 == 0 and w2.shape[1] % 256 == 0
-                and dy.shape[1] % 64 == 0 and pre.is_contiguous()):
+                and dy.shape[1] % 64 == 0 and dy.shape[1] >= 128
+                and pre.is_contiguous()):
             _pick_backend(("mlp_bh", dy.shape[0], dy.shape[1],
                            w2.shape[1]), _hand_dgelu, _composed)()
         else:
