@@ -11,6 +11,7 @@ layout of models/gpt2.py), so no changes to the training forward."""
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -71,6 +72,48 @@ class Generator:
         x = ops.layernorm(x, self.m.lnf_g, self.m.lnf_b, self.cfg.ln_eps)
         return ops.linear(x[:, -1:], self.m.wte)   # last position only
 
+    # -- static-shape greedy decode step (hipGraph-capturable) -----------
+    #
+    # Eager per-token decode is LAUNCH-bound (~1000 kernels/token across
+    # 24 layers). This step runs on fixed buffers only: embeds tok at the
+    # device-side position cur_t, writes K/V into the cache at cur_t
+    # (index_copy_, no host sync), attends over the FULL S_max cache with
+    # a kpos > cur_t mask (exp(-inf) = 0 makes it bitwise equal to
+    # slicing), argmaxes, feeds the next token back into tok, and
+    # advances cur_t — so a captured graph replays autonomously.
+
+    def _decode_step(self, tok, cur_t, kc, vc, kpos):
+        cfg = self.cfg
+        B = tok.shape[0]
+        H, d = cfg.n_head, self.m.wte.shape[1]
+        D = d // H
+        scale = 1.0 / math.sqrt(D)
+        x = ops.embedding(tok.reshape(-1), self.m.wte).reshape(B, 1, d) \
+            + ops.embedding(cur_t, self.m.wpe)
+        mask = (kpos > cur_t)                       # [S_max] device bool
+        for li, blk in enumerate(self.m.blocks):
+            h = ops.layernorm(x, blk.ln1_g, blk.ln1_b, cfg.ln_eps)
+            qkv = ops.linear(h, blk.w_qkv, blk.b_qkv)
+            q, k, v = qkv.split(d, dim=-1)
+            qh = q.reshape(B, 1, H, D).transpose(1, 2)
+            kc[li].index_copy_(2, cur_t,
+                               k.reshape(B, 1, H, D).transpose(1, 2))
+            vc[li].index_copy_(2, cur_t,
+                               v.reshape(B, 1, H, D).transpose(1, 2))
+            scores = ops.matmul(qh.contiguous(),
+                                kc[li].transpose(-1, -2)) * scale
+            scores = scores.masked_fill(mask, float("-inf"))
+            p = torch.softmax(scores.float(), dim=-1).to(x.dtype)
+            a = ops.matmul(p, vc[li])
+            a = a.transpose(1, 2).reshape(B, 1, d)
+            x = x + ops.linear(a, blk.w_proj, blk.b_proj)
+            h = ops.layernorm(x, blk.ln2_g, blk.ln2_b, cfg.ln_eps)
+            h = ops.linear(h, blk.w_fc, blk.b_fc, act="gelu")
+            x = x + ops.linear(h, blk.w_out, blk.b_out)
+        logits = self._logits(x).float()[:, -1, :cfg.vocab_size]
+        tok.copy_(logits.argmax(-1, keepdim=True))
+        cur_t.add_(1)
+
     @torch.no_grad()
     def generate(self, ids: torch.Tensor, max_new_tokens: int,
                  temperature: float = 0.0, top_k: int = 0,
@@ -94,6 +137,45 @@ class Generator:
                                                            self.m.wpe)
         x = self._stack(x, kc, vc, 0)
         cur = S0
+
+        # greedy decode on GPU: static-buffer steps, hipGraph-captured
+        # after two eager warmups (the eager per-token loop is
+        # launch-bound; replaying one recorded step removes ~1000
+        # launches/token). TEPDIST_DECODE_GRAPH=0 forces eager.
+        if (temperature == 0 and ids.is_cuda and S_max > S0
+                and max_new_tokens >= 8
+                and os.environ.get("TEPDIST_DECODE_GRAPH", "1") != "0"):
+            logits = self._logits(x).float()[:, -1, :cfg.vocab_size]
+            tok = logits.argmax(-1, keepdim=True)
+            outs = [tok.clone()]
+            cur_t = torch.full((1,), S0, device=dev, dtype=torch.long)
+            kpos = torch.arange(S_max, device=dev)
+            n_rest = min(max_new_tokens, S_max - S0) - 1
+            graph = None
+            done = 0
+            try:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    for _ in range(min(2, n_rest)):
+                        self._decode_step(tok, cur_t, kc, vc, kpos)
+                        outs.append(tok.clone())
+                        done += 1
+                torch.cuda.current_stream().wait_stream(side)
+                if done < n_rest:
+                    graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(graph):
+                        self._decode_step(tok, cur_t, kc, vc, kpos)
+            except Exception:
+                graph = None           # capture failed: stay eager
+            while done < n_rest:
+                if graph is not None:
+                    graph.replay()
+                else:
+                    self._decode_step(tok, cur_t, kc, vc, kpos)
+                outs.append(tok.clone())
+                done += 1
+            return torch.cat([ids] + outs, dim=1)
         for _ in range(max_new_tokens):
             if cur >= S_max:
                 break
