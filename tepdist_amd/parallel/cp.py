@@ -83,8 +83,12 @@ class ContextParallelGPT2(GPT2):
         if labels is None or not dist.is_initialized():
             return out
         # global token-mean loss: weight each rank's local mean by its
-        # valid-token count, all-reduce (autograd-aware) the weighted sum
+        # valid-token count, all-reduce (autograd-aware) the weighted sum.
+        # A shard whose labels are ALL ignored contributes exactly 0 (its
+        # local mean may be nan — mean over zero tokens).
         n = (labels.reshape(-1) != -1).sum().to(out.dtype)
         tot = n.clone()
         dist.all_reduce(tot, group=self.cp_group)
-        return reduce_from_group(out * (n / tot), self.cp_group)
+        local = torch.where(n > 0, out * (n / tot.clamp_min(1)),
+                            torch.zeros_like(out))
+        return reduce_from_group(local, self.cp_group)
