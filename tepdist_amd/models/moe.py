@@ -146,8 +146,8 @@ class MoELayer(nn.Module):
         ys = []
         for e in range(self.e_local):
             xe = recv[:, e].reshape(-1, d).contiguous()
-            h = ops.linear(xe, self.w1[e], self.b1[e], act="gelu")
-            ys.append(ops.linear(h, self.w2[e], self.b2[e]))
+            ys.append(ops.mlp(xe, self.w1[e], self.b1[e],
+                              self.w2[e], self.b2[e]))
         Y = torch.stack(ys, dim=0)                       # [e_local, src*C, d]
         Y = Y.reshape(self.e_local, -1, C, d).transpose(0, 1)  # [src, e_l, C, d]
 
