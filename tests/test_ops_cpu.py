@@ -155,3 +155,51 @@ def test_adamw_matches_torch():
         opt_ref.step()
         opt_ours.step()
     torch.testing.assert_close(p_ours.data, p_ref.data, rtol=1e-5, atol=1e-6)
+
+
+def test_mlp_op_matches_composed():
+    """ops.mlp (fused on GPU) composes to exactly linear(gelu)+linear on
+    CPU — values and all five grads."""
+    import torch
+
+    from tepdist_amd import ops
+    torch.manual_seed(0)
+    T, d = 16, 8
+    x = torch.randn(T, d, requires_grad=True)
+    w1 = torch.randn(4 * d, d, requires_grad=True)
+    b1 = torch.randn(4 * d, requires_grad=True)
+    w2 = torch.randn(d, 4 * d, requires_grad=True)
+    b2 = torch.randn(d, requires_grad=True)
+    y = ops.mlp(x, w1, b1, w2, b2)
+    y.sum().backward()
+    x2, w12, b12, w22, b22 = (t.detach().clone().requires_grad_()
+                              for t in (x, w1, b1, w2, b2))
+    ref = ops.linear(ops.linear(x2, w12, b12, act="gelu"), w22, b22)
+    ref.sum().backward()
+    torch.testing.assert_close(y, ref)
+    for a, b in ((x, x2), (w1, w12), (b1, b12), (w2, w22), (b2, b22)):
+        torch.testing.assert_close(a.grad, b.grad)
+
+
+def test_add_layernorm_op_cpu():
+    """ops.add_layernorm returns (sum, normed) with correct grads to both
+    consumers on the CPU composed path."""
+    import torch
+
+    from tepdist_amd import ops
+    torch.manual_seed(1)
+    x = torch.randn(8, 16, requires_grad=True)
+    r = torch.randn(8, 16, requires_grad=True)
+    g = torch.ones(16, requires_grad=True)
+    b = torch.zeros(16, requires_grad=True)
+    s, y = ops.add_layernorm(x, r, g, b)
+    (s.pow(2).sum() + y.sum()).backward()
+    x2, r2, g2, b2 = (t.detach().clone().requires_grad_()
+                      for t in (x, r, g, b))
+    s2 = x2 + r2
+    y2 = ops.layernorm(s2, g2, b2)
+    (s2.pow(2).sum() + y2.sum()).backward()
+    torch.testing.assert_close(s, s2)
+    torch.testing.assert_close(y, y2)
+    torch.testing.assert_close(x.grad, x2.grad)
+    torch.testing.assert_close(g.grad, g2.grad)
