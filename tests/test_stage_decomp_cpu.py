@@ -219,10 +219,11 @@ def _pp4_worker(rank, world, port):
         dist.destroy_process_group()
 
 
+@pytest.mark.parametrize("world", [3, 4])
 @pytest.mark.timeout(600)
-def test_planned_pipeline_pp4():
+def test_planned_pipeline_deep(world):
     port = free_port()
-    mp.spawn(_pp4_worker, args=(4, port), nprocs=4, join=True)
+    mp.spawn(_pp4_worker, args=(world, port), nprocs=world, join=True)
 
 
 def test_micro_count_respects_mesh_feasibility():
