@@ -15,9 +15,10 @@ CommDevManager turns mesh rounds into RCCL process groups, and each rank
 runs its transformed graph through the interpreter over the CDNA4 kernel
 layer (runtime/planned.py; reference: ExecuteRPCPlan runs the planner's
 compiled sub-modules, service_rt.cc:530-671). Plans with pipeline stages
-fall back to the hand 1F1B engine (stage decomposition of the planned
-graph is the remaining gap). `--parallel dp|tp<N>|pp<N>` selects the
-hand-parallelized model classes for A/B comparison.
+run the generic stage decomposition of the planned graph through the
+task-list executor (scheduled 1F1B order, gc_plan buffer release).
+`--parallel dp|tp<N>|pp<N>` selects the hand-parallelized model classes
+for A/B comparison.
 
 Metric/config per BASELINE.json: tokens/sec (whole node), GPT-2
 auto-parallel, synthetic data, random-init weights, bf16."""
