@@ -290,3 +290,35 @@ def test_fuse_mlp_respects_collectives():
     # the collective (all_gather of the split output) sits between the
     # pair -> the gelu linear survives unfused
     assert fc_t and mlps == 0
+
+
+@pytest.mark.timeout(600)
+def test_planned_path_learns():
+    """End-to-end learning sanity: the planned-graph module trained with
+    the fused-optimizer surface memorizes a fixed batch (loss falls by
+    >40% in 15 steps) — catches sign/scale bugs no single-step
+    equivalence test can."""
+    import torch
+
+    from tepdist_amd.ir.capture import gpt2_ir
+    from tepdist_amd.models.configs import GPT2Config
+    from tepdist_amd.planner.auto_parallel import AutoParallel
+    from tepdist_amd.runtime.planned import PlannedModule
+    from tepdist_amd.train.optim import AdamW
+    cfg = GPT2Config(name="t", n_layer=2, n_embd=64, n_head=4,
+                     vocab_size=256, n_ctx=64)
+    g = gpt2_ir(cfg, batch=4, seq=32)
+    mod = PlannedModule(g, AutoParallel(g, 1).run())
+    opt = AdamW(mod.parameters(), lr=3e-3)
+    gen = torch.Generator().manual_seed(5)
+    ids = torch.randint(0, 256, (4, 33), generator=gen)
+    first = last = None
+    for step in range(15):
+        opt.zero_grad()
+        loss = mod(ids[:, :-1], ids[:, 1:])
+        loss.backward()
+        opt.step()
+        if step == 0:
+            first = float(loss.detach())
+        last = float(loss.detach())
+    assert last < first * 0.6, (first, last)
