@@ -83,7 +83,7 @@ def _learn_worker(rank, world, port):
         ids = torch.randint(0, 128, (2, 33), generator=g)
         li, ll, pos = m.shard_inputs(ids[:, :-1], ids[:, 1:].contiguous())
         first = last = None
-        for step in range(15):
+        for step in range(25):
             opt.zero_grad()
             red.reset()
             red.arm()
@@ -94,7 +94,7 @@ def _learn_worker(rank, world, port):
             if step == 0:
                 first = float(loss.detach())
             last = float(loss.detach())
-        assert last < first * 0.6, (rank, first, last)
+        assert last < first * 0.7, (rank, first, last)
     finally:
         dist.destroy_process_group()
 
